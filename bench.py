@@ -1,0 +1,239 @@
+#!/usr/bin/env python3
+"""infinistore-amd flagship benchmark — the BASELINE.json headline metric:
+put/get throughput (GB/s, aggregate over all server GPUs) and round-trip
+latency on 128 KB KV-cache blocks.
+
+One *step* = write `--blocks` 128 KB blocks from a GPU tensor into the store
+(via the local IPC path + batched HIP gather kernel) + sync, then read them
+all back + sync. `value` is the whole-job aggregate GB/s of payload moved
+(put + get) across all N GPUs.
+
+Launch (driver contract):
+    python bench.py --gpus N --steps K --warmup W
+For N>1 the driver uses torch.distributed.run with one rank per GPU; rank r
+runs a client on GPU r, rank 0 additionally hosts the in-process server
+sharded over all N GPUs (weak scaling: per-GPU work is fixed).
+"""
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+import uuid
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--block-kb", type=int, default=128)
+    p.add_argument("--blocks", type=int, default=2048,
+                   help="blocks per rank per step (2048 x 128KB = 256 MB)")
+    p.add_argument("--pool-gb", type=int, default=8, help="pool GB per shard")
+    p.add_argument("--port", type=int, default=0)
+    p.add_argument("--cpu", action="store_true",
+                   help="CPU-only mode (TCP fabric, DRAM pool) for dev boxes")
+    p.add_argument("--latency-ops", type=int, default=200,
+                   help="single-block round-trips for the p50/p99 measurement")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(args.gpus, world)
+
+    have_gpu = torch.cuda.is_available() and not args.cpu
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+
+        dist = tdist
+        dist.init_process_group(backend="gloo")  # host-side coordination only
+
+    import infinistore_amd as ifs
+
+    port = args.port or (23000 + (os.getpid() % 1000) if world == 1 else 23999)
+
+    # Rank 0 hosts the server, sharded over all visible GPUs.
+    if rank == 0:
+        scfg = ifs.ServerConfig(
+            service_port=port,
+            manage_port=port + 1,
+            prealloc_size=args.pool_gb,
+            minimal_allocate_size=args.block_kb,
+            cpu_only=not have_gpu,
+            devices=list(range(n_gpus)) if have_gpu else [],
+        )
+        ifs.register_server(scfg)
+    if dist:
+        dist.barrier()
+
+    block_bytes = args.block_kb * 1024
+    elems_per_block = block_bytes // 2  # bf16
+    total_elems = args.blocks * elems_per_block
+    dev = f"cuda:{local_rank}" if have_gpu else "cpu"
+
+    if have_gpu:
+        torch.cuda.set_device(local_rank)
+    src = torch.randn(total_elems, dtype=torch.bfloat16, device=dev)
+    dst = torch.zeros_like(src)
+
+    ccfg = ifs.ClientConfig(
+        host_addr="127.0.0.1",
+        service_port=port,
+        connection_type=ifs.TYPE_LOCAL_GPU if have_gpu else ifs.TYPE_RDMA,
+        link_type="TCP",
+    )
+    conn = ifs.InfinityConnection(ccfg)
+    conn.connect()
+    if not have_gpu:
+        conn.register_mr(src)
+        conn.register_mr(dst)
+
+    run_id = uuid.uuid4().hex[:8]
+    offsets = [i * elems_per_block for i in range(args.blocks)]
+
+    def step_keys(step):
+        return [f"r{rank}-s{step}-{run_id}-{i}" for i in range(args.blocks)]
+
+    def do_put(keys):
+        if have_gpu:
+            conn.local_gpu_write_cache(src, list(zip(keys, offsets)), elems_per_block)
+            conn.sync()
+        else:
+            blocks = conn.allocate_rdma(keys, block_bytes)
+            conn.rdma_write_cache(src, offsets, elems_per_block, blocks)
+            conn.sync()
+
+    def do_get(keys):
+        conn.read_cache(dst, list(zip(keys, offsets)), elems_per_block)
+        conn.sync()
+
+    def purge_all():
+        if dist:
+            dist.barrier()
+        if rank == 0:
+            ifs.purge_kv_map()
+        if dist:
+            dist.barrier()
+
+    def sync_all():
+        if have_gpu:
+            torch.cuda.synchronize()
+        if dist:
+            dist.barrier()
+
+    # ---- correctness spot-check + warmup ----
+    for w in range(args.warmup):
+        keys = [f"warm-{k}" for k in step_keys(w)]
+        do_put(keys)
+        do_get(keys)
+    if not torch.equal(src.cpu(), dst.cpu()):
+        print(json.dumps({"error": "data mismatch in warmup"}))
+        sys.exit(1)
+    purge_all()
+
+    # ---- timed region ----
+    sync_all()
+    t0 = time.perf_counter()
+    put_time = 0.0
+    get_time = 0.0
+    for s in range(args.steps):
+        keys = step_keys(s)
+        tp = time.perf_counter()
+        do_put(keys)
+        put_time += time.perf_counter() - tp
+        tg = time.perf_counter()
+        do_get(keys)
+        get_time += time.perf_counter() - tg
+    sync_all()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks; bytes summed over ranks
+    if dist:
+        t = torch.tensor([elapsed, put_time, get_time], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed, put_time, get_time = t.tolist()
+
+    purge_all()
+
+    # ---- latency phase (single-block round trips) ----
+    lat_put, lat_get = [], []
+    lkeys = [f"lat-{run_id}-{i}" for i in range(args.latency_ops)]
+    for i in range(args.latency_ops):
+        t1 = time.perf_counter()
+        if have_gpu:
+            conn.local_gpu_write_cache(src, [(lkeys[i], 0)], elems_per_block)
+            conn.sync()
+        else:
+            blk = conn.allocate_rdma([lkeys[i]], block_bytes)
+            conn.rdma_write_cache(src, [0], elems_per_block, blk)
+            conn.sync()
+        t2 = time.perf_counter()
+        conn.read_cache(dst, [(lkeys[i], 0)], elems_per_block)
+        conn.sync()
+        t3 = time.perf_counter()
+        lat_put.append((t2 - t1) * 1e6)
+        lat_get.append((t3 - t2) * 1e6)
+    purge_all()
+
+    def pct(v, q):
+        return statistics.quantiles(v, n=100)[q - 1] if len(v) >= 10 else max(v)
+
+    bytes_per_step_rank = args.blocks * block_bytes
+    total_gb = 2.0 * bytes_per_step_rank * args.steps * world / 1e9  # put+get
+    gbps = total_gb / elapsed
+    put_gbps = bytes_per_step_rank * args.steps * world / 1e9 / put_time
+    get_gbps = bytes_per_step_rank * args.steps * world / 1e9 / get_time
+
+    if rank == 0:
+        result = {
+            "metric": "put_get_GBps",
+            "value": round(gbps, 3),
+            "unit": "GB/s",
+            "n_gpus": world if have_gpu else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "kvcache-store",
+                "block_kb": args.block_kb,
+                "blocks_per_rank_per_step": args.blocks,
+                "global_batch": args.blocks * world,
+                "seq_len": 0,
+                "parallelism": f"shard{world}",
+                "path": "local_gpu_ipc" if have_gpu else "tcp_fabric_cpu",
+                "put_GBps": round(put_gbps, 3),
+                "get_GBps": round(get_gbps, 3),
+                "p50_put_us": round(pct(lat_put, 50), 1),
+                "p99_put_us": round(pct(lat_put, 99), 1),
+                "p50_get_us": round(pct(lat_get, 50), 1),
+                "p99_get_us": round(pct(lat_get, 99), 1),
+            },
+        }
+        print(json.dumps(result))
+
+    conn.close()
+    if dist:
+        dist.barrier()
+    if rank == 0:
+        ifs.unregister_server()
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,422 @@
+#include "client.h"
+
+#include <arpa/inet.h>
+#include <netdb.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <chrono>
+#include <cstring>
+
+#include "../core/log.h"
+#include "../core/utils.h"
+#include "../gpu/gpu.h"
+
+namespace ifs {
+
+ClientConn::~ClientConn() { close_conn(); }
+
+void ClientConn::close_conn() {
+    {
+        std::lock_guard<std::mutex> lk(q_mu_);
+        worker_stop_ = true;
+    }
+    q_cv_.notify_all();
+    if (worker_.joinable()) worker_.join();
+    worker_started_ = false;
+    if (fd_ >= 0) {
+        ::close(fd_);
+        fd_ = -1;
+    }
+    connected_ = false;
+    rdma_connected_ = false;
+}
+
+int ClientConn::init_connection(const ClientConfigC& cfg) {
+    set_log_level(cfg.log_level.c_str());
+    if (connected_) return -1;
+    struct addrinfo hints{}, *res = nullptr;
+    hints.ai_family = AF_INET;
+    hints.ai_socktype = SOCK_STREAM;
+    char port[16];
+    snprintf(port, sizeof(port), "%d", cfg.service_port);
+    if (getaddrinfo(cfg.host_addr.c_str(), port, &hints, &res) != 0 || !res) {
+        ERROR("getaddrinfo(%s) failed", cfg.host_addr.c_str());
+        return -1;
+    }
+    int fd = ::socket(res->ai_family, res->ai_socktype, res->ai_protocol);
+    if (fd < 0) {
+        freeaddrinfo(res);
+        return -1;
+    }
+    if (::connect(fd, res->ai_addr, res->ai_addrlen) != 0) {
+        ERROR("connect %s:%d failed: %s", cfg.host_addr.c_str(), cfg.service_port,
+              strerror(errno));
+        ::close(fd);
+        freeaddrinfo(res);
+        return -1;
+    }
+    freeaddrinfo(res);
+    int one = 1;
+    setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+    fd_ = fd;
+    connected_ = true;
+    return 0;
+}
+
+int ClientConn::setup_rdma(const ClientConfigC& cfg) {
+    (void)cfg;
+    if (!connected_) return -1;
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!send_req(OP_RDMA_EXCHANGE, nullptr, 0)) return -1;
+    int code = 0;
+    if (!recv_status(&code) || code != FINISH) return -1;
+    std::vector<uint8_t> tag;
+    if (!recv_payload(&tag) || tag.size() != 4) return -1;
+    if (memcmp(tag.data(), "TCPF", 4) != 0) {
+        ERROR("unknown fabric tag");
+        return -1;
+    }
+    rdma_connected_ = true;
+    // Spawn the async worker lazily on first connect.
+    if (!worker_started_) {
+        worker_stop_ = false;
+        worker_ = std::thread([this] { worker_main(); });
+        worker_started_ = true;
+    }
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// framing helpers (io_mu_ must be held)
+// ---------------------------------------------------------------------------
+bool ClientConn::send_req(char op, const uint8_t* body, size_t n) {
+    Header h{kMagic, op, static_cast<uint32_t>(n)};
+    if (!send_exact(fd_, &h, sizeof(h))) return false;
+    if (n && !send_exact(fd_, body, n)) return false;
+    return true;
+}
+
+bool ClientConn::recv_status(int* code) { return recv_exact(fd_, code, 4); }
+
+bool ClientConn::recv_payload(std::vector<uint8_t>* out) {
+    uint32_t len = 0;
+    if (!recv_exact(fd_, &len, 4)) return false;
+    out->resize(len);
+    if (len && !recv_exact(fd_, out->data(), len)) return false;
+    return true;
+}
+
+// ---------------------------------------------------------------------------
+// local (IPC) path
+// ---------------------------------------------------------------------------
+int ClientConn::rw_local(char op, const std::vector<std::pair<std::string, uint64_t>>& blocks,
+                         int block_size, uintptr_t ptr, int device_id) {
+    if (!connected_) return -1;
+    if (!gpu::available()) {
+        ERROR("local path requires a GPU");
+        return -1;
+    }
+    gpu::IpcHandle handle;
+    uint64_t base_offset = 0;
+    if (!gpu::ipc_export(reinterpret_cast<void*>(ptr), &handle, &base_offset)) return -1;
+
+    LocalMetaMsg msg;
+    msg.device = device_id;
+    msg.block_size = block_size;
+    msg.base_offset = base_offset;
+    msg.ipc_handle.assign(handle.bytes, handle.bytes + gpu::kIpcHandleSize);
+    msg.blocks.reserve(blocks.size());
+    for (auto& b : blocks) msg.blocks.push_back({b.first, b.second});
+    auto body = build_local_meta(msg);
+
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!send_req(op, body.data(), body.size())) return -1;
+    int code = 0;
+    if (!recv_status(&code)) return -1;
+    if (code != TASK_ACCEPTED && code != FINISH) {
+        WARN("rw_local op=%c -> %d", op, code);
+        return -code;
+    }
+    return 0;
+}
+
+int ClientConn::sync_local() {
+    if (!connected_) return -1;
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!send_req(OP_SYNC, nullptr, 0)) return -1;
+    int remain = -1;
+    if (!recv_status(&remain)) return -1;
+    return remain;
+}
+
+// ---------------------------------------------------------------------------
+// RDMA-semantics path over the TCP fabric
+// ---------------------------------------------------------------------------
+int ClientConn::register_mr(uintptr_t ptr, size_t size) {
+    bool dev = false;
+    if (gpu::available()) {
+        // classify host vs device like the reference's cudaPointerGetAttributes
+        // check (libinfinistore.cpp:1168)
+        dev = gpu::is_device_pointer(reinterpret_cast<const void*>(ptr));
+    }
+    std::lock_guard<std::mutex> lk(region_mu_);
+    regions_.push_back({ptr, size, dev});
+    return 1;
+}
+
+bool ClientConn::is_device_ptr(uintptr_t ptr) {
+    {
+        std::lock_guard<std::mutex> lk(region_mu_);
+        for (auto& r : regions_) {
+            if (ptr >= r.ptr && ptr < r.ptr + r.size) return r.device;
+        }
+    }
+    if (gpu::available()) return gpu::is_device_pointer(reinterpret_cast<const void*>(ptr));
+    return false;
+}
+
+std::vector<RemoteBlockOut> ClientConn::do_allocate(const std::vector<std::string>& keys,
+                                                    int block_size) {
+    RemoteMetaMsg msg;
+    msg.keys = keys;
+    msg.block_size = block_size;
+    msg.op = OP_RDMA_ALLOCATE;
+    auto body = build_remote_meta(msg);
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!send_req(OP_RDMA_ALLOCATE, body.data(), body.size())) return {};
+    int code = 0;
+    if (!recv_status(&code) || code != FINISH) {
+        WARN("allocate -> %d", code);
+        return {};
+    }
+    std::vector<uint8_t> payload;
+    if (!recv_payload(&payload)) return {};
+    std::vector<RemoteBlockWire> wire;
+    if (!parse_allocate_response(payload.data(), payload.size(), &wire)) return {};
+    std::vector<RemoteBlockOut> out;
+    out.reserve(wire.size());
+    for (auto& w : wire) out.push_back({w.rkey, w.remote_addr});
+    return out;
+}
+
+std::vector<RemoteBlockOut> ClientConn::allocate_rdma(const std::vector<std::string>& keys,
+                                                      int block_size) {
+    if (!rdma_connected_) return {};
+    return do_allocate(keys, block_size);
+}
+
+int ClientConn::allocate_rdma_async(const std::vector<std::string>& keys, int block_size,
+                                    std::function<void(std::vector<RemoteBlockOut>)> cb) {
+    if (!rdma_connected_) return -1;
+    inflight_.fetch_add(1);
+    enqueue([this, keys, block_size, cb = std::move(cb)] {
+        auto res = do_allocate(keys, block_size);
+        cb(std::move(res));
+    });
+    return 0;
+}
+
+int ClientConn::do_w_rdma(const uint64_t* offsets, size_t n_offsets, int block_size,
+                          const RemoteBlockOut* blocks, size_t n_blocks, uintptr_t base_ptr) {
+    if (n_offsets != n_blocks) return -1;
+    bool dev = is_device_ptr(base_ptr);
+    size_t bs = static_cast<size_t>(block_size);
+
+    // Collect non-FAKE blocks (duplicates are skipped — first write wins).
+    std::vector<size_t> idx;
+    idx.reserve(n_blocks);
+    for (size_t i = 0; i < n_blocks; i++) {
+        if (!is_fake_remote_block(blocks[i].rkey, blocks[i].remote_addr)) idx.push_back(i);
+    }
+
+    // Chunked pipelined puts: assemble [n][bs][addrs][payload] bodies of up
+    // to ~8 MB, stream them all, then collect the acks.
+    const size_t kChunkBytes = 8u << 20;
+    size_t per = std::max<size_t>(1, kChunkBytes / (bs + 8));
+    size_t n_chunks = 0;
+    std::vector<uint64_t> all_addrs;
+    {
+        std::unique_lock<std::mutex> lk(io_mu_);
+        for (size_t start = 0; start < idx.size(); start += per) {
+            size_t take = std::min(per, idx.size() - start);
+            std::vector<uint8_t> body(8 + take * 8 + take * bs);
+            uint32_t n32 = static_cast<uint32_t>(take), bs32 = static_cast<uint32_t>(bs);
+            memcpy(body.data(), &n32, 4);
+            memcpy(body.data() + 4, &bs32, 4);
+            uint64_t* addrs = reinterpret_cast<uint64_t*>(body.data() + 8);
+            uint8_t* payload = body.data() + 8 + take * 8;
+            for (size_t j = 0; j < take; j++) {
+                size_t i = idx[start + j];
+                addrs[j] = blocks[i].remote_addr;
+                all_addrs.push_back(blocks[i].remote_addr);
+                const void* src = reinterpret_cast<const void*>(base_ptr + offsets[i]);
+                if (dev) {
+                    if (!gpu::memcpy_d2h(payload + j * bs, src, bs)) return -1;
+                } else {
+                    memcpy(payload + j * bs, src, bs);
+                }
+            }
+            if (!send_req(OP_TCP_PUT, body.data(), body.size())) return -1;
+            n_chunks++;
+        }
+        for (size_t ci = 0; ci < n_chunks; ci++) {
+            int code = 0;
+            if (!recv_status(&code)) return -1;
+            if (code != TASK_ACCEPTED && code != FINISH) return -code;
+        }
+        if (!all_addrs.empty()) {
+            // Commit; the server ACKs after flipping committed=true, so a
+            // successful return here means readers will see the data.
+            RemoteMetaMsg cm;
+            cm.op = OP_RDMA_WRITE_COMMIT;
+            cm.block_size = block_size;
+            cm.remote_addrs = std::move(all_addrs);
+            auto cbody = build_remote_meta(cm);
+            if (!send_req(OP_RDMA_WRITE_COMMIT, cbody.data(), cbody.size())) return -1;
+            int code = 0;
+            if (!recv_status(&code) || code != FINISH) return -1;
+        }
+    }
+    return 0;
+}
+
+int ClientConn::w_rdma(const uint64_t* offsets, size_t n_offsets, int block_size,
+                       const RemoteBlockOut* blocks, size_t n_blocks, uintptr_t base_ptr) {
+    if (!rdma_connected_) return -1;
+    return do_w_rdma(offsets, n_offsets, block_size, blocks, n_blocks, base_ptr);
+}
+
+int ClientConn::w_rdma_async(const uint64_t* offsets, size_t n_offsets, int block_size,
+                             const RemoteBlockOut* blocks, size_t n_blocks, uintptr_t base_ptr,
+                             std::function<void()> cb) {
+    if (!rdma_connected_) return -1;
+    std::vector<uint64_t> offs(offsets, offsets + n_offsets);
+    std::vector<RemoteBlockOut> blks(blocks, blocks + n_blocks);
+    inflight_.fetch_add(1);
+    enqueue([this, offs = std::move(offs), block_size, blks = std::move(blks), base_ptr,
+             cb = std::move(cb)] {
+        do_w_rdma(offs.data(), offs.size(), block_size, blks.data(), blks.size(), base_ptr);
+        cb();
+    });
+    return 0;
+}
+
+int ClientConn::do_r_rdma(const std::vector<std::pair<std::string, uint64_t>>& blocks,
+                          int block_size, uintptr_t base_ptr) {
+    bool dev = is_device_ptr(base_ptr);
+    size_t bs = static_cast<size_t>(block_size);
+    RemoteMetaMsg msg;
+    msg.block_size = block_size;
+    msg.op = OP_RDMA_READ;
+    for (auto& b : blocks) msg.keys.push_back(b.first);
+    auto body = build_remote_meta(msg);
+
+    std::vector<uint8_t> payload;
+    {
+        std::lock_guard<std::mutex> lk(io_mu_);
+        if (!send_req(OP_TCP_GET, body.data(), body.size())) return -1;
+        int code = 0;
+        if (!recv_status(&code)) return -1;
+        if (code != FINISH) {
+            WARN("read -> %d", code);
+            return -code;
+        }
+        if (!recv_payload(&payload)) return -1;
+    }
+    if (payload.size() < blocks.size() * bs) return -1;
+    for (size_t i = 0; i < blocks.size(); i++) {
+        void* dst = reinterpret_cast<void*>(base_ptr + blocks[i].second);
+        const void* src = payload.data() + i * bs;
+        if (dev) {
+            if (!gpu::memcpy_h2d(dst, src, bs)) return -1;
+        } else {
+            memcpy(dst, src, bs);
+        }
+    }
+    return 0;
+}
+
+int ClientConn::r_rdma(const std::vector<std::pair<std::string, uint64_t>>& blocks, int block_size,
+                       uintptr_t base_ptr) {
+    if (!rdma_connected_) return -1;
+    return do_r_rdma(blocks, block_size, base_ptr);
+}
+
+int ClientConn::r_rdma_async(const std::vector<std::pair<std::string, uint64_t>>& blocks,
+                             int block_size, uintptr_t base_ptr, std::function<void()> cb) {
+    if (!rdma_connected_) return -1;
+    inflight_.fetch_add(1);
+    enqueue([this, blocks, block_size, base_ptr, cb = std::move(cb)] {
+        do_r_rdma(blocks, block_size, base_ptr);
+        cb();
+    });
+    return 0;
+}
+
+int ClientConn::sync_rdma() {
+    std::unique_lock<std::mutex> lk(q_mu_);
+    bool ok = drain_cv_.wait_for(lk, std::chrono::seconds(10),
+                                 [this] { return inflight_.load() == 0; });
+    return ok ? 0 : -1;
+}
+
+// ---------------------------------------------------------------------------
+// queries
+// ---------------------------------------------------------------------------
+int ClientConn::check_exist(const std::string& key) {
+    if (!connected_) return -1;
+    std::vector<uint8_t> body(4 + key.size());
+    uint32_t len = static_cast<uint32_t>(key.size());
+    memcpy(body.data(), &len, 4);
+    memcpy(body.data() + 4, key.data(), key.size());
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!send_req(OP_CHECK_EXIST, body.data(), body.size())) return -1;
+    int code = -1;
+    if (!recv_status(&code)) return -1;
+    return code;  // 0 exists, 1 not
+}
+
+int ClientConn::get_match_last_index(const std::vector<std::string>& keys) {
+    if (!connected_) return -1;
+    auto body = build_match_request(keys);
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!send_req(OP_GET_MATCH_LAST_IDX, body.data(), body.size())) return -1;
+    int idx = -1;
+    if (!recv_status(&idx)) return -1;
+    return idx;
+}
+
+// ---------------------------------------------------------------------------
+// async worker
+// ---------------------------------------------------------------------------
+void ClientConn::enqueue(std::function<void()> fn) {
+    {
+        std::lock_guard<std::mutex> lk(q_mu_);
+        q_.push_back(std::move(fn));
+    }
+    q_cv_.notify_one();
+}
+
+void ClientConn::worker_main() {
+    for (;;) {
+        std::function<void()> fn;
+        {
+            std::unique_lock<std::mutex> lk(q_mu_);
+            q_cv_.wait(lk, [this] { return worker_stop_ || !q_.empty(); });
+            if (worker_stop_ && q_.empty()) return;
+            fn = std::move(q_.front());
+            q_.pop_front();
+        }
+        fn();
+        if (inflight_.fetch_sub(1) == 1) {
+            std::lock_guard<std::mutex> lk(q_mu_);
+            drain_cv_.notify_all();
+        }
+    }
+}
+
+}  // namespace ifs

@@ -1,0 +1,125 @@
+// infinistore-amd client connection.
+//
+// API parity with the reference's Connection (/root/reference/src/
+// libinfinistore.h:34-122): TCP control plane, local-GPU IPC path,
+// allocate/write/commit/read with RDMA semantics, async variants with
+// callbacks, sync. Data plane here is the TCP fabric (inline payloads) —
+// chosen at OP_RDMA_EXCHANGE time, so an ibverbs fabric can be negotiated
+// instead when rdma-core + a NIC exist. Unlike the reference, the commit is
+// ACKed by the server before the client's write callback fires, closing the
+// reference's sync-before-commit race (libinfinistore.cpp:403-410).
+#pragma once
+
+#include <atomic>
+#include <condition_variable>
+#include <cstdint>
+#include <deque>
+#include <functional>
+#include <mutex>
+#include <string>
+#include <thread>
+#include <utility>
+#include <vector>
+
+#include "../core/protocol.h"
+
+namespace ifs {
+
+struct ClientConfigC {
+    std::string host_addr;
+    int service_port = 0;
+    std::string connection_type = "RDMA";  // "RDMA" | "LOCAL_GPU"
+    std::string dev_name;                  // verbs fabric only
+    int ib_port = 1;
+    std::string link_type = "Ethernet";
+    std::string log_level = "warning";
+};
+
+struct RemoteBlockOut {
+    uint32_t rkey;
+    uint64_t remote_addr;
+};
+
+class ClientConn {
+   public:
+    ClientConn() = default;
+    ~ClientConn();
+    ClientConn(const ClientConn&) = delete;
+    ClientConn& operator=(const ClientConn&) = delete;
+
+    int init_connection(const ClientConfigC& cfg);  // blocking TCP connect
+    int setup_rdma(const ClientConfigC& cfg);       // fabric negotiation
+    void close_conn();
+
+    // ---- local (IPC) path ----
+    // blocks: (key, byte offset into the tensor). op: 'W' or 'R'.
+    int rw_local(char op, const std::vector<std::pair<std::string, uint64_t>>& blocks,
+                 int block_size, uintptr_t ptr, int device_id);
+    int sync_local();
+
+    // ---- RDMA-semantics path ----
+    int register_mr(uintptr_t ptr, size_t size);
+    std::vector<RemoteBlockOut> allocate_rdma(const std::vector<std::string>& keys,
+                                              int block_size);
+    int allocate_rdma_async(const std::vector<std::string>& keys, int block_size,
+                            std::function<void(std::vector<RemoteBlockOut>)> cb);
+    int w_rdma(const uint64_t* offsets, size_t n_offsets, int block_size,
+               const RemoteBlockOut* blocks, size_t n_blocks, uintptr_t base_ptr);
+    int w_rdma_async(const uint64_t* offsets, size_t n_offsets, int block_size,
+                     const RemoteBlockOut* blocks, size_t n_blocks, uintptr_t base_ptr,
+                     std::function<void()> cb);
+    int r_rdma(const std::vector<std::pair<std::string, uint64_t>>& blocks, int block_size,
+               uintptr_t base_ptr);
+    int r_rdma_async(const std::vector<std::pair<std::string, uint64_t>>& blocks, int block_size,
+                     uintptr_t base_ptr, std::function<void()> cb);
+    int sync_rdma();  // wait for all async ops to drain
+
+    // ---- queries ----
+    int check_exist(const std::string& key);
+    int get_match_last_index(const std::vector<std::string>& keys);
+
+    bool rdma_connected() const { return rdma_connected_; }
+
+   private:
+    // synchronous framed request/response (io_mu_ held)
+    bool send_req(char op, const uint8_t* body, size_t n);
+    bool recv_status(int* code);
+    bool recv_payload(std::vector<uint8_t>* out);  // u32 len + bytes
+
+    int do_w_rdma(const uint64_t* offsets, size_t n_offsets, int block_size,
+                  const RemoteBlockOut* blocks, size_t n_blocks, uintptr_t base_ptr);
+    int do_r_rdma(const std::vector<std::pair<std::string, uint64_t>>& blocks, int block_size,
+                  uintptr_t base_ptr);
+    std::vector<RemoteBlockOut> do_allocate(const std::vector<std::string>& keys, int block_size);
+
+    // Is [ptr, ptr+n) device memory? (classified at register_mr time; falls
+    // back to a hipPointerGetAttributes probe.)
+    bool is_device_ptr(uintptr_t ptr);
+
+    void worker_main();
+    void enqueue(std::function<void()> fn);
+
+    int fd_ = -1;
+    bool connected_ = false;
+    bool rdma_connected_ = false;
+    std::mutex io_mu_;
+
+    struct Region {
+        uintptr_t ptr;
+        size_t size;
+        bool device;
+    };
+    std::vector<Region> regions_;
+    std::mutex region_mu_;
+
+    std::thread worker_;
+    std::deque<std::function<void()>> q_;
+    std::mutex q_mu_;
+    std::condition_variable q_cv_;
+    std::condition_variable drain_cv_;
+    std::atomic<int> inflight_{0};
+    bool worker_stop_ = false;
+    bool worker_started_ = false;
+};
+
+}  // namespace ifs

@@ -1,0 +1,91 @@
+// HIP/ROCm layer for infinistore-amd (MI355X-native).
+//
+// Replaces the reference's CUDA-runtime call sites with an
+// MI355X-first design: the pool lives in hipMalloc'd HBM3E, per-block
+// hipMemcpyAsync loops are replaced by ONE batched gather/scatter HIP kernel
+// launch per request (csrc/gpu/kernels.hip), and completion runs on pooled
+// HIP streams with event FIFOs per shard.
+//
+// Everything here degrades gracefully when no GPU is present (CPU-only
+// server mode, BASELINE config 1): `available()` is false and callers fall
+// back to host arenas + memcpy.
+#pragma once
+
+#include <cstddef>
+#include <cstdint>
+#include <string>
+
+namespace ifs {
+namespace gpu {
+
+bool available();
+int device_count();
+std::string device_name(int dev);
+size_t device_total_mem(int dev);
+
+// --- memory ----------------------------------------------------------------
+void* alloc_device(int dev, size_t bytes);           // hipMalloc on device
+void free_device(void* p);
+void* alloc_host_pinned(size_t bytes);               // hipHostMalloc (falls back to aligned_alloc)
+void free_host_pinned(void* p);
+bool memcpy_d2h(void* dst, const void* src, size_t n);
+bool memcpy_h2d(void* dst, const void* src, size_t n);
+bool memcpy_d2d(void* dst, const void* src, size_t n);  // same or peer device
+// Async variants on a stream (hipMemcpyAsync).
+using Stream = void*;
+bool memcpy_h2d_async(void* dst, const void* src, size_t n, Stream s);
+bool memcpy_d2h_async(void* dst, const void* src, size_t n, Stream s);
+bool memcpy_any_async(void* dst, const void* src, size_t n, Stream s);
+
+// --- IPC -------------------------------------------------------------------
+constexpr size_t kIpcHandleSize = 64;  // == HIP_IPC_HANDLE_SIZE
+struct IpcHandle {
+    uint8_t bytes[kIpcHandleSize];
+};
+
+// Export `ptr`'s allocation; also reports the offset of ptr inside that
+// allocation so importers can reconstruct the exact address (works with the
+// PyTorch caching allocator, unlike the reference which needs
+// PYTORCH_NO_CUDA_MEMORY_CACHING).
+bool ipc_export(const void* ptr, IpcHandle* handle, uint64_t* base_offset);
+void* ipc_open(const IpcHandle& handle, int src_device);
+bool ipc_close(void* base);
+
+// Enable xGMI peer access dev -> peer (idempotent). Returns false if the
+// pair cannot peer.
+bool enable_peer_access(int dev, int peer);
+
+// True if ptr is device (HBM) memory — hipPointerGetAttributes probe, the
+// HIP analog of the reference's host-vs-device MR classification.
+bool is_device_pointer(const void* ptr);
+
+// --- streams / events (opaque wrappers so non-HIP TUs can hold them) -------
+using Event = void*;
+Stream stream_create(int dev);
+void stream_destroy(Stream s);
+bool stream_sync(Stream s);
+Event event_create(int dev);
+void event_destroy(Event e);
+bool event_record(Event e, Stream s);
+bool event_sync(Event e);      // blocking wait
+bool event_query(Event e);     // true if complete
+bool set_device(int dev);
+
+// --- batched block copy ----------------------------------------------------
+// One launch copies n_blocks of `bytes_per_block` from src_ptrs[i] to
+// dst_ptrs[i]. The pointer arrays must be in device-visible memory on the
+// launch device. Chooses the vectorized 16 B/lane kernel when every pointer
+// and the size are 16-byte aligned, else a byte-granular fallback.
+bool launch_copy_blocks(int dev, Stream s, const uint64_t* dev_src_ptrs,
+                        const uint64_t* dev_dst_ptrs, int n_blocks, size_t bytes_per_block,
+                        bool aligned16);
+
+// --- block fingerprint ------------------------------------------------------
+// 64-bit position-salted fingerprint per block -> out_hashes[i] (device mem).
+bool launch_hash_blocks(int dev, Stream s, const uint64_t* dev_ptrs, int n_blocks,
+                        size_t bytes_per_block, uint64_t* dev_out_hashes);
+
+const char* last_error();
+
+}  // namespace gpu
+}  // namespace ifs

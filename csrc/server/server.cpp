@@ -1,0 +1,733 @@
+#include "server.h"
+
+#include <algorithm>
+#include <cstring>
+
+#include "../core/log.h"
+
+namespace ifs {
+
+// ---------------------------------------------------------------------------
+// Connection state
+// ---------------------------------------------------------------------------
+struct Server::Conn : RefCounted {
+    Server* srv = nullptr;
+    uv_tcp_t tcp;
+    bool closed = false;
+
+    // read state machine
+    enum State { kHeader, kBody } state = kHeader;
+    Header hdr{};
+    std::vector<uint8_t> buf;  // accumulated bytes
+
+    // local path: in-flight async copy count (client polls OP_SYNC).
+    std::atomic<int> remain{0};
+
+    // cached IPC mappings: handle bytes -> base pointer (closed on disconnect)
+    std::map<std::vector<uint8_t>, std::pair<void*, int>> ipc_cache;  // base, src_dev
+
+    // TCP-fabric: blocks allocated for this conn, not yet committed.
+    std::unordered_map<uint64_t, Ref<BlockEntry>> pending_rdma;
+
+    ~Conn() override {
+        for (auto& kv : ipc_cache) {
+            if (gpu::available()) gpu::ipc_close(kv.second.first);
+        }
+    }
+};
+
+namespace {
+
+struct WriteReq {
+    uv_write_t req;
+    std::vector<uint8_t> data;
+};
+
+void send_buf(Server::Conn* c, std::vector<uint8_t> data);
+
+void conn_close(Server::Conn* c) {
+    if (c->closed) return;
+    c->closed = true;
+    uv_close(reinterpret_cast<uv_handle_t*>(&c->tcp), [](uv_handle_t* h) {
+        auto* c = static_cast<Server::Conn*>(h->data);
+        c->unref();
+    });
+}
+
+void send_status(Server::Conn* c, int code) {
+    std::vector<uint8_t> v(4);
+    memcpy(v.data(), &code, 4);
+    send_buf(c, std::move(v));
+}
+
+void send_status_payload(Server::Conn* c, int code, const uint8_t* payload, size_t n) {
+    std::vector<uint8_t> v(8 + n);
+    uint32_t len = static_cast<uint32_t>(n);
+    memcpy(v.data(), &code, 4);
+    memcpy(v.data() + 4, &len, 4);
+    if (n) memcpy(v.data() + 8, payload, n);
+    send_buf(c, std::move(v));
+}
+
+void send_buf(Server::Conn* c, std::vector<uint8_t> data) {
+    if (c->closed) return;
+    auto* wr = new WriteReq();
+    wr->data = std::move(data);
+    wr->req.data = wr;
+    uv_buf_t b = uv_buf_init(reinterpret_cast<char*>(wr->data.data()),
+                             static_cast<unsigned>(wr->data.size()));
+    int r = uv_write(&wr->req, reinterpret_cast<uv_stream_t*>(&c->tcp), &b, 1,
+                     [](uv_write_t* req, int status) {
+                         auto* wr = static_cast<WriteReq*>(req->data);
+                         delete wr;
+                         if (status < 0) DEBUG("uv_write status %d", status);
+                     });
+    if (r != 0) {
+        delete wr;
+        conn_close(c);
+    }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// Server lifecycle
+// ---------------------------------------------------------------------------
+Server::Server(const ServerOptions& opt) : opt_(opt) {
+    set_log_level(opt.log_level.c_str());
+    if (opt_.devices.empty()) {
+        ShardOptions so;
+        so.device = -1;
+        so.pool_bytes = opt_.prealloc_bytes;
+        so.block_granule = opt_.block_granule;
+        so.auto_extend = opt_.auto_extend;
+        so.extend_bytes = opt_.extend_bytes;
+        shards_.emplace_back(new Shard(so));
+    } else {
+        for (int dev : opt_.devices) {
+            ShardOptions so;
+            so.device = dev;
+            so.pool_bytes = opt_.prealloc_bytes;
+            so.block_granule = opt_.block_granule;
+            so.n_streams = opt_.n_streams;
+            so.auto_extend = opt_.auto_extend;
+            so.extend_bytes = opt_.extend_bytes;
+            shards_.emplace_back(new Shard(so));
+        }
+    }
+}
+
+Server::~Server() { stop(); }
+
+bool Server::start() {
+    if (running_.load()) return true;
+    for (auto& s : shards_) {
+        if (!s->init()) {
+            ERROR("shard init failed");
+            return false;
+        }
+    }
+    // Enable xGMI peer access between all shard devices up front (cross-GPU
+    // reads/writes run the copy kernel on the pool device against peer VAs).
+    if (gpu::available()) {
+        int n = gpu::device_count();
+        for (auto& s : shards_) {
+            if (!s->on_gpu()) continue;
+            for (int peer = 0; peer < n; peer++) {
+                if (peer != s->device()) gpu::enable_peer_access(s->device(), peer);
+            }
+        }
+    }
+
+    uv_loop_init(&loop_);
+    uv_async_init(&loop_, &post_async_, &Server::on_post_async);
+    post_async_.data = this;
+    uv_async_init(&loop_, &stop_async_, &Server::on_stop_async);
+    stop_async_.data = this;
+
+    uv_tcp_init(&loop_, &listener_);
+    listener_.data = this;
+    struct sockaddr_in addr;
+    uv_ip4_addr("0.0.0.0", opt_.service_port, &addr);
+    int r = uv_tcp_bind(&listener_, reinterpret_cast<const struct sockaddr*>(&addr), 0);
+    if (r == 0)
+        r = uv_listen(reinterpret_cast<uv_stream_t*>(&listener_), 512,
+                      &Server::on_new_connection);
+    if (r != 0) {
+        ERROR("bind/listen 0.0.0.0:%d failed: %s", opt_.service_port, uv_strerror(r));
+        uv_close(reinterpret_cast<uv_handle_t*>(&listener_), nullptr);
+        uv_close(reinterpret_cast<uv_handle_t*>(&post_async_), nullptr);
+        uv_close(reinterpret_cast<uv_handle_t*>(&stop_async_), nullptr);
+        uv_run(&loop_, UV_RUN_NOWAIT);
+        uv_loop_close(&loop_);
+        return false;
+    }
+    running_.store(true);
+    stop_requested_.store(false);
+    loop_thread_ = std::thread([this] { loop_main(); });
+    INFO("server listening on 0.0.0.0:%d with %zu shard(s)", opt_.service_port, shards_.size());
+    return true;
+}
+
+void Server::loop_main() {
+    uv_run(&loop_, UV_RUN_DEFAULT);
+    uv_loop_close(&loop_);
+    running_.store(false);
+}
+
+void Server::stop() {
+    if (!running_.load()) return;
+    stop_requested_.store(true);
+    uv_async_send(&stop_async_);
+    if (loop_thread_.joinable()) loop_thread_.join();
+    // Drop all stored blocks.
+    purge();
+}
+
+void Server::on_stop_async(uv_async_t* h) {
+    auto* srv = static_cast<Server*>(h->data);
+    // Close all connections, listener and asyncs; loop exits when no handles.
+    for (auto* c : srv->conns_) conn_close(c);
+    srv->conns_.clear();
+    uv_close(reinterpret_cast<uv_handle_t*>(&srv->listener_), nullptr);
+    uv_close(reinterpret_cast<uv_handle_t*>(&srv->post_async_), nullptr);
+    uv_close(reinterpret_cast<uv_handle_t*>(&srv->stop_async_), nullptr);
+}
+
+void Server::post(std::function<void()> fn) {
+    {
+        std::lock_guard<std::mutex> lk(post_mu_);
+        posted_.push_back(std::move(fn));
+    }
+    uv_async_send(&post_async_);
+}
+
+void Server::on_post_async(uv_async_t* h) {
+    auto* srv = static_cast<Server*>(h->data);
+    std::vector<std::function<void()>> fns;
+    {
+        std::lock_guard<std::mutex> lk(srv->post_mu_);
+        fns.swap(srv->posted_);
+    }
+    for (auto& f : fns) f();
+}
+
+// ---------------------------------------------------------------------------
+// Accept + read state machine
+// ---------------------------------------------------------------------------
+void Server::on_new_connection(uv_stream_t* server, int status) {
+    auto* srv = static_cast<Server*>(server->data);
+    if (status < 0) {
+        WARN("accept error: %s", uv_strerror(status));
+        return;
+    }
+    auto* c = new Conn();
+    c->srv = srv;
+    uv_tcp_init(&srv->loop_, &c->tcp);
+    c->tcp.data = c;
+    if (uv_accept(server, reinterpret_cast<uv_stream_t*>(&c->tcp)) != 0) {
+        conn_close(c);
+        return;
+    }
+    uv_tcp_nodelay(&c->tcp, 1);
+    srv->conns_.push_back(c);
+    uv_read_start(
+        reinterpret_cast<uv_stream_t*>(&c->tcp),
+        [](uv_handle_t*, size_t suggested, uv_buf_t* buf) {
+            buf->base = static_cast<char*>(malloc(suggested));
+            buf->len = suggested;
+        },
+        [](uv_stream_t* stream, ssize_t nread, const uv_buf_t* buf) {
+            auto* c = static_cast<Conn*>(stream->data);
+            Server* srv = c->srv;
+            if (nread < 0) {
+                free(buf->base);
+                auto& v = srv->conns_;
+                v.erase(std::remove(v.begin(), v.end(), c), v.end());
+                conn_close(c);
+                return;
+            }
+            if (nread == 0) {
+                free(buf->base);
+                return;
+            }
+            c->buf.insert(c->buf.end(), reinterpret_cast<uint8_t*>(buf->base),
+                          reinterpret_cast<uint8_t*>(buf->base) + nread);
+            free(buf->base);
+            // Drain as many complete requests as are buffered.
+            size_t consumed = 0;
+            for (;;) {
+                if (c->state == Conn::kHeader) {
+                    if (c->buf.size() - consumed < sizeof(Header)) break;
+                    memcpy(&c->hdr, c->buf.data() + consumed, sizeof(Header));
+                    consumed += sizeof(Header);
+                    if (c->hdr.magic != kMagic) {
+                        WARN("bad magic from client; closing");
+                        auto& v = srv->conns_;
+                        v.erase(std::remove(v.begin(), v.end(), c), v.end());
+                        conn_close(c);
+                        return;
+                    }
+                    size_t cap = (c->hdr.op == OP_TCP_PUT) ? (256u << 20) : kProtocolBufferSize;
+                    if (c->hdr.body_size > cap) {
+                        WARN("body too large (%u) for op %c", c->hdr.body_size, c->hdr.op);
+                        auto& v = srv->conns_;
+                        v.erase(std::remove(v.begin(), v.end(), c), v.end());
+                        conn_close(c);
+                        return;
+                    }
+                    c->state = Conn::kBody;
+                }
+                if (c->state == Conn::kBody) {
+                    if (c->buf.size() - consumed < c->hdr.body_size) break;
+                    std::vector<uint8_t> body(c->buf.data() + consumed,
+                                              c->buf.data() + consumed + c->hdr.body_size);
+                    consumed += c->hdr.body_size;
+                    c->state = Conn::kHeader;
+                    srv->handle_request(c, c->hdr.op, std::move(body));
+                    if (c->closed) return;
+                }
+            }
+            if (consumed) c->buf.erase(c->buf.begin(), c->buf.begin() + consumed);
+        });
+}
+
+// ---------------------------------------------------------------------------
+// Dispatch
+// ---------------------------------------------------------------------------
+void Server::handle_request(Conn* c, char op, std::vector<uint8_t> body) {
+    DEBUG("request op=%s body=%zu", op_name(op).c_str(), body.size());
+    switch (op) {
+        case OP_W: {
+            LocalMetaMsg msg;
+            if (!parse_local_meta(body.data(), body.size(), &msg)) return send_status(c, INVALID_REQ);
+            return op_local_write(c, msg);
+        }
+        case OP_R: {
+            LocalMetaMsg msg;
+            if (!parse_local_meta(body.data(), body.size(), &msg)) return send_status(c, INVALID_REQ);
+            return op_local_read(c, msg);
+        }
+        case OP_SYNC:
+            return op_sync(c);
+        case OP_RDMA_EXCHANGE:
+            return op_exchange(c, body);
+        case OP_RDMA_ALLOCATE: {
+            RemoteMetaMsg msg;
+            if (!parse_remote_meta(body.data(), body.size(), &msg))
+                return send_status(c, INVALID_REQ);
+            return op_allocate(c, msg);
+        }
+        case OP_TCP_PUT:
+            return op_tcp_put(c, body);
+        case OP_TCP_GET: {
+            RemoteMetaMsg msg;
+            if (!parse_remote_meta(body.data(), body.size(), &msg))
+                return send_status(c, INVALID_REQ);
+            return op_tcp_get(c, msg);
+        }
+        case OP_RDMA_WRITE_COMMIT: {
+            RemoteMetaMsg msg;
+            if (!parse_remote_meta(body.data(), body.size(), &msg))
+                return send_status(c, INVALID_REQ);
+            return op_commit(c, msg);
+        }
+        case OP_CHECK_EXIST:
+            return op_check_exist(c, body);
+        case OP_GET_MATCH_LAST_IDX:
+            return op_match_index(c, body);
+        default:
+            WARN("unknown op '%c'", op);
+            return send_status(c, INVALID_REQ);
+    }
+}
+
+Shard* Server::shard_for_device(int device) {
+    for (auto& s : shards_)
+        if (s->device() == device) return s.get();
+    return shards_[static_cast<size_t>(device < 0 ? 0 : device) % shards_.size()].get();
+}
+
+Shard* Server::shard_least_used() {
+    Shard* best = shards_[0].get();
+    size_t best_used = SIZE_MAX;
+    for (auto& s : shards_) {
+        size_t total = s->total_blocks();
+        size_t used = s->used_blocks();
+        size_t free_blocks = total - used;
+        (void)free_blocks;
+        if (used < best_used) {
+            best_used = used;
+            best = s.get();
+        }
+    }
+    return best;
+}
+
+void Server::maybe_extend(Shard* s) {
+    if (!s->need_extend()) return;
+    int expect = 0;
+    if (!extending_.compare_exchange_strong(expect, 1)) return;
+    std::thread([this, s] {
+        INFO("extending pool on shard dev=%d", s->device());
+        s->extend();
+        extending_.store(0);
+    }).detach();
+}
+
+// ---- local (IPC) path -----------------------------------------------------
+void Server::op_local_write(Conn* c, const LocalMetaMsg& msg) {
+    if (!gpu::available()) return send_status(c, SYSTEM_ERROR);
+    if (msg.ipc_handle.size() != gpu::kIpcHandleSize || msg.block_size <= 0)
+        return send_status(c, INVALID_REQ);
+
+    // Resolve (cached) client mapping.
+    void* base = nullptr;
+    auto it = c->ipc_cache.find(msg.ipc_handle);
+    if (it != c->ipc_cache.end()) {
+        base = it->second.first;
+    } else {
+        gpu::IpcHandle h;
+        memcpy(h.bytes, msg.ipc_handle.data(), gpu::kIpcHandleSize);
+        base = gpu::ipc_open(h, msg.device);
+        if (!base) return send_status(c, INTERNAL_ERROR);
+        c->ipc_cache.emplace(msg.ipc_handle, std::make_pair(base, msg.device));
+    }
+    uint8_t* client_ptr = static_cast<uint8_t*>(base) + msg.base_offset;
+
+    Shard* shard = shard_for_device(msg.device);
+    size_t page = static_cast<size_t>(msg.block_size);
+
+    Shard::CopyJob job;
+    job.bytes_per_block = page;
+    std::vector<Ref<BlockEntry>> new_entries;
+    std::vector<std::string> new_keys;
+    {
+        std::lock_guard<std::mutex> lk(kv_mu_);
+        for (auto& b : msg.blocks) {
+            if (kv_.count(b.key)) continue;  // dedup: first write wins
+            void* dst = nullptr;
+            int pool_idx = -1;
+            bool ok = shard->allocate(page, 1, [&](void* p, int idx) {
+                dst = p;
+                pool_idx = idx;
+            });
+            if (!ok) {
+                for (auto& k : new_keys) kv_.erase(k);  // roll back this request
+                return send_status(c, OUT_OF_MEMORY);
+            }
+            auto* e = new BlockEntry();
+            e->ptr = dst;
+            e->size = page;
+            e->pool_idx = pool_idx;
+            e->shard = shard;
+            e->committed = false;
+            Ref<BlockEntry> ref(e);
+            kv_.emplace(b.key, ref);
+            new_entries.push_back(ref);
+            new_keys.push_back(b.key);
+            job.src.push_back(reinterpret_cast<uint64_t>(client_ptr + b.offset));
+            job.dst.push_back(reinterpret_cast<uint64_t>(dst));
+        }
+    }
+
+    n_writes_.fetch_add(1);
+    bytes_in_.fetch_add(job.src.size() * page);
+    maybe_extend(shard);
+
+    if (job.src.empty()) {
+        // everything was a duplicate — nothing to copy
+        return send_status(c, TASK_ACCEPTED);
+    }
+
+    c->remain.fetch_add(1);
+    c->ref();
+    auto entries = std::make_shared<std::vector<Ref<BlockEntry>>>(std::move(new_entries));
+    auto keys = std::make_shared<std::vector<std::string>>(std::move(new_keys));
+    job.done = [this, c, entries, keys](bool ok) {
+        post([this, c, entries, keys, ok] {
+            if (ok) {
+                for (auto& e : *entries) e->committed = true;
+            } else {
+                std::lock_guard<std::mutex> lk(kv_mu_);
+                for (size_t i = 0; i < keys->size(); i++) {
+                    auto it = kv_.find((*keys)[i]);
+                    if (it != kv_.end() && it->second.get() == (*entries)[i].get()) kv_.erase(it);
+                }
+            }
+            c->remain.fetch_sub(1);
+            c->unref();
+        });
+    };
+    if (!shard->submit_copy(std::move(job))) {
+        c->remain.fetch_sub(1);
+        c->unref();
+        return send_status(c, INTERNAL_ERROR);
+    }
+    send_status(c, TASK_ACCEPTED);
+}
+
+void Server::op_local_read(Conn* c, const LocalMetaMsg& msg) {
+    if (!gpu::available()) return send_status(c, SYSTEM_ERROR);
+    if (msg.ipc_handle.size() != gpu::kIpcHandleSize || msg.block_size <= 0)
+        return send_status(c, INVALID_REQ);
+
+    void* base = nullptr;
+    auto it = c->ipc_cache.find(msg.ipc_handle);
+    if (it != c->ipc_cache.end()) {
+        base = it->second.first;
+    } else {
+        gpu::IpcHandle h;
+        memcpy(h.bytes, msg.ipc_handle.data(), gpu::kIpcHandleSize);
+        base = gpu::ipc_open(h, msg.device);
+        if (!base) return send_status(c, INTERNAL_ERROR);
+        c->ipc_cache.emplace(msg.ipc_handle, std::make_pair(base, msg.device));
+    }
+    uint8_t* client_ptr = static_cast<uint8_t*>(base) + msg.base_offset;
+    size_t page = static_cast<size_t>(msg.block_size);
+
+    // Group blocks by owning shard (keys may live on different GPUs).
+    std::map<Shard*, Shard::CopyJob> jobs;
+    auto held = std::make_shared<std::vector<Ref<BlockEntry>>>();
+    {
+        std::lock_guard<std::mutex> lk(kv_mu_);
+        for (auto& b : msg.blocks) {
+            auto kit = kv_.find(b.key);
+            if (kit == kv_.end() || !kit->second->committed) {
+                return send_status(c, KEY_NOT_FOUND);
+            }
+            BlockEntry* e = kit->second.get();
+            auto& job = jobs[e->shard];
+            job.bytes_per_block = page;
+            job.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
+            job.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.offset));
+            held->push_back(kit->second);
+        }
+    }
+    n_reads_.fetch_add(1);
+    bytes_out_.fetch_add(msg.blocks.size() * page);
+    if (jobs.empty()) return send_status(c, TASK_ACCEPTED);
+
+    c->remain.fetch_add(1);
+    c->ref();
+    auto pending = std::make_shared<std::atomic<int>>(static_cast<int>(jobs.size()));
+    for (auto& [shard, job] : jobs) {
+        Shard::CopyJob j = std::move(job);
+        j.done = [this, c, held, pending](bool ok) {
+            (void)ok;
+            if (pending->fetch_sub(1) == 1) {
+                post([this, c, held] {
+                    c->remain.fetch_sub(1);
+                    c->unref();
+                });
+            }
+        };
+        if (!shard->submit_copy(std::move(j))) {
+            if (pending->fetch_sub(1) == 1) {
+                c->remain.fetch_sub(1);
+                c->unref();
+                return send_status(c, INTERNAL_ERROR);
+            }
+        }
+    }
+    send_status(c, TASK_ACCEPTED);
+}
+
+void Server::op_sync(Conn* c) { send_status(c, c->remain.load()); }
+
+// ---- TCP fabric (RDMA-semantics) -----------------------------------------
+void Server::op_exchange(Conn* c, const std::vector<uint8_t>& body) {
+    (void)body;
+    // Fabric negotiation: this build serves the TCP data fabric ("TCPF").
+    // A verbs fabric would parse rdma_conn_info_t here instead.
+    const char tag[4] = {'T', 'C', 'P', 'F'};
+    send_status_payload(c, FINISH, reinterpret_cast<const uint8_t*>(tag), 4);
+}
+
+void Server::op_allocate(Conn* c, const RemoteMetaMsg& msg) {
+    if (msg.block_size <= 0 || msg.keys.empty()) return send_status(c, INVALID_REQ);
+    size_t page = static_cast<size_t>(msg.block_size);
+    Shard* shard = shard_least_used();
+    std::vector<RemoteBlockWire> blocks;
+    blocks.reserve(msg.keys.size());
+    std::vector<std::pair<std::string, Ref<BlockEntry>>> created;
+    {
+        std::lock_guard<std::mutex> lk(kv_mu_);
+        for (auto& key : msg.keys) {
+            if (kv_.count(key)) {
+                blocks.push_back({0, 0, 0});  // FAKE block: dup key, client skips
+                continue;
+            }
+            void* ptr = nullptr;
+            int pool_idx = -1;
+            bool ok = shard->allocate(page, 1, [&](void* p, int idx) {
+                ptr = p;
+                pool_idx = idx;
+            });
+            if (!ok) {
+                for (auto& ckv : created) kv_.erase(ckv.first);
+                return send_status(c, OUT_OF_MEMORY);
+            }
+            auto* e = new BlockEntry();
+            e->ptr = ptr;
+            e->size = page;
+            e->pool_idx = pool_idx;
+            e->shard = shard;
+            Ref<BlockEntry> ref(e);
+            kv_.emplace(key, ref);
+            created.push_back({key, ref});
+            c->pending_rdma.emplace(reinterpret_cast<uint64_t>(ptr), ref);
+            blocks.push_back({static_cast<uint32_t>(shard->device() + 1), 0,
+                              reinterpret_cast<uint64_t>(ptr)});
+        }
+    }
+    maybe_extend(shard);
+    auto payload = build_allocate_response(blocks);
+    send_status_payload(c, FINISH, payload.data(), payload.size());
+}
+
+void Server::op_tcp_put(Conn* c, const std::vector<uint8_t>& body) {
+    // Layout: [u32 n][u32 block_size][u64 addr x n][payload n*block_size]
+    if (body.size() < 8) return send_status(c, INVALID_REQ);
+    uint32_t n, bs;
+    memcpy(&n, body.data(), 4);
+    memcpy(&bs, body.data() + 4, 4);
+    size_t need = 8 + static_cast<size_t>(n) * 8 + static_cast<size_t>(n) * bs;
+    if (n == 0 || bs == 0 || body.size() < need) return send_status(c, INVALID_REQ);
+    const uint64_t* addrs = reinterpret_cast<const uint64_t*>(body.data() + 8);
+    const uint8_t* payload = body.data() + 8 + static_cast<size_t>(n) * 8;
+
+    for (uint32_t i = 0; i < n; i++) {
+        auto it = c->pending_rdma.find(addrs[i]);
+        if (it == c->pending_rdma.end()) {
+            WARN("tcp_put: unknown addr %llx", (unsigned long long)addrs[i]);
+            return send_status(c, INVALID_REQ);
+        }
+        BlockEntry* e = it->second.get();
+        if (e->size < bs) return send_status(c, INVALID_REQ);
+        const uint8_t* src = payload + static_cast<size_t>(i) * bs;
+        if (e->shard->on_gpu()) {
+            if (!gpu::memcpy_h2d(e->ptr, src, bs)) return send_status(c, INTERNAL_ERROR);
+        } else {
+            memcpy(e->ptr, src, bs);
+        }
+    }
+    n_put_.fetch_add(1);
+    bytes_in_.fetch_add(static_cast<size_t>(n) * bs);
+    send_status(c, TASK_ACCEPTED);
+}
+
+void Server::op_commit(Conn* c, const RemoteMetaMsg& msg) {
+    for (uint64_t addr : msg.remote_addrs) {
+        auto it = c->pending_rdma.find(addr);
+        if (it == c->pending_rdma.end()) continue;
+        it->second->committed = true;
+        c->pending_rdma.erase(it);
+    }
+    send_status(c, FINISH);
+}
+
+void Server::op_tcp_get(Conn* c, const RemoteMetaMsg& msg) {
+    if (msg.block_size <= 0 || msg.keys.empty()) return send_status(c, INVALID_REQ);
+    size_t page = static_cast<size_t>(msg.block_size);
+    std::vector<Ref<BlockEntry>> entries;
+    {
+        std::lock_guard<std::mutex> lk(kv_mu_);
+        for (auto& key : msg.keys) {
+            auto it = kv_.find(key);
+            if (it == kv_.end() || !it->second->committed) return send_status(c, KEY_NOT_FOUND);
+            entries.push_back(it->second);
+        }
+    }
+    std::vector<uint8_t> payload(entries.size() * page);
+    for (size_t i = 0; i < entries.size(); i++) {
+        BlockEntry* e = entries[i].get();
+        size_t nbytes = std::min(page, e->size);
+        if (e->shard->on_gpu()) {
+            if (!gpu::memcpy_d2h(payload.data() + i * page, e->ptr, nbytes))
+                return send_status(c, INTERNAL_ERROR);
+        } else {
+            memcpy(payload.data() + i * page, e->ptr, nbytes);
+        }
+    }
+    n_get_.fetch_add(1);
+    bytes_out_.fetch_add(payload.size());
+    send_status_payload(c, FINISH, payload.data(), payload.size());
+}
+
+// ---- queries ---------------------------------------------------------------
+void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body) {
+    // body: u32 len + key bytes (simple framing; no flatbuffer needed)
+    if (body.size() < 4) return send_status(c, INVALID_REQ);
+    uint32_t len;
+    memcpy(&len, body.data(), 4);
+    if (body.size() < 4 + len) return send_status(c, INVALID_REQ);
+    std::string key(reinterpret_cast<const char*>(body.data() + 4), len);
+    bool exists;
+    {
+        std::lock_guard<std::mutex> lk(kv_mu_);
+        auto it = kv_.find(key);
+        exists = it != kv_.end() && it->second->committed;
+    }
+    send_status(c, exists ? 0 : 1);
+}
+
+void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body) {
+    std::vector<std::string> keys;
+    if (!parse_match_request(body.data(), body.size(), &keys) || keys.empty())
+        return send_status(c, -1);
+    // Binary search for the last present index, assuming the prefix property
+    // (keys[0..i] present iff i <= match). Requires committed entries —
+    // divergence from the reference, which counts uncommitted keys as
+    // present (infinistore.cpp:1097); an uncommitted key cannot be read, so
+    // reporting it as a hit would make the subsequent read_cache fail.
+    std::lock_guard<std::mutex> lk(kv_mu_);
+    auto present = [&](size_t i) {
+        auto it = kv_.find(keys[i]);
+        return it != kv_.end() && it->second->committed;
+    };
+    long left = 0, right = static_cast<long>(keys.size());
+    while (left < right) {
+        long mid = (left + right) / 2;
+        if (present(static_cast<size_t>(mid)))
+            left = mid + 1;
+        else
+            right = mid;
+    }
+    send_status(c, static_cast<int>(left - 1));
+}
+
+// ---------------------------------------------------------------------------
+// Management plane
+// ---------------------------------------------------------------------------
+size_t Server::kvmap_len() {
+    std::lock_guard<std::mutex> lk(kv_mu_);
+    return kv_.size();
+}
+
+size_t Server::purge() {
+    std::lock_guard<std::mutex> lk(kv_mu_);
+    size_t n = kv_.size();
+    kv_.clear();
+    return n;
+}
+
+std::string Server::stats_json() {
+    char buf[1024];
+    size_t used = 0, total = 0;
+    for (auto& s : shards_) {
+        used += s->used_blocks();
+        total += s->total_blocks();
+    }
+    snprintf(buf, sizeof(buf),
+             "{\"kv_len\": %zu, \"shards\": %zu, \"used_blocks\": %zu, \"total_blocks\": %zu, "
+             "\"writes\": %llu, \"reads\": %llu, \"puts\": %llu, \"gets\": %llu, "
+             "\"bytes_in\": %llu, \"bytes_out\": %llu}",
+             kvmap_len(), shards_.size(), used, total,
+             (unsigned long long)n_writes_.load(), (unsigned long long)n_reads_.load(),
+             (unsigned long long)n_put_.load(), (unsigned long long)n_get_.load(),
+             (unsigned long long)bytes_in_.load(), (unsigned long long)bytes_out_.load());
+    return buf;
+}
+
+}  // namespace ifs

@@ -1,0 +1,231 @@
+#include "shard.h"
+
+#include <algorithm>
+#include <cstdlib>
+#include <cstring>
+
+#include "../core/log.h"
+
+namespace ifs {
+
+Shard::Shard(const ShardOptions& opt) : opt_(opt) {}
+
+Shard::~Shard() {
+    {
+        std::lock_guard<std::mutex> lk(task_mu_);
+        stopping_ = true;
+    }
+    task_cv_.notify_all();
+    if (completion_thread_.joinable()) completion_thread_.join();
+    for (auto& sc : streams_) {
+        for (auto& sl : sc.slots) {
+            if (sl.event) gpu::event_destroy(sl.event);
+            if (sl.h_src) gpu::free_host_pinned(sl.h_src);
+            if (sl.h_dst) gpu::free_host_pinned(sl.h_dst);
+            if (sl.d_src) gpu::free_device(sl.d_src);
+            if (sl.d_dst) gpu::free_device(sl.d_dst);
+        }
+        if (sc.stream) gpu::stream_destroy(sc.stream);
+    }
+    // MM frees arenas via its free_fns.
+}
+
+bool Shard::init() {
+    if (inited_) return true;
+    size_t desc_bytes = opt_.max_descs_per_slot * sizeof(uint64_t);
+    if (on_gpu()) {
+        void* arena = gpu::alloc_device(opt_.device, opt_.pool_bytes);
+        if (!arena) {
+            ERROR("shard %d: hipMalloc %zu MB failed", opt_.device, opt_.pool_bytes >> 20);
+            return false;
+        }
+        {
+            std::lock_guard<std::mutex> lk(alloc_mu_);
+            mm_.add_pool(arena, opt_.pool_bytes, opt_.block_granule,
+                         [](void* p, size_t) { gpu::free_device(p); });
+        }
+        streams_.resize(opt_.n_streams);
+        for (auto& sc : streams_) {
+            sc.stream = gpu::stream_create(opt_.device);
+            if (!sc.stream) return false;
+            sc.slots.resize(opt_.slots_per_stream);
+            for (auto& sl : sc.slots) {
+                sl.h_src = static_cast<uint64_t*>(gpu::alloc_host_pinned(desc_bytes));
+                sl.h_dst = static_cast<uint64_t*>(gpu::alloc_host_pinned(desc_bytes));
+                sl.d_src = static_cast<uint64_t*>(gpu::alloc_device(opt_.device, desc_bytes));
+                sl.d_dst = static_cast<uint64_t*>(gpu::alloc_device(opt_.device, desc_bytes));
+                sl.event = gpu::event_create(opt_.device);
+                if (!sl.h_src || !sl.h_dst || !sl.d_src || !sl.d_dst || !sl.event) return false;
+            }
+        }
+        completion_thread_ = std::thread([this] { completion_loop(); });
+    } else {
+        void* arena = nullptr;
+        if (posix_memalign(&arena, 4096, opt_.pool_bytes) != 0) {
+            ERROR("cpu shard: alloc %zu MB failed", opt_.pool_bytes >> 20);
+            return false;
+        }
+        std::lock_guard<std::mutex> lk(alloc_mu_);
+        mm_.add_pool(arena, opt_.pool_bytes, opt_.block_granule, [](void* p, size_t) { free(p); });
+    }
+    inited_ = true;
+    return true;
+}
+
+bool Shard::allocate(size_t size, size_t n, const AllocationCallback& cb) {
+    std::lock_guard<std::mutex> lk(alloc_mu_);
+    return mm_.allocate(size, n, cb);
+}
+
+bool Shard::deallocate(void* ptr, size_t size, int pool_idx) {
+    std::lock_guard<std::mutex> lk(alloc_mu_);
+    return mm_.deallocate(ptr, size, pool_idx);
+}
+
+size_t Shard::used_blocks() {
+    std::lock_guard<std::mutex> lk(alloc_mu_);
+    return mm_.used_blocks();
+}
+
+size_t Shard::total_blocks() {
+    std::lock_guard<std::mutex> lk(alloc_mu_);
+    return mm_.total_blocks();
+}
+
+bool Shard::contains(const void* p) {
+    std::lock_guard<std::mutex> lk(alloc_mu_);
+    for (size_t i = 0; i < mm_.num_pools(); i++)
+        if (mm_.pool(static_cast<int>(i))->contains(const_cast<void*>(p))) return true;
+    return false;
+}
+
+bool Shard::need_extend() {
+    std::lock_guard<std::mutex> lk(alloc_mu_);
+    return opt_.auto_extend && mm_.need_extend();
+}
+
+bool Shard::extend() {
+    if (on_gpu()) {
+        void* arena = gpu::alloc_device(opt_.device, opt_.extend_bytes);
+        if (!arena) return false;
+        std::lock_guard<std::mutex> lk(alloc_mu_);
+        mm_.add_pool(arena, opt_.extend_bytes, opt_.block_granule,
+                     [](void* p, size_t) { gpu::free_device(p); });
+    } else {
+        void* arena = nullptr;
+        if (posix_memalign(&arena, 4096, opt_.extend_bytes) != 0) return false;
+        std::lock_guard<std::mutex> lk(alloc_mu_);
+        mm_.add_pool(arena, opt_.extend_bytes, opt_.block_granule,
+                     [](void* p, size_t) { free(p); });
+    }
+    return true;
+}
+
+Shard::Slot* Shard::acquire_slot(StreamCtx& sc) {
+    std::unique_lock<std::mutex> lk(task_mu_);
+    for (;;) {
+        for (auto& sl : sc.slots) {
+            if (!sl.busy) {
+                sl.busy = true;
+                return &sl;
+            }
+        }
+        slot_cv_.wait(lk);
+        if (stopping_) return nullptr;
+    }
+}
+
+bool Shard::submit_copy(CopyJob&& job) {
+    size_t n = job.src.size();
+    if (n != job.dst.size() || job.bytes_per_block == 0) return false;
+    if (n == 0) {
+        if (job.done) job.done(true);
+        return true;
+    }
+
+    if (!on_gpu()) {
+        // CPU shard: copies run inline on the caller (loop) thread.
+        for (size_t i = 0; i < n; i++)
+            memcpy(reinterpret_cast<void*>(job.dst[i]), reinterpret_cast<const void*>(job.src[i]),
+                   job.bytes_per_block);
+        if (job.done) job.done(true);
+        return true;
+    }
+
+    // Alignment check for the vectorized kernel.
+    bool aligned = job.bytes_per_block % 16 == 0;
+    if (aligned) {
+        for (size_t i = 0; i < n && aligned; i++)
+            aligned = (job.src[i] % 16 == 0) && (job.dst[i] % 16 == 0);
+    }
+
+    std::lock_guard<std::mutex> submit_lk(submit_mu_);
+    StreamCtx& sc = streams_[next_stream_];
+    next_stream_ = (next_stream_ + 1) % static_cast<int>(streams_.size());
+
+    // Chunk over slot capacity; the done callback fires after the final chunk
+    // (chunks on one stream complete in order).
+    size_t cap = opt_.max_descs_per_slot;
+    size_t off = 0;
+    while (off < n) {
+        size_t take = std::min(cap, n - off);
+        Slot* slot = acquire_slot(sc);
+        if (!slot) return false;  // shutting down
+        memcpy(slot->h_src, job.src.data() + off, take * sizeof(uint64_t));
+        memcpy(slot->h_dst, job.dst.data() + off, take * sizeof(uint64_t));
+        bool last = off + take >= n;
+        bool ok = gpu::set_device(opt_.device);
+        // Async H2D of the descriptor arrays, then one kernel launch.
+        ok = ok && gpu::memcpy_h2d_async(slot->d_src, slot->h_src, take * sizeof(uint64_t),
+                                         sc.stream);
+        ok = ok && gpu::memcpy_h2d_async(slot->d_dst, slot->h_dst, take * sizeof(uint64_t),
+                                         sc.stream);
+        ok = ok && gpu::launch_copy_blocks(opt_.device, sc.stream, slot->d_src, slot->d_dst,
+                                           static_cast<int>(take), job.bytes_per_block, aligned);
+        ok = ok && gpu::event_record(slot->event, sc.stream);
+        if (!ok) {
+            // Report the failure through the done callback (even if earlier
+            // chunks were already queued) and stop submitting.
+            ERROR("shard %d submit failed: %s", opt_.device, gpu::last_error());
+            {
+                std::lock_guard<std::mutex> lk(task_mu_);
+                slot->busy = false;
+                tasks_.push_back({&sc, nullptr, std::move(job.done)});
+            }
+            task_cv_.notify_one();
+            return true;
+        }
+        {
+            std::lock_guard<std::mutex> lk(task_mu_);
+            tasks_.push_back(
+                {&sc, slot, last ? std::move(job.done) : std::function<void(bool)>()});
+        }
+        task_cv_.notify_one();
+        off += take;
+    }
+    return true;
+}
+
+void Shard::completion_loop() {
+    gpu::set_device(opt_.device);
+    for (;;) {
+        PendingTask t;
+        {
+            std::unique_lock<std::mutex> lk(task_mu_);
+            task_cv_.wait(lk, [this] { return stopping_ || !tasks_.empty(); });
+            if (stopping_ && tasks_.empty()) return;
+            t = std::move(tasks_.front());
+            tasks_.pop_front();
+        }
+        bool ok = t.slot != nullptr;
+        if (t.slot) {
+            ok = gpu::event_sync(t.slot->event);
+            std::lock_guard<std::mutex> lk(task_mu_);
+            t.slot->busy = false;
+        }
+        slot_cv_.notify_all();
+        if (t.done) t.done(ok);
+    }
+}
+
+}  // namespace ifs

@@ -1,0 +1,113 @@
+// Per-GPU pool shard for infinistore-amd.
+//
+// The reference drives one pinned-host-DRAM pool from a single libuv thread
+// (/root/reference/src/infinistore.cpp:1, mempool.cpp:29-44). This build
+// shards the pool across MI355X GPUs: each shard owns an HBM3E arena
+// (hipMalloc), a bitmap allocator, a small pool of HIP streams with
+// descriptor-staging slots, and a completion thread that waits on HIP events
+// FIFO and hands results back to the event loop (SURVEY.md §7 hard part 4).
+// In CPU mode (no GPU) a shard is a host-DRAM arena and copies run inline.
+#pragma once
+
+#include <condition_variable>
+#include <cstdint>
+#include <deque>
+#include <functional>
+#include <mutex>
+#include <thread>
+#include <vector>
+
+#include "../core/mempool.h"
+#include "../gpu/gpu.h"
+
+namespace ifs {
+
+struct ShardOptions {
+    int device = -1;          // GPU ordinal; -1 = CPU shard
+    size_t pool_bytes = 0;    // initial arena size
+    size_t block_granule = 64 << 10;  // bitmap granule (minimal_allocate_size)
+    int n_streams = 4;
+    int slots_per_stream = 4;
+    size_t max_descs_per_slot = 65536;  // 64K blocks -> 1 MiB of u64 ptrs/side
+    bool auto_extend = false;
+    size_t extend_bytes = 10ull << 30;
+};
+
+class Shard {
+   public:
+    // Descriptor list: uniform-size block copies src[i] -> dst[i].
+    struct CopyJob {
+        std::vector<uint64_t> src;
+        std::vector<uint64_t> dst;
+        size_t bytes_per_block = 0;
+        // Completion callback; invoked exactly once from the shard completion
+        // thread (GPU) or inline (CPU shard). ok=false means the copy failed.
+        std::function<void(bool ok)> done;
+    };
+
+    explicit Shard(const ShardOptions& opt);
+    ~Shard();
+    Shard(const Shard&) = delete;
+    Shard& operator=(const Shard&) = delete;
+
+    bool init();  // allocate arena, streams, start completion thread
+
+    int device() const { return opt_.device; }
+    bool on_gpu() const { return opt_.device >= 0; }
+
+    // Allocator (thread-safe).
+    bool allocate(size_t size, size_t n, const AllocationCallback& cb);
+    bool deallocate(void* ptr, size_t size, int pool_idx);
+    size_t used_blocks();
+    size_t total_blocks();
+    bool contains(const void* p);
+
+    // Submit a batched copy; returns false if the job could not be launched
+    // (done() is NOT called in that case).
+    bool submit_copy(CopyJob&& job);
+
+    // Extend pool by one arena (called off the hot path).
+    bool extend();
+    bool need_extend();
+
+   private:
+    struct Slot {
+        uint64_t* h_src = nullptr;  // pinned staging
+        uint64_t* h_dst = nullptr;
+        uint64_t* d_src = nullptr;  // device descriptor buffers
+        uint64_t* d_dst = nullptr;
+        gpu::Event event = nullptr;
+        bool busy = false;
+    };
+    struct StreamCtx {
+        gpu::Stream stream = nullptr;
+        std::vector<Slot> slots;
+        int next_slot = 0;
+    };
+    struct PendingTask {
+        StreamCtx* sc;
+        Slot* slot;
+        std::function<void(bool)> done;  // may be empty for chunked sub-jobs
+    };
+
+    Slot* acquire_slot(StreamCtx& sc);
+    void completion_loop();
+
+    ShardOptions opt_;
+    MM mm_;
+    std::mutex alloc_mu_;
+
+    std::vector<StreamCtx> streams_;
+    int next_stream_ = 0;
+    std::mutex submit_mu_;
+
+    std::deque<PendingTask> tasks_;
+    std::mutex task_mu_;
+    std::condition_variable task_cv_;
+    std::condition_variable slot_cv_;
+    std::thread completion_thread_;
+    bool stopping_ = false;
+    bool inited_ = false;
+};
+
+}  // namespace ifs
