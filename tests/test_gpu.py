@@ -1,0 +1,258 @@
+"""GPU (MI355X) tests: local IPC path with the batched HIP gather/scatter
+kernel, HBM pool shards, fingerprint kernel numerics, and fabric interop.
+All tests here are @pytest.mark.gpu and need ROCm hardware."""
+
+import json
+import os
+import socket
+import subprocess
+import sys
+import time
+import uuid
+
+import pytest
+import torch
+
+import infinistore_amd as ifs
+
+from conftest import free_port, make_client
+
+pytestmark = pytest.mark.gpu
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _wait_port(port, timeout=60):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        try:
+            s = socket.create_connection(("127.0.0.1", port), timeout=1)
+            s.close()
+            return True
+        except OSError:
+            time.sleep(0.3)
+    return False
+
+
+@pytest.fixture(scope="module")
+def gpu_server():
+    """Server subprocess with HBM pool shards on all GPUs (the local IPC path
+    requires separate client/server processes)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    port = free_port()
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "infinistore_amd.server",
+            "--service-port", str(port),
+            "--manage-port", str(free_port()),
+            "--prealloc-size", "4",
+            "--minimal-allocate-size", "64",
+            "--no-manage",
+            "--log-level", "info",
+        ],
+        cwd=REPO,
+    )
+    assert _wait_port(port), "server did not come up"
+    yield port
+    proc.terminate()
+    proc.wait(timeout=20)
+
+
+def local_conn(port):
+    cfg = ifs.ClientConfig(
+        host_addr="127.0.0.1", service_port=port, connection_type=ifs.TYPE_LOCAL_GPU
+    )
+    c = ifs.InfinityConnection(cfg)
+    c.connect()
+    return c
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16, torch.float16])
+def test_local_roundtrip_dtypes(gpu_server, dtype):
+    conn = local_conn(gpu_server)
+    try:
+        n = 64 * 1024
+        page = 16 * 1024
+        src = torch.randn(n, device="cuda:0").to(dtype)
+        dst = torch.zeros(n, dtype=dtype, device="cuda:0")
+        keys = [f"{uuid.uuid4()}-{i}" for i in range(n // page)]
+        offs = [i * page for i in range(n // page)]
+        conn.local_gpu_write_cache(src, list(zip(keys, offs)), page)
+        conn.sync()
+        conn.read_cache(dst, list(zip(keys, offs)), page)
+        conn.sync()
+        assert torch.equal(src, dst)
+    finally:
+        conn.close()
+
+
+def test_local_roundtrip_caching_allocator(gpu_server):
+    """The base_offset protocol extension: tensors allocated by the torch
+    caching allocator (NOT at an allocation base) must round-trip."""
+    conn = local_conn(gpu_server)
+    try:
+        pad = torch.empty(1000, device="cuda:0")  # shifts the next alloc
+        src = torch.randn(32768, device="cuda:0")
+        dst = torch.zeros_like(src)
+        del pad
+        key = f"off-{uuid.uuid4()}"
+        conn.local_gpu_write_cache(src, [(key, 0)], 32768)
+        conn.sync()
+        conn.read_cache(dst, [(key, 0)], 32768)
+        conn.sync()
+        assert torch.equal(src, dst)
+    finally:
+        conn.close()
+
+
+def test_local_many_blocks_128kb(gpu_server):
+    """The headline shape: many 128 KB blocks in one request -> one kernel."""
+    conn = local_conn(gpu_server)
+    try:
+        page_elems = 65536  # 128 KB bf16
+        nb = 512            # 64 MB
+        src = torch.randn(nb * page_elems, dtype=torch.bfloat16, device="cuda:0")
+        dst = torch.zeros_like(src)
+        pre = uuid.uuid4().hex
+        keys = [f"{pre}-{i}" for i in range(nb)]
+        offs = [i * page_elems for i in range(nb)]
+        conn.local_gpu_write_cache(src, list(zip(keys, offs)), page_elems)
+        conn.sync()
+        conn.read_cache(dst, list(zip(keys, offs)), page_elems)
+        conn.sync()
+        assert torch.equal(src, dst)
+    finally:
+        conn.close()
+
+
+def test_local_dedup_and_queries(gpu_server):
+    conn = local_conn(gpu_server)
+    try:
+        a = torch.full((8192,), 1.0, device="cuda:0")
+        b = torch.full((8192,), 2.0, device="cuda:0")
+        out = torch.zeros(8192, device="cuda:0")
+        key = f"dup-{uuid.uuid4()}"
+        conn.local_gpu_write_cache(a, [(key, 0)], 8192)
+        conn.sync()
+        conn.local_gpu_write_cache(b, [(key, 0)], 8192)  # silently dropped
+        conn.sync()
+        conn.read_cache(out, [(key, 0)], 8192)
+        conn.sync()
+        assert torch.equal(out, a)
+        assert conn.check_exist(key)
+        assert conn.get_match_last_index([key, "missing-x"]) == 0
+    finally:
+        conn.close()
+
+
+def test_interop_rdma_write_local_read(gpu_server):
+    """Write via the fabric path (CPU tensor), read via the local GPU path —
+    the reference's CPU-RDMA -> local-GPU interop case."""
+    wconn = make_client(gpu_server)
+    rconn = local_conn(gpu_server)
+    try:
+        src = torch.randn(32768)  # CPU
+        dst = torch.zeros(32768, device="cuda:0")
+        key = f"interop-{uuid.uuid4()}"
+        wconn.register_mr(src)
+        blocks = wconn.allocate_rdma([key], 32768 * 4)
+        wconn.rdma_write_cache(src, [0], 32768, blocks)
+        wconn.sync()
+        rconn.read_cache(dst, [(key, 0)], 32768)
+        rconn.sync()
+        assert torch.equal(src, dst.cpu())
+    finally:
+        wconn.close()
+        rconn.close()
+
+
+def test_gpu_tensor_fabric_roundtrip(gpu_server):
+    """RDMA-semantics path with a GPU tensor on the client side (d2h staging
+    on the client, h2d into the HBM pool on the server)."""
+    conn = make_client(gpu_server)
+    try:
+        src = torch.randn(65536, device="cuda:0")
+        dst = torch.zeros(65536, device="cuda:0")
+        conn.register_mr(src)
+        conn.register_mr(dst)
+        pre = uuid.uuid4().hex
+        keys = [f"{pre}-{i}" for i in range(4)]
+        offs = [i * 16384 for i in range(4)]
+        blocks = conn.allocate_rdma(keys, 16384 * 4)
+        conn.rdma_write_cache(src, offs, 16384, blocks)
+        conn.sync()
+        conn.read_cache(dst, list(zip(keys, offs)), 16384)
+        conn.sync()
+        assert torch.equal(src, dst)
+    finally:
+        conn.close()
+
+
+def _mix64(x):
+    M = (1 << 64) - 1
+    x = (x + 0x9E3779B97F4A7C15) & M
+    x = ((x ^ (x >> 30)) * 0xBF58476D1CE4E5B9) & M
+    x = ((x ^ (x >> 27)) * 0x94D049BB133111EB) & M
+    return x ^ (x >> 31)
+
+
+def _fingerprint_ref(data: bytes, block_size: int):
+    """Pure-Python reference of the hash_blocks kernel."""
+    M = (1 << 64) - 1
+    out = []
+    for b0 in range(0, len(data), block_size):
+        blk = data[b0 : b0 + block_size]
+        h = 0
+        nw = len(blk) // 8
+        for i in range(nw):
+            w = int.from_bytes(blk[i * 8 : i * 8 + 8], "little")
+            h ^= _mix64(w ^ ((i * 0xFF51AFD7ED558CCD) & M))
+        tb = len(blk) - nw * 8
+        if tb:
+            tail = int.from_bytes(blk[nw * 8 :], "little")
+            h ^= _mix64(tail ^ ((nw * 0xFF51AFD7ED558CCD) & M))
+        out.append(_mix64(h ^ block_size))
+    return out
+
+
+def test_fingerprint_kernel_vs_cpu_reference():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    torch.manual_seed(0)
+    t = torch.randn(16384, device="cuda:0")  # 64 KB fp32
+    bs = 16384  # bytes per block -> 4 blocks
+    got = ifs.fingerprint_blocks(t, [0, 4096, 8192, 12288], 4096)
+    ref = _fingerprint_ref(t.cpu().numpy().tobytes(), bs)
+    assert got == ref
+    # determinism + sensitivity
+    again = ifs.fingerprint_blocks(t, [0, 4096, 8192, 12288], 4096)
+    assert again == got
+    t2 = t.clone()
+    t2[0] += 1.0
+    other = ifs.fingerprint_blocks(t2, [0, 4096, 8192, 12288], 4096)
+    assert other[0] != got[0] and other[1:] == got[1:]
+
+
+def test_native_extension_is_loaded():
+    """Guard against a silent eager/PyTorch fallback: the HIP extension must
+    actually be the thing serving GPU ops."""
+    from infinistore_amd import _native
+
+    assert _native.gpu_available()
+    assert _native.__file__.endswith(".so")
+    assert "infinistore_amd" in _native.__file__
+
+
+def test_bench_smoke():
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "3", "--warmup", "1",
+         "--blocks", "256", "--latency-ops", "30", "--pool-gb", "2"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=420,
+    )
+    assert r.returncode == 0, r.stdout + r.stderr
+    line = [l for l in r.stdout.strip().splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["metric"] == "put_get_GBps" and out["value"] > 0
+    assert out["config"]["path"] == "local_gpu_ipc"
