@@ -122,12 +122,16 @@ int ClientConn::rw_local(char op, const std::vector<std::pair<std::string, uint6
     }
     gpu::IpcHandle handle;
     uint64_t base_offset = 0;
-    if (!gpu::ipc_export(reinterpret_cast<void*>(ptr), &handle, &base_offset)) return -1;
+    bool have_handle =
+        gpu::ipc_export(reinterpret_cast<void*>(ptr), &handle, &base_offset);
+    if (!have_handle) memset(handle.bytes, 0, gpu::kIpcHandleSize);
 
     LocalMetaMsg msg;
     msg.device = device_id;
     msg.block_size = block_size;
     msg.base_offset = base_offset;
+    msg.pid = static_cast<int32_t>(getpid());
+    msg.base_ptr = ptr - base_offset;
     msg.ipc_handle.assign(handle.bytes, handle.bytes + gpu::kIpcHandleSize);
     msg.blocks.reserve(blocks.size());
     for (auto& b : blocks) msg.blocks.push_back({b.first, b.second});

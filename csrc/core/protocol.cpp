@@ -43,6 +43,8 @@ std::vector<uint8_t> build_local_meta(const LocalMetaMsg& m) {
     b.add_scalar<int32_t>(2, m.block_size, 0);
     b.add_offset(3, blocks_vec);
     b.add_scalar<uint64_t>(4, m.base_offset, 0);
+    b.add_scalar<int32_t>(5, m.pid, 0);
+    b.add_scalar<uint64_t>(6, m.base_ptr, 0);
     auto root = b.end_table();
     b.finish(root);
     return b.release();
@@ -57,6 +59,8 @@ bool parse_local_meta(const uint8_t* buf, size_t len, LocalMetaMsg* out) {
     out->ipc_handle = t.scalar_vector<uint8_t>(1);
     out->block_size = t.scalar<int32_t>(2, 0);
     out->base_offset = t.scalar<uint64_t>(4, 0);
+    out->pid = t.scalar<int32_t>(5, 0);
+    out->base_ptr = t.scalar<uint64_t>(6, 0);
     out->blocks.clear();
     size_t n = t.vec_len(3);
     out->blocks.reserve(n);

@@ -87,6 +87,9 @@ struct LocalMetaMsg {
     int32_t block_size = 0;
     std::vector<KeyOffset> blocks;
     uint64_t base_offset = 0;  // extension, field id 4
+    int32_t pid = 0;           // extension, field id 5: client pid — enables a
+    uint64_t base_ptr = 0;     // field id 6: same-process fast path (an IPC
+                               // handle cannot be opened in its own process)
 };
 
 struct RemoteMetaMsg {

@@ -20,8 +20,12 @@ struct Server::Conn : RefCounted {
     Header hdr{};
     std::vector<uint8_t> buf;  // accumulated bytes
 
-    // local path: in-flight async copy count (client polls OP_SYNC).
+    // local path: in-flight async copy count. OP_SYNC blocks server-side:
+    // when remain>0 the response is deferred until the count drains to zero
+    // (the reference instead has the client poll with sleeps,
+    // lib.py:578-592 — ~0.5 ms of added latency per op there).
     std::atomic<int> remain{0};
+    bool sync_waiting = false;  // loop thread only
 
     // cached IPC mappings: handle bytes -> base pointer (closed on disconnect)
     std::map<std::vector<uint8_t>, std::pair<void*, int>> ipc_cache;  // base, src_dev
@@ -376,23 +380,29 @@ void Server::maybe_extend(Shard* s) {
 }
 
 // ---- local (IPC) path -----------------------------------------------------
+namespace {
+// Resolve the client's allocation base: same-process fast path (an IPC
+// handle cannot be opened inside the exporting process) or the per-conn
+// cached hipIpcOpenMemHandle mapping.
+void* resolve_client_base(Server::Conn* c, const LocalMetaMsg& msg) {
+    if (msg.pid != 0 && msg.pid == static_cast<int32_t>(getpid()) && msg.base_ptr != 0)
+        return reinterpret_cast<void*>(msg.base_ptr);
+    auto it = c->ipc_cache.find(msg.ipc_handle);
+    if (it != c->ipc_cache.end()) return it->second.first;
+    gpu::IpcHandle h;
+    memcpy(h.bytes, msg.ipc_handle.data(), gpu::kIpcHandleSize);
+    void* base = gpu::ipc_open(h, msg.device);
+    if (base) c->ipc_cache.emplace(msg.ipc_handle, std::make_pair(base, msg.device));
+    return base;
+}
+}  // namespace
+
 void Server::op_local_write(Conn* c, const LocalMetaMsg& msg) {
     if (!gpu::available()) return send_status(c, SYSTEM_ERROR);
     if (msg.ipc_handle.size() != gpu::kIpcHandleSize || msg.block_size <= 0)
         return send_status(c, INVALID_REQ);
-
-    // Resolve (cached) client mapping.
-    void* base = nullptr;
-    auto it = c->ipc_cache.find(msg.ipc_handle);
-    if (it != c->ipc_cache.end()) {
-        base = it->second.first;
-    } else {
-        gpu::IpcHandle h;
-        memcpy(h.bytes, msg.ipc_handle.data(), gpu::kIpcHandleSize);
-        base = gpu::ipc_open(h, msg.device);
-        if (!base) return send_status(c, INTERNAL_ERROR);
-        c->ipc_cache.emplace(msg.ipc_handle, std::make_pair(base, msg.device));
-    }
+    void* base = resolve_client_base(c, msg);
+    if (!base) return send_status(c, INTERNAL_ERROR);
     uint8_t* client_ptr = static_cast<uint8_t*>(base) + msg.base_offset;
 
     Shard* shard = shard_for_device(msg.device);
@@ -404,30 +414,36 @@ void Server::op_local_write(Conn* c, const LocalMetaMsg& msg) {
     std::vector<std::string> new_keys;
     {
         std::lock_guard<std::mutex> lk(kv_mu_);
+        // Pass 1: dedup (first write wins), collect the keys to store.
+        std::vector<const KeyOffset*> fresh;
+        fresh.reserve(msg.blocks.size());
         for (auto& b : msg.blocks) {
-            if (kv_.count(b.key)) continue;  // dedup: first write wins
-            void* dst = nullptr;
-            int pool_idx = -1;
-            bool ok = shard->allocate(page, 1, [&](void* p, int idx) {
-                dst = p;
-                pool_idx = idx;
-            });
-            if (!ok) {
-                for (auto& k : new_keys) kv_.erase(k);  // roll back this request
-                return send_status(c, OUT_OF_MEMORY);
-            }
+            if (!kv_.count(b.key)) fresh.push_back(&b);
+        }
+        // Pass 2: one batched allocator call for all pages.
+        std::vector<std::pair<void*, int>> slots;
+        slots.reserve(fresh.size());
+        bool ok = fresh.empty() ||
+                  shard->allocate(page, fresh.size(),
+                                  [&](void* p, int idx) { slots.push_back({p, idx}); });
+        if (!ok) return send_status(c, OUT_OF_MEMORY);
+        new_entries.reserve(fresh.size());
+        new_keys.reserve(fresh.size());
+        job.src.reserve(fresh.size());
+        job.dst.reserve(fresh.size());
+        for (size_t i = 0; i < fresh.size(); i++) {
             auto* e = new BlockEntry();
-            e->ptr = dst;
+            e->ptr = slots[i].first;
             e->size = page;
-            e->pool_idx = pool_idx;
+            e->pool_idx = slots[i].second;
             e->shard = shard;
             e->committed = false;
             Ref<BlockEntry> ref(e);
-            kv_.emplace(b.key, ref);
+            kv_.emplace(fresh[i]->key, ref);
             new_entries.push_back(ref);
-            new_keys.push_back(b.key);
-            job.src.push_back(reinterpret_cast<uint64_t>(client_ptr + b.offset));
-            job.dst.push_back(reinterpret_cast<uint64_t>(dst));
+            new_keys.push_back(fresh[i]->key);
+            job.src.push_back(reinterpret_cast<uint64_t>(client_ptr + fresh[i]->offset));
+            job.dst.push_back(reinterpret_cast<uint64_t>(slots[i].first));
         }
     }
 
@@ -455,13 +471,11 @@ void Server::op_local_write(Conn* c, const LocalMetaMsg& msg) {
                     if (it != kv_.end() && it->second.get() == (*entries)[i].get()) kv_.erase(it);
                 }
             }
-            c->remain.fetch_sub(1);
-            c->unref();
+            finish_task(c);
         });
     };
     if (!shard->submit_copy(std::move(job))) {
-        c->remain.fetch_sub(1);
-        c->unref();
+        finish_task(c);
         return send_status(c, INTERNAL_ERROR);
     }
     send_status(c, TASK_ACCEPTED);
@@ -471,18 +485,8 @@ void Server::op_local_read(Conn* c, const LocalMetaMsg& msg) {
     if (!gpu::available()) return send_status(c, SYSTEM_ERROR);
     if (msg.ipc_handle.size() != gpu::kIpcHandleSize || msg.block_size <= 0)
         return send_status(c, INVALID_REQ);
-
-    void* base = nullptr;
-    auto it = c->ipc_cache.find(msg.ipc_handle);
-    if (it != c->ipc_cache.end()) {
-        base = it->second.first;
-    } else {
-        gpu::IpcHandle h;
-        memcpy(h.bytes, msg.ipc_handle.data(), gpu::kIpcHandleSize);
-        base = gpu::ipc_open(h, msg.device);
-        if (!base) return send_status(c, INTERNAL_ERROR);
-        c->ipc_cache.emplace(msg.ipc_handle, std::make_pair(base, msg.device));
-    }
+    void* base = resolve_client_base(c, msg);
+    if (!base) return send_status(c, INTERNAL_ERROR);
     uint8_t* client_ptr = static_cast<uint8_t*>(base) + msg.base_offset;
     size_t page = static_cast<size_t>(msg.block_size);
 
@@ -516,16 +520,12 @@ void Server::op_local_read(Conn* c, const LocalMetaMsg& msg) {
         j.done = [this, c, held, pending](bool ok) {
             (void)ok;
             if (pending->fetch_sub(1) == 1) {
-                post([this, c, held] {
-                    c->remain.fetch_sub(1);
-                    c->unref();
-                });
+                post([this, c, held] { finish_task(c); });
             }
         };
         if (!shard->submit_copy(std::move(j))) {
             if (pending->fetch_sub(1) == 1) {
-                c->remain.fetch_sub(1);
-                c->unref();
+                finish_task(c);
                 return send_status(c, INTERNAL_ERROR);
             }
         }
@@ -533,7 +533,22 @@ void Server::op_local_read(Conn* c, const LocalMetaMsg& msg) {
     send_status(c, TASK_ACCEPTED);
 }
 
-void Server::op_sync(Conn* c) { send_status(c, c->remain.load()); }
+void Server::op_sync(Conn* c) {
+    if (c->remain.load() == 0) {
+        send_status(c, 0);
+    } else {
+        c->sync_waiting = true;  // answered by finish_task when remain drains
+    }
+}
+
+// Runs on the loop thread after a local-path copy completes.
+void Server::finish_task(Conn* c) {
+    if (c->remain.fetch_sub(1) == 1 && c->sync_waiting) {
+        c->sync_waiting = false;
+        send_status(c, 0);
+    }
+    c->unref();
+}
 
 // ---- TCP fabric (RDMA-semantics) -----------------------------------------
 void Server::op_exchange(Conn* c, const std::vector<uint8_t>& body) {

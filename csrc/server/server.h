@@ -102,6 +102,7 @@ class Server {
     void op_local_write(Conn* c, const LocalMetaMsg& msg);
     void op_local_read(Conn* c, const LocalMetaMsg& msg);
     void op_sync(Conn* c);
+    void finish_task(Conn* c);  // remain-- (+ deferred sync reply)
     void op_exchange(Conn* c, const std::vector<uint8_t>& body);
     void op_allocate(Conn* c, const RemoteMetaMsg& msg);
     void op_tcp_put(Conn* c, const std::vector<uint8_t>& body);
