@@ -80,6 +80,11 @@ bool launch_copy_blocks(int dev, Stream s, const uint64_t* dev_src_ptrs,
                         const uint64_t* dev_dst_ptrs, int n_blocks, size_t bytes_per_block,
                         bool aligned16);
 
+// Small batches (n <= 16, 16B-aligned): descriptors passed as kernel args
+// from HOST arrays — no device staging needed.
+bool launch_copy_blocks_inline(int dev, Stream s, const uint64_t* src_ptrs,
+                               const uint64_t* dst_ptrs, int n_blocks, size_t bytes_per_block);
+
 // --- block fingerprint ------------------------------------------------------
 // 64-bit position-salted fingerprint per block -> out_hashes[i] (device mem).
 bool launch_hash_blocks(int dev, Stream s, const uint64_t* dev_ptrs, int n_blocks,

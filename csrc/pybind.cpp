@@ -49,6 +49,7 @@ struct ServerConfigPy {
     bool auto_increase = false;
     std::vector<int> devices;         // GPU ordinals to shard over; empty=auto
     bool cpu_only = false;            // force CPU pool even if GPUs exist
+    int cpu_shards = 1;               // CPU-mode shard count
 };
 
 bool start_server(const ServerConfigPy& cfg) {
@@ -62,6 +63,7 @@ bool start_server(const ServerConfigPy& cfg) {
     opt.block_granule = static_cast<size_t>(cfg.minimal_allocate_size) << 10;
     opt.auto_extend = cfg.auto_increase;
     opt.n_streams = cfg.num_stream > 0 ? cfg.num_stream : 4;
+    opt.cpu_shards = cfg.cpu_shards;
     opt.log_level = cfg.log_level;
     if (!cfg.cpu_only && gpu::available()) {
         if (!cfg.devices.empty()) {
@@ -139,7 +141,8 @@ PYBIND11_MODULE(_native, m) {
         .def_readwrite("num_stream", &ServerConfigPy::num_stream)
         .def_readwrite("auto_increase", &ServerConfigPy::auto_increase)
         .def_readwrite("devices", &ServerConfigPy::devices)
-        .def_readwrite("cpu_only", &ServerConfigPy::cpu_only);
+        .def_readwrite("cpu_only", &ServerConfigPy::cpu_only)
+        .def_readwrite("cpu_shards", &ServerConfigPy::cpu_shards);
 
     // ---- client connection ----
     py::class_<ClientConn>(m, "Connection")

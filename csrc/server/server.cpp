@@ -100,13 +100,15 @@ void send_buf(Server::Conn* c, std::vector<uint8_t> data) {
 Server::Server(const ServerOptions& opt) : opt_(opt) {
     set_log_level(opt.log_level.c_str());
     if (opt_.devices.empty()) {
-        ShardOptions so;
-        so.device = -1;
-        so.pool_bytes = opt_.prealloc_bytes;
-        so.block_granule = opt_.block_granule;
-        so.auto_extend = opt_.auto_extend;
-        so.extend_bytes = opt_.extend_bytes;
-        shards_.emplace_back(new Shard(so));
+        for (int i = 0; i < std::max(1, opt_.cpu_shards); i++) {
+            ShardOptions so;
+            so.device = -1;
+            so.pool_bytes = opt_.prealloc_bytes;
+            so.block_granule = opt_.block_granule;
+            so.auto_extend = opt_.auto_extend;
+            so.extend_bytes = opt_.extend_bytes;
+            shards_.emplace_back(new Shard(so));
+        }
     } else {
         for (int dev : opt_.devices) {
             ShardOptions so;

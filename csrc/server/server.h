@@ -49,7 +49,8 @@ struct ServerOptions {
     size_t block_granule = 64 << 10;      // minimal_allocate_size
     bool auto_extend = false;
     size_t extend_bytes = 10ull << 30;
-    std::vector<int> devices;  // GPU ordinals to shard over; empty => CPU shard
+    std::vector<int> devices;  // GPU ordinals to shard over; empty => CPU shard(s)
+    int cpu_shards = 1;        // CPU-mode shard count (tests the routing path)
     int n_streams = 4;
     std::string log_level = "warning";
 };

@@ -163,6 +163,24 @@ bool Shard::submit_copy(CopyJob&& job) {
     StreamCtx& sc = streams_[next_stream_];
     next_stream_ = (next_stream_ + 1) % static_cast<int>(streams_.size());
 
+    // Small aligned batches: descriptors ride in the kernel arguments.
+    if (aligned && n <= 16) {
+        Slot* slot = acquire_slot(sc);
+        if (!slot) return false;
+        bool ok = gpu::set_device(opt_.device) &&
+                  gpu::launch_copy_blocks_inline(opt_.device, sc.stream, job.src.data(),
+                                                 job.dst.data(), static_cast<int>(n),
+                                                 job.bytes_per_block) &&
+                  gpu::event_record(slot->event, sc.stream);
+        {
+            std::lock_guard<std::mutex> lk(task_mu_);
+            if (!ok) slot->busy = false;
+            tasks_.push_back({&sc, ok ? slot : nullptr, std::move(job.done)});
+        }
+        task_cv_.notify_one();
+        return true;
+    }
+
     // Chunk over slot capacity; the done callback fires after the final chunk
     // (chunks on one stream complete in order).
     size_t cap = opt_.max_descs_per_slot;

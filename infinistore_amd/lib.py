@@ -104,6 +104,7 @@ class ServerConfig(_native.ServerConfig):
         self.auto_increase = kwargs.get("auto_increase", False)
         self.devices = kwargs.get("devices", [])
         self.cpu_only = kwargs.get("cpu_only", False)
+        self.cpu_shards = kwargs.get("cpu_shards", 1)
 
     def __repr__(self):
         return (

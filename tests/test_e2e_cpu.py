@@ -244,3 +244,29 @@ def test_server_restart(ports):
         put_get_roundtrip(conn, 1024, 512)
         conn.close()
         ifs.unregister_server()
+
+
+def test_multi_shard_routing(ports):
+    """Two CPU shards: allocation spreads via least-used routing; reads must
+    gather correctly across shards (the xGMI cross-shard path, CPU-modeled)."""
+    service_port, manage_port = ports
+    cfg = ifs.ServerConfig(
+        service_port=service_port,
+        manage_port=manage_port,
+        prealloc_size=1,
+        minimal_allocate_size=16,
+        cpu_only=True,
+        cpu_shards=2,
+    )
+    ifs.register_server(cfg)
+    try:
+        conn = make_client(service_port)
+        # Interleave many small allocations so both shards receive keys.
+        put_get_roundtrip(conn, 32 * 1024, 256)  # 128 pages
+        import json
+
+        stats = json.loads(ifs.get_server_stats())
+        assert stats["shards"] == 2
+        conn.close()
+    finally:
+        ifs.unregister_server()
