@@ -384,6 +384,16 @@ int ClientConn::check_exist(const std::string& key) {
     return code;  // 0 exists, 1 not
 }
 
+int ClientConn::delete_keys(const std::vector<std::string>& keys) {
+    if (!connected_) return -1;
+    auto body = build_match_request(keys);
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!send_req(OP_DELETE, body.data(), body.size())) return -1;
+    int n = -1;
+    if (!recv_status(&n)) return -1;
+    return n;
+}
+
 int ClientConn::get_match_last_index(const std::vector<std::string>& keys) {
     if (!connected_) return -1;
     auto body = build_match_request(keys);

@@ -77,6 +77,9 @@ class ClientConn {
     // ---- queries ----
     int check_exist(const std::string& key);
     int get_match_last_index(const std::vector<std::string>& keys);
+    // Delete keys; returns the number removed (extension: engine-driven
+    // eviction — the reference only offers wholesale purge).
+    int delete_keys(const std::vector<std::string>& keys);
 
     bool rdma_connected() const { return rdma_connected_; }
 

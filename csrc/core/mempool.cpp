@@ -58,9 +58,20 @@ void MemoryPool::mark(size_t start, size_t nb, bool used) {
 
 bool MemoryPool::find_run(size_t nb, size_t* out_start) {
     // Two passes: from cursor to end, then from 0 to cursor.
-    for (int pass = 0; pass < 2; pass++) {
-        size_t w_begin = pass == 0 ? cursor_ : 0;
-        size_t w_end = pass == 0 ? n_words_ : cursor_;
+    if (find_run_in(nb, cursor_, n_words_, n_blocks_, out_start)) {
+        cursor_ = *out_start / 64;
+        return true;
+    }
+    if (find_run_in(nb, 0, cursor_, n_blocks_, out_start)) {
+        cursor_ = *out_start / 64;
+        return true;
+    }
+    return false;
+}
+
+bool MemoryPool::find_run_in(size_t nb, size_t w_begin, size_t w_end, size_t block_limit,
+                             size_t* out_start) {
+    {
         size_t run_start = 0, run_len = 0;
         for (size_t w = w_begin; w < w_end;) {
             // Skip fully-used words fast via the summary (only when we are
@@ -77,9 +88,8 @@ bool MemoryPool::find_run(size_t nb, size_t* out_start) {
             if (word == 0) {
                 if (run_len == 0) run_start = w * 64;
                 run_len += 64;
-                if (run_len >= nb) {
+                if (run_len >= nb && run_start + nb <= block_limit) {
                     *out_start = run_start;
-                    cursor_ = w;
                     return true;
                 }
                 w++;
@@ -94,9 +104,8 @@ bool MemoryPool::find_run(size_t nb, size_t* out_start) {
                 } else {
                     if (run_len == 0) run_start = idx;
                     run_len++;
-                    if (run_len >= nb) {
+                    if (run_len >= nb && run_start + nb <= block_limit) {
                         *out_start = run_start;
-                        cursor_ = w;
                         return true;
                     }
                 }
@@ -115,6 +124,43 @@ void* MemoryPool::allocate(size_t size) {
     mark(start, nb, true);
     used_blocks_ += nb;
     return static_cast<uint8_t*>(base_) + start * block_size_;
+}
+
+void* MemoryPool::allocate_below(size_t size, size_t limit_block) {
+    if (size == 0 || size > size_) return nullptr;
+    size_t nb = (size + block_size_ - 1) / block_size_;
+    size_t start;
+    size_t w_end = std::min(n_words_, (limit_block + 63) / 64);
+    if (!find_run_in(nb, 0, w_end, limit_block, &start)) return nullptr;
+    mark(start, nb, true);
+    used_blocks_ += nb;
+    return static_cast<uint8_t*>(base_) + start * block_size_;
+}
+
+size_t MemoryPool::high_water() const {
+    for (size_t w = n_words_; w-- > 0;) {
+        uint64_t bits = bits_[w];
+        if (w == n_words_ - 1) {
+            // Mask out the synthetic tail-used bits of a partial last word.
+            size_t tail = n_words_ * 64 - n_blocks_;
+            if (tail) bits &= ~(~0ull << (64 - tail));
+        }
+        if (bits) return w * 64 + (64 - static_cast<size_t>(__builtin_clzll(bits)));
+    }
+    return 0;
+}
+
+size_t MemoryPool::largest_free_run() const {
+    size_t best = 0, run = 0;
+    for (size_t i = 0; i < n_blocks_; i++) {
+        if (bits_[i / 64] & (1ull << (i % 64))) {
+            run = 0;
+        } else {
+            run++;
+            if (run > best) best = run;
+        }
+    }
+    return best;
 }
 
 bool MemoryPool::deallocate(void* ptr, size_t size) {

@@ -32,8 +32,19 @@ class MemoryPool {
     // Allocate `size` bytes (rounded up to whole blocks, contiguous).
     // Returns nullptr if no run of free blocks is large enough.
     void* allocate(size_t size);
+    // Compaction helper: allocate only from block range [0, limit_block).
+    void* allocate_below(size_t size, size_t limit_block);
     // Returns false on invalid pointer / double free.
     bool deallocate(void* ptr, size_t size);
+
+    size_t block_index_of(void* ptr) const {
+        return (reinterpret_cast<uintptr_t>(ptr) - reinterpret_cast<uintptr_t>(base_)) /
+               block_size_;
+    }
+    // Highest used block index + 1 (0 if empty) — the pool's "high-water mark".
+    size_t high_water() const;
+    // Largest contiguous free run, in blocks.
+    size_t largest_free_run() const;
 
     bool contains(void* ptr) const {
         auto p = reinterpret_cast<uintptr_t>(ptr);
@@ -55,6 +66,8 @@ class MemoryPool {
     bool run_is_free(size_t start, size_t nb) const;
     void mark(size_t start, size_t nb, bool used);
     bool find_run(size_t nb, size_t* out_start);
+    bool find_run_in(size_t nb, size_t w_begin, size_t w_end, size_t block_limit,
+                     size_t* out_start);
 
     void* base_;
     size_t size_;
@@ -100,6 +113,9 @@ class MM {
     size_t used_blocks() const;
     size_t num_pools() const { return pools_.size(); }
     const MemoryPool* pool(int idx) const {
+        return idx >= 0 && static_cast<size_t>(idx) < pools_.size() ? pools_[idx].get() : nullptr;
+    }
+    MemoryPool* pool_mut(int idx) {
         return idx >= 0 && static_cast<size_t>(idx) < pools_.size() ? pools_[idx].get() : nullptr;
     }
 

@@ -15,6 +15,7 @@ std::string op_name(char op) {
         case OP_GET_MATCH_LAST_IDX: return "match_last_index";
         case OP_TCP_PUT: return "tcp_put";
         case OP_TCP_GET: return "tcp_get";
+        case OP_DELETE: return "delete";
         default: return "unknown";
     }
 }

@@ -48,6 +48,7 @@ constexpr char OP_GET_MATCH_LAST_IDX = 'M';
 // Extensions (TCP data fabric):
 constexpr char OP_TCP_PUT = 'P';  // inline block data put (emulated RDMA_WRITE)
 constexpr char OP_TCP_GET = 'G';  // inline block data get (emulated server push)
+constexpr char OP_DELETE = 'X';   // delete keys (extension: engine-driven eviction)
 
 std::string op_name(char op);
 

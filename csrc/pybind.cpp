@@ -87,6 +87,9 @@ void stop_server() {
 size_t purge_kv_map_py() { return g_server ? g_server->purge() : 0; }
 size_t get_kvmap_len_py() { return g_server ? g_server->kvmap_len() : 0; }
 std::string server_stats_py() { return g_server ? g_server->stats_json() : "{}"; }
+std::pair<size_t, size_t> server_compact_py() {
+    return g_server ? g_server->compact() : std::make_pair<size_t, size_t>(0, 0);
+}
 
 // GPU fingerprint helper: hash n blocks of a device tensor in one kernel
 // launch (block i at base + offsets[i], each `block_size` bytes).
@@ -237,7 +240,8 @@ PYBIND11_MODULE(_native, m) {
         .def("sync_rdma", &ClientConn::sync_rdma, py::call_guard<py::gil_scoped_release>())
         .def("check_exist", &ClientConn::check_exist, py::call_guard<py::gil_scoped_release>())
         .def("get_match_last_index", &ClientConn::get_match_last_index,
-             py::call_guard<py::gil_scoped_release>());
+             py::call_guard<py::gil_scoped_release>())
+        .def("delete_keys", &ClientConn::delete_keys, py::call_guard<py::gil_scoped_release>());
 
     // ---- server ----
     m.def("start_server", &start_server, py::call_guard<py::gil_scoped_release>());
@@ -245,6 +249,7 @@ PYBIND11_MODULE(_native, m) {
     m.def("purge_kv_map", &purge_kv_map_py);
     m.def("get_kvmap_len", &get_kvmap_len_py);
     m.def("server_stats", &server_stats_py);
+    m.def("server_compact", &server_compact_py, py::call_guard<py::gil_scoped_release>());
 
     // ---- logging ----
     m.def("set_log_level", [](const std::string& lvl) { set_log_level(lvl.c_str()); });

@@ -2,6 +2,7 @@
 
 #include <algorithm>
 #include <cstring>
+#include <future>
 
 #include "../core/log.h"
 
@@ -342,6 +343,8 @@ void Server::handle_request(Conn* c, char op, std::vector<uint8_t> body) {
             return op_check_exist(c, body);
         case OP_GET_MATCH_LAST_IDX:
             return op_match_index(c, body);
+        case OP_DELETE:
+            return op_delete(c, body);
         default:
             WARN("unknown op '%c'", op);
             return send_status(c, INVALID_REQ);
@@ -714,9 +717,86 @@ void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body) {
     send_status(c, static_cast<int>(left - 1));
 }
 
+void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
+    std::vector<std::string> keys;
+    if (!parse_match_request(body.data(), body.size(), &keys))
+        return send_status(c, INVALID_REQ);
+    int n = 0;
+    {
+        std::lock_guard<std::mutex> lk(kv_mu_);
+        for (auto& k : keys) n += static_cast<int>(kv_.erase(k));
+    }
+    send_status(c, n);
+}
+
 // ---------------------------------------------------------------------------
 // Management plane
 // ---------------------------------------------------------------------------
+std::pair<size_t, size_t> Server::compact() {
+    if (!running_.load()) return {0, 0};
+    std::promise<std::pair<size_t, size_t>> prom;
+    auto fut = prom.get_future();
+    post([this, &prom] {
+        size_t moved = 0, bytes = 0;
+        for (auto& shard_up : shards_) {
+            Shard* shard = shard_up.get();
+            // Collect committed, idle (refcount==1: only the map holds them)
+            // entries on this shard. New reads cannot start while this runs —
+            // it executes on the loop thread.
+            std::vector<std::pair<void*, size_t>> movable;
+            std::vector<Ref<BlockEntry>> owners;
+            {
+                std::lock_guard<std::mutex> lk(kv_mu_);
+                for (auto& kvp : kv_) {
+                    BlockEntry* e = kvp.second.get();
+                    if (e->shard == shard && e->committed && e->ref_count() == 1)
+                        movable.push_back({e->ptr, e->size});
+                }
+            }
+            if (movable.empty()) continue;
+            auto moves = shard->plan_compaction(movable);
+            if (moves.empty()) continue;
+
+            std::map<size_t, Shard::CopyJob> by_size;  // one job per page size
+            for (auto& m : moves) {
+                auto& j = by_size[m.size];
+                j.bytes_per_block = m.size;
+                j.src.push_back(reinterpret_cast<uint64_t>(m.old_ptr));
+                j.dst.push_back(reinterpret_cast<uint64_t>(m.new_ptr));
+            }
+            for (auto& [sz, j2] : by_size) {
+                std::promise<bool> cp;
+                auto cf = cp.get_future();
+                Shard::CopyJob jj = std::move(j2);
+                jj.done = [&cp](bool ok) { cp.set_value(ok); };
+                if (!shard->submit_copy(std::move(jj))) {
+                    cp.set_value(false);
+                }
+                cf.wait();  // completion thread fulfills; loop thread is idle here
+            }
+            // Swap pointers in the index and free the old slots.
+            std::map<void*, Shard::Move*> by_old;
+            for (auto& m : moves) by_old[m.old_ptr] = &m;
+            {
+                std::lock_guard<std::mutex> lk(kv_mu_);
+                for (auto& kvp : kv_) {
+                    BlockEntry* e = kvp.second.get();
+                    auto it = by_old.find(e->ptr);
+                    if (it == by_old.end() || e->shard != shard) continue;
+                    Shard::Move* m = it->second;
+                    e->ptr = m->new_ptr;
+                    e->pool_idx = m->pool_idx;
+                    shard->deallocate(m->old_ptr, m->size, m->pool_idx);
+                    moved++;
+                    bytes += m->size;
+                }
+            }
+        }
+        prom.set_value({moved, bytes});
+    });
+    return fut.get();
+}
+
 size_t Server::kvmap_len() {
     std::lock_guard<std::mutex> lk(kv_mu_);
     return kv_.size();

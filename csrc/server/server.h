@@ -85,6 +85,9 @@ class Server {
     size_t kvmap_len();
     size_t purge();
     std::string stats_json();
+    // Defragment the pools: move committed, idle blocks to lower addresses
+    // using the batched copy kernel. Returns (moved_blocks, moved_bytes).
+    std::pair<size_t, size_t> compact();
     int num_shards() const { return static_cast<int>(shards_.size()); }
 
     struct Conn;  // connection state (public: file-local helpers use it)
@@ -111,6 +114,7 @@ class Server {
     void op_commit(Conn* c, const RemoteMetaMsg& msg);
     void op_check_exist(Conn* c, const std::vector<uint8_t>& body);
     void op_match_index(Conn* c, const std::vector<uint8_t>& body);
+    void op_delete(Conn* c, const std::vector<uint8_t>& body);
 
     Shard* shard_for_device(int device);
     Shard* shard_least_used();

@@ -121,6 +121,41 @@ bool Shard::extend() {
     return true;
 }
 
+std::vector<Shard::Move> Shard::plan_compaction(
+    const std::vector<std::pair<void*, size_t>>& movable) {
+    std::lock_guard<std::mutex> lk(alloc_mu_);
+    // Move highest addresses first so freed tail space cascades.
+    auto sorted = movable;
+    std::sort(sorted.begin(), sorted.end(),
+              [](const auto& a, const auto& b) { return a.first > b.first; });
+    std::vector<Move> moves;
+    for (auto& [ptr, size] : sorted) {
+        MemoryPool* owner = nullptr;
+        for (size_t i = 0; i < mm_.num_pools(); i++) {
+            MemoryPool* p = mm_.pool_mut(static_cast<int>(i));
+            if (p->contains(ptr)) {
+                owner = p;
+                break;
+            }
+        }
+        if (!owner) continue;
+        size_t idx = owner->block_index_of(ptr);
+        void* np = owner->allocate_below(size, idx);
+        if (np) moves.push_back({ptr, np, size, owner->pool_idx()});
+    }
+    return moves;
+}
+
+size_t Shard::largest_free_run_bytes() {
+    std::lock_guard<std::mutex> lk(alloc_mu_);
+    size_t best = 0;
+    for (size_t i = 0; i < mm_.num_pools(); i++) {
+        const MemoryPool* p = mm_.pool(static_cast<int>(i));
+        best = std::max(best, p->largest_free_run() * p->block_size());
+    }
+    return best;
+}
+
 Shard::Slot* Shard::acquire_slot(StreamCtx& sc) {
     std::unique_lock<std::mutex> lk(task_mu_);
     for (;;) {

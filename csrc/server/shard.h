@@ -70,6 +70,19 @@ class Shard {
     bool extend();
     bool need_extend();
 
+    // Compaction planning: for each (ptr, size), try to find a lower slot in
+    // the same pool; marks the new slots used. Returns the planned moves.
+    struct Move {
+        void* old_ptr;
+        void* new_ptr;
+        size_t size;
+        int pool_idx;
+    };
+    std::vector<Move> plan_compaction(const std::vector<std::pair<void*, size_t>>& movable);
+
+    // Fragmentation stats (largest contiguous free run in bytes, per pool sum).
+    size_t largest_free_run_bytes();
+
    private:
     struct Slot {
         uint64_t* h_src = nullptr;  // pinned staging

@@ -22,6 +22,7 @@ from .lib import (
     purge_kv_map,
     get_kvmap_len,
     get_server_stats,
+    compact_pool,
     fingerprint_blocks,
 )
 
@@ -42,6 +43,7 @@ __all__ = [
     "purge_kv_map",
     "get_kvmap_len",
     "get_server_stats",
+    "compact_pool",
     "fingerprint_blocks",
 ]
 

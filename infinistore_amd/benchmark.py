@@ -1,0 +1,209 @@
+"""infinistore-amd benchmark harness.
+
+Superset of the reference's harness (/root/reference/infinistore/benchmark.py:
+same workload shape — `--size` MB written as uuid-keyed pages of `--block-size`
+KB in `--steps` layer-batches, then read back, printing write/read MB/s) plus
+the measurements BASELINE.md requires that the reference lacks: p50/p99
+round-trip latency and a concurrent-client saturation mode.
+
+Examples:
+    python -m infinistore_amd.benchmark --server 127.0.0.1 --port 22345 \
+        --size 1024 --block-size 128 --local-gpu
+    python -m infinistore_amd.benchmark ... --clients 8   # saturation
+"""
+
+import argparse
+import json
+import statistics
+import time
+import uuid
+
+import torch
+
+from . import lib
+
+
+def parse_args():
+    p = argparse.ArgumentParser(description="infinistore-amd benchmark")
+    p.add_argument("--server", default="127.0.0.1")
+    p.add_argument("--port", type=int, default=22345)
+    p.add_argument("--size", type=int, default=128, help="total MB per iteration")
+    p.add_argument("--block-size", type=int, default=32, help="page size in KB")
+    p.add_argument("--steps", type=int, default=32,
+                   help="layer-batches per iteration (simulates per-layer writes)")
+    p.add_argument("--iteration", type=int, default=3)
+    p.add_argument("--local-gpu", action="store_true",
+                   help="use the local IPC path (default: fabric path)")
+    p.add_argument("--src-gpu", type=int, default=0)
+    p.add_argument("--dst-gpu", type=int, default=0)
+    p.add_argument("--latency-ops", type=int, default=0,
+                   help="additionally measure N single-page round-trips")
+    p.add_argument("--clients", type=int, default=1,
+                   help="concurrent client processes (saturation mode)")
+    p.add_argument("--verify", action="store_true")
+    p.add_argument("--json", action="store_true", help="print a JSON summary")
+    return p.parse_args()
+
+
+def make_conn(args, local):
+    cfg = lib.ClientConfig(
+        host_addr=args.server,
+        service_port=args.port,
+        connection_type=lib.TYPE_LOCAL_GPU if local else lib.TYPE_RDMA,
+        link_type="TCP",
+    )
+    conn = lib.InfinityConnection(cfg)
+    conn.connect()
+    return conn
+
+
+def run_once(args, conn, local, device):
+    block_bytes = args.block_size << 10
+    total_bytes = args.size << 20
+    n_blocks = total_bytes // block_bytes
+    elems = total_bytes // 4
+    page_elems = block_bytes // 4
+
+    src = torch.rand(elems, dtype=torch.float32, device=device)
+    dst_dev = f"cuda:{args.dst_gpu}" if local else device
+    dst = torch.zeros(elems, dtype=torch.float32, device=dst_dev)
+    if not local:
+        conn.register_mr(src)
+        conn.register_mr(dst)
+
+    run = uuid.uuid4().hex
+    keys = [f"{run}-{i}" for i in range(n_blocks)]
+    offsets = [i * page_elems for i in range(n_blocks)]
+    per_step = max(1, n_blocks // args.steps)
+
+    # ---- write (layer-by-layer batches, like prefill) ----
+    t0 = time.perf_counter()
+    for s0 in range(0, n_blocks, per_step):
+        sl = slice(s0, min(s0 + per_step, n_blocks))
+        if local:
+            conn.local_gpu_write_cache(src, list(zip(keys[sl], offsets[sl])), page_elems)
+        else:
+            blocks = conn.allocate_rdma(keys[sl], block_bytes)
+            conn.rdma_write_cache(src, offsets[sl], page_elems, blocks)
+    conn.sync()
+    w_time = time.perf_counter() - t0
+
+    # ---- read ----
+    t0 = time.perf_counter()
+    conn.read_cache(dst, list(zip(keys, offsets)), page_elems)
+    conn.sync()
+    r_time = time.perf_counter() - t0
+
+    if args.verify:
+        assert torch.equal(src.cpu(), dst.cpu()), "verification failed"
+
+    return total_bytes / w_time / 1e6, total_bytes / r_time / 1e6
+
+
+def run_latency(args, conn, local, device):
+    block_bytes = args.block_size << 10
+    page_elems = block_bytes // 4
+    src = torch.rand(page_elems, dtype=torch.float32, device=device)
+    dst = torch.zeros_like(src)
+    if not local:
+        conn.register_mr(src)
+        conn.register_mr(dst)
+    put_us, get_us = [], []
+    run = uuid.uuid4().hex
+    for i in range(args.latency_ops):
+        key = f"lat-{run}-{i}"
+        t0 = time.perf_counter()
+        if local:
+            conn.local_gpu_write_cache(src, [(key, 0)], page_elems)
+        else:
+            blocks = conn.allocate_rdma([key], block_bytes)
+            conn.rdma_write_cache(src, [0], page_elems, blocks)
+        conn.sync()
+        t1 = time.perf_counter()
+        conn.read_cache(dst, [(key, 0)], page_elems)
+        conn.sync()
+        t2 = time.perf_counter()
+        put_us.append((t1 - t0) * 1e6)
+        get_us.append((t2 - t1) * 1e6)
+    return put_us, get_us
+
+
+def pct(v, q):
+    if not v:
+        return 0.0
+    return statistics.quantiles(v, n=100)[q - 1] if len(v) >= 10 else max(v)
+
+
+def _worker(args_dict, q):
+    ns = argparse.Namespace(**args_dict)
+    local = ns.local_gpu and torch.cuda.is_available()
+    device = f"cuda:{ns.src_gpu}" if local else "cpu"
+    conn = make_conn(ns, local)
+    try:
+        w, r = run_once(ns, conn, local, device)
+        q.put((w, r))
+    finally:
+        conn.close()
+
+
+def main():
+    args = parse_args()
+    local = args.local_gpu and torch.cuda.is_available()
+    device = f"cuda:{args.src_gpu}" if local else "cpu"
+
+    if args.clients > 1:
+        import multiprocessing
+
+        ctx = multiprocessing.get_context("spawn")
+        q = ctx.Queue()
+        t0 = time.perf_counter()
+        procs = [
+            ctx.Process(target=_worker, args=(vars(args), q))
+            for _ in range(args.clients)
+        ]
+        for p in procs:
+            p.start()
+        results = [q.get(timeout=600) for _ in procs]
+        for p in procs:
+            p.join()
+        wall = time.perf_counter() - t0
+        agg = args.clients * (args.size << 20) * 2 / wall / 1e6
+        print(f"saturation: {args.clients} clients, aggregate {agg:.2f} MB/s "
+              f"(per-client write {statistics.mean(w for w, _ in results):.2f} MB/s, "
+              f"read {statistics.mean(r for _, r in results):.2f} MB/s)")
+        return
+
+    conn = make_conn(args, local)
+    try:
+        writes, reads = [], []
+        for it in range(args.iteration):
+            w, r = run_once(args, conn, local, device)
+            writes.append(w)
+            reads.append(r)
+            print(f"[iter {it}] write cache: {w:.2f} MB/s, read cache: {r:.2f} MB/s")
+        summary = {
+            "write_MBps": round(statistics.mean(writes), 2),
+            "read_MBps": round(statistics.mean(reads), 2),
+            "block_kb": args.block_size,
+            "size_mb": args.size,
+            "path": "local_gpu" if local else "fabric",
+        }
+        if args.latency_ops:
+            put_us, get_us = run_latency(args, conn, local, device)
+            summary.update(
+                p50_put_us=round(pct(put_us, 50), 1),
+                p99_put_us=round(pct(put_us, 99), 1),
+                p50_get_us=round(pct(get_us, 50), 1),
+                p99_get_us=round(pct(get_us, 99), 1),
+            )
+            print(f"latency: put p50 {summary['p50_put_us']} us "
+                  f"p99 {summary['p99_put_us']} us; get p50 {summary['p50_get_us']} us "
+                  f"p99 {summary['p99_get_us']} us")
+        if args.json:
+            print(json.dumps(summary))
+    finally:
+        conn.close()
+
+
+if __name__ == "__main__":
+    main()

@@ -167,6 +167,13 @@ def get_server_stats():
     return _native.server_stats()
 
 
+def compact_pool():
+    """Defragment the (in-process) server's pools by moving committed idle
+    blocks to lower addresses with the batched copy kernel. Returns
+    (moved_blocks, moved_bytes). Extension over the reference."""
+    return _native.server_compact()
+
+
 def register_server(*args):
     """Start the in-process server.
 
@@ -449,6 +456,15 @@ class InfinityConnection:
         ret = self.conn.get_match_last_index(keys)
         if ret < 0:
             raise Exception("can't find a match")
+        return ret
+
+    def delete_keys(self, keys: List[str]) -> int:
+        """Delete keys from the store; returns the number removed.
+        Extension over the reference (which only offers wholesale purge) —
+        lets the inference engine evict cold prefixes."""
+        ret = self.conn.delete_keys(keys)
+        if ret < 0:
+            raise Exception("Failed to delete keys")
         return ret
 
     # -- internal -------------------------------------------------------------

@@ -49,6 +49,12 @@ def make_app(config: ServerConfig):
 
         return json.loads(lib.get_server_stats())
 
+    @app.post("/compact")
+    async def compact():
+        loop = asyncio.get_running_loop()
+        moved, moved_bytes = await loop.run_in_executor(None, lib.compact_pool)
+        return {"moved_blocks": moved, "moved_bytes": moved_bytes}
+
     @app.post("/selftest/{port}")
     async def selftest(port: int):
         # Loopback roundtrip through the full client stack (CPU tensor).

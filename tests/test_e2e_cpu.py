@@ -270,3 +270,58 @@ def test_multi_shard_routing(ports):
         conn.close()
     finally:
         ifs.unregister_server()
+
+
+def test_delete_keys(cpu_server):
+    conn = make_client(cpu_server)
+    try:
+        keys = put_get_roundtrip(conn, 2048, 512)  # 4 keys
+        assert conn.delete_keys(keys[:2]) == 2
+        assert not conn.check_exist(keys[0])
+        assert conn.check_exist(keys[2])
+        assert conn.delete_keys(["never-there"]) == 0
+        dst = torch.zeros(512)
+        conn.register_mr(dst)
+        with pytest.raises(Exception):
+            conn.read_cache(dst, [(keys[0], 0)], 512)
+    finally:
+        conn.close()
+
+
+def test_compaction_defragments(ports):
+    """Fill the pool with small pages, delete every other key, verify a big
+    allocation fails, compact, verify it then succeeds."""
+    service_port, manage_port = ports
+    cfg = ifs.ServerConfig(
+        service_port=service_port,
+        manage_port=manage_port,
+        prealloc_size=1,  # 1 GB
+        minimal_allocate_size=1024,  # 1 MB granule -> 1024 blocks
+        cpu_only=True,
+    )
+    ifs.register_server(cfg)
+    try:
+        conn = make_client(service_port)
+        page = 1 << 20  # 1 MB pages
+        n = 1024        # fill the entire 1 GB pool
+        src = torch.zeros(page // 4, dtype=torch.float32)
+        conn.register_mr(src)
+        keys = [f"frag-{i}" for i in range(n)]
+        blocks = conn.allocate_rdma(keys, page)
+        assert len(blocks) == n
+        # write+commit them all (1 MB each, chunked internally)
+        for i in range(n):
+            conn.rdma_write_cache(src, [0], page // 4, [blocks[i]])
+        conn.sync()
+        # free every other key -> 512 MB free but fragmented into 1 MB holes
+        conn.delete_keys(keys[0::2])
+        big = 16 << 20  # 16 MB needs 16 contiguous blocks
+        with pytest.raises(Exception):
+            conn.allocate_rdma(["big-page"], big)
+        moved, moved_bytes = ifs.compact_pool()
+        assert moved > 0 and moved_bytes == moved * page
+        got = conn.allocate_rdma(["big-page"], big)
+        assert len(got) == 1 and got[0][1] != 0
+        conn.close()
+    finally:
+        ifs.unregister_server()
