@@ -115,6 +115,21 @@ bool ClientConn::recv_payload(std::vector<uint8_t>* out) {
 // ---------------------------------------------------------------------------
 int ClientConn::rw_local(char op, const std::vector<std::pair<std::string, uint64_t>>& blocks,
                          int block_size, uintptr_t ptr, int device_id) {
+    std::string blob;
+    std::vector<uint64_t> offs;
+    offs.reserve(blocks.size());
+    for (size_t i = 0; i < blocks.size(); i++) {
+        if (i) blob.push_back('\0');
+        blob.append(blocks[i].first);
+        offs.push_back(blocks[i].second);
+    }
+    return rw_local_packed(op, blob.data(), blob.size(), offs.data(), blocks.size(), block_size,
+                           ptr, device_id);
+}
+
+int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
+                                const uint64_t* offsets, size_t n, int block_size,
+                                uintptr_t ptr, int device_id) {
     if (!connected_) return -1;
     if (!gpu::available()) {
         ERROR("local path requires a GPU");
@@ -126,19 +141,23 @@ int ClientConn::rw_local(char op, const std::vector<std::pair<std::string, uint6
         gpu::ipc_export(reinterpret_cast<void*>(ptr), &handle, &base_offset);
     if (!have_handle) memset(handle.bytes, 0, gpu::kIpcHandleSize);
 
-    LocalMetaMsg msg;
-    msg.device = device_id;
-    msg.block_size = block_size;
-    msg.base_offset = base_offset;
-    msg.pid = static_cast<int32_t>(getpid());
-    msg.base_ptr = ptr - base_offset;
-    msg.ipc_handle.assign(handle.bytes, handle.bytes + gpu::kIpcHandleSize);
-    msg.blocks.reserve(blocks.size());
-    for (auto& b : blocks) msg.blocks.push_back({b.first, b.second});
-    auto body = build_local_meta(msg);
+    PackedLocalHdr h;
+    h.device = device_id;
+    h.pid = static_cast<int32_t>(getpid());
+    h.base_ptr = ptr - base_offset;
+    h.base_offset = base_offset;
+    h.block_size = static_cast<uint32_t>(block_size);
+    h.n_blocks = static_cast<uint32_t>(n);
+    memcpy(h.ipc, handle.bytes, gpu::kIpcHandleSize);
 
+    std::vector<uint8_t> body(sizeof(h) + n * 8 + blob_len);
+    memcpy(body.data(), &h, sizeof(h));
+    memcpy(body.data() + sizeof(h), offsets, n * 8);
+    memcpy(body.data() + sizeof(h) + n * 8, keys_blob, blob_len);
+
+    char wire_op = (op == 'W') ? OP_W_FAST : OP_R_FAST;
     std::lock_guard<std::mutex> lk(io_mu_);
-    if (!send_req(op, body.data(), body.size())) return -1;
+    if (!send_req(wire_op, body.data(), body.size())) return -1;
     int code = 0;
     if (!recv_status(&code)) return -1;
     if (code != TASK_ACCEPTED && code != FINISH) {

@@ -55,6 +55,11 @@ class ClientConn {
     // blocks: (key, byte offset into the tensor). op: 'W' or 'R'.
     int rw_local(char op, const std::vector<std::pair<std::string, uint64_t>>& blocks,
                  int block_size, uintptr_t ptr, int device_id);
+    // Fast path: keys as one NUL-separated blob + an offsets array — built
+    // straight into the packed wire format (OP_W_FAST / OP_R_FAST).
+    int rw_local_packed(char op, const char* keys_blob, size_t blob_len,
+                        const uint64_t* offsets, size_t n, int block_size, uintptr_t ptr,
+                        int device_id);
     int sync_local();
 
     // ---- RDMA-semantics path ----

@@ -49,6 +49,25 @@ constexpr char OP_GET_MATCH_LAST_IDX = 'M';
 constexpr char OP_TCP_PUT = 'P';  // inline block data put (emulated RDMA_WRITE)
 constexpr char OP_TCP_GET = 'G';  // inline block data get (emulated server push)
 constexpr char OP_DELETE = 'X';   // delete keys (extension: engine-driven eviction)
+// Packed fast-path local ops (extension): same semantics as OP_W/OP_R but a
+// flat binary layout the hot path can build/parse at memcpy speed — the
+// flatbuffers LocalMetaRequest ops remain accepted for wire compatibility.
+// Body: PackedLocalHdr, u64 offsets[n], NUL-separated key bytes (n keys).
+constexpr char OP_W_FAST = 'w';
+constexpr char OP_R_FAST = 'r';
+
+#pragma pack(push, 1)
+struct PackedLocalHdr {
+    int32_t device;
+    int32_t pid;
+    uint64_t base_ptr;
+    uint64_t base_offset;
+    uint32_t block_size;
+    uint32_t n_blocks;
+    uint8_t ipc[64];
+};
+#pragma pack(pop)
+static_assert(sizeof(PackedLocalHdr) == 96, "packed local header size");
 
 std::string op_name(char op);
 

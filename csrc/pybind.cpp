@@ -162,6 +162,24 @@ PYBIND11_MODULE(_native, m) {
                 py::gil_scoped_release rel;
                 return c.rw_local(op.empty() ? 'W' : op[0], blocks, block_size, ptr, device);
             })
+        .def(
+            "rw_local_fast",
+            [](ClientConn& c, const std::string& op, py::bytes keys_blob, py::bytes offsets,
+               size_t n, int block_size, uintptr_t ptr, int device) {
+                char* kb;
+                Py_ssize_t kb_len;
+                PyBytes_AsStringAndSize(keys_blob.ptr(), &kb, &kb_len);
+                char* ob;
+                Py_ssize_t ob_len;
+                PyBytes_AsStringAndSize(offsets.ptr(), &ob, &ob_len);
+                if (static_cast<size_t>(ob_len) != n * 8)
+                    throw std::runtime_error("offsets must be n uint64");
+                py::gil_scoped_release rel;
+                return c.rw_local_packed(op.empty() ? 'W' : op[0], kb,
+                                         static_cast<size_t>(kb_len),
+                                         reinterpret_cast<const uint64_t*>(ob), n, block_size,
+                                         ptr, device);
+            })
         .def("sync_local", &ClientConn::sync_local, py::call_guard<py::gil_scoped_release>())
         .def("register_mr", &ClientConn::register_mr, py::call_guard<py::gil_scoped_release>())
         .def(

@@ -302,18 +302,76 @@ void Server::on_new_connection(uv_stream_t* server, int status) {
 // ---------------------------------------------------------------------------
 // Dispatch
 // ---------------------------------------------------------------------------
+namespace {
+
+// View over the flatbuffers LocalMetaRequest (msg must outlive the view).
+Server::LocalView to_view(const LocalMetaMsg& msg) {
+    Server::LocalView v;
+    v.device = msg.device;
+    v.pid = msg.pid;
+    v.base_ptr = msg.base_ptr;
+    v.base_offset = msg.base_offset;
+    v.block_size = msg.block_size;
+    v.ipc = msg.ipc_handle.data();
+    v.ipc_len = msg.ipc_handle.size();
+    v.blocks.reserve(msg.blocks.size());
+    for (auto& b : msg.blocks) v.blocks.push_back({std::string_view(b.key), b.offset});
+    return v;
+}
+
+// Packed fast-path body: PackedLocalHdr, u64 offsets[n], NUL-separated keys.
+bool parse_packed_local(const std::vector<uint8_t>& body, Server::LocalView* v) {
+    if (body.size() < sizeof(PackedLocalHdr)) return false;
+    PackedLocalHdr h;
+    memcpy(&h, body.data(), sizeof(h));
+    size_t n = h.n_blocks;
+    size_t off_end = sizeof(h) + n * 8;
+    if (body.size() < off_end) return false;
+    v->device = h.device;
+    v->pid = h.pid;
+    v->base_ptr = h.base_ptr;
+    v->base_offset = h.base_offset;
+    v->block_size = h.block_size;
+    v->ipc = body.data() + offsetof(PackedLocalHdr, ipc);
+    v->ipc_len = 64;
+    const uint64_t* offs = reinterpret_cast<const uint64_t*>(body.data() + sizeof(h));
+    const char* kp = reinterpret_cast<const char*>(body.data() + off_end);
+    const char* kend = reinterpret_cast<const char*>(body.data() + body.size());
+    v->blocks.reserve(n);
+    for (size_t i = 0; i < n; i++) {
+        const char* nul = static_cast<const char*>(memchr(kp, 0, kend - kp));
+        const char* ke = nul ? nul : kend;
+        if (kp > kend || (i + 1 < n && !nul)) return false;
+        v->blocks.push_back({std::string_view(kp, ke - kp), offs[i]});
+        kp = nul ? nul + 1 : kend;
+    }
+    return true;
+}
+
+}  // namespace
+
 void Server::handle_request(Conn* c, char op, std::vector<uint8_t> body) {
     DEBUG("request op=%s body=%zu", op_name(op).c_str(), body.size());
     switch (op) {
         case OP_W: {
             LocalMetaMsg msg;
             if (!parse_local_meta(body.data(), body.size(), &msg)) return send_status(c, INVALID_REQ);
-            return op_local_write(c, msg);
+            return op_local_write(c, to_view(msg));
         }
         case OP_R: {
             LocalMetaMsg msg;
             if (!parse_local_meta(body.data(), body.size(), &msg)) return send_status(c, INVALID_REQ);
-            return op_local_read(c, msg);
+            return op_local_read(c, to_view(msg));
+        }
+        case OP_W_FAST: {
+            LocalView v;
+            if (!parse_packed_local(body, &v)) return send_status(c, INVALID_REQ);
+            return op_local_write(c, v);
+        }
+        case OP_R_FAST: {
+            LocalView v;
+            if (!parse_packed_local(body, &v)) return send_status(c, INVALID_REQ);
+            return op_local_read(c, v);
         }
         case OP_SYNC:
             return op_sync(c);
@@ -389,22 +447,23 @@ namespace {
 // Resolve the client's allocation base: same-process fast path (an IPC
 // handle cannot be opened inside the exporting process) or the per-conn
 // cached hipIpcOpenMemHandle mapping.
-void* resolve_client_base(Server::Conn* c, const LocalMetaMsg& msg) {
+void* resolve_client_base(Server::Conn* c, const Server::LocalView& msg) {
     if (msg.pid != 0 && msg.pid == static_cast<int32_t>(getpid()) && msg.base_ptr != 0)
         return reinterpret_cast<void*>(msg.base_ptr);
-    auto it = c->ipc_cache.find(msg.ipc_handle);
+    std::vector<uint8_t> key(msg.ipc, msg.ipc + msg.ipc_len);
+    auto it = c->ipc_cache.find(key);
     if (it != c->ipc_cache.end()) return it->second.first;
     gpu::IpcHandle h;
-    memcpy(h.bytes, msg.ipc_handle.data(), gpu::kIpcHandleSize);
+    memcpy(h.bytes, msg.ipc, gpu::kIpcHandleSize);
     void* base = gpu::ipc_open(h, msg.device);
-    if (base) c->ipc_cache.emplace(msg.ipc_handle, std::make_pair(base, msg.device));
+    if (base) c->ipc_cache.emplace(std::move(key), std::make_pair(base, msg.device));
     return base;
 }
 }  // namespace
 
-void Server::op_local_write(Conn* c, const LocalMetaMsg& msg) {
+void Server::op_local_write(Conn* c, const LocalView& msg) {
     if (!gpu::available()) return send_status(c, SYSTEM_ERROR);
-    if (msg.ipc_handle.size() != gpu::kIpcHandleSize || msg.block_size <= 0)
+    if (msg.ipc_len != gpu::kIpcHandleSize || msg.block_size <= 0)
         return send_status(c, INVALID_REQ);
     void* base = resolve_client_base(c, msg);
     if (!base) return send_status(c, INTERNAL_ERROR);
@@ -420,10 +479,10 @@ void Server::op_local_write(Conn* c, const LocalMetaMsg& msg) {
     {
         std::lock_guard<std::mutex> lk(kv_mu_);
         // Pass 1: dedup (first write wins), collect the keys to store.
-        std::vector<const KeyOffset*> fresh;
+        std::vector<const std::pair<std::string_view, uint64_t>*> fresh;
         fresh.reserve(msg.blocks.size());
         for (auto& b : msg.blocks) {
-            if (!kv_.count(b.key)) fresh.push_back(&b);
+            if (kv_.find(b.first) == kv_.end()) fresh.push_back(&b);
         }
         // Pass 2: one batched allocator call for all pages.
         std::vector<std::pair<void*, int>> slots;
@@ -444,10 +503,10 @@ void Server::op_local_write(Conn* c, const LocalMetaMsg& msg) {
             e->shard = shard;
             e->committed = false;
             Ref<BlockEntry> ref(e);
-            kv_.emplace(fresh[i]->key, ref);
+            kv_.emplace(std::string(fresh[i]->first), ref);
             new_entries.push_back(ref);
-            new_keys.push_back(fresh[i]->key);
-            job.src.push_back(reinterpret_cast<uint64_t>(client_ptr + fresh[i]->offset));
+            new_keys.emplace_back(fresh[i]->first);
+            job.src.push_back(reinterpret_cast<uint64_t>(client_ptr + fresh[i]->second));
             job.dst.push_back(reinterpret_cast<uint64_t>(slots[i].first));
         }
     }
@@ -486,9 +545,9 @@ void Server::op_local_write(Conn* c, const LocalMetaMsg& msg) {
     send_status(c, TASK_ACCEPTED);
 }
 
-void Server::op_local_read(Conn* c, const LocalMetaMsg& msg) {
+void Server::op_local_read(Conn* c, const LocalView& msg) {
     if (!gpu::available()) return send_status(c, SYSTEM_ERROR);
-    if (msg.ipc_handle.size() != gpu::kIpcHandleSize || msg.block_size <= 0)
+    if (msg.ipc_len != gpu::kIpcHandleSize || msg.block_size <= 0)
         return send_status(c, INVALID_REQ);
     void* base = resolve_client_base(c, msg);
     if (!base) return send_status(c, INTERNAL_ERROR);
@@ -498,10 +557,11 @@ void Server::op_local_read(Conn* c, const LocalMetaMsg& msg) {
     // Group blocks by owning shard (keys may live on different GPUs).
     std::map<Shard*, Shard::CopyJob> jobs;
     auto held = std::make_shared<std::vector<Ref<BlockEntry>>>();
+    held->reserve(msg.blocks.size());
     {
         std::lock_guard<std::mutex> lk(kv_mu_);
         for (auto& b : msg.blocks) {
-            auto kit = kv_.find(b.key);
+            auto kit = kv_.find(b.first);
             if (kit == kv_.end() || !kit->second->committed) {
                 return send_status(c, KEY_NOT_FOUND);
             }
@@ -509,7 +569,7 @@ void Server::op_local_read(Conn* c, const LocalMetaMsg& msg) {
             auto& job = jobs[e->shard];
             job.bytes_per_block = page;
             job.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
-            job.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.offset));
+            job.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.second));
             held->push_back(kit->second);
         }
     }

@@ -31,6 +31,7 @@
 #include <memory>
 #include <mutex>
 #include <string>
+#include <string_view>
 #include <thread>
 #include <unordered_map>
 #include <vector>
@@ -92,8 +93,21 @@ class Server {
 
     struct Conn;  // connection state (public: file-local helpers use it)
 
-   private:
+    // Uniform view of a local-path request (flatbuffers or packed fast-path
+    // format). Key views point into the request body / parsed message and
+    // are only valid during the handling call.
+    struct LocalView {
+        int32_t device = 0;
+        int32_t pid = 0;
+        uint64_t base_ptr = 0;
+        uint64_t base_offset = 0;
+        int64_t block_size = 0;
+        const uint8_t* ipc = nullptr;
+        size_t ipc_len = 0;
+        std::vector<std::pair<std::string_view, uint64_t>> blocks;
+    };
 
+   private:
     // ---- loop-thread plumbing ----
     void loop_main();
     void post(std::function<void()> fn);  // run fn on the loop thread
@@ -103,8 +117,8 @@ class Server {
 
     // ---- request handling (loop thread) ----
     void handle_request(Conn* c, char op, std::vector<uint8_t> body);
-    void op_local_write(Conn* c, const LocalMetaMsg& msg);
-    void op_local_read(Conn* c, const LocalMetaMsg& msg);
+    void op_local_write(Conn* c, const LocalView& msg);
+    void op_local_read(Conn* c, const LocalView& msg);
     void op_sync(Conn* c);
     void finish_task(Conn* c);  // remain-- (+ deferred sync reply)
     void op_exchange(Conn* c, const std::vector<uint8_t>& body);
@@ -135,8 +149,16 @@ class Server {
     std::mutex post_mu_;
     std::vector<std::function<void()>> posted_;
 
+    // Heterogeneous-lookup map (C++20): hot-path lookups by string_view
+    // avoid a std::string allocation per key.
+    struct SvHash {
+        using is_transparent = void;
+        size_t operator()(std::string_view s) const noexcept {
+            return std::hash<std::string_view>{}(s);
+        }
+    };
     std::mutex kv_mu_;
-    std::unordered_map<std::string, Ref<BlockEntry>> kv_;
+    std::unordered_map<std::string, Ref<BlockEntry>, SvHash, std::equal_to<>> kv_;
 
     std::vector<Conn*> conns_;  // loop thread only
 

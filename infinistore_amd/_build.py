@@ -52,7 +52,7 @@ def _pybind11_include() -> str:
 def _common_flags():
     return [
         "-O3",
-        "-std=c++17",
+        "-std=c++20",
         "-fPIC",
         "-D__HIP_PLATFORM_AMD__",
         f"-I{CSRC}",

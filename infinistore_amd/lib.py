@@ -22,6 +22,7 @@ import time
 import asyncio
 from typing import List, Tuple
 
+import numpy as np
 import torch
 
 from . import _build
@@ -291,6 +292,18 @@ class InfinityConnection:
         self.rdma_connected = False
 
     # -- local (IPC) path ---------------------------------------------------
+    @staticmethod
+    def _pack_blocks(blocks, element_size):
+        """(keys_blob, offsets_bytes, n): keys NUL-joined, offsets in bytes as
+        u64 — the packed fast-path arguments (avoids per-tuple marshalling)."""
+        n = len(blocks)
+        if n == 0:
+            return b"", b"", 0
+        keys, offsets = zip(*blocks)
+        blob = "\x00".join(keys).encode()
+        offs = np.fromiter(offsets, dtype=np.uint64, count=n) * np.uint64(element_size)
+        return blob, offs.tobytes(), n
+
     def local_gpu_write_cache(
         self, cache: torch.Tensor, blocks: List[Tuple[str, int]], page_size: int
     ):
@@ -298,10 +311,12 @@ class InfinityConnection:
         self._verify(cache)
         assert self.local_connected
         element_size = cache.element_size()
-        blocks_in_bytes = [(key, offset * element_size) for key, offset in blocks]
-        ret = self.conn.rw_local(
+        blob, offs, n = self._pack_blocks(blocks, element_size)
+        ret = self.conn.rw_local_fast(
             self.OP_W,
-            blocks_in_bytes,
+            blob,
+            offs,
+            n,
             page_size * element_size,
             cache.data_ptr(),
             _remap_device_id(cache),
@@ -314,16 +329,19 @@ class InfinityConnection:
         """Read pages into `cache` (offsets in elements)."""
         self._verify(cache)
         element_size = cache.element_size()
-        blocks_in_bytes = [(key, offset * element_size) for key, offset in blocks]
         if self.local_connected:
-            ret = self.conn.rw_local(
+            blob, offs, n = self._pack_blocks(blocks, element_size)
+            ret = self.conn.rw_local_fast(
                 self.OP_R,
-                blocks_in_bytes,
+                blob,
+                offs,
+                n,
                 page_size * element_size,
                 cache.data_ptr(),
                 _remap_device_id(cache),
             )
         elif self.rdma_connected:
+            blocks_in_bytes = [(key, offset * element_size) for key, offset in blocks]
             ret = self.conn.r_rdma(
                 blocks_in_bytes, page_size * element_size, cache.data_ptr()
             )
