@@ -137,9 +137,24 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
     }
     gpu::IpcHandle handle;
     uint64_t base_offset = 0;
-    bool have_handle =
-        gpu::ipc_export(reinterpret_cast<void*>(ptr), &handle, &base_offset);
-    if (!have_handle) memset(handle.bytes, 0, gpu::kIpcHandleSize);
+    {
+        std::lock_guard<std::mutex> lk(ipc_mu_);
+        auto it = ipc_export_cache_.find(ptr);
+        if (it != ipc_export_cache_.end()) {
+            memcpy(handle.bytes, it->second.handle, gpu::kIpcHandleSize);
+            base_offset = it->second.base_offset;
+        } else {
+            bool have =
+                gpu::ipc_export(reinterpret_cast<void*>(ptr), &handle, &base_offset);
+            if (!have) memset(handle.bytes, 0, gpu::kIpcHandleSize);
+            if (ipc_export_cache_.size() > 4096) ipc_export_cache_.clear();
+            IpcExport ent;
+            memcpy(ent.handle, handle.bytes, gpu::kIpcHandleSize);
+            ent.base_offset = base_offset;
+            ent.have_handle = have;
+            ipc_export_cache_.emplace(ptr, ent);
+        }
+    }
 
     PackedLocalHdr h;
     h.device = device_id;

@@ -18,6 +18,7 @@
 #include <mutex>
 #include <string>
 #include <thread>
+#include <unordered_map>
 #include <utility>
 #include <vector>
 
@@ -119,6 +120,16 @@ class ClientConn {
     };
     std::vector<Region> regions_;
     std::mutex region_mu_;
+
+    // Cache of IPC exports keyed by exact data pointer (KV tensors are
+    // long-lived and re-exported every request otherwise).
+    struct IpcExport {
+        uint8_t handle[64];
+        uint64_t base_offset;
+        bool have_handle;
+    };
+    std::unordered_map<uintptr_t, IpcExport> ipc_export_cache_;
+    std::mutex ipc_mu_;
 
     std::thread worker_;
     std::deque<std::function<void()>> q_;
