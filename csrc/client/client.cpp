@@ -163,6 +163,10 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
     h.base_offset = base_offset;
     h.block_size = static_cast<uint32_t>(block_size);
     h.n_blocks = static_cast<uint32_t>(n);
+    // Reads complete in one round trip (response deferred to completion);
+    // writes stay async so uploads overlap compute (the prefill pattern).
+    h.flags = (op == 'R') ? kLocalFlagSyncResponse : 0;
+    h.rsvd = 0;
     memcpy(h.ipc, handle.bytes, gpu::kIpcHandleSize);
 
     std::vector<uint8_t> body(sizeof(h) + n * 8 + blob_len);
@@ -179,15 +183,20 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
         WARN("rw_local op=%c -> %d", op, code);
         return -code;
     }
+    if (op == 'W') local_dirty_ = true;  // a sync round trip is needed
     return 0;
 }
 
 int ClientConn::sync_local() {
     if (!connected_) return -1;
     std::lock_guard<std::mutex> lk(io_mu_);
+    // Reads complete before their response (kLocalFlagSyncResponse), so the
+    // sync round trip is only needed after writes.
+    if (!local_dirty_) return 0;
     if (!send_req(OP_SYNC, nullptr, 0)) return -1;
     int remain = -1;
     if (!recv_status(&remain)) return -1;
+    if (remain == 0) local_dirty_ = false;
     return remain;
 }
 

@@ -111,6 +111,7 @@ class ClientConn {
     int fd_ = -1;
     bool connected_ = false;
     bool rdma_connected_ = false;
+    bool local_dirty_ = false;  // writes since last drained sync (io_mu_)
     std::mutex io_mu_;
 
     struct Region {

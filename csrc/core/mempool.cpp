@@ -126,6 +126,18 @@ void* MemoryPool::allocate(size_t size) {
     return static_cast<uint8_t*>(base_) + start * block_size_;
 }
 
+void* MemoryPool::allocate_contiguous(size_t size, size_t n) {
+    if (size == 0 || n == 0) return nullptr;
+    size_t nb = (size + block_size_ - 1) / block_size_;
+    size_t total = nb * n;
+    if (total > n_blocks_) return nullptr;
+    size_t start;
+    if (!find_run(total, &start)) return nullptr;
+    mark(start, total, true);
+    used_blocks_ += total;
+    return static_cast<uint8_t*>(base_) + start * block_size_;
+}
+
 void* MemoryPool::allocate_below(size_t size, size_t limit_block) {
     if (size == 0 || size > size_) return nullptr;
     size_t nb = (size + block_size_ - 1) / block_size_;
@@ -207,6 +219,20 @@ int MM::add_pool(void* base, size_t size, size_t block_size, ArenaFree free_fn) 
 }
 
 bool MM::allocate(size_t size, size_t n, const AllocationCallback& cb) {
+    // Batch fast path: one contiguous run for the whole request.
+    if (n > 1) {
+        for (auto& pool : pools_) {
+            void* base = pool->allocate_contiguous(size, n);
+            if (base) {
+                size_t stride =
+                    ((size + pool->block_size() - 1) / pool->block_size()) * pool->block_size();
+                for (size_t i = 0; i < n; i++)
+                    cb(static_cast<uint8_t*>(base) + i * stride, pool->pool_idx());
+                return true;
+            }
+        }
+        // fall through to per-page allocation across pools
+    }
     struct Undo {
         void* ptr;
         int pool;

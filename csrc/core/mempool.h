@@ -32,6 +32,10 @@ class MemoryPool {
     // Allocate `size` bytes (rounded up to whole blocks, contiguous).
     // Returns nullptr if no run of free blocks is large enough.
     void* allocate(size_t size);
+    // Batch fast path: allocate n pages of `size` as ONE contiguous run
+    // (one bitmap scan instead of n). Returns the base or nullptr; pages sit
+    // at base + i*ceil(size/block)*block and may be freed individually.
+    void* allocate_contiguous(size_t size, size_t n);
     // Compaction helper: allocate only from block range [0, limit_block).
     void* allocate_below(size_t size, size_t limit_block);
     // Returns false on invalid pointer / double free.

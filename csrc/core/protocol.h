@@ -64,10 +64,16 @@ struct PackedLocalHdr {
     uint64_t base_offset;
     uint32_t block_size;
     uint32_t n_blocks;
+    uint32_t flags;
+    uint32_t rsvd;
     uint8_t ipc[64];
 };
 #pragma pack(pop)
-static_assert(sizeof(PackedLocalHdr) == 96, "packed local header size");
+static_assert(sizeof(PackedLocalHdr) == 104, "packed local header size");
+
+// PackedLocalHdr.flags: defer the response until the copy completes (one
+// round trip instead of request-ack + sync).
+constexpr uint32_t kLocalFlagSyncResponse = 1;
 
 std::string op_name(char op);
 

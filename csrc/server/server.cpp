@@ -332,6 +332,7 @@ bool parse_packed_local(const std::vector<uint8_t>& body, Server::LocalView* v) 
     v->base_ptr = h.base_ptr;
     v->base_offset = h.base_offset;
     v->block_size = h.block_size;
+    v->flags = h.flags;
     v->ipc = body.data() + offsetof(PackedLocalHdr, ipc);
     v->ipc_len = 64;
     const uint64_t* offs = reinterpret_cast<const uint64_t*>(body.data() + sizeof(h));
@@ -575,17 +576,23 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
     }
     n_reads_.fetch_add(1);
     bytes_out_.fetch_add(msg.blocks.size() * page);
-    if (jobs.empty()) return send_status(c, TASK_ACCEPTED);
+    bool sync_resp = (msg.flags & kLocalFlagSyncResponse) != 0;
+    if (jobs.empty()) return send_status(c, sync_resp ? FINISH : TASK_ACCEPTED);
 
     c->remain.fetch_add(1);
     c->ref();
     auto pending = std::make_shared<std::atomic<int>>(static_cast<int>(jobs.size()));
+    auto all_ok = std::make_shared<std::atomic<bool>>(true);
     for (auto& [shard, job] : jobs) {
         Shard::CopyJob j = std::move(job);
-        j.done = [this, c, held, pending](bool ok) {
-            (void)ok;
+        j.done = [this, c, held, pending, all_ok, sync_resp](bool ok) {
+            if (!ok) all_ok->store(false);
             if (pending->fetch_sub(1) == 1) {
-                post([this, c, held] { finish_task(c); });
+                post([this, c, held, all_ok, sync_resp] {
+                    if (sync_resp)
+                        send_status(c, all_ok->load() ? FINISH : INTERNAL_ERROR);
+                    finish_task(c);
+                });
             }
         };
         if (!shard->submit_copy(std::move(j))) {
@@ -595,7 +602,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
             }
         }
     }
-    send_status(c, TASK_ACCEPTED);
+    if (!sync_resp) send_status(c, TASK_ACCEPTED);
 }
 
 void Server::op_sync(Conn* c) {

@@ -102,6 +102,7 @@ class Server {
         uint64_t base_ptr = 0;
         uint64_t base_offset = 0;
         int64_t block_size = 0;
+        uint32_t flags = 0;
         const uint8_t* ipc = nullptr;
         size_t ipc_len = 0;
         std::vector<std::pair<std::string_view, uint64_t>> blocks;
