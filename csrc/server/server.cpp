@@ -1,6 +1,7 @@
 #include "server.h"
 
 #include <algorithm>
+#include <chrono>
 #include <cstring>
 #include <future>
 
@@ -169,6 +170,7 @@ bool Server::start() {
         uv_loop_close(&loop_);
         return false;
     }
+    kv_.reserve(1u << 20);  // avoid rehash storms during bulk prefill writes
     running_.store(true);
     stop_requested_.store(false);
     loop_thread_ = std::thread([this] { loop_main(); });
@@ -353,6 +355,23 @@ bool parse_packed_local(const std::vector<uint8_t>& body, Server::LocalView* v) 
 
 void Server::handle_request(Conn* c, char op, std::vector<uint8_t> body) {
     DEBUG("request op=%s body=%zu", op_name(op).c_str(), body.size());
+    struct Timer {
+        Server* s;
+        char op;
+        std::chrono::steady_clock::time_point t0 = std::chrono::steady_clock::now();
+        ~Timer() {
+            auto us = std::chrono::duration_cast<std::chrono::microseconds>(
+                          std::chrono::steady_clock::now() - t0)
+                          .count();
+            auto& st = s->op_stats_[static_cast<uint8_t>(op) & 127];
+            st.count.fetch_add(1, std::memory_order_relaxed);
+            st.total_us.fetch_add(static_cast<uint64_t>(us), std::memory_order_relaxed);
+            uint64_t prev = st.max_us.load(std::memory_order_relaxed);
+            while (static_cast<uint64_t>(us) > prev &&
+                   !st.max_us.compare_exchange_weak(prev, static_cast<uint64_t>(us))) {
+            }
+        }
+    } timer{this, op};
     switch (op) {
         case OP_W: {
             LocalMetaMsg msg;
@@ -878,20 +897,38 @@ size_t Server::purge() {
 
 std::string Server::stats_json() {
     char buf[1024];
-    size_t used = 0, total = 0;
+    size_t used = 0, total = 0, frag = 0;
     for (auto& s : shards_) {
         used += s->used_blocks();
         total += s->total_blocks();
+        frag = std::max(frag, s->largest_free_run_bytes());
     }
     snprintf(buf, sizeof(buf),
              "{\"kv_len\": %zu, \"shards\": %zu, \"used_blocks\": %zu, \"total_blocks\": %zu, "
              "\"writes\": %llu, \"reads\": %llu, \"puts\": %llu, \"gets\": %llu, "
-             "\"bytes_in\": %llu, \"bytes_out\": %llu}",
+             "\"bytes_in\": %llu, \"bytes_out\": %llu, \"largest_free_run_bytes\": %zu, "
+             "\"op_us\": {",
              kvmap_len(), shards_.size(), used, total,
              (unsigned long long)n_writes_.load(), (unsigned long long)n_reads_.load(),
              (unsigned long long)n_put_.load(), (unsigned long long)n_get_.load(),
-             (unsigned long long)bytes_in_.load(), (unsigned long long)bytes_out_.load());
-    return buf;
+             (unsigned long long)bytes_in_.load(), (unsigned long long)bytes_out_.load(), frag);
+    std::string out(buf);
+    bool first = true;
+    for (int i = 0; i < 128; i++) {
+        uint64_t cnt = op_stats_[i].count.load();
+        if (!cnt) continue;
+        char entry[160];
+        snprintf(entry, sizeof(entry),
+                 "%s\"%s\": {\"count\": %llu, \"avg_us\": %.1f, \"max_us\": %llu}",
+                 first ? "" : ", ", op_name(static_cast<char>(i)).c_str(),
+                 (unsigned long long)cnt,
+                 static_cast<double>(op_stats_[i].total_us.load()) / cnt,
+                 (unsigned long long)op_stats_[i].max_us.load());
+        out += entry;
+        first = false;
+    }
+    out += "}}";
+    return out;
 }
 
 }  // namespace ifs

@@ -167,6 +167,16 @@ class Server {
     std::atomic<uint64_t> n_writes_{0}, n_reads_{0}, n_put_{0}, n_get_{0};
     std::atomic<uint64_t> bytes_in_{0}, bytes_out_{0};
     std::atomic<int> extending_{0};
+
+    // per-op handler timing (loop-thread time, µs) — the per-op latency log
+    // the reference keeps via INFO prints (infinistore.cpp:1162-1166),
+    // aggregated instead of logged per request.
+    struct OpStat {
+        std::atomic<uint64_t> count{0};
+        std::atomic<uint64_t> total_us{0};
+        std::atomic<uint64_t> max_us{0};
+    };
+    OpStat op_stats_[128];
 };
 
 }  // namespace ifs
