@@ -98,15 +98,18 @@ def main():
         conn.register_mr(src)
         conn.register_mr(dst)
 
+    import numpy as np
+
     run_id = uuid.uuid4().hex[:8]
     offsets = [i * elems_per_block for i in range(args.blocks)]
+    offsets_np = np.asarray(offsets, dtype=np.uint64)
 
     def step_keys(step):
         return [f"r{rank}-s{step}-{run_id}-{i}" for i in range(args.blocks)]
 
     def do_put(keys):
         if have_gpu:
-            conn.local_gpu_write_cache(src, list(zip(keys, offsets)), elems_per_block)
+            conn.write_pages(src, keys, offsets_np, elems_per_block)
             conn.sync()
         else:
             blocks = conn.allocate_rdma(keys, block_bytes)
@@ -114,8 +117,12 @@ def main():
             conn.sync()
 
     def do_get(keys):
-        conn.read_cache(dst, list(zip(keys, offsets)), elems_per_block)
-        conn.sync()
+        if have_gpu:
+            conn.read_pages(dst, keys, offsets_np, elems_per_block)
+            conn.sync()
+        else:
+            conn.read_cache(dst, list(zip(keys, offsets)), elems_per_block)
+            conn.sync()
 
     def purge_all():
         if dist:
@@ -230,6 +237,7 @@ def main():
     if dist:
         dist.barrier()
     if rank == 0:
+        print("server stats:", ifs.get_server_stats(), file=sys.stderr)
         ifs.unregister_server()
     if dist:
         dist.destroy_process_group()

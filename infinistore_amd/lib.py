@@ -325,6 +325,43 @@ class InfinityConnection:
             raise Exception(f"Failed to write to infinistore, ret = {ret}")
         return 0
 
+    # -- vectorized page API (extension) ------------------------------------
+    # Same semantics as local_gpu_write_cache/read_cache but takes keys and
+    # offsets as parallel sequences; `offsets` may be a reusable
+    # np.ndarray(dtype=uint64) of ELEMENT offsets, skipping per-call tuple
+    # marshalling on the hot path.
+    def write_pages(self, cache: torch.Tensor, keys: List[str], offsets, page_size: int):
+        self._verify(cache)
+        assert self.local_connected, "write_pages uses the local GPU path"
+        es = cache.element_size()
+        blob = "\x00".join(keys).encode()
+        offs = np.asarray(offsets, dtype=np.uint64) * np.uint64(es)
+        ret = self.conn.rw_local_fast(
+            self.OP_W, blob, offs.tobytes(), len(keys), page_size * es,
+            cache.data_ptr(), _remap_device_id(cache),
+        )
+        if ret < 0:
+            raise Exception(f"Failed to write to infinistore, ret = {ret}")
+        return 0
+
+    def read_pages(self, cache: torch.Tensor, keys: List[str], offsets, page_size: int):
+        self._verify(cache)
+        es = cache.element_size()
+        if self.local_connected:
+            blob = "\x00".join(keys).encode()
+            offs = np.asarray(offsets, dtype=np.uint64) * np.uint64(es)
+            ret = self.conn.rw_local_fast(
+                self.OP_R, blob, offs.tobytes(), len(keys), page_size * es,
+                cache.data_ptr(), _remap_device_id(cache),
+            )
+        elif self.rdma_connected:
+            blocks = [(k, int(o) * es) for k, o in zip(keys, offsets)]
+            ret = self.conn.r_rdma(blocks, page_size * es, cache.data_ptr())
+        else:
+            raise Exception("Not connected to any instance")
+        if ret < 0:
+            raise Exception(f"Failed to read from infinistore, ret = {ret}")
+
     def read_cache(self, cache: torch.Tensor, blocks: List[Tuple[str, int]], page_size: int):
         """Read pages into `cache` (offsets in elements)."""
         self._verify(cache)
