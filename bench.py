@@ -40,6 +40,9 @@ def parse_args():
                    help="CPU-only mode (TCP fabric, DRAM pool) for dev boxes")
     p.add_argument("--latency-ops", type=int, default=200,
                    help="single-block round-trips for the p50/p99 measurement")
+    p.add_argument("--cross", action="store_true",
+                   help="each rank reads keys written by the next rank "
+                        "(forces xGMI cross-shard traffic)")
     return p.parse_args()
 
 
@@ -83,6 +86,9 @@ def main():
 
     if have_gpu:
         torch.cuda.set_device(local_rank)
+    # Same seed on every rank: src contents are identical across ranks, so
+    # cross-rank reads (--cross) can still be verified against local src.
+    torch.manual_seed(12345)
     src = torch.randn(total_elems, dtype=torch.bfloat16, device=dev)
     dst = torch.zeros_like(src)
 
@@ -101,11 +107,18 @@ def main():
     import numpy as np
 
     run_id = uuid.uuid4().hex[:8]
+    if dist:  # one shared run id so ranks can name each other's keys
+        obj = [run_id]
+        dist.broadcast_object_list(obj, src=0)
+        run_id = obj[0]
     offsets = [i * elems_per_block for i in range(args.blocks)]
     offsets_np = np.asarray(offsets, dtype=np.uint64)
 
-    def step_keys(step):
-        return [f"r{rank}-s{step}-{run_id}-{i}" for i in range(args.blocks)]
+    read_rank = (rank + 1) % world if (args.cross and world > 1) else rank
+
+    def step_keys(step, owner=None):
+        owner = rank if owner is None else owner
+        return [f"r{owner}-s{step}-{run_id}-{i}" for i in range(args.blocks)]
 
     def do_put(keys):
         if have_gpu:
@@ -138,11 +151,14 @@ def main():
         if dist:
             dist.barrier()
 
+    cross = args.cross and world > 1
+
     # ---- correctness spot-check + warmup ----
     for w in range(args.warmup):
-        keys = [f"warm-{k}" for k in step_keys(w)]
-        do_put(keys)
-        do_get(keys)
+        do_put([f"warm-{k}" for k in step_keys(w)])
+        if cross:
+            dist.barrier()
+        do_get([f"warm-{k}" for k in step_keys(w, read_rank)])
     if not torch.equal(src.cpu(), dst.cpu()):
         print(json.dumps({"error": "data mismatch in warmup"}))
         sys.exit(1)
@@ -154,12 +170,13 @@ def main():
     put_time = 0.0
     get_time = 0.0
     for s in range(args.steps):
-        keys = step_keys(s)
         tp = time.perf_counter()
-        do_put(keys)
+        do_put(step_keys(s))
         put_time += time.perf_counter() - tp
+        if cross:
+            dist.barrier()  # readers wait for the writer of their keys
         tg = time.perf_counter()
-        do_get(keys)
+        do_get(step_keys(s, read_rank))
         get_time += time.perf_counter() - tg
     sync_all()
     elapsed = time.perf_counter() - t0
@@ -174,7 +191,7 @@ def main():
 
     # ---- latency phase (single-block round trips) ----
     lat_put, lat_get = [], []
-    lkeys = [f"lat-{run_id}-{i}" for i in range(args.latency_ops)]
+    lkeys = [f"lat-r{rank}-{run_id}-{i}" for i in range(args.latency_ops)]
     for i in range(args.latency_ops):
         t1 = time.perf_counter()
         if have_gpu:

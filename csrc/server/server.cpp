@@ -1,5 +1,7 @@
 #include "server.h"
 
+#include "../core/crash.h"
+
 #include <algorithm>
 #include <chrono>
 #include <cstring>
@@ -129,6 +131,7 @@ Server::~Server() { stop(); }
 
 bool Server::start() {
     if (running_.load()) return true;
+    install_crash_handlers();
     for (auto& s : shards_) {
         if (!s->init()) {
             ERROR("shard init failed");

@@ -256,3 +256,51 @@ def test_bench_smoke():
     out = json.loads(line)
     assert out["metric"] == "put_get_GBps" and out["value"] > 0
     assert out["config"]["path"] == "local_gpu_ipc"
+
+
+def test_cross_gpu_read(gpu_server):
+    """Disaggregated pattern over xGMI: writer on GPU 0, reader on GPU 1 —
+    the pool shard's copy kernel pushes into peer memory. Runs on the
+    multi-GPU round-end box; skipped on 1-GPU boxes."""
+    if torch.cuda.device_count() < 2:
+        pytest.skip("needs >= 2 GPUs")
+    wconn = local_conn(gpu_server)
+    rconn = local_conn(gpu_server)
+    try:
+        n = 256 * 1024
+        page = 64 * 1024
+        src = torch.randn(n, device="cuda:0")
+        dst = torch.zeros(n, device="cuda:1")
+        pre = uuid.uuid4().hex
+        keys = [f"{pre}-{i}" for i in range(n // page)]
+        offs = [i * page for i in range(n // page)]
+        wconn.local_gpu_write_cache(src, list(zip(keys, offs)), page)
+        wconn.sync()
+        rconn.read_cache(dst, list(zip(keys, offs)), page)
+        rconn.sync()
+        assert torch.equal(src.cpu(), dst.cpu())
+    finally:
+        wconn.close()
+        rconn.close()
+
+
+def test_cross_gpu_write(gpu_server):
+    """Writer tensor on GPU 1, shard affinity routes to shard 1; reader on
+    GPU 0 pulls across xGMI."""
+    if torch.cuda.device_count() < 2:
+        pytest.skip("needs >= 2 GPUs")
+    wconn = local_conn(gpu_server)
+    rconn = local_conn(gpu_server)
+    try:
+        n = 64 * 1024
+        src = torch.randn(n, device="cuda:1")
+        dst = torch.zeros(n, device="cuda:0")
+        key = f"xw-{uuid.uuid4()}"
+        wconn.local_gpu_write_cache(src, [(key, 0)], n)
+        wconn.sync()
+        rconn.read_cache(dst, [(key, 0)], n)
+        rconn.sync()
+        assert torch.equal(src.cpu(), dst.cpu())
+    finally:
+        wconn.close()
+        rconn.close()
