@@ -42,6 +42,10 @@ def parse_args():
                    help="concurrent client processes (saturation mode)")
     p.add_argument("--verify", action="store_true")
     p.add_argument("--json", action="store_true", help="print a JSON summary")
+    p.add_argument("--spawn-server", action="store_true",
+                   help="start a local server subprocess for the benchmark")
+    p.add_argument("--prealloc-size", type=int, default=8,
+                   help="pool GB per shard for --spawn-server")
     return p.parse_args()
 
 
@@ -146,8 +150,33 @@ def _worker(args_dict, q):
         conn.close()
 
 
+def _spawn_server(args):
+    import socket
+    import subprocess
+    import sys
+    import time as _t
+
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "infinistore_amd.server",
+         "--service-port", str(args.port),
+         "--manage-port", str(args.port + 1),
+         "--prealloc-size", str(args.prealloc_size),
+         "--minimal-allocate-size", str(args.block_size),
+         "--no-manage"],
+    )
+    t0 = _t.time()
+    while _t.time() - t0 < 60:
+        try:
+            socket.create_connection(("127.0.0.1", args.port), timeout=1).close()
+            return proc
+        except OSError:
+            _t.sleep(0.3)
+    raise RuntimeError("spawned server did not come up")
+
+
 def main():
     args = parse_args()
+    server_proc = _spawn_server(args) if args.spawn_server else None
     local = args.local_gpu and torch.cuda.is_available()
     device = f"cuda:{args.src_gpu}" if local else "cpu"
 
@@ -171,6 +200,9 @@ def main():
         print(f"saturation: {args.clients} clients, aggregate {agg:.2f} MB/s "
               f"(per-client write {statistics.mean(w for w, _ in results):.2f} MB/s, "
               f"read {statistics.mean(r for _, r in results):.2f} MB/s)")
+        if server_proc is not None:
+            server_proc.terminate()
+            server_proc.wait(timeout=20)
         return
 
     conn = make_conn(args, local)
@@ -203,6 +235,9 @@ def main():
             print(json.dumps(summary))
     finally:
         conn.close()
+        if server_proc is not None:
+            server_proc.terminate()
+            server_proc.wait(timeout=20)
 
 
 if __name__ == "__main__":
