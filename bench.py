@@ -54,6 +54,8 @@ def main():
     n_gpus = max(args.gpus, world)
 
     have_gpu = torch.cuda.is_available() and not args.cpu
+    if have_gpu:
+        local_rank = local_rank % torch.cuda.device_count()
     dist = None
     if world > 1:
         import torch.distributed as tdist
@@ -67,13 +69,14 @@ def main():
 
     # Rank 0 hosts the server, sharded over all visible GPUs.
     if rank == 0:
+        n_shards = min(n_gpus, torch.cuda.device_count()) if have_gpu else 0
         scfg = ifs.ServerConfig(
             service_port=port,
             manage_port=port + 1,
             prealloc_size=args.pool_gb,
             minimal_allocate_size=args.block_kb,
             cpu_only=not have_gpu,
-            devices=list(range(n_gpus)) if have_gpu else [],
+            devices=list(range(n_shards)) if have_gpu else [],
         )
         ifs.register_server(scfg)
     if dist:

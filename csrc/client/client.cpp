@@ -62,6 +62,11 @@ int ClientConn::init_connection(const ClientConfigC& cfg) {
     freeaddrinfo(res);
     int one = 1;
     setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+    // Failure detection: a dead/stuck server surfaces as an op error after
+    // 60 s instead of hanging the caller forever (the reference's client
+    // blocks indefinitely on its sockets).
+    struct timeval tv{60, 0};
+    setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
     fd_ = fd;
     connected_ = true;
     return 0;

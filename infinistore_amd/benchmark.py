@@ -40,6 +40,13 @@ def parse_args():
                    help="additionally measure N single-page round-trips")
     p.add_argument("--clients", type=int, default=1,
                    help="concurrent client processes (saturation mode)")
+    p.add_argument("--shape", choices=["llama3-8b", "llama3-70b"],
+                   help="preset: per-layer paged-KV shapes (16-token pages, "
+                        "8 KV heads x 128 dim, bf16 -> 64 KB pages; 8B=32 "
+                        "layers, 70B=80 layers); --steps becomes n_layers and "
+                        "--size is derived from --seq-len x --batch")
+    p.add_argument("--seq-len", type=int, default=8192)
+    p.add_argument("--batch", type=int, default=1)
     p.add_argument("--verify", action="store_true")
     p.add_argument("--json", action="store_true", help="print a JSON summary")
     p.add_argument("--spawn-server", action="store_true",
@@ -174,8 +181,28 @@ def _spawn_server(args):
     raise RuntimeError("spawned server did not come up")
 
 
+SHAPES = {
+    # (n_layers, kv_page_bytes): 16-token pages, 8 KV heads, head_dim 128,
+    # bf16, K+V -> 2*16*8*128*2 = 64 KiB per layer-page (Llama-3 GQA).
+    "llama3-8b": (32, 64 << 10),
+    "llama3-70b": (80, 64 << 10),
+}
+
+
+def apply_shape(args):
+    n_layers, page_bytes = SHAPES[args.shape]
+    pages_per_layer = (args.seq_len + 15) // 16 * args.batch
+    args.block_size = page_bytes >> 10
+    args.steps = n_layers  # one layer-batch of pages per step (prefill order)
+    args.size = n_layers * pages_per_layer * page_bytes >> 20
+    print(f"shape {args.shape}: {n_layers} layers x {pages_per_layer} pages "
+          f"x {page_bytes >> 10} KB = {args.size} MB per iteration")
+
+
 def main():
     args = parse_args()
+    if args.shape:
+        apply_shape(args)
     server_proc = _spawn_server(args) if args.spawn_server else None
     local = args.local_gpu and torch.cuda.is_available()
     device = f"cuda:{args.src_gpu}" if local else "cpu"
