@@ -748,20 +748,29 @@ void Server::op_tcp_get(Conn* c, const RemoteMetaMsg& msg) {
             entries.push_back(it->second);
         }
     }
-    std::vector<uint8_t> payload(entries.size() * page);
+    // Build the framed response (status + len + blocks) in ONE buffer: no
+    // intermediate payload copy (large gets were memory-bound on redundant
+    // buffer fills).
+    size_t total = entries.size() * page;
+    std::vector<uint8_t> resp(8 + total);
+    int code = FINISH;
+    uint32_t len32 = static_cast<uint32_t>(total);
+    memcpy(resp.data(), &code, 4);
+    memcpy(resp.data() + 4, &len32, 4);
+    uint8_t* out = resp.data() + 8;
     for (size_t i = 0; i < entries.size(); i++) {
         BlockEntry* e = entries[i].get();
         size_t nbytes = std::min(page, e->size);
         if (e->shard->on_gpu()) {
-            if (!gpu::memcpy_d2h(payload.data() + i * page, e->ptr, nbytes))
+            if (!gpu::memcpy_d2h(out + i * page, e->ptr, nbytes))
                 return send_status(c, INTERNAL_ERROR);
         } else {
-            memcpy(payload.data() + i * page, e->ptr, nbytes);
+            memcpy(out + i * page, e->ptr, nbytes);
         }
     }
     n_get_.fetch_add(1);
-    bytes_out_.fetch_add(payload.size());
-    send_status_payload(c, FINISH, payload.data(), payload.size());
+    bytes_out_.fetch_add(total);
+    send_buf(c, std::move(resp));
 }
 
 // ---- queries ---------------------------------------------------------------
