@@ -168,21 +168,48 @@ def main():
     purge_all()
 
     # ---- timed region ----
+    debug_t = {"keygen": 0.0, "put_req": 0.0, "put_sync": 0.0, "get_req": 0.0,
+               "get_sync": 0.0}
+    debug = os.environ.get("IFS_BENCH_DEBUG")
+
     sync_all()
     t0 = time.perf_counter()
     put_time = 0.0
     get_time = 0.0
     for s in range(args.steps):
         tp = time.perf_counter()
-        do_put(step_keys(s))
+        if debug and have_gpu:
+            keys = step_keys(s)
+            ta = time.perf_counter()
+            conn.write_pages(src, keys, offsets_np, elems_per_block)
+            tb = time.perf_counter()
+            conn.sync()
+            tc = time.perf_counter()
+            debug_t["keygen"] += ta - tp
+            debug_t["put_req"] += tb - ta
+            debug_t["put_sync"] += tc - tb
+        else:
+            do_put(step_keys(s))
         put_time += time.perf_counter() - tp
         if cross:
             dist.barrier()  # readers wait for the writer of their keys
         tg = time.perf_counter()
-        do_get(step_keys(s, read_rank))
+        if debug and have_gpu:
+            keys = step_keys(s, read_rank)
+            ta = time.perf_counter()
+            conn.read_pages(dst, keys, offsets_np, elems_per_block)
+            tb = time.perf_counter()
+            conn.sync()
+            debug_t["get_req"] += tb - ta
+            debug_t["get_sync"] += time.perf_counter() - tb
+        else:
+            do_get(step_keys(s, read_rank))
         get_time += time.perf_counter() - tg
     sync_all()
     elapsed = time.perf_counter() - t0
+    if debug:
+        per = {k: round(v / args.steps * 1e6, 1) for k, v in debug_t.items()}
+        print(f"rank {rank} per-step us: {per}", file=sys.stderr)
 
     # max over ranks; bytes summed over ranks
     if dist:
