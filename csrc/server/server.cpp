@@ -561,11 +561,11 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
             finish_task(c);
         });
     };
-    if (!shard->submit_copy(std::move(job))) {
-        finish_task(c);
-        return send_status(c, INTERNAL_ERROR);
-    }
+    // Respond before submitting: the client's next action (more writes or a
+    // sync) overlaps the descriptor upload + kernel launch. A submit failure
+    // (shutdown) is reported through the rollback path; readers see 404.
     send_status(c, TASK_ACCEPTED);
+    if (!shard->submit_copy(std::move(job))) finish_task(c);
 }
 
 void Server::op_local_read(Conn* c, const LocalView& msg) {
