@@ -1,0 +1,119 @@
+"""Concurrency / robustness stress tests (CPU, TCP fabric)."""
+
+import socket
+import struct
+import threading
+import uuid
+
+import pytest
+import torch
+
+import infinistore_amd as ifs
+from conftest import make_client
+
+
+def test_many_threads_one_server(cpu_server):
+    """8 threads × separate connections hammer mixed ops concurrently."""
+    errors = []
+
+    def worker(tid):
+        try:
+            conn = make_client(cpu_server)
+            src = torch.full((2048,), float(tid))
+            dst = torch.zeros(2048)
+            conn.register_mr(src)
+            conn.register_mr(dst)
+            pre = f"t{tid}-{uuid.uuid4().hex}"
+            for it in range(5):
+                keys = [f"{pre}-{it}-{i}" for i in range(4)]
+                blocks = conn.allocate_rdma(keys, 512 * 4)
+                conn.rdma_write_cache(src, [0, 512, 1024, 1536], 512, blocks)
+                conn.sync()
+                conn.read_cache(dst, list(zip(keys, [0, 512, 1024, 1536])), 512)
+                conn.sync()
+                assert torch.equal(src, dst)
+                assert conn.get_match_last_index(keys) == 3
+                conn.delete_keys(keys[2:])
+                assert not conn.check_exist(keys[3])
+            conn.close()
+        except Exception as e:  # pragma: no cover
+            errors.append(f"t{tid}: {e}")
+
+    threads = [threading.Thread(target=worker, args=(t,)) for t in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+    assert not errors, errors
+
+
+def test_async_ops_interleaved(cpu_server):
+    """Async writes from the worker thread interleaved with sync queries on
+    the caller thread (the vLLM usage pattern)."""
+    import asyncio
+
+    async def run():
+        cfg = ifs.ClientConfig(
+            host_addr="127.0.0.1", service_port=cpu_server,
+            connection_type=ifs.TYPE_RDMA, link_type="TCP",
+        )
+        conn = ifs.InfinityConnection(cfg)
+        await conn.connect_async()
+        try:
+            src = torch.arange(8192, dtype=torch.float32)
+            conn.register_mr(src)
+            pre = uuid.uuid4().hex
+            futs = []
+            for it in range(8):
+                keys = [f"{pre}-{it}-{i}" for i in range(4)]
+                blocks = await conn.allocate_rdma_async(keys, 512 * 4)
+                futs.append(conn.rdma_write_cache_async(
+                    src, [0, 512, 1024, 1536], 512, blocks))
+                conn.check_exist(f"{pre}-0-0")  # interleaved sync query
+            await asyncio.gather(*futs)
+            conn.sync()
+            assert conn.check_exist(f"{pre}-7-3")
+        finally:
+            conn.close()
+
+    asyncio.run(run())
+
+
+def test_bad_magic_closes_connection(cpu_server):
+    s = socket.create_connection(("127.0.0.1", cpu_server), timeout=5)
+    s.sendall(struct.pack("<IcI", 0x12345678, b"S", 0))
+    # server should close on bad magic
+    s.settimeout(5)
+    assert s.recv(4) == b""
+    s.close()
+
+
+def test_oversized_body_rejected(cpu_server):
+    s = socket.create_connection(("127.0.0.1", cpu_server), timeout=5)
+    s.sendall(struct.pack("<IcI", 0xDEADBEEF, b"M", 1 << 30))
+    s.settimeout(5)
+    assert s.recv(4) == b""
+    s.close()
+
+
+def test_partial_header_then_disconnect(cpu_server):
+    # half a header, then hang up — the server must survive.
+    s = socket.create_connection(("127.0.0.1", cpu_server), timeout=5)
+    s.sendall(b"\xef\xbe\xad")
+    s.close()
+    # server still serves afterwards
+    conn = make_client(cpu_server)
+    assert conn.check_exist("still-alive-" + uuid.uuid4().hex) is False
+    conn.close()
+
+
+def test_garbage_flatbuffer_body(cpu_server):
+    s = socket.create_connection(("127.0.0.1", cpu_server), timeout=5)
+    body = b"\xff" * 64
+    s.sendall(struct.pack("<IcI", 0xDEADBEEF, b"M", len(body)) + body)
+    s.settimeout(5)
+    code = struct.unpack("<i", s.recv(4))[0]
+    assert code < 0 or code == 400  # error, but connection alive
+    s.sendall(struct.pack("<IcI", 0xDEADBEEF, b"S", 0))
+    assert struct.unpack("<i", s.recv(4))[0] == 0
+    s.close()
