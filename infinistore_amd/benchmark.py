@@ -145,14 +145,18 @@ def pct(v, q):
     return statistics.quantiles(v, n=100)[q - 1] if len(v) >= 10 else max(v)
 
 
-def _worker(args_dict, q):
+def _worker(args_dict, q, barrier, wid):
     ns = argparse.Namespace(**args_dict)
     local = ns.local_gpu and torch.cuda.is_available()
-    device = f"cuda:{ns.src_gpu}" if local else "cpu"
+    n_dev = torch.cuda.device_count() if torch.cuda.is_available() else 1
+    device = f"cuda:{(ns.src_gpu + wid) % n_dev}" if local else "cpu"
     conn = make_conn(ns, local)
     try:
+        barrier.wait(timeout=300)  # start all clients together (steady state)
+        t0 = time.perf_counter()
         w, r = run_once(ns, conn, local, device)
-        q.put((w, r))
+        wall = time.perf_counter() - t0
+        q.put((w, r, wall))
     finally:
         conn.close()
 
@@ -212,21 +216,21 @@ def main():
 
         ctx = multiprocessing.get_context("spawn")
         q = ctx.Queue()
-        t0 = time.perf_counter()
+        barrier = ctx.Barrier(args.clients)
         procs = [
-            ctx.Process(target=_worker, args=(vars(args), q))
-            for _ in range(args.clients)
+            ctx.Process(target=_worker, args=(vars(args), q, barrier, wid))
+            for wid in range(args.clients)
         ]
         for p in procs:
             p.start()
         results = [q.get(timeout=600) for _ in procs]
         for p in procs:
             p.join()
-        wall = time.perf_counter() - t0
+        wall = max(r[2] for r in results)  # steady-state window (post-barrier)
         agg = args.clients * (args.size << 20) * 2 / wall / 1e6
         print(f"saturation: {args.clients} clients, aggregate {agg:.2f} MB/s "
-              f"(per-client write {statistics.mean(w for w, _ in results):.2f} MB/s, "
-              f"read {statistics.mean(r for _, r in results):.2f} MB/s)")
+              f"(per-client write {statistics.mean(r[0] for r in results):.2f} MB/s, "
+              f"read {statistics.mean(r[1] for r in results):.2f} MB/s)")
         if server_proc is not None:
             server_proc.terminate()
             server_proc.wait(timeout=20)
