@@ -14,9 +14,11 @@
 #include "../core/log.h"
 #include "../core/utils.h"
 #include "../gpu/gpu.h"
+#include "client_verbs.h"
 
 namespace ifs {
 
+ClientConn::ClientConn() = default;
 ClientConn::~ClientConn() { close_conn(); }
 
 void ClientConn::close_conn() {
@@ -27,6 +29,7 @@ void ClientConn::close_conn() {
     q_cv_.notify_all();
     if (worker_.joinable()) worker_.join();
     worker_started_ = false;
+    verbs_.reset();
     if (fd_ >= 0) {
         ::close(fd_);
         fd_ = -1;
@@ -73,17 +76,23 @@ int ClientConn::init_connection(const ClientConfigC& cfg) {
 }
 
 int ClientConn::setup_rdma(const ClientConfigC& cfg) {
-    (void)cfg;
     if (!connected_) return -1;
     std::lock_guard<std::mutex> lk(io_mu_);
-    if (!send_req(OP_RDMA_EXCHANGE, nullptr, 0)) return -1;
-    int code = 0;
-    if (!recv_status(&code) || code != FINISH) return -1;
-    std::vector<uint8_t> tag;
-    if (!recv_payload(&tag) || tag.size() != 4) return -1;
-    if (memcmp(tag.data(), "TCPF", 4) != 0) {
-        ERROR("unknown fabric tag");
-        return -1;
+    // Prefer the verbs fabric when rdma-core + a NIC are available on both
+    // ends; otherwise (or on any handshake failure) the TCP data fabric
+    // serves the same API.
+    bool attempted = false;
+    verbs_ = VerbsClient::establish(fd_, cfg, &attempted);
+    if (!verbs_ && !attempted) {
+        if (!send_req(OP_RDMA_EXCHANGE, nullptr, 0)) return -1;
+        int code = 0;
+        if (!recv_status(&code) || code != FINISH) return -1;
+        std::vector<uint8_t> tag;
+        if (!recv_payload(&tag) || tag.size() < 4) return -1;
+        if (memcmp(tag.data(), "TCPF", 4) != 0) {
+            ERROR("unknown fabric tag");
+            return -1;
+        }
     }
     rdma_connected_ = true;
     // Spawn the async worker lazily on first connect.
@@ -215,6 +224,7 @@ int ClientConn::register_mr(uintptr_t ptr, size_t size) {
         // check (libinfinistore.cpp:1168)
         dev = gpu::is_device_pointer(reinterpret_cast<const void*>(ptr));
     }
+    if (verbs_ && !verbs_->register_mr(reinterpret_cast<void*>(ptr), size, dev)) return -1;
     std::lock_guard<std::mutex> lk(region_mu_);
     regions_.push_back({ptr, size, dev});
     return 1;
@@ -233,6 +243,13 @@ bool ClientConn::is_device_ptr(uintptr_t ptr) {
 
 std::vector<RemoteBlockOut> ClientConn::do_allocate(const std::vector<std::string>& keys,
                                                     int block_size) {
+    if (verbs_) {
+        auto res = verbs_->allocate(keys, block_size);
+        std::vector<RemoteBlockOut> out;
+        out.reserve(res.size());
+        for (auto& r : res) out.push_back({r.first, r.second});
+        return out;
+    }
     RemoteMetaMsg msg;
     msg.keys = keys;
     msg.block_size = block_size;
@@ -275,6 +292,14 @@ int ClientConn::allocate_rdma_async(const std::vector<std::string>& keys, int bl
 int ClientConn::do_w_rdma(const uint64_t* offsets, size_t n_offsets, int block_size,
                           const RemoteBlockOut* blocks, size_t n_blocks, uintptr_t base_ptr) {
     if (n_offsets != n_blocks) return -1;
+    if (verbs_) {
+        std::vector<std::pair<uint32_t, uint64_t>> blks;
+        blks.reserve(n_blocks);
+        for (size_t i = 0; i < n_blocks; i++)
+            blks.push_back({blocks[i].rkey, blocks[i].remote_addr});
+        return verbs_->write_blocks(offsets, n_offsets, block_size, blks.data(), blks.size(),
+                                    base_ptr);
+    }
     bool dev = is_device_ptr(base_ptr);
     size_t bs = static_cast<size_t>(block_size);
 
@@ -359,6 +384,7 @@ int ClientConn::w_rdma_async(const uint64_t* offsets, size_t n_offsets, int bloc
 
 int ClientConn::do_r_rdma(const std::vector<std::pair<std::string, uint64_t>>& blocks,
                           int block_size, uintptr_t base_ptr) {
+    if (verbs_) return verbs_->read_blocks(blocks, block_size, base_ptr);
     bool dev = is_device_ptr(base_ptr);
     size_t bs = static_cast<size_t>(block_size);
     RemoteMetaMsg msg;

@@ -15,6 +15,7 @@
 #include <cstdint>
 #include <deque>
 #include <functional>
+#include <memory>
 #include <mutex>
 #include <string>
 #include <thread>
@@ -25,6 +26,8 @@
 #include "../core/protocol.h"
 
 namespace ifs {
+
+class VerbsClient;  // client_verbs.h
 
 struct ClientConfigC {
     std::string host_addr;
@@ -43,7 +46,7 @@ struct RemoteBlockOut {
 
 class ClientConn {
    public:
-    ClientConn() = default;
+    ClientConn();  // out-of-line: members hold incomplete types (VerbsClient)
     ~ClientConn();
     ClientConn(const ClientConn&) = delete;
     ClientConn& operator=(const ClientConn&) = delete;
@@ -113,6 +116,7 @@ class ClientConn {
     bool rdma_connected_ = false;
     bool local_dirty_ = false;  // writes since last drained sync (io_mu_)
     std::mutex io_mu_;
+    std::unique_ptr<VerbsClient> verbs_;  // non-null when the verbs fabric won
 
     struct Region {
         uintptr_t ptr;

@@ -59,6 +59,11 @@ bool enable_peer_access(int dev, int peer);
 // HIP analog of the reference's host-vs-device MR classification.
 bool is_device_pointer(const void* ptr);
 
+// Export a dmabuf fd covering [ptr, ptr+size) of device memory (the amdgpu
+// path for registering HBM with an RDMA NIC: ibv_reg_dmabuf_mr). Returns
+// the fd or -1; *offset is the start offset inside the dmabuf.
+int export_dmabuf(void* ptr, size_t size, uint64_t* offset);
+
 // --- streams / events (opaque wrappers so non-HIP TUs can hold them) -------
 using Event = void*;
 Stream stream_create(int dev);

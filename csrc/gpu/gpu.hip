@@ -168,6 +168,24 @@ bool is_device_pointer(const void* ptr) {
     return attr.type == hipMemoryTypeDevice;
 }
 
+int export_dmabuf(void* ptr, size_t size, uint64_t* offset) {
+    *offset = 0;
+#if HIP_VERSION >= 60000000 || defined(hipMemRangeHandleTypeDmaBufFd)
+    int fd = -1;
+    hipError_t e = hipMemGetHandleForAddressRange(&fd, reinterpret_cast<hipDeviceptr_t>(ptr),
+                                                  size, hipMemRangeHandleTypeDmaBufFd, 0);
+    if (e != hipSuccess) {
+        fail("hipMemGetHandleForAddressRange", e);
+        return -1;
+    }
+    return fd;
+#else
+    (void)ptr;
+    (void)size;
+    return -1;
+#endif
+}
+
 bool enable_peer_access(int dev, int peer) {
     if (dev == peer) return true;
     int can = 0;

@@ -15,6 +15,7 @@
 
 #include "client/client.h"
 #include "core/log.h"
+#include "fabric/wr_flow.h"
 #include "core/protocol.h"
 #include "core/mempool.h"
 #include "gpu/gpu.h"
@@ -65,6 +66,9 @@ bool start_server(const ServerConfigPy& cfg) {
     opt.n_streams = cfg.num_stream > 0 ? cfg.num_stream : 4;
     opt.cpu_shards = cfg.cpu_shards;
     opt.log_level = cfg.log_level;
+    opt.dev_name = cfg.dev_name;
+    opt.ib_port = cfg.ib_port;
+    opt.link_type = cfg.link_type;
     if (!cfg.cpu_only && gpu::available()) {
         if (!cfg.devices.empty()) {
             opt.devices = cfg.devices;
@@ -349,6 +353,35 @@ PYBIND11_MODULE(_native, m) {
         if (!parse_match_request(reinterpret_cast<const uint8_t*>(s.data()), s.size(), &keys))
             throw std::runtime_error("parse failed");
         return keys;
+    });
+
+    // WR flow-control simulation (chain split + outstanding cap + overflow).
+    m.def("_dbg_wrflow_sim", [](int n_wrs, int batch, int cap, int complete_after) {
+        std::deque<size_t> inflight_chains;
+        std::vector<size_t> posted_sizes;
+        long peak = 0;
+        WrFlow* flow_ptr = nullptr;
+        WrFlow flow(
+            [&](const WrChain& ch) {
+                posted_sizes.push_back(ch.wrs.size());
+                inflight_chains.push_back(ch.wrs.size());
+                if (flow_ptr && flow_ptr->outstanding() > peak) peak = flow_ptr->outstanding();
+                return true;
+            },
+            batch, cap);
+        flow_ptr = &flow;
+        (void)complete_after;
+        std::vector<WrDesc> wrs(static_cast<size_t>(n_wrs), WrDesc{0, 0, 0, 0, 0});
+        flow.submit(std::move(wrs), true, 42, 7);
+        size_t parked_peak = flow.parked();
+        // Deliver completions FIFO; draining may post more chains.
+        while (!inflight_chains.empty()) {
+            size_t n = inflight_chains.front();
+            inflight_chains.pop_front();
+            flow.on_chain_complete(n);
+        }
+        return py::make_tuple(posted_sizes, peak, flow.outstanding(), parked_peak,
+                              flow.parked());
     });
 
     // Mempool (host-backed) for allocator unit tests.

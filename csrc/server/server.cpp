@@ -11,38 +11,6 @@
 
 namespace ifs {
 
-// ---------------------------------------------------------------------------
-// Connection state
-// ---------------------------------------------------------------------------
-struct Server::Conn : RefCounted {
-    Server* srv = nullptr;
-    uv_tcp_t tcp;
-    bool closed = false;
-
-    // read state machine
-    enum State { kHeader, kBody } state = kHeader;
-    Header hdr{};
-    std::vector<uint8_t> buf;  // accumulated bytes
-
-    // local path: in-flight async copy count. OP_SYNC blocks server-side:
-    // when remain>0 the response is deferred until the count drains to zero
-    // (the reference instead has the client poll with sleeps,
-    // lib.py:578-592 — ~0.5 ms of added latency per op there).
-    std::atomic<int> remain{0};
-    bool sync_waiting = false;  // loop thread only
-
-    // cached IPC mappings: handle bytes -> base pointer (closed on disconnect)
-    std::map<std::vector<uint8_t>, std::pair<void*, int>> ipc_cache;  // base, src_dev
-
-    // TCP-fabric: blocks allocated for this conn, not yet committed.
-    std::unordered_map<uint64_t, Ref<BlockEntry>> pending_rdma;
-
-    ~Conn() override {
-        for (auto& kv : ipc_cache) {
-            if (gpu::available()) gpu::ipc_close(kv.second.first);
-        }
-    }
-};
 
 namespace {
 
@@ -199,7 +167,10 @@ void Server::stop() {
 void Server::on_stop_async(uv_async_t* h) {
     auto* srv = static_cast<Server*>(h->data);
     // Close all connections, listener and asyncs; loop exits when no handles.
-    for (auto* c : srv->conns_) conn_close(c);
+    for (auto* c : srv->conns_) {
+        srv->verbs_teardown(c);
+        conn_close(c);
+    }
     srv->conns_.clear();
     uv_close(reinterpret_cast<uv_handle_t*>(&srv->listener_), nullptr);
     uv_close(reinterpret_cast<uv_handle_t*>(&srv->post_async_), nullptr);
@@ -256,6 +227,7 @@ void Server::on_new_connection(uv_stream_t* server, int status) {
                 free(buf->base);
                 auto& v = srv->conns_;
                 v.erase(std::remove(v.begin(), v.end(), c), v.end());
+                srv->verbs_teardown(c);
                 conn_close(c);
                 return;
             }
@@ -277,6 +249,7 @@ void Server::on_new_connection(uv_stream_t* server, int status) {
                         WARN("bad magic from client; closing");
                         auto& v = srv->conns_;
                         v.erase(std::remove(v.begin(), v.end(), c), v.end());
+                        srv->verbs_teardown(c);
                         conn_close(c);
                         return;
                     }
@@ -285,6 +258,7 @@ void Server::on_new_connection(uv_stream_t* server, int status) {
                         WARN("body too large (%u) for op %c", c->hdr.body_size, c->hdr.op);
                         auto& v = srv->conns_;
                         v.erase(std::remove(v.begin(), v.end(), c), v.end());
+                        srv->verbs_teardown(c);
                         conn_close(c);
                         return;
                     }
@@ -644,25 +618,32 @@ void Server::finish_task(Conn* c) {
     c->unref();
 }
 
-// ---- TCP fabric (RDMA-semantics) -----------------------------------------
+// ---- fabric negotiation + RDMA-semantics ops ------------------------------
 void Server::op_exchange(Conn* c, const std::vector<uint8_t>& body) {
-    (void)body;
-    // Fabric negotiation: this build serves the TCP data fabric ("TCPF").
-    // A verbs fabric would parse rdma_conn_info_t here instead.
+    // A body carrying ConnInfo requests the verbs fabric; succeed only when
+    // rdma-core + an active NIC are present, else fall back to the TCP data
+    // fabric ("TCPF" tag, payloads inline on this socket).
+    if (body.size() >= sizeof(vf::ConnInfo)) {
+        std::vector<uint8_t> reply;
+        if (verbs_handshake(c, body, &reply)) {
+            return send_status_payload(c, FINISH, reply.data(), reply.size());
+        }
+    }
     const char tag[4] = {'T', 'C', 'P', 'F'};
     send_status_payload(c, FINISH, reinterpret_cast<const uint8_t*>(tag), 4);
 }
 
-void Server::op_allocate(Conn* c, const RemoteMetaMsg& msg) {
-    if (msg.block_size <= 0 || msg.keys.empty()) return send_status(c, INVALID_REQ);
-    size_t page = static_cast<size_t>(msg.block_size);
+std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
+                                                     const std::vector<std::string>& keys,
+                                                     size_t page, int* status) {
+    *status = FINISH;
     Shard* shard = shard_least_used();
     std::vector<RemoteBlockWire> blocks;
-    blocks.reserve(msg.keys.size());
-    std::vector<std::pair<std::string, Ref<BlockEntry>>> created;
+    blocks.reserve(keys.size());
+    std::vector<std::string> created;
     {
         std::lock_guard<std::mutex> lk(kv_mu_);
-        for (auto& key : msg.keys) {
+        for (auto& key : keys) {
             if (kv_.count(key)) {
                 blocks.push_back({0, 0, 0});  // FAKE block: dup key, client skips
                 continue;
@@ -674,8 +655,9 @@ void Server::op_allocate(Conn* c, const RemoteMetaMsg& msg) {
                 pool_idx = idx;
             });
             if (!ok) {
-                for (auto& ckv : created) kv_.erase(ckv.first);
-                return send_status(c, OUT_OF_MEMORY);
+                for (auto& k : created) kv_.erase(k);
+                *status = OUT_OF_MEMORY;
+                return {};
             }
             auto* e = new BlockEntry();
             e->ptr = ptr;
@@ -684,13 +666,42 @@ void Server::op_allocate(Conn* c, const RemoteMetaMsg& msg) {
             e->shard = shard;
             Ref<BlockEntry> ref(e);
             kv_.emplace(key, ref);
-            created.push_back({key, ref});
+            created.push_back(key);
             c->pending_rdma.emplace(reinterpret_cast<uint64_t>(ptr), ref);
             blocks.push_back({static_cast<uint32_t>(shard->device() + 1), 0,
                               reinterpret_cast<uint64_t>(ptr)});
         }
     }
     maybe_extend(shard);
+    return blocks;
+}
+
+void Server::commit_addrs(Conn* c, const std::vector<uint64_t>& addrs) {
+    for (uint64_t addr : addrs) {
+        auto it = c->pending_rdma.find(addr);
+        if (it == c->pending_rdma.end()) continue;
+        it->second->committed = true;
+        c->pending_rdma.erase(it);
+    }
+}
+
+bool Server::collect_read_entries(const std::vector<std::string>& keys,
+                                  std::vector<Ref<BlockEntry>>* out) {
+    std::lock_guard<std::mutex> lk(kv_mu_);
+    out->reserve(keys.size());
+    for (auto& key : keys) {
+        auto it = kv_.find(key);
+        if (it == kv_.end() || !it->second->committed) return false;
+        out->push_back(it->second);
+    }
+    return true;
+}
+
+void Server::op_allocate(Conn* c, const RemoteMetaMsg& msg) {
+    if (msg.block_size <= 0 || msg.keys.empty()) return send_status(c, INVALID_REQ);
+    int status = FINISH;
+    auto blocks = allocate_blocks(c, msg.keys, static_cast<size_t>(msg.block_size), &status);
+    if (status != FINISH) return send_status(c, status);
     auto payload = build_allocate_response(blocks);
     send_status_payload(c, FINISH, payload.data(), payload.size());
 }
@@ -727,12 +738,7 @@ void Server::op_tcp_put(Conn* c, const std::vector<uint8_t>& body) {
 }
 
 void Server::op_commit(Conn* c, const RemoteMetaMsg& msg) {
-    for (uint64_t addr : msg.remote_addrs) {
-        auto it = c->pending_rdma.find(addr);
-        if (it == c->pending_rdma.end()) continue;
-        it->second->committed = true;
-        c->pending_rdma.erase(it);
-    }
+    commit_addrs(c, msg.remote_addrs);
     send_status(c, FINISH);
 }
 
@@ -740,14 +746,7 @@ void Server::op_tcp_get(Conn* c, const RemoteMetaMsg& msg) {
     if (msg.block_size <= 0 || msg.keys.empty()) return send_status(c, INVALID_REQ);
     size_t page = static_cast<size_t>(msg.block_size);
     std::vector<Ref<BlockEntry>> entries;
-    {
-        std::lock_guard<std::mutex> lk(kv_mu_);
-        for (auto& key : msg.keys) {
-            auto it = kv_.find(key);
-            if (it == kv_.end() || !it->second->committed) return send_status(c, KEY_NOT_FOUND);
-            entries.push_back(it->second);
-        }
-    }
+    if (!collect_read_entries(msg.keys, &entries)) return send_status(c, KEY_NOT_FOUND);
     // Build the framed response (status + len + blocks) in ONE buffer: no
     // intermediate payload copy (large gets were memory-bound on redundant
     // buffer fills).

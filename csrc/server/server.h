@@ -39,6 +39,7 @@
 #include "../core/mempool.h"
 #include "../core/protocol.h"
 #include "../core/utils.h"
+#include "../fabric/verbs_fabric.h"
 #include "../gpu/gpu.h"
 #include "shard.h"
 
@@ -54,6 +55,10 @@ struct ServerOptions {
     int cpu_shards = 1;        // CPU-mode shard count (tests the routing path)
     int n_streams = 4;
     std::string log_level = "warning";
+    // verbs fabric (used when rdma-core + an active NIC are present)
+    std::string dev_name;
+    int ib_port = 1;
+    std::string link_type = "Ethernet";
 };
 
 class Server;
@@ -91,7 +96,7 @@ class Server {
     std::pair<size_t, size_t> compact();
     int num_shards() const { return static_cast<int>(shards_.size()); }
 
-    struct Conn;  // connection state (public: file-local helpers use it)
+    struct Conn;  // defined below (file-local helpers + server_verbs use it)
 
     // Uniform view of a local-path request (flatbuffers or packed fast-path
     // format). Key views point into the request body / parsed message and
@@ -108,7 +113,28 @@ class Server {
         std::vector<std::pair<std::string_view, uint64_t>> blocks;
     };
 
+    // ---- data-plane helpers shared by the TCP and verbs fabrics ----
+    // Allocate pages for keys (dedup -> FAKE blocks); fills *status on error.
+    std::vector<RemoteBlockWire> allocate_blocks(Conn* c,
+                                                 const std::vector<std::string>& keys,
+                                                 size_t page, int* status);
+    void commit_addrs(Conn* c, const std::vector<uint64_t>& addrs);
+    // Collect committed entries for keys; false => missing/uncommitted key.
+    bool collect_read_entries(const std::vector<std::string>& keys,
+                              std::vector<Ref<BlockEntry>>* out);
+
+    // ---- verbs fabric seam (server_verbs.cpp) ----
+    struct VerbsPeer;
+    // Attempt a verbs handshake for OP_RDMA_EXCHANGE; fills the reply
+    // payload ("VRBS" + ConnInfo) on success.
+    bool verbs_handshake(Conn* c, const std::vector<uint8_t>& body,
+                         std::vector<uint8_t>* reply);
+    void verbs_teardown(Conn* c);  // loop thread; stops the CQ poll
+    uv_loop_t* loop() { return &loop_; }
+
    private:
+    friend struct VerbsPeer;
+
     // ---- loop-thread plumbing ----
     void loop_main();
     void post(std::function<void()> fn);  // run fn on the loop thread
@@ -137,6 +163,7 @@ class Server {
 
     ServerOptions opt_;
     std::vector<std::unique_ptr<Shard>> shards_;
+    std::unique_ptr<vf::Driver> vdrv_;  // lazy; created at first verbs handshake
 
     uv_loop_t loop_;
     uv_tcp_t listener_;
@@ -177,6 +204,41 @@ class Server {
         std::atomic<uint64_t> max_us{0};
     };
     OpStat op_stats_[128];
+};
+
+
+// Connection state (one per accepted TCP client).
+struct Server::Conn : RefCounted {
+    Server* srv = nullptr;
+    uv_tcp_t tcp;
+    bool closed = false;
+
+    // read state machine
+    enum State { kHeader, kBody } state = kHeader;
+    Header hdr{};
+    std::vector<uint8_t> buf;  // accumulated bytes
+
+    // local path: in-flight async copy count. OP_SYNC blocks server-side:
+    // when remain>0 the response is deferred until the count drains to zero
+    // (the reference instead has the client poll with sleeps,
+    // lib.py:578-592 — ~0.5 ms of added latency per op there).
+    std::atomic<int> remain{0};
+    bool sync_waiting = false;  // loop thread only
+
+    // cached IPC mappings: handle bytes -> base pointer (closed on disconnect)
+    std::map<std::vector<uint8_t>, std::pair<void*, int>> ipc_cache;  // base, src_dev
+
+    // fabric: blocks allocated for this conn, not yet committed.
+    std::unordered_map<uint64_t, Ref<BlockEntry>> pending_rdma;
+
+    // verbs fabric peer (owned; torn down on the loop thread).
+    Server::VerbsPeer* verbs = nullptr;
+
+    ~Conn() override {
+        for (auto& kv : ipc_cache) {
+            if (gpu::available()) gpu::ipc_close(kv.second.first);
+        }
+    }
 };
 
 }  // namespace ifs

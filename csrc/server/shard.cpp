@@ -146,6 +146,14 @@ std::vector<Shard::Move> Shard::plan_compaction(
     return moves;
 }
 
+void Shard::for_each_arena(const std::function<void(void*, size_t, bool)>& fn) {
+    std::lock_guard<std::mutex> lk(alloc_mu_);
+    for (size_t i = 0; i < mm_.num_pools(); i++) {
+        const MemoryPool* p = mm_.pool(static_cast<int>(i));
+        fn(p->base(), p->size(), on_gpu());
+    }
+}
+
 size_t Shard::largest_free_run_bytes() {
     std::lock_guard<std::mutex> lk(alloc_mu_);
     size_t best = 0;

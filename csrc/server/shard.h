@@ -83,6 +83,9 @@ class Shard {
     // Fragmentation stats (largest contiguous free run in bytes, per pool sum).
     size_t largest_free_run_bytes();
 
+    // Enumerate pool arenas (for fabric MR registration).
+    void for_each_arena(const std::function<void(void*, size_t, bool on_gpu)>& fn);
+
    private:
     struct Slot {
         uint64_t* h_src = nullptr;  // pinned staging

@@ -26,9 +26,12 @@ SOURCES = [
     CSRC / "core" / "protocol.cpp",
     CSRC / "core" / "mempool.cpp",
     CSRC / "gpu" / "gpu.hip",
+    CSRC / "fabric" / "verbs_fabric.cpp",
     CSRC / "server" / "shard.cpp",
     CSRC / "server" / "server.cpp",
+    CSRC / "server" / "server_verbs.cpp",
     CSRC / "client" / "client.cpp",
+    CSRC / "client" / "client_verbs.cpp",
     CSRC / "pybind.cpp",
 ]
 
