@@ -11,6 +11,41 @@
 
 namespace ifs {
 
+// ---------------------------------------------------------------------------
+// BlockEntry slab allocator (entries churn by the thousands per request)
+// ---------------------------------------------------------------------------
+namespace {
+struct EntrySlab {
+    std::mutex mu;
+    std::vector<void*> free_list;
+    std::vector<std::unique_ptr<char[]>> chunks;
+    static constexpr size_t kPerChunk = 4096;
+
+    void* take(size_t n) {
+        std::lock_guard<std::mutex> lk(mu);
+        if (free_list.empty()) {
+            chunks.emplace_back(new char[n * kPerChunk]);
+            char* base = chunks.back().get();
+            free_list.reserve(kPerChunk);
+            for (size_t i = 0; i < kPerChunk; i++) free_list.push_back(base + i * n);
+        }
+        void* p = free_list.back();
+        free_list.pop_back();
+        return p;
+    }
+    void give(void* p) {
+        std::lock_guard<std::mutex> lk(mu);
+        free_list.push_back(p);
+    }
+};
+EntrySlab& entry_slab() {
+    static EntrySlab s;
+    return s;
+}
+}  // namespace
+
+void* BlockEntry::operator new(size_t n) { return entry_slab().take(n); }
+void BlockEntry::operator delete(void* p) { entry_slab().give(p); }
 
 namespace {
 

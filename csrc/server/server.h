@@ -65,6 +65,8 @@ class Server;
 
 // One stored block. Refcounted: the kv map holds one ref; in-flight reads
 // hold another so purge cannot free memory under an active copy.
+// Slab-allocated: prefill writes create thousands of entries per request,
+// so class-level new/delete run on a chunked freelist instead of malloc.
 struct BlockEntry : RefCounted {
     void* ptr = nullptr;
     size_t size = 0;
@@ -74,6 +76,8 @@ struct BlockEntry : RefCounted {
     ~BlockEntry() override {
         if (shard && ptr) shard->deallocate(ptr, size, pool_idx);
     }
+    static void* operator new(size_t n);
+    static void operator delete(void* p);
 };
 
 class Server {
