@@ -39,8 +39,10 @@ struct EntrySlab {
     }
 };
 EntrySlab& entry_slab() {
-    static EntrySlab s;
-    return s;
+    // Intentionally leaked: entries may still be freed during interpreter
+    // teardown after static destructors would have run.
+    static EntrySlab* s = new EntrySlab();
+    return *s;
 }
 }  // namespace
 
