@@ -51,6 +51,7 @@ struct ServerConfigPy {
     std::vector<int> devices;         // GPU ordinals to shard over; empty=auto
     bool cpu_only = false;            // force CPU pool even if GPUs exist
     int cpu_shards = 1;               // CPU-mode shard count
+    bool auto_evict = false;          // LRU-evict on allocation failure
 };
 
 bool start_server(const ServerConfigPy& cfg) {
@@ -69,6 +70,7 @@ bool start_server(const ServerConfigPy& cfg) {
     opt.dev_name = cfg.dev_name;
     opt.ib_port = cfg.ib_port;
     opt.link_type = cfg.link_type;
+    opt.auto_evict = cfg.auto_evict;
     if (!cfg.cpu_only && gpu::available()) {
         if (!cfg.devices.empty()) {
             opt.devices = cfg.devices;
@@ -149,7 +151,8 @@ PYBIND11_MODULE(_native, m) {
         .def_readwrite("auto_increase", &ServerConfigPy::auto_increase)
         .def_readwrite("devices", &ServerConfigPy::devices)
         .def_readwrite("cpu_only", &ServerConfigPy::cpu_only)
-        .def_readwrite("cpu_shards", &ServerConfigPy::cpu_shards);
+        .def_readwrite("cpu_shards", &ServerConfigPy::cpu_shards)
+        .def_readwrite("auto_evict", &ServerConfigPy::auto_evict);
 
     // ---- client connection ----
     py::class_<ClientConn>(m, "Connection")

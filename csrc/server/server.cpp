@@ -465,6 +465,29 @@ Shard* Server::shard_least_used() {
     return best;
 }
 
+size_t Server::evict_lru_locked(Shard* shard, size_t bytes) {
+    // Candidates: committed, idle (only the map holds a ref), on this shard.
+    std::vector<std::pair<uint64_t, const std::string*>> cands;
+    for (auto& kvp : kv_) {
+        BlockEntry* e = kvp.second.get();
+        if (e->shard == shard && e->committed && e->ref_count() == 1)
+            cands.push_back({e->last_access, &kvp.first});
+    }
+    std::sort(cands.begin(), cands.end(),
+              [](const auto& a, const auto& b) { return a.first < b.first; });
+    size_t freed = 0;
+    for (auto& [tick, key] : cands) {
+        if (freed >= bytes) break;
+        auto it = kv_.find(*key);
+        if (it == kv_.end()) continue;
+        freed += it->second->size;
+        kv_.erase(it);
+        n_evicted_.fetch_add(1);
+    }
+    if (freed) DEBUG("auto-evicted %zu bytes from shard dev=%d", freed, shard->device());
+    return freed;
+}
+
 void Server::maybe_extend(Shard* s) {
     if (!s->need_extend()) return;
     int expect = 0;
@@ -521,9 +544,16 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
         // Pass 2: one batched allocator call for all pages.
         std::vector<std::pair<void*, int>> slots;
         slots.reserve(fresh.size());
-        bool ok = fresh.empty() ||
-                  shard->allocate(page, fresh.size(),
-                                  [&](void* p, int idx) { slots.push_back({p, idx}); });
+        auto try_alloc = [&] {
+            return fresh.empty() ||
+                   shard->allocate(page, fresh.size(),
+                                   [&](void* p, int idx) { slots.push_back({p, idx}); });
+        };
+        bool ok = try_alloc();
+        if (!ok && opt_.auto_evict &&
+            evict_lru_locked(shard, page * fresh.size()) > 0) {
+            ok = try_alloc();
+        }
         if (!ok) return send_status(c, OUT_OF_MEMORY);
         new_entries.reserve(fresh.size());
         new_keys.reserve(fresh.size());
@@ -536,6 +566,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
             e->pool_idx = slots[i].second;
             e->shard = shard;
             e->committed = false;
+            e->last_access = tick();
             Ref<BlockEntry> ref(e);
             kv_.emplace(std::string(fresh[i]->first), ref);
             new_entries.push_back(ref);
@@ -592,6 +623,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
     std::map<Shard*, Shard::CopyJob> jobs;
     auto held = std::make_shared<std::vector<Ref<BlockEntry>>>();
     held->reserve(msg.blocks.size());
+    uint64_t read_tick = tick();
     {
         std::lock_guard<std::mutex> lk(kv_mu_);
         for (auto& b : msg.blocks) {
@@ -600,6 +632,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
                 return send_status(c, KEY_NOT_FOUND);
             }
             BlockEntry* e = kit->second.get();
+            e->last_access = read_tick;
             auto& job = jobs[e->shard];
             job.bytes_per_block = page;
             job.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
@@ -687,10 +720,14 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
             }
             void* ptr = nullptr;
             int pool_idx = -1;
-            bool ok = shard->allocate(page, 1, [&](void* p, int idx) {
-                ptr = p;
-                pool_idx = idx;
-            });
+            auto try_alloc = [&] {
+                return shard->allocate(page, 1, [&](void* p, int idx) {
+                    ptr = p;
+                    pool_idx = idx;
+                });
+            };
+            bool ok = try_alloc();
+            if (!ok && opt_.auto_evict && evict_lru_locked(shard, page) > 0) ok = try_alloc();
             if (!ok) {
                 for (auto& k : created) kv_.erase(k);
                 *status = OUT_OF_MEMORY;
@@ -701,6 +738,7 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
             e->size = page;
             e->pool_idx = pool_idx;
             e->shard = shard;
+            e->last_access = tick();
             Ref<BlockEntry> ref(e);
             kv_.emplace(key, ref);
             created.push_back(key);
@@ -726,9 +764,11 @@ bool Server::collect_read_entries(const std::vector<std::string>& keys,
                                   std::vector<Ref<BlockEntry>>* out) {
     std::lock_guard<std::mutex> lk(kv_mu_);
     out->reserve(keys.size());
+    uint64_t t = tick();
     for (auto& key : keys) {
         auto it = kv_.find(key);
         if (it == kv_.end() || !it->second->committed) return false;
+        it->second->last_access = t;
         out->push_back(it->second);
     }
     return true;

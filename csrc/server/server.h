@@ -59,6 +59,11 @@ struct ServerOptions {
     std::string dev_name;
     int ib_port = 1;
     std::string link_type = "Ethernet";
+    // Opt-in LRU eviction: when an allocation fails, evict the
+    // least-recently-accessed committed idle keys and retry (extension — the
+    // reference only fails the write; engines can also evict explicitly via
+    // delete_keys).
+    bool auto_evict = false;
 };
 
 class Server;
@@ -73,6 +78,7 @@ struct BlockEntry : RefCounted {
     int pool_idx = -1;
     Shard* shard = nullptr;
     bool committed = false;
+    uint64_t last_access = 0;  // LRU tick (auto_evict)
     ~BlockEntry() override {
         if (shard && ptr) shard->deallocate(ptr, size, pool_idx);
     }
@@ -164,6 +170,12 @@ class Server {
     Shard* shard_for_device(int device);
     Shard* shard_least_used();
     void maybe_extend(Shard* s);
+    // Evict >= `bytes` of LRU committed idle entries on `shard`.
+    // Caller must hold kv_mu_. Returns bytes freed.
+    size_t evict_lru_locked(Shard* shard, size_t bytes);
+    uint64_t tick() { return access_tick_.fetch_add(1, std::memory_order_relaxed); }
+    std::atomic<uint64_t> access_tick_{1};
+    std::atomic<uint64_t> n_evicted_{0};
 
     ServerOptions opt_;
     std::vector<std::unique_ptr<Shard>> shards_;

@@ -106,6 +106,7 @@ class ServerConfig(_native.ServerConfig):
         self.devices = kwargs.get("devices", [])
         self.cpu_only = kwargs.get("cpu_only", False)
         self.cpu_shards = kwargs.get("cpu_shards", 1)
+        self.auto_evict = kwargs.get("auto_evict", False)
 
     def __repr__(self):
         return (

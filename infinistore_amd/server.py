@@ -106,6 +106,8 @@ def parse_args():
                    help="HIP streams per GPU shard")
     p.add_argument("--auto-increase", action="store_true",
                    help="extend the pool automatically when nearly full")
+    p.add_argument("--auto-evict", action="store_true",
+                   help="LRU-evict committed idle keys when the pool is full")
     p.add_argument("--devices", default="",
                    help="comma-separated GPU ordinals to shard over (default: all)")
     p.add_argument("--cpu-only", action="store_true",
@@ -147,6 +149,7 @@ def main():
         minimal_allocate_size=args.minimal_allocate_size,
         num_stream=args.num_stream,
         auto_increase=args.auto_increase,
+        auto_evict=args.auto_evict,
         devices=devices,
         cpu_only=args.cpu_only,
         dev_name=args.dev_name,

@@ -325,3 +325,79 @@ def test_compaction_defragments(ports):
         conn.close()
     finally:
         ifs.unregister_server()
+
+
+def test_read_pages_vectorized_api(cpu_server):
+    """read_pages (vectorized extension API) over the fabric path."""
+    import numpy as np
+
+    conn = make_client(cpu_server)
+    try:
+        src = torch.randn(4096)
+        dst = torch.zeros(4096)
+        conn.register_mr(src)
+        conn.register_mr(dst)
+        keys = _keys(4)
+        offsets = [0, 1024, 2048, 3072]
+        blocks = conn.allocate_rdma(keys, 1024 * 4)
+        conn.rdma_write_cache(src, offsets, 1024, blocks)
+        conn.sync()
+        conn.read_pages(dst, keys, np.asarray(offsets, dtype=np.uint64), 1024)
+        conn.sync()
+        assert torch.equal(src, dst)
+    finally:
+        conn.close()
+
+
+def test_auto_evict_lru(ports):
+    """With auto_evict on, a full pool evicts least-recently-accessed
+    committed keys instead of failing; recently-read keys survive."""
+    service_port, manage_port = ports
+    cfg = ifs.ServerConfig(
+        service_port=service_port,
+        manage_port=manage_port,
+        prealloc_size=1,  # 1 GB
+        minimal_allocate_size=1024,  # 1 MB granule -> 1024 blocks
+        cpu_only=True,
+        auto_evict=True,
+    )
+    ifs.register_server(cfg)
+    try:
+        conn = make_client(service_port)
+        page = 1 << 20
+        src = torch.zeros(page // 4)
+        dst = torch.zeros(page // 4)
+        conn.register_mr(src)
+        conn.register_mr(dst)
+
+        def put(keys):
+            blocks = conn.allocate_rdma(keys, page)
+            for b in blocks:
+                conn.rdma_write_cache(src, [0], page // 4, [b])
+            conn.sync()
+
+        put([f"old-{i}" for i in range(512)])
+        put([f"mid-{i}" for i in range(512)])  # pool now full
+        # touch the old keys so they become most-recently-used
+        conn.read_cache(dst, [("old-0", 0)], page // 4)
+        conn.sync()
+        for i in range(1, 512, 64):
+            conn.read_cache(dst, [(f"old-{i}", 0)], page // 4)
+        conn.sync()
+        # new writes must succeed by evicting the LRU keys (the untouched
+        # old-* ones are oldest; the freshly-read ones must survive)
+        put([f"new-{i}" for i in range(128)])
+        assert conn.check_exist("new-0") and conn.check_exist("new-127")
+        assert conn.check_exist("old-0")  # recently read -> survived
+        evicted_old = sum(
+            0 if conn.check_exist(f"old-{i}") else 1 for i in range(512)
+        )
+        evicted_mid = sum(
+            0 if conn.check_exist(f"mid-{i}") else 1 for i in range(512)
+        )
+        assert evicted_old + evicted_mid >= 128
+        # untouched old keys are evicted before the (younger) mid keys
+        assert evicted_old >= evicted_mid
+        conn.close()
+    finally:
+        ifs.unregister_server()

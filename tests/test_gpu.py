@@ -304,3 +304,30 @@ def test_cross_gpu_write(gpu_server):
     finally:
         wconn.close()
         rconn.close()
+
+
+def test_benchmark_harness_gpu(gpu_server):
+    """Reference-style harness end-to-end on the GPU local path."""
+    r = subprocess.run(
+        [sys.executable, "-m", "infinistore_amd.benchmark",
+         "--port", str(gpu_server), "--local-gpu", "--size", "64",
+         "--block-size", "128", "--iteration", "1", "--latency-ops", "10",
+         "--json", "--verify"],
+        cwd=REPO, capture_output=True, text=True, timeout=300,
+    )
+    assert r.returncode == 0, r.stdout + r.stderr
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["path"] == "local_gpu" and out["write_MBps"] > 0
+
+
+def test_example_client_run(gpu_server):
+    """The sync client example's core roundtrip against a live server."""
+    cfg = ifs.ClientConfig(host_addr="127.0.0.1", service_port=gpu_server,
+                           connection_type=ifs.TYPE_LOCAL_GPU)
+    conn = ifs.InfinityConnection(cfg)
+    conn.connect()
+    from infinistore_amd.example import client as ex
+
+    ex.run(conn, "cuda:0", "cuda:0")
+    conn.close()
