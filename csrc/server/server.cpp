@@ -155,13 +155,16 @@ bool Server::start() {
         }
     }
 
-    uv_loop_init(&loop_);
-    uv_async_init(&loop_, &post_async_, &Server::on_post_async);
-    post_async_.data = this;
-    uv_async_init(&loop_, &stop_async_, &Server::on_stop_async);
-    stop_async_.data = this;
+    // Bring up the main IO loop (listener) and the worker IO loops.
+    main_io_.srv = this;
+    main_io_.is_main = true;
+    uv_loop_init(&main_io_.loop);
+    uv_async_init(&main_io_.loop, &main_io_.post_async, &IoLoop::on_post);
+    main_io_.post_async.data = &main_io_;
+    uv_async_init(&main_io_.loop, &main_io_.stop_async, &IoLoop::on_stop);
+    main_io_.stop_async.data = &main_io_;
 
-    uv_tcp_init(&loop_, &listener_);
+    uv_tcp_init(&main_io_.loop, &listener_);
     listener_.data = this;
     struct sockaddr_in addr;
     uv_ip4_addr("0.0.0.0", opt_.service_port, &addr);
@@ -172,62 +175,81 @@ bool Server::start() {
     if (r != 0) {
         ERROR("bind/listen 0.0.0.0:%d failed: %s", opt_.service_port, uv_strerror(r));
         uv_close(reinterpret_cast<uv_handle_t*>(&listener_), nullptr);
-        uv_close(reinterpret_cast<uv_handle_t*>(&post_async_), nullptr);
-        uv_close(reinterpret_cast<uv_handle_t*>(&stop_async_), nullptr);
-        uv_run(&loop_, UV_RUN_NOWAIT);
-        uv_loop_close(&loop_);
+        uv_close(reinterpret_cast<uv_handle_t*>(&main_io_.post_async), nullptr);
+        uv_close(reinterpret_cast<uv_handle_t*>(&main_io_.stop_async), nullptr);
+        uv_run(&main_io_.loop, UV_RUN_NOWAIT);
+        uv_loop_close(&main_io_.loop);
         return false;
+    }
+    for (int i = 0; i < std::max(0, opt_.io_threads); i++) {
+        auto w = std::make_unique<IoLoop>();
+        w->srv = this;
+        uv_loop_init(&w->loop);
+        uv_async_init(&w->loop, &w->post_async, &IoLoop::on_post);
+        w->post_async.data = w.get();
+        uv_async_init(&w->loop, &w->stop_async, &IoLoop::on_stop);
+        w->stop_async.data = w.get();
+        workers_.push_back(std::move(w));
     }
     kv_.reserve(1u << 20);  // avoid rehash storms during bulk prefill writes
     running_.store(true);
     stop_requested_.store(false);
-    loop_thread_ = std::thread([this] { loop_main(); });
-    INFO("server listening on 0.0.0.0:%d with %zu shard(s)", opt_.service_port, shards_.size());
+    main_io_.start();
+    for (auto& w : workers_) w->start();
+    INFO("server listening on 0.0.0.0:%d with %zu shard(s), %zu IO loops",
+         opt_.service_port, shards_.size(), workers_.size() + 1);
     return true;
 }
 
-void Server::loop_main() {
-    uv_run(&loop_, UV_RUN_DEFAULT);
-    uv_loop_close(&loop_);
-    running_.store(false);
+void Server::IoLoop::start() {
+    thread = std::thread([this] {
+        uv_run(&loop, UV_RUN_DEFAULT);
+        uv_loop_close(&loop);
+    });
 }
 
 void Server::stop() {
     if (!running_.load()) return;
     stop_requested_.store(true);
-    uv_async_send(&stop_async_);
-    if (loop_thread_.joinable()) loop_thread_.join();
+    for (auto& w : workers_) w->request_stop();
+    main_io_.request_stop();
+    for (auto& w : workers_)
+        if (w->thread.joinable()) w->thread.join();
+    if (main_io_.thread.joinable()) main_io_.thread.join();
+    workers_.clear();
+    running_.store(false);
     // Drop all stored blocks.
     purge();
 }
 
-void Server::on_stop_async(uv_async_t* h) {
-    auto* srv = static_cast<Server*>(h->data);
-    // Close all connections, listener and asyncs; loop exits when no handles.
-    for (auto* c : srv->conns_) {
-        srv->verbs_teardown(c);
+void Server::IoLoop::on_stop(uv_async_t* h) {
+    auto* io = static_cast<IoLoop*>(h->data);
+    // Close this loop's connections and handles; the loop exits when none
+    // remain.
+    for (auto* c : io->conns) {
+        io->srv->verbs_teardown(c);
         conn_close(c);
     }
-    srv->conns_.clear();
-    uv_close(reinterpret_cast<uv_handle_t*>(&srv->listener_), nullptr);
-    uv_close(reinterpret_cast<uv_handle_t*>(&srv->post_async_), nullptr);
-    uv_close(reinterpret_cast<uv_handle_t*>(&srv->stop_async_), nullptr);
+    io->conns.clear();
+    if (io->is_main) uv_close(reinterpret_cast<uv_handle_t*>(&io->srv->listener_), nullptr);
+    uv_close(reinterpret_cast<uv_handle_t*>(&io->post_async), nullptr);
+    uv_close(reinterpret_cast<uv_handle_t*>(&io->stop_async), nullptr);
 }
 
-void Server::post(std::function<void()> fn) {
+void Server::IoLoop::post(std::function<void()> fn) {
     {
-        std::lock_guard<std::mutex> lk(post_mu_);
-        posted_.push_back(std::move(fn));
+        std::lock_guard<std::mutex> lk(post_mu);
+        posted.push_back(std::move(fn));
     }
-    uv_async_send(&post_async_);
+    uv_async_send(&post_async);
 }
 
-void Server::on_post_async(uv_async_t* h) {
-    auto* srv = static_cast<Server*>(h->data);
+void Server::IoLoop::on_post(uv_async_t* h) {
+    auto* io = static_cast<IoLoop*>(h->data);
     std::vector<std::function<void()>> fns;
     {
-        std::lock_guard<std::mutex> lk(srv->post_mu_);
-        fns.swap(srv->posted_);
+        std::lock_guard<std::mutex> lk(io->post_mu);
+        fns.swap(io->posted);
     }
     for (auto& f : fns) f();
 }
@@ -241,16 +263,49 @@ void Server::on_new_connection(uv_stream_t* server, int status) {
         WARN("accept error: %s", uv_strerror(status));
         return;
     }
+    // Accept on the main loop, then hand the fd to a worker loop (round
+    // robin) so request handling scales across IO threads.
+    uv_tcp_t* tmp = new uv_tcp_t();
+    uv_tcp_init(&srv->main_io_.loop, tmp);
+    tmp->data = nullptr;
+    if (uv_accept(server, reinterpret_cast<uv_stream_t*>(tmp)) != 0) {
+        uv_close(reinterpret_cast<uv_handle_t*>(tmp),
+                 [](uv_handle_t* h) { delete reinterpret_cast<uv_tcp_t*>(h); });
+        return;
+    }
+    uv_os_fd_t fd;
+    if (uv_fileno(reinterpret_cast<uv_handle_t*>(tmp), &fd) != 0) {
+        uv_close(reinterpret_cast<uv_handle_t*>(tmp),
+                 [](uv_handle_t* h) { delete reinterpret_cast<uv_tcp_t*>(h); });
+        return;
+    }
+    int fd2 = dup(fd);
+    uv_close(reinterpret_cast<uv_handle_t*>(tmp),
+             [](uv_handle_t* h) { delete reinterpret_cast<uv_tcp_t*>(h); });
+    IoLoop* io = &srv->main_io_;
+    if (!srv->workers_.empty()) {
+        uint32_t i = srv->next_worker_.fetch_add(1) %
+                     static_cast<uint32_t>(srv->workers_.size());
+        io = srv->workers_[i].get();
+    }
+    if (io == &srv->main_io_)
+        srv->adopt_fd(io, fd2);
+    else
+        io->post([srv, io, fd2] { srv->adopt_fd(io, fd2); });
+}
+
+void Server::adopt_fd(IoLoop* io, int fd) {
     auto* c = new Conn();
-    c->srv = srv;
-    uv_tcp_init(&srv->loop_, &c->tcp);
+    c->srv = this;
+    c->owner = io;
+    uv_tcp_init(&io->loop, &c->tcp);
     c->tcp.data = c;
-    if (uv_accept(server, reinterpret_cast<uv_stream_t*>(&c->tcp)) != 0) {
+    if (uv_tcp_open(&c->tcp, fd) != 0) {
         conn_close(c);
         return;
     }
     uv_tcp_nodelay(&c->tcp, 1);
-    srv->conns_.push_back(c);
+    io->conns.push_back(c);
     uv_read_start(
         reinterpret_cast<uv_stream_t*>(&c->tcp),
         [](uv_handle_t*, size_t suggested, uv_buf_t* buf) {
@@ -262,7 +317,7 @@ void Server::on_new_connection(uv_stream_t* server, int status) {
             Server* srv = c->srv;
             if (nread < 0) {
                 free(buf->base);
-                auto& v = srv->conns_;
+                auto& v = c->owner->conns;
                 v.erase(std::remove(v.begin(), v.end(), c), v.end());
                 srv->verbs_teardown(c);
                 conn_close(c);
@@ -284,7 +339,7 @@ void Server::on_new_connection(uv_stream_t* server, int status) {
                     consumed += sizeof(Header);
                     if (c->hdr.magic != kMagic) {
                         WARN("bad magic from client; closing");
-                        auto& v = srv->conns_;
+                        auto& v = c->owner->conns;
                         v.erase(std::remove(v.begin(), v.end(), c), v.end());
                         srv->verbs_teardown(c);
                         conn_close(c);
@@ -293,7 +348,7 @@ void Server::on_new_connection(uv_stream_t* server, int status) {
                     size_t cap = (c->hdr.op == OP_TCP_PUT) ? (256u << 20) : kProtocolBufferSize;
                     if (c->hdr.body_size > cap) {
                         WARN("body too large (%u) for op %c", c->hdr.body_size, c->hdr.op);
-                        auto& v = srv->conns_;
+                        auto& v = c->owner->conns;
                         v.erase(std::remove(v.begin(), v.end(), c), v.end());
                         srv->verbs_teardown(c);
                         conn_close(c);
@@ -533,47 +588,61 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
     job.bytes_per_block = page;
     std::vector<Ref<BlockEntry>> new_entries;
     std::vector<std::string> new_keys;
+    std::vector<uint64_t> fresh_offs;
     {
+        // Short critical section: dedup (first write wins) + placeholder
+        // inserts. Allocation and job building run outside kv_mu_ so
+        // concurrent IO threads don't serialize on the index for the whole
+        // request (uncommitted entries are invisible to reads).
+        uint64_t t = tick();
         std::lock_guard<std::mutex> lk(kv_mu_);
-        // Pass 1: dedup (first write wins), collect the keys to store.
-        std::vector<const std::pair<std::string_view, uint64_t>*> fresh;
-        fresh.reserve(msg.blocks.size());
+        new_entries.reserve(msg.blocks.size());
+        new_keys.reserve(msg.blocks.size());
+        fresh_offs.reserve(msg.blocks.size());
         for (auto& b : msg.blocks) {
-            if (kv_.find(b.first) == kv_.end()) fresh.push_back(&b);
-        }
-        // Pass 2: one batched allocator call for all pages.
-        std::vector<std::pair<void*, int>> slots;
-        slots.reserve(fresh.size());
-        auto try_alloc = [&] {
-            return fresh.empty() ||
-                   shard->allocate(page, fresh.size(),
-                                   [&](void* p, int idx) { slots.push_back({p, idx}); });
-        };
-        bool ok = try_alloc();
-        if (!ok && opt_.auto_evict &&
-            evict_lru_locked(shard, page * fresh.size()) > 0) {
-            ok = try_alloc();
-        }
-        if (!ok) return send_status(c, OUT_OF_MEMORY);
-        new_entries.reserve(fresh.size());
-        new_keys.reserve(fresh.size());
-        job.src.reserve(fresh.size());
-        job.dst.reserve(fresh.size());
-        for (size_t i = 0; i < fresh.size(); i++) {
+            if (kv_.find(b.first) != kv_.end()) continue;
             auto* e = new BlockEntry();
-            e->ptr = slots[i].first;
             e->size = page;
-            e->pool_idx = slots[i].second;
             e->shard = shard;
             e->committed = false;
-            e->last_access = tick();
+            e->last_access = t;
             Ref<BlockEntry> ref(e);
-            kv_.emplace(std::string(fresh[i]->first), ref);
+            kv_.emplace(std::string(b.first), ref);
             new_entries.push_back(ref);
-            new_keys.emplace_back(fresh[i]->first);
-            job.src.push_back(reinterpret_cast<uint64_t>(client_ptr + fresh[i]->second));
-            job.dst.push_back(reinterpret_cast<uint64_t>(slots[i].first));
+            new_keys.emplace_back(b.first);
+            fresh_offs.push_back(b.second);
         }
+    }
+    // Batched allocation (shard allocator lock only).
+    size_t n_fresh = new_entries.size();
+    std::vector<std::pair<void*, int>> slots;
+    slots.reserve(n_fresh);
+    auto try_alloc = [&] {
+        return n_fresh == 0 ||
+               shard->allocate(page, n_fresh,
+                               [&](void* p, int idx) { slots.push_back({p, idx}); });
+    };
+    bool alloc_ok = try_alloc();
+    if (!alloc_ok && opt_.auto_evict) {
+        std::lock_guard<std::mutex> lk(kv_mu_);
+        if (evict_lru_locked(shard, page * n_fresh) > 0) alloc_ok = try_alloc();
+    }
+    if (!alloc_ok) {
+        std::lock_guard<std::mutex> lk(kv_mu_);
+        for (size_t i = 0; i < n_fresh; i++) {
+            auto it = kv_.find(new_keys[i]);
+            if (it != kv_.end() && it->second.get() == new_entries[i].get()) kv_.erase(it);
+        }
+        return send_status(c, OUT_OF_MEMORY);
+    }
+    job.src.reserve(n_fresh);
+    job.dst.reserve(n_fresh);
+    for (size_t i = 0; i < n_fresh; i++) {
+        BlockEntry* e = new_entries[i].get();
+        e->ptr = slots[i].first;
+        e->pool_idx = slots[i].second;
+        job.src.push_back(reinterpret_cast<uint64_t>(client_ptr + fresh_offs[i]));
+        job.dst.push_back(reinterpret_cast<uint64_t>(slots[i].first));
     }
 
     n_writes_.fetch_add(1);
@@ -590,7 +659,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
     auto entries = std::make_shared<std::vector<Ref<BlockEntry>>>(std::move(new_entries));
     auto keys = std::make_shared<std::vector<std::string>>(std::move(new_keys));
     job.done = [this, c, entries, keys](bool ok) {
-        post([this, c, entries, keys, ok] {
+        c->owner->post([this, c, entries, keys, ok] {
             if (ok) {
                 for (auto& e : *entries) e->committed = true;
             } else {
@@ -654,7 +723,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
         j.done = [this, c, held, pending, all_ok, sync_resp](bool ok) {
             if (!ok) all_ok->store(false);
             if (pending->fetch_sub(1) == 1) {
-                post([this, c, held, all_ok, sync_resp] {
+                c->owner->post([this, c, held, all_ok, sync_resp] {
                     if (sync_resp)
                         send_status(c, all_ok->load() ? FINISH : INTERNAL_ERROR);
                     finish_task(c);
@@ -908,67 +977,56 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
 // ---------------------------------------------------------------------------
 std::pair<size_t, size_t> Server::compact() {
     if (!running_.load()) return {0, 0};
-    std::promise<std::pair<size_t, size_t>> prom;
-    auto fut = prom.get_future();
-    post([this, &prom] {
-        size_t moved = 0, bytes = 0;
-        for (auto& shard_up : shards_) {
-            Shard* shard = shard_up.get();
-            // Collect committed, idle (refcount==1: only the map holds them)
-            // entries on this shard. New reads cannot start while this runs —
-            // it executes on the loop thread.
-            std::vector<std::pair<void*, size_t>> movable;
-            std::vector<Ref<BlockEntry>> owners;
-            {
-                std::lock_guard<std::mutex> lk(kv_mu_);
-                for (auto& kvp : kv_) {
-                    BlockEntry* e = kvp.second.get();
-                    if (e->shard == shard && e->committed && e->ref_count() == 1)
-                        movable.push_back({e->ptr, e->size});
-                }
-            }
-            if (movable.empty()) continue;
-            auto moves = shard->plan_compaction(movable);
-            if (moves.empty()) continue;
-
-            std::map<size_t, Shard::CopyJob> by_size;  // one job per page size
-            for (auto& m : moves) {
-                auto& j = by_size[m.size];
-                j.bytes_per_block = m.size;
-                j.src.push_back(reinterpret_cast<uint64_t>(m.old_ptr));
-                j.dst.push_back(reinterpret_cast<uint64_t>(m.new_ptr));
-            }
-            for (auto& [sz, j2] : by_size) {
-                std::promise<bool> cp;
-                auto cf = cp.get_future();
-                Shard::CopyJob jj = std::move(j2);
-                jj.done = [&cp](bool ok) { cp.set_value(ok); };
-                if (!shard->submit_copy(std::move(jj))) {
-                    cp.set_value(false);
-                }
-                cf.wait();  // completion thread fulfills; loop thread is idle here
-            }
-            // Swap pointers in the index and free the old slots.
-            std::map<void*, Shard::Move*> by_old;
-            for (auto& m : moves) by_old[m.old_ptr] = &m;
-            {
-                std::lock_guard<std::mutex> lk(kv_mu_);
-                for (auto& kvp : kv_) {
-                    BlockEntry* e = kvp.second.get();
-                    auto it = by_old.find(e->ptr);
-                    if (it == by_old.end() || e->shard != shard) continue;
-                    Shard::Move* m = it->second;
-                    e->ptr = m->new_ptr;
-                    e->pool_idx = m->pool_idx;
-                    shard->deallocate(m->old_ptr, m->size, m->pool_idx);
-                    moved++;
-                    bytes += m->size;
-                }
-            }
+    // Maintenance op: kv_mu_ is held for the WHOLE plan+copy+swap so no
+    // request thread can start a read against a block while it moves
+    // (requests block on the mutex for the few ms this takes).
+    size_t moved = 0, bytes = 0;
+    std::lock_guard<std::mutex> lk(kv_mu_);
+    for (auto& shard_up : shards_) {
+        Shard* shard = shard_up.get();
+        std::vector<std::pair<void*, size_t>> movable;
+        for (auto& kvp : kv_) {
+            BlockEntry* e = kvp.second.get();
+            if (e->shard == shard && e->committed && e->ref_count() == 1)
+                movable.push_back({e->ptr, e->size});
         }
-        prom.set_value({moved, bytes});
-    });
-    return fut.get();
+        if (movable.empty()) continue;
+        auto moves = shard->plan_compaction(movable);
+        if (moves.empty()) continue;
+
+        std::map<size_t, Shard::CopyJob> by_size;  // one job per page size
+        for (auto& m : moves) {
+            auto& j = by_size[m.size];
+            j.bytes_per_block = m.size;
+            j.src.push_back(reinterpret_cast<uint64_t>(m.old_ptr));
+            j.dst.push_back(reinterpret_cast<uint64_t>(m.new_ptr));
+        }
+        for (auto& [sz, j2] : by_size) {
+            std::promise<bool> cp;
+            auto cf = cp.get_future();
+            Shard::CopyJob jj = std::move(j2);
+            jj.done = [&cp](bool ok) { cp.set_value(ok); };
+            if (!shard->submit_copy(std::move(jj))) {
+                cp.set_value(false);
+            }
+            cf.wait();  // shard completion thread fulfills (never takes kv_mu_)
+        }
+        // Swap pointers in the index and free the old slots.
+        std::map<void*, Shard::Move*> by_old;
+        for (auto& m : moves) by_old[m.old_ptr] = &m;
+        for (auto& kvp : kv_) {
+            BlockEntry* e = kvp.second.get();
+            auto it = by_old.find(e->ptr);
+            if (it == by_old.end() || e->shard != shard) continue;
+            Shard::Move* m = it->second;
+            e->ptr = m->new_ptr;
+            e->pool_idx = m->pool_idx;
+            shard->deallocate(m->old_ptr, m->size, m->pool_idx);
+            moved++;
+            bytes += m->size;
+        }
+    }
+    return {moved, bytes};
 }
 
 size_t Server::kvmap_len() {

@@ -54,6 +54,10 @@ struct ServerOptions {
     std::vector<int> devices;  // GPU ordinals to shard over; empty => CPU shard(s)
     int cpu_shards = 1;        // CPU-mode shard count (tests the routing path)
     int n_streams = 4;
+    // IO worker loops in addition to the accept loop. One loop thread
+    // saturates near ~10k heavy requests/s; 8 client ranks at full rate need
+    // parallel request handling (SURVEY.md §7 hard part 4). 0 = single loop.
+    int io_threads = 3;
     std::string log_level = "warning";
     // verbs fabric (used when rdma-core + an active NIC are present)
     std::string dev_name;
@@ -139,18 +143,36 @@ class Server {
     // payload ("VRBS" + ConnInfo) on success.
     bool verbs_handshake(Conn* c, const std::vector<uint8_t>& body,
                          std::vector<uint8_t>* reply);
-    void verbs_teardown(Conn* c);  // loop thread; stops the CQ poll
-    uv_loop_t* loop() { return &loop_; }
+    void verbs_teardown(Conn* c);  // owner-loop thread; stops the CQ poll
+
+    // One libuv loop + thread. The main loop owns the listener; accepted
+    // connections are handed (by fd) round-robin to worker loops so request
+    // parsing/dispatch scales across cores. All cross-loop work goes through
+    // post().
+    struct IoLoop {
+        uv_loop_t loop;
+        uv_async_t post_async;
+        uv_async_t stop_async;
+        std::thread thread;
+        std::mutex post_mu;
+        std::vector<std::function<void()>> posted;
+        Server* srv = nullptr;
+        std::vector<Conn*> conns;  // touched only from this loop's thread
+        bool is_main = false;
+
+        void post(std::function<void()> fn);
+        void start();
+        void request_stop() { uv_async_send(&stop_async); }
+        static void on_post(uv_async_t* h);
+        static void on_stop(uv_async_t* h);
+    };
 
    private:
     friend struct VerbsPeer;
 
-    // ---- loop-thread plumbing ----
-    void loop_main();
-    void post(std::function<void()> fn);  // run fn on the loop thread
-    static void on_post_async(uv_async_t* h);
-    static void on_stop_async(uv_async_t* h);
     static void on_new_connection(uv_stream_t* server, int status);
+    void adopt_fd(IoLoop* io, int fd);  // runs on io's thread
+    void post(std::function<void()> fn) { main_io_.post(std::move(fn)); }
 
     // ---- request handling (loop thread) ----
     void handle_request(Conn* c, char op, std::vector<uint8_t> body);
@@ -181,17 +203,12 @@ class Server {
     std::vector<std::unique_ptr<Shard>> shards_;
     std::unique_ptr<vf::Driver> vdrv_;  // lazy; created at first verbs handshake
 
-    uv_loop_t loop_;
+    IoLoop main_io_;
+    std::vector<std::unique_ptr<IoLoop>> workers_;
+    std::atomic<uint32_t> next_worker_{0};
     uv_tcp_t listener_;
-    uv_async_t post_async_;
-    uv_async_t stop_async_;
-    std::thread loop_thread_;
     std::atomic<bool> running_{false};
     std::atomic<bool> stop_requested_{false};
-    bool start_ok_ = false;
-
-    std::mutex post_mu_;
-    std::vector<std::function<void()>> posted_;
 
     // Heterogeneous-lookup map (C++20): hot-path lookups by string_view
     // avoid a std::string allocation per key.
@@ -203,8 +220,6 @@ class Server {
     };
     std::mutex kv_mu_;
     std::unordered_map<std::string, Ref<BlockEntry>, SvHash, std::equal_to<>> kv_;
-
-    std::vector<Conn*> conns_;  // loop thread only
 
     // stats
     std::atomic<uint64_t> n_writes_{0}, n_reads_{0}, n_put_{0}, n_get_{0};
@@ -226,6 +241,7 @@ class Server {
 // Connection state (one per accepted TCP client).
 struct Server::Conn : RefCounted {
     Server* srv = nullptr;
+    Server::IoLoop* owner = nullptr;  // the loop thread serving this conn
     uv_tcp_t tcp;
     bool closed = false;
 

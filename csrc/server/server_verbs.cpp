@@ -146,7 +146,7 @@ bool Server::verbs_handshake(Conn* c, const std::vector<uint8_t>& body,
     for (int i = 0; i < vf::kRecvBufs; i++) peer->ep.post_recv_buf(i);
     for (int i = 0; i < vf::kSendBufs; i++) peer->free_send.push_back(i);
 
-    uv_poll_init(&loop_, &peer->poll, peer->ep.comp_fd());
+    uv_poll_init(&c->owner->loop, &peer->poll, peer->ep.comp_fd());
     peer->poll.data = peer;
     uv_poll_start(&peer->poll, UV_READABLE, [](uv_poll_t* h, int status, int) {
         if (status < 0) return;
