@@ -69,11 +69,18 @@ def main():
 
     # Rank 0 hosts the server, sharded over all visible GPUs.
     if rank == 0:
-        n_shards = min(n_gpus, torch.cuda.device_count()) if have_gpu else 0
+        n_shards = min(n_gpus, torch.cuda.device_count()) if have_gpu else 1
+        # Keys accumulate across the timed steps (purged only between
+        # phases): size each shard for the whole run, including rank
+        # oversubscription (more ranks than GPUs on small test boxes).
+        ranks_per_shard = max(1, (world + n_shards - 1) // n_shards)
+        need_bytes = (args.steps + args.warmup) * args.blocks * args.block_kb * 1024
+        need_gb = (need_bytes * ranks_per_shard * 5) // (4 << 30) + 1  # +25% slack
+        pool_gb = max(args.pool_gb, int(need_gb))
         scfg = ifs.ServerConfig(
             service_port=port,
             manage_port=port + 1,
-            prealloc_size=args.pool_gb,
+            prealloc_size=pool_gb,
             minimal_allocate_size=args.block_kb,
             cpu_only=not have_gpu,
             devices=list(range(n_shards)) if have_gpu else [],
