@@ -319,11 +319,12 @@ class Table {
 
     std::string string_field(int field_id) const { return str_at(indirect(field_id)); }
 
-    // Vector helpers
+    // Vector helpers (length clamped: a valid element needs >= 1 byte).
     size_t vec_len(int field_id) const {
         size_t p = indirect(field_id);
         if (!p) return 0;
-        return r_.read<uint32_t>(p);
+        uint32_t n = r_.read<uint32_t>(p);
+        return static_cast<size_t>(n) > r_.len() ? 0 : n;
     }
     size_t vec_data(int field_id) const {
         size_t p = indirect(field_id);
@@ -348,6 +349,8 @@ class Table {
         size_t p = indirect(field_id);
         if (!p) return out;
         uint32_t n = r_.read<uint32_t>(p);
+        // Malformed-length guard: each element needs >= 4 bytes of buffer.
+        if (static_cast<size_t>(n) > r_.len() / 4) return out;
         out.reserve(n);
         for (uint32_t i = 0; i < n; i++) {
             size_t slot = p + 4 + static_cast<size_t>(i) * 4;
