@@ -473,7 +473,7 @@ void Server::handle_request(Conn* c, char op, std::vector<uint8_t> body) {
             return op_allocate(c, msg);
         }
         case OP_TCP_PUT:
-            return op_tcp_put(c, body);
+            return op_tcp_put(c, std::move(body));
         case OP_TCP_GET: {
             RemoteMetaMsg msg;
             if (!parse_remote_meta(body.data(), body.size(), &msg))
@@ -852,7 +852,7 @@ void Server::op_allocate(Conn* c, const RemoteMetaMsg& msg) {
     send_status_payload(c, FINISH, payload.data(), payload.size());
 }
 
-void Server::op_tcp_put(Conn* c, const std::vector<uint8_t>& body) {
+void Server::op_tcp_put(Conn* c, std::vector<uint8_t> body) {
     // Layout: [u32 n][u32 block_size][u64 addr x n][payload n*block_size]
     if (body.size() < 8) return send_status(c, INVALID_REQ);
     uint32_t n, bs;
@@ -861,8 +861,13 @@ void Server::op_tcp_put(Conn* c, const std::vector<uint8_t>& body) {
     size_t need = 8 + static_cast<size_t>(n) * 8 + static_cast<size_t>(n) * bs;
     if (n == 0 || bs == 0 || body.size() < need) return send_status(c, INVALID_REQ);
     const uint64_t* addrs = reinterpret_cast<const uint64_t*>(body.data() + 8);
-    const uint8_t* payload = body.data() + 8 + static_cast<size_t>(n) * 8;
+    size_t payload_off = 8 + static_cast<size_t>(n) * 8;
 
+    // Validate addresses and group by owning shard; copies then run on the
+    // shards' fabric worker threads (the IO loop never blocks on memcpy).
+    std::map<Shard*, Shard::FabricJob> jobs;
+    std::vector<Ref<BlockEntry>> held;
+    held.reserve(n);
     for (uint32_t i = 0; i < n; i++) {
         auto it = c->pending_rdma.find(addrs[i]);
         if (it == c->pending_rdma.end()) {
@@ -871,19 +876,53 @@ void Server::op_tcp_put(Conn* c, const std::vector<uint8_t>& body) {
         }
         BlockEntry* e = it->second.get();
         if (e->size < bs) return send_status(c, INVALID_REQ);
-        const uint8_t* src = payload + static_cast<size_t>(i) * bs;
-        if (e->shard->on_gpu()) {
-            if (!gpu::memcpy_h2d(e->ptr, src, bs)) return send_status(c, INTERNAL_ERROR);
-        } else {
-            memcpy(e->ptr, src, bs);
-        }
+        auto& j = jobs[e->shard];
+        j.is_put = true;
+        j.bytes_per_block = bs;
+        j.block_ptrs.push_back(reinterpret_cast<uint64_t>(e->ptr));
+        j.host_offsets.push_back(payload_off + static_cast<size_t>(i) * bs);
+        held.push_back(it->second);
     }
     n_put_.fetch_add(1);
     bytes_in_.fetch_add(static_cast<size_t>(n) * bs);
-    send_status(c, TASK_ACCEPTED);
+
+    auto shared_body = std::make_shared<std::vector<uint8_t>>(std::move(body));
+    auto pending = std::make_shared<std::atomic<int>>(static_cast<int>(jobs.size()));
+    auto all_ok = std::make_shared<std::atomic<bool>>(true);
+    auto held_sp = std::make_shared<std::vector<Ref<BlockEntry>>>(std::move(held));
+    c->fabric_inflight.fetch_add(1);
+    c->ref();
+    for (auto& [shard, job] : jobs) {
+        Shard::FabricJob j = std::move(job);
+        j.host = shared_body;
+        j.done = [this, c, pending, all_ok, held_sp](bool ok) {
+            if (!ok) all_ok->store(false);
+            if (pending->fetch_sub(1) == 1) {
+                c->owner->post([this, c, all_ok, held_sp] {
+                    send_status(c, all_ok->load() ? TASK_ACCEPTED : INTERNAL_ERROR);
+                    if (c->fabric_inflight.fetch_sub(1) == 1) {
+                        // Drain commits that arrived while copies were in flight.
+                        for (auto& addrs2 : c->deferred_commits) {
+                            commit_addrs(c, addrs2);
+                            send_status(c, FINISH);
+                        }
+                        c->deferred_commits.clear();
+                    }
+                    c->unref();
+                });
+            }
+        };
+        shard->submit_fabric(std::move(j));
+    }
 }
 
 void Server::op_commit(Conn* c, const RemoteMetaMsg& msg) {
+    if (c->fabric_inflight.load() > 0) {
+        // Puts still copying on the fabric workers: commit (and its ACK)
+        // must wait so readers never see committed-but-unwritten blocks.
+        c->deferred_commits.push_back(msg.remote_addrs);
+        return;
+    }
     commit_addrs(c, msg.remote_addrs);
     send_status(c, FINISH);
 }
@@ -893,29 +932,46 @@ void Server::op_tcp_get(Conn* c, const RemoteMetaMsg& msg) {
     size_t page = static_cast<size_t>(msg.block_size);
     std::vector<Ref<BlockEntry>> entries;
     if (!collect_read_entries(msg.keys, &entries)) return send_status(c, KEY_NOT_FOUND);
-    // Build the framed response (status + len + blocks) in ONE buffer: no
-    // intermediate payload copy (large gets were memory-bound on redundant
-    // buffer fills).
+    // The framed response (status + len + blocks) is ONE buffer, filled by
+    // the shards' fabric workers; the response is sent on completion.
     size_t total = entries.size() * page;
-    std::vector<uint8_t> resp(8 + total);
+    auto resp = std::make_shared<std::vector<uint8_t>>(8 + total);
     int code = FINISH;
     uint32_t len32 = static_cast<uint32_t>(total);
-    memcpy(resp.data(), &code, 4);
-    memcpy(resp.data() + 4, &len32, 4);
-    uint8_t* out = resp.data() + 8;
+    memcpy(resp->data(), &code, 4);
+    memcpy(resp->data() + 4, &len32, 4);
+    std::map<Shard*, Shard::FabricJob> jobs;
     for (size_t i = 0; i < entries.size(); i++) {
         BlockEntry* e = entries[i].get();
-        size_t nbytes = std::min(page, e->size);
-        if (e->shard->on_gpu()) {
-            if (!gpu::memcpy_d2h(out + i * page, e->ptr, nbytes))
-                return send_status(c, INTERNAL_ERROR);
-        } else {
-            memcpy(out + i * page, e->ptr, nbytes);
-        }
+        auto& j = jobs[e->shard];
+        j.is_put = false;
+        j.bytes_per_block = page;
+        j.block_ptrs.push_back(reinterpret_cast<uint64_t>(e->ptr));
+        j.host_offsets.push_back(8 + i * page);
     }
     n_get_.fetch_add(1);
     bytes_out_.fetch_add(total);
-    send_buf(c, std::move(resp));
+    auto pending = std::make_shared<std::atomic<int>>(static_cast<int>(jobs.size()));
+    auto all_ok = std::make_shared<std::atomic<bool>>(true);
+    auto held_sp = std::make_shared<std::vector<Ref<BlockEntry>>>(std::move(entries));
+    c->ref();
+    for (auto& [shard, job] : jobs) {
+        Shard::FabricJob j = std::move(job);
+        j.host = resp;
+        j.done = [this, c, resp, pending, all_ok, held_sp](bool ok) {
+            if (!ok) all_ok->store(false);
+            if (pending->fetch_sub(1) == 1) {
+                c->owner->post([this, c, resp, all_ok] {
+                    if (all_ok->load())
+                        send_buf(c, std::move(*resp));
+                    else
+                        send_status(c, INTERNAL_ERROR);
+                    c->unref();
+                });
+            }
+        };
+        shard->submit_fabric(std::move(j));
+    }
 }
 
 // ---- queries ---------------------------------------------------------------

@@ -182,7 +182,7 @@ class Server {
     void finish_task(Conn* c);  // remain-- (+ deferred sync reply)
     void op_exchange(Conn* c, const std::vector<uint8_t>& body);
     void op_allocate(Conn* c, const RemoteMetaMsg& msg);
-    void op_tcp_put(Conn* c, const std::vector<uint8_t>& body);
+    void op_tcp_put(Conn* c, std::vector<uint8_t> body);
     void op_tcp_get(Conn* c, const RemoteMetaMsg& msg);
     void op_commit(Conn* c, const RemoteMetaMsg& msg);
     void op_check_exist(Conn* c, const std::vector<uint8_t>& body);
@@ -262,6 +262,9 @@ struct Server::Conn : RefCounted {
 
     // fabric: blocks allocated for this conn, not yet committed.
     std::unordered_map<uint64_t, Ref<BlockEntry>> pending_rdma;
+    // async fabric puts in flight (owner loop defers commits behind them)
+    std::atomic<int> fabric_inflight{0};
+    std::vector<std::vector<uint64_t>> deferred_commits;  // owner loop only
 
     // verbs fabric peer (owned; torn down on the loop thread).
     Server::VerbsPeer* verbs = nullptr;

@@ -16,6 +16,11 @@ Shard::~Shard() {
         stopping_ = true;
     }
     task_cv_.notify_all();
+    {
+        std::lock_guard<std::mutex> lk(fabric_mu_);
+    }
+    fabric_cv_.notify_all();
+    if (fabric_thread_.joinable()) fabric_thread_.join();
     if (completion_thread_.joinable()) completion_thread_.join();
     for (auto& sc : streams_) {
         for (auto& sl : sc.slots) {
@@ -68,6 +73,7 @@ bool Shard::init() {
         std::lock_guard<std::mutex> lk(alloc_mu_);
         mm_.add_pool(arena, opt_.pool_bytes, opt_.block_granule, [](void* p, size_t) { free(p); });
     }
+    fabric_thread_ = std::thread([this] { fabric_loop(); });
     inited_ = true;
     return true;
 }
@@ -286,6 +292,52 @@ void Shard::completion_loop() {
         }
         slot_cv_.notify_all();
         if (t.done) t.done(ok);
+    }
+}
+
+
+void Shard::submit_fabric(FabricJob&& job) {
+    {
+        std::lock_guard<std::mutex> lk(fabric_mu_);
+        fabric_q_.push_back(std::move(job));
+    }
+    fabric_cv_.notify_one();
+}
+
+void Shard::fabric_loop() {
+    if (on_gpu()) gpu::set_device(opt_.device);
+    for (;;) {
+        FabricJob job;
+        {
+            std::unique_lock<std::mutex> lk(fabric_mu_);
+            fabric_cv_.wait(lk, [this] { return stopping_ || !fabric_q_.empty(); });
+            if (stopping_ && fabric_q_.empty()) return;
+            if (fabric_q_.empty()) continue;
+            job = std::move(fabric_q_.front());
+            fabric_q_.pop_front();
+        }
+        bool ok = job.host != nullptr;
+        size_t bs = job.bytes_per_block;
+        if (job.is_put) {
+            for (size_t i = 0; i < job.block_ptrs.size() && ok; i++) {
+                const uint8_t* src = job.host->data() + job.host_offsets[i];
+                void* dst = reinterpret_cast<void*>(job.block_ptrs[i]);
+                if (on_gpu())
+                    ok = gpu::memcpy_h2d(dst, src, bs);
+                else
+                    memcpy(dst, src, bs);
+            }
+        } else {
+            for (size_t i = 0; i < job.block_ptrs.size() && ok; i++) {
+                uint8_t* dst = job.host->data() + job.host_offsets[i];
+                const void* src = reinterpret_cast<const void*>(job.block_ptrs[i]);
+                if (on_gpu())
+                    ok = gpu::memcpy_d2h(dst, src, bs);
+                else
+                    memcpy(dst, src, bs);
+            }
+        }
+        if (job.done) job.done(ok);
     }
 }
 

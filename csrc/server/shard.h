@@ -9,10 +9,12 @@
 // In CPU mode (no GPU) a shard is a host-DRAM arena and copies run inline.
 #pragma once
 
+#include <atomic>
 #include <condition_variable>
 #include <cstdint>
 #include <deque>
 #include <functional>
+#include <memory>
 #include <mutex>
 #include <thread>
 #include <vector>
@@ -66,6 +68,20 @@ class Shard {
     // (done() is NOT called in that case).
     bool submit_copy(CopyJob&& job);
 
+    // Fabric IO (TCP data path): bulk host<->pool copies run on a dedicated
+    // worker thread so IO loops never block on hipMemcpy/memcpy.
+    struct FabricJob {
+        bool is_put = false;
+        std::shared_ptr<std::vector<uint8_t>> host;  // source (put) / dest (get)
+        std::vector<uint64_t> block_ptrs;            // pool block addresses
+        std::vector<size_t> host_offsets;            // absolute offsets into *host
+        size_t bytes_per_block = 0;
+        // Invoked on the worker thread; captured state (body buffers, block
+        // refs) is released when the job is destroyed after the call.
+        std::function<void(bool)> done;
+    };
+    void submit_fabric(FabricJob&& job);
+
     // Extend pool by one arena (called off the hot path).
     bool extend();
     bool need_extend();
@@ -108,6 +124,7 @@ class Shard {
 
     Slot* acquire_slot(StreamCtx& sc);
     void completion_loop();
+    void fabric_loop();
 
     ShardOptions opt_;
     MM mm_;
@@ -122,8 +139,13 @@ class Shard {
     std::condition_variable task_cv_;
     std::condition_variable slot_cv_;
     std::thread completion_thread_;
-    bool stopping_ = false;
+    std::atomic<bool> stopping_{false};
     bool inited_ = false;
+
+    std::deque<FabricJob> fabric_q_;
+    std::mutex fabric_mu_;
+    std::condition_variable fabric_cv_;
+    std::thread fabric_thread_;
 };
 
 }  // namespace ifs
