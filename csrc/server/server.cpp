@@ -237,6 +237,10 @@ void Server::IoLoop::on_stop(uv_async_t* h) {
 }
 
 void Server::IoLoop::post(std::function<void()> fn) {
+    // After stop is requested the async handles are being closed; drop the
+    // callback (captured refs release on destruction) instead of touching a
+    // dying handle.
+    if (srv && srv->stop_requested_.load(std::memory_order_acquire)) return;
     {
         std::lock_guard<std::mutex> lk(post_mu);
         posted.push_back(std::move(fn));
@@ -295,6 +299,10 @@ void Server::on_new_connection(uv_stream_t* server, int status) {
 }
 
 void Server::adopt_fd(IoLoop* io, int fd) {
+    if (stop_requested_.load()) {
+        ::close(fd);
+        return;
+    }
     auto* c = new Conn();
     c->srv = this;
     c->owner = io;
