@@ -43,9 +43,6 @@ def parse_args():
     p.add_argument("--cross", action="store_true",
                    help="each rank reads keys written by the next rank "
                         "(forces xGMI cross-shard traffic)")
-    p.add_argument("--overlap", action="store_true",
-                   help="steady-state serving: a writer thread puts step s "
-                        "while a reader thread gets step s-1 (two connections)")
     return p.parse_args()
 
 
@@ -183,61 +180,6 @@ def main():
     purge_all()
 
     # ---- timed region ----
-    if args.overlap and have_gpu and not cross:
-        # Steady-state serving: writer and reader run concurrently on two
-        # connections (the prefill-node + decode-node pattern on one box).
-        import concurrent.futures
-
-        rconn = ifs.InfinityConnection(ccfg)
-        rconn.connect()
-        rdst = torch.zeros_like(dst)
-
-        def put_step(s):
-            conn.write_pages(src, put_keys[s], offsets_np, elems_per_block)
-            conn.sync()
-
-        def get_step(s):
-            rconn.read_pages(rdst, put_keys[s], offsets_np, elems_per_block)
-            rconn.sync()
-
-        put_step(0)  # pipeline prologue
-        sync_all()
-        t0 = time.perf_counter()
-        with concurrent.futures.ThreadPoolExecutor(max_workers=2) as ex:
-            for s in range(1, args.steps):
-                fw = ex.submit(put_step, s)
-                fr = ex.submit(get_step, s - 1)
-                fw.result()
-                fr.result()
-        get_step(args.steps - 1)
-        sync_all()
-        elapsed = time.perf_counter() - t0
-        total_gb = 2.0 * args.blocks * block_bytes * args.steps * world / 1e9
-        gbps = total_gb / elapsed
-        rconn.close()
-        purge_all()
-        if rank == 0:
-            print(json.dumps({
-                "metric": "put_get_GBps", "value": round(gbps, 3), "unit": "GB/s",
-                "n_gpus": world, "steps": args.steps, "warmup": args.warmup,
-                "ms_per_step": round(elapsed / args.steps * 1000, 3),
-                "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
-                "dtype": "bf16", "data": "synthetic",
-                "config": {"model": "kvcache-store", "block_kb": args.block_kb,
-                           "blocks_per_rank_per_step": args.blocks,
-                           "global_batch": args.blocks * world, "seq_len": 0,
-                           "parallelism": f"shard{world}",
-                           "path": "local_gpu_ipc_overlap"},
-            }))
-        conn.close()
-        if dist:
-            dist.barrier()
-        if rank == 0:
-            ifs.unregister_server()
-        if dist:
-            dist.destroy_process_group()
-        return
-
     debug_t = {"keygen": 0.0, "put_req": 0.0, "put_sync": 0.0, "get_req": 0.0,
                "get_sync": 0.0}
     debug = os.environ.get("IFS_BENCH_DEBUG")
