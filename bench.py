@@ -36,6 +36,10 @@ def parse_args():
                    help="blocks per rank per step (2048 x 128KB = 256 MB)")
     p.add_argument("--pool-gb", type=int, default=8, help="pool GB per shard")
     p.add_argument("--port", type=int, default=0)
+    p.add_argument("--server-addr", default="127.0.0.1",
+                   help="connect to an external server instead of hosting one "
+                        "(multi-node runs; the remote server must be started "
+                        "separately)")
     p.add_argument("--cpu", action="store_true",
                    help="CPU-only mode (TCP fabric, DRAM pool) for dev boxes")
     p.add_argument("--latency-ops", type=int, default=200,
@@ -67,8 +71,9 @@ def main():
 
     port = args.port or (23000 + (os.getpid() % 1000) if world == 1 else 23999)
 
+    external = args.server_addr != "127.0.0.1"
     # Rank 0 hosts the server, sharded over all visible GPUs.
-    if rank == 0:
+    if rank == 0 and not external:
         n_shards = min(n_gpus, torch.cuda.device_count()) if have_gpu else 1
         # Keys accumulate across the timed steps (purged only between
         # phases): size each shard for the whole run, including rank
@@ -102,15 +107,16 @@ def main():
     src = torch.randn(total_elems, dtype=torch.bfloat16, device=dev)
     dst = torch.zeros_like(src)
 
+    use_local_path = have_gpu and not external
     ccfg = ifs.ClientConfig(
-        host_addr="127.0.0.1",
+        host_addr=args.server_addr,
         service_port=port,
-        connection_type=ifs.TYPE_LOCAL_GPU if have_gpu else ifs.TYPE_RDMA,
-        link_type="TCP",
+        connection_type=ifs.TYPE_LOCAL_GPU if use_local_path else ifs.TYPE_RDMA,
+        link_type="TCP" if not external else "Ethernet",
     )
     conn = ifs.InfinityConnection(ccfg)
     conn.connect()
-    if not have_gpu:
+    if not use_local_path:
         conn.register_mr(src)
         conn.register_mr(dst)
 
@@ -131,7 +137,7 @@ def main():
         return [f"r{owner}-s{step}-{run_id}-{i}" for i in range(args.blocks)]
 
     def do_put(keys):
-        if have_gpu:
+        if use_local_path:
             conn.write_pages(src, keys, offsets_np, elems_per_block)
             conn.sync()
         else:
@@ -140,7 +146,7 @@ def main():
             conn.sync()
 
     def do_get(keys):
-        if have_gpu:
+        if use_local_path:
             conn.read_pages(dst, keys, offsets_np, elems_per_block)
             conn.sync()
         else:
@@ -150,7 +156,7 @@ def main():
     def purge_all():
         if dist:
             dist.barrier()
-        if rank == 0:
+        if rank == 0 and not external:
             ifs.purge_kv_map()
         if dist:
             dist.barrier()
@@ -190,7 +196,7 @@ def main():
     get_time = 0.0
     for s in range(args.steps):
         tp = time.perf_counter()
-        if debug and have_gpu:
+        if debug and use_local_path:
             conn.write_pages(src, put_keys[s], offsets_np, elems_per_block)
             tb = time.perf_counter()
             conn.sync()
@@ -203,7 +209,7 @@ def main():
         if cross:
             dist.barrier()  # readers wait for the writer of their keys
         tg = time.perf_counter()
-        if debug and have_gpu:
+        if debug and use_local_path:
             conn.read_pages(dst, get_keys[s], offsets_np, elems_per_block)
             tb = time.perf_counter()
             conn.sync()
@@ -231,7 +237,7 @@ def main():
     lkeys = [f"lat-r{rank}-{run_id}-{i}" for i in range(args.latency_ops)]
     for i in range(args.latency_ops):
         t1 = time.perf_counter()
-        if have_gpu:
+        if use_local_path:
             conn.local_gpu_write_cache(src, [(lkeys[i], 0)], elems_per_block)
             conn.sync()
         else:
@@ -276,7 +282,7 @@ def main():
                 "global_batch": args.blocks * world,
                 "seq_len": 0,
                 "parallelism": f"shard{world}",
-                "path": "local_gpu_ipc" if have_gpu else "tcp_fabric_cpu",
+                "path": "local_gpu_ipc" if use_local_path else ("fabric_remote" if external else "tcp_fabric_cpu"),
                 "put_GBps": round(put_gbps, 3),
                 "get_GBps": round(get_gbps, 3),
                 "p50_put_us": round(pct(lat_put, 50), 1),
@@ -290,7 +296,7 @@ def main():
     conn.close()
     if dist:
         dist.barrier()
-    if rank == 0:
+    if rank == 0 and not external:
         print("server stats:", ifs.get_server_stats(), file=sys.stderr)
         ifs.unregister_server()
     if dist:

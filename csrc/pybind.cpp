@@ -52,6 +52,7 @@ struct ServerConfigPy {
     bool cpu_only = false;            // force CPU pool even if GPUs exist
     int cpu_shards = 1;               // CPU-mode shard count
     bool auto_evict = false;          // LRU-evict on allocation failure
+    int io_threads = 3;               // worker IO loops (0 = single loop)
 };
 
 bool start_server(const ServerConfigPy& cfg) {
@@ -71,6 +72,7 @@ bool start_server(const ServerConfigPy& cfg) {
     opt.ib_port = cfg.ib_port;
     opt.link_type = cfg.link_type;
     opt.auto_evict = cfg.auto_evict;
+    opt.io_threads = cfg.io_threads;
     if (!cfg.cpu_only && gpu::available()) {
         if (!cfg.devices.empty()) {
             opt.devices = cfg.devices;
@@ -152,7 +154,8 @@ PYBIND11_MODULE(_native, m) {
         .def_readwrite("devices", &ServerConfigPy::devices)
         .def_readwrite("cpu_only", &ServerConfigPy::cpu_only)
         .def_readwrite("cpu_shards", &ServerConfigPy::cpu_shards)
-        .def_readwrite("auto_evict", &ServerConfigPy::auto_evict);
+        .def_readwrite("auto_evict", &ServerConfigPy::auto_evict)
+        .def_readwrite("io_threads", &ServerConfigPy::io_threads);
 
     // ---- client connection ----
     py::class_<ClientConn>(m, "Connection")

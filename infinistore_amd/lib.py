@@ -107,6 +107,7 @@ class ServerConfig(_native.ServerConfig):
         self.cpu_only = kwargs.get("cpu_only", False)
         self.cpu_shards = kwargs.get("cpu_shards", 1)
         self.auto_evict = kwargs.get("auto_evict", False)
+        self.io_threads = kwargs.get("io_threads", 3)
 
     def __repr__(self):
         return (

@@ -104,6 +104,8 @@ def parse_args():
                    help="allocation granule in KB")
     p.add_argument("--num-stream", type=int, default=4,
                    help="HIP streams per GPU shard")
+    p.add_argument("--io-threads", type=int, default=3,
+                   help="worker IO loop threads (0 = single loop)")
     p.add_argument("--auto-increase", action="store_true",
                    help="extend the pool automatically when nearly full")
     p.add_argument("--auto-evict", action="store_true",
@@ -148,6 +150,7 @@ def main():
         prealloc_size=args.prealloc_size,
         minimal_allocate_size=args.minimal_allocate_size,
         num_stream=args.num_stream,
+        io_threads=args.io_threads,
         auto_increase=args.auto_increase,
         auto_evict=args.auto_evict,
         devices=devices,
