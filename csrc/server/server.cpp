@@ -739,6 +739,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
             }
         };
         if (!shard->submit_copy(std::move(j))) {
+            all_ok->store(false);
             if (pending->fetch_sub(1) == 1) {
                 finish_task(c);
                 return send_status(c, INTERNAL_ERROR);
