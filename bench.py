@@ -138,8 +138,7 @@ def main():
 
     def do_put(keys):
         if use_local_path:
-            conn.write_pages(src, keys, offsets_np, elems_per_block)
-            conn.sync()
+            conn.write_pages(src, keys, offsets_np, elems_per_block, sync=True)
         else:
             blocks = conn.allocate_rdma(keys, block_bytes)
             conn.rdma_write_cache(src, offsets, elems_per_block, blocks)
@@ -197,7 +196,7 @@ def main():
     for s in range(args.steps):
         tp = time.perf_counter()
         if debug and use_local_path:
-            conn.write_pages(src, put_keys[s], offsets_np, elems_per_block)
+            conn.write_pages(src, put_keys[s], offsets_np, elems_per_block, sync=True)
             tb = time.perf_counter()
             conn.sync()
             tc = time.perf_counter()

@@ -138,12 +138,12 @@ int ClientConn::rw_local(char op, const std::vector<std::pair<std::string, uint6
         offs.push_back(blocks[i].second);
     }
     return rw_local_packed(op, blob.data(), blob.size(), offs.data(), blocks.size(), block_size,
-                           ptr, device_id);
+                           ptr, device_id, false);
 }
 
 int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
                                 const uint64_t* offsets, size_t n, int block_size,
-                                uintptr_t ptr, int device_id) {
+                                uintptr_t ptr, int device_id, bool sync_response) {
     if (!connected_) return -1;
     if (!gpu::available()) {
         ERROR("local path requires a GPU");
@@ -178,8 +178,9 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
     h.block_size = static_cast<uint32_t>(block_size);
     h.n_blocks = static_cast<uint32_t>(n);
     // Reads complete in one round trip (response deferred to completion);
-    // writes stay async so uploads overlap compute (the prefill pattern).
-    h.flags = (op == 'R') ? kLocalFlagSyncResponse : 0;
+    // writes default to async so uploads overlap compute (prefill pattern),
+    // with an opt-in single-round-trip mode (write_pages(sync=True)).
+    h.flags = (op == 'R' || sync_response) ? kLocalFlagSyncResponse : 0;
     h.rsvd = 0;
     memcpy(h.ipc, handle.bytes, gpu::kIpcHandleSize);
 
@@ -197,7 +198,7 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
         WARN("rw_local op=%c -> %d", op, code);
         return -code;
     }
-    if (op == 'W') local_dirty_ = true;  // a sync round trip is needed
+    if (op == 'W' && !sync_response) local_dirty_ = true;  // sync RTT needed
     return 0;
 }
 

@@ -63,7 +63,7 @@ class ClientConn {
     // straight into the packed wire format (OP_W_FAST / OP_R_FAST).
     int rw_local_packed(char op, const char* keys_blob, size_t blob_len,
                         const uint64_t* offsets, size_t n, int block_size, uintptr_t ptr,
-                        int device_id);
+                        int device_id, bool sync_response = false);
     int sync_local();
 
     // ---- RDMA-semantics path ----

@@ -175,7 +175,7 @@ PYBIND11_MODULE(_native, m) {
         .def(
             "rw_local_fast",
             [](ClientConn& c, const std::string& op, py::bytes keys_blob, py::bytes offsets,
-               size_t n, int block_size, uintptr_t ptr, int device) {
+               size_t n, int block_size, uintptr_t ptr, int device, bool sync_response) {
                 char* kb;
                 Py_ssize_t kb_len;
                 PyBytes_AsStringAndSize(keys_blob.ptr(), &kb, &kb_len);
@@ -188,8 +188,11 @@ PYBIND11_MODULE(_native, m) {
                 return c.rw_local_packed(op.empty() ? 'W' : op[0], kb,
                                          static_cast<size_t>(kb_len),
                                          reinterpret_cast<const uint64_t*>(ob), n, block_size,
-                                         ptr, device);
-            })
+                                         ptr, device, sync_response);
+            },
+            py::arg("op"), py::arg("keys_blob"), py::arg("offsets"), py::arg("n"),
+            py::arg("block_size"), py::arg("ptr"), py::arg("device"),
+            py::arg("sync_response") = false)
         .def("sync_local", &ClientConn::sync_local, py::call_guard<py::gil_scoped_release>())
         .def("register_mr", &ClientConn::register_mr, py::call_guard<py::gil_scoped_release>())
         .def(

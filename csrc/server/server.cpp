@@ -659,15 +659,16 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
 
     if (job.src.empty()) {
         // everything was a duplicate — nothing to copy
-        return send_status(c, TASK_ACCEPTED);
+        return send_status(c, (msg.flags & kLocalFlagSyncResponse) ? FINISH : TASK_ACCEPTED);
     }
 
     c->remain.fetch_add(1);
     c->ref();
+    bool sync_resp = (msg.flags & kLocalFlagSyncResponse) != 0;
     auto entries = std::make_shared<std::vector<Ref<BlockEntry>>>(std::move(new_entries));
     auto keys = std::make_shared<std::vector<std::string>>(std::move(new_keys));
-    job.done = [this, c, entries, keys](bool ok) {
-        c->owner->post([this, c, entries, keys, ok] {
+    job.done = [this, c, entries, keys, sync_resp](bool ok) {
+        c->owner->post([this, c, entries, keys, ok, sync_resp] {
             if (ok) {
                 for (auto& e : *entries) e->committed = true;
             } else {
@@ -677,13 +678,14 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
                     if (it != kv_.end() && it->second.get() == (*entries)[i].get()) kv_.erase(it);
                 }
             }
+            if (sync_resp) send_status(c, ok ? FINISH : INTERNAL_ERROR);
             finish_task(c);
         });
     };
-    // Respond before submitting: the client's next action (more writes or a
-    // sync) overlaps the descriptor upload + kernel launch. A submit failure
-    // (shutdown) is reported through the rollback path; readers see 404.
-    send_status(c, TASK_ACCEPTED);
+    // Async write: respond before submitting so the client's next request
+    // overlaps the kernel (the prefill overlap pattern). Sync-response write
+    // (flags&1): one round trip, response sent on completion instead.
+    if (!sync_resp) send_status(c, TASK_ACCEPTED);
     if (!shard->submit_copy(std::move(job))) finish_task(c);
 }
 

@@ -332,7 +332,11 @@ class InfinityConnection:
     # offsets as parallel sequences; `offsets` may be a reusable
     # np.ndarray(dtype=uint64) of ELEMENT offsets, skipping per-call tuple
     # marshalling on the hot path.
-    def write_pages(self, cache: torch.Tensor, keys: List[str], offsets, page_size: int):
+    def write_pages(self, cache: torch.Tensor, keys: List[str], offsets, page_size: int,
+                    sync: bool = False):
+        """sync=True completes the write in ONE round trip (the response is
+        sent when the copy finishes); sync=False (default) returns after the
+        server accepts, so uploads overlap compute — call sync() later."""
         self._verify(cache)
         assert self.local_connected, "write_pages uses the local GPU path"
         es = cache.element_size()
@@ -340,7 +344,7 @@ class InfinityConnection:
         offs = np.asarray(offsets, dtype=np.uint64) * np.uint64(es)
         ret = self.conn.rw_local_fast(
             self.OP_W, blob, offs.tobytes(), len(keys), page_size * es,
-            cache.data_ptr(), _remap_device_id(cache),
+            cache.data_ptr(), _remap_device_id(cache), sync,
         )
         if ret < 0:
             raise Exception(f"Failed to write to infinistore, ret = {ret}")
