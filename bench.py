@@ -252,6 +252,8 @@ def main():
     purge_all()
 
     def pct(v, q):
+        if not v:
+            return 0.0
         return statistics.quantiles(v, n=100)[q - 1] if len(v) >= 10 else max(v)
 
     bytes_per_step_rank = args.blocks * block_bytes

@@ -190,10 +190,19 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
     memcpy(body.data() + sizeof(h) + n * 8, keys_blob, blob_len);
 
     char wire_op = (op == 'W') ? OP_W_FAST : OP_R_FAST;
+    static const bool dbg = getenv("IFS_CLIENT_DEBUG") != nullptr;
+    auto t0 = std::chrono::steady_clock::now();
     std::lock_guard<std::mutex> lk(io_mu_);
     if (!send_req(wire_op, body.data(), body.size())) return -1;
+    auto t1 = std::chrono::steady_clock::now();
     int code = 0;
     if (!recv_status(&code)) return -1;
+    if (dbg && n > 64) {
+        auto t2 = std::chrono::steady_clock::now();
+        fprintf(stderr, "[cdbg] op=%c n=%zu send=%.0fus wait=%.0fus\n", op, n,
+                std::chrono::duration<double, std::micro>(t1 - t0).count(),
+                std::chrono::duration<double, std::micro>(t2 - t1).count());
+    }
     if (code != TASK_ACCEPTED && code != FINISH) {
         WARN("rw_local op=%c -> %d", op, code);
         return -code;

@@ -530,21 +530,22 @@ Shard* Server::shard_least_used() {
 
 size_t Server::evict_lru_locked(Shard* shard, size_t bytes) {
     // Candidates: committed, idle (only the map holds a ref), on this shard.
-    std::vector<std::pair<uint64_t, const std::string*>> cands;
-    for (auto& kvp : kv_) {
-        BlockEntry* e = kvp.second.get();
+    // Key views stay valid across erase (the arena is append-only).
+    std::vector<std::pair<uint64_t, std::string_view>> cands;
+    kv_.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
+        BlockEntry* e = val.get();
         if (e->shard == shard && e->committed && e->ref_count() == 1)
-            cands.push_back({e->last_access, &kvp.first});
-    }
+            cands.push_back({e->last_access, key});
+    });
     std::sort(cands.begin(), cands.end(),
               [](const auto& a, const auto& b) { return a.first < b.first; });
     size_t freed = 0;
-    for (auto& [tick, key] : cands) {
+    for (auto& [tick_v, key] : cands) {
         if (freed >= bytes) break;
-        auto it = kv_.find(*key);
-        if (it == kv_.end()) continue;
-        freed += it->second->size;
-        kv_.erase(it);
+        Ref<BlockEntry>* v = kv_.find(key);
+        if (!v) continue;
+        freed += (*v)->size;
+        kv_.erase(key);
         n_evicted_.fetch_add(1);
     }
     if (freed) DEBUG("auto-evicted %zu bytes from shard dev=%d", freed, shard->device());
@@ -608,15 +609,16 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
         new_keys.reserve(msg.blocks.size());
         fresh_offs.reserve(msg.blocks.size());
         for (auto& b : msg.blocks) {
-            if (kv_.find(b.first) != kv_.end()) continue;
             auto* e = new BlockEntry();
             e->size = page;
             e->shard = shard;
             e->committed = false;
             e->last_access = t;
             Ref<BlockEntry> ref(e);
-            kv_.emplace(std::string(b.first), ref);
-            new_entries.push_back(ref);
+            bool inserted = false;
+            kv_.emplace(b.first, ref, &inserted);
+            if (!inserted) continue;  // dedup: first write wins (e freed via ref)
+            new_entries.push_back(std::move(ref));
             new_keys.emplace_back(b.first);
             fresh_offs.push_back(b.second);
         }
@@ -638,8 +640,8 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
     if (!alloc_ok) {
         std::lock_guard<std::mutex> lk(kv_mu_);
         for (size_t i = 0; i < n_fresh; i++) {
-            auto it = kv_.find(new_keys[i]);
-            if (it != kv_.end() && it->second.get() == new_entries[i].get()) kv_.erase(it);
+            Ref<BlockEntry>* v = kv_.find(new_keys[i]);
+            if (v && v->get() == new_entries[i].get()) kv_.erase(new_keys[i]);
         }
         return send_status(c, OUT_OF_MEMORY);
     }
@@ -665,17 +667,26 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
     c->remain.fetch_add(1);
     c->ref();
     bool sync_resp = (msg.flags & kLocalFlagSyncResponse) != 0;
+    static const bool sdbg = getenv("IFS_SERVER_DEBUG") != nullptr;
+    auto t_start = std::chrono::steady_clock::now();
     auto entries = std::make_shared<std::vector<Ref<BlockEntry>>>(std::move(new_entries));
     auto keys = std::make_shared<std::vector<std::string>>(std::move(new_keys));
-    job.done = [this, c, entries, keys, sync_resp](bool ok) {
+    job.done = [this, c, entries, keys, sync_resp, t_start](bool ok) {
+        if (sdbg && entries->size() > 64) {
+            auto us = std::chrono::duration<double, std::micro>(
+                          std::chrono::steady_clock::now() - t_start)
+                          .count();
+            fprintf(stderr, "[sdbg] write n=%zu submit->complete=%.0fus\n", entries->size(),
+                    us);
+        }
         c->owner->post([this, c, entries, keys, ok, sync_resp] {
             if (ok) {
                 for (auto& e : *entries) e->committed = true;
             } else {
                 std::lock_guard<std::mutex> lk(kv_mu_);
                 for (size_t i = 0; i < keys->size(); i++) {
-                    auto it = kv_.find((*keys)[i]);
-                    if (it != kv_.end() && it->second.get() == (*entries)[i].get()) kv_.erase(it);
+                    Ref<BlockEntry>* v = kv_.find((*keys)[i]);
+                    if (v && v->get() == (*entries)[i].get()) kv_.erase((*keys)[i]);
                 }
             }
             if (sync_resp) send_status(c, ok ? FINISH : INTERNAL_ERROR);
@@ -706,17 +717,17 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
     {
         std::lock_guard<std::mutex> lk(kv_mu_);
         for (auto& b : msg.blocks) {
-            auto kit = kv_.find(b.first);
-            if (kit == kv_.end() || !kit->second->committed) {
+            Ref<BlockEntry>* v = kv_.find(b.first);
+            if (!v || !(*v)->committed) {
                 return send_status(c, KEY_NOT_FOUND);
             }
-            BlockEntry* e = kit->second.get();
+            BlockEntry* e = v->get();
             e->last_access = read_tick;
             auto& job = jobs[e->shard];
             job.bytes_per_block = page;
             job.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
             job.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.second));
-            held->push_back(kit->second);
+            held->push_back(*v);
         }
     }
     n_reads_.fetch_add(1);
@@ -730,7 +741,15 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
     auto all_ok = std::make_shared<std::atomic<bool>>(true);
     for (auto& [shard, job] : jobs) {
         Shard::CopyJob j = std::move(job);
-        j.done = [this, c, held, pending, all_ok, sync_resp](bool ok) {
+        static const bool sdbg2 = getenv("IFS_SERVER_DEBUG") != nullptr;
+        auto t_start2 = std::chrono::steady_clock::now();
+        j.done = [this, c, held, pending, all_ok, sync_resp, t_start2](bool ok) {
+            if (sdbg2 && held->size() > 64) {
+                auto us = std::chrono::duration<double, std::micro>(
+                              std::chrono::steady_clock::now() - t_start2)
+                              .count();
+                fprintf(stderr, "[sdbg] read n=%zu submit->complete=%.0fus\n", held->size(), us);
+            }
             if (!ok) all_ok->store(false);
             if (pending->fetch_sub(1) == 1) {
                 c->owner->post([this, c, held, all_ok, sync_resp] {
@@ -794,7 +813,7 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
     {
         std::lock_guard<std::mutex> lk(kv_mu_);
         for (auto& key : keys) {
-            if (kv_.count(key)) {
+            if (kv_.find(key) != nullptr) {
                 blocks.push_back({0, 0, 0});  // FAKE block: dup key, client skips
                 continue;
             }
@@ -820,7 +839,8 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
             e->shard = shard;
             e->last_access = tick();
             Ref<BlockEntry> ref(e);
-            kv_.emplace(key, ref);
+            bool ins = false;
+            kv_.emplace(key, ref, &ins);
             created.push_back(key);
             c->pending_rdma.emplace(reinterpret_cast<uint64_t>(ptr), ref);
             blocks.push_back({static_cast<uint32_t>(shard->device() + 1), 0,
@@ -846,10 +866,10 @@ bool Server::collect_read_entries(const std::vector<std::string>& keys,
     out->reserve(keys.size());
     uint64_t t = tick();
     for (auto& key : keys) {
-        auto it = kv_.find(key);
-        if (it == kv_.end() || !it->second->committed) return false;
-        it->second->last_access = t;
-        out->push_back(it->second);
+        Ref<BlockEntry>* v = kv_.find(key);
+        if (!v || !(*v)->committed) return false;
+        (*v)->last_access = t;
+        out->push_back(*v);
     }
     return true;
 }
@@ -996,8 +1016,8 @@ void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body) {
     bool exists;
     {
         std::lock_guard<std::mutex> lk(kv_mu_);
-        auto it = kv_.find(key);
-        exists = it != kv_.end() && it->second->committed;
+        Ref<BlockEntry>* v = kv_.find(key);
+        exists = v && (*v)->committed;
     }
     send_status(c, exists ? 0 : 1);
 }
@@ -1013,8 +1033,8 @@ void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body) {
     // reporting it as a hit would make the subsequent read_cache fail.
     std::lock_guard<std::mutex> lk(kv_mu_);
     auto present = [&](size_t i) {
-        auto it = kv_.find(keys[i]);
-        return it != kv_.end() && it->second->committed;
+        Ref<BlockEntry>* v = kv_.find(keys[i]);
+        return v && (*v)->committed;
     };
     long left = 0, right = static_cast<long>(keys.size());
     while (left < right) {
@@ -1034,7 +1054,7 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
     int n = 0;
     {
         std::lock_guard<std::mutex> lk(kv_mu_);
-        for (auto& k : keys) n += static_cast<int>(kv_.erase(k));
+        for (auto& k : keys) n += kv_.erase(k) ? 1 : 0;
     }
     send_status(c, n);
 }
@@ -1052,11 +1072,11 @@ std::pair<size_t, size_t> Server::compact() {
     for (auto& shard_up : shards_) {
         Shard* shard = shard_up.get();
         std::vector<std::pair<void*, size_t>> movable;
-        for (auto& kvp : kv_) {
-            BlockEntry* e = kvp.second.get();
+        kv_.for_each([&](std::string_view, Ref<BlockEntry>& val) {
+            BlockEntry* e = val.get();
             if (e->shard == shard && e->committed && e->ref_count() == 1)
                 movable.push_back({e->ptr, e->size});
-        }
+        });
         if (movable.empty()) continue;
         auto moves = shard->plan_compaction(movable);
         if (moves.empty()) continue;
@@ -1081,17 +1101,17 @@ std::pair<size_t, size_t> Server::compact() {
         // Swap pointers in the index and free the old slots.
         std::map<void*, Shard::Move*> by_old;
         for (auto& m : moves) by_old[m.old_ptr] = &m;
-        for (auto& kvp : kv_) {
-            BlockEntry* e = kvp.second.get();
+        kv_.for_each([&](std::string_view, Ref<BlockEntry>& val) {
+            BlockEntry* e = val.get();
             auto it = by_old.find(e->ptr);
-            if (it == by_old.end() || e->shard != shard) continue;
+            if (it == by_old.end() || e->shard != shard) return;
             Shard::Move* m = it->second;
             e->ptr = m->new_ptr;
             e->pool_idx = m->pool_idx;
             shard->deallocate(m->old_ptr, m->size, m->pool_idx);
             moved++;
             bytes += m->size;
-        }
+        });
     }
     return {moved, bytes};
 }

@@ -41,6 +41,7 @@
 #include "../core/utils.h"
 #include "../fabric/verbs_fabric.h"
 #include "../gpu/gpu.h"
+#include "kvmap.h"
 #include "shard.h"
 
 namespace ifs {
@@ -210,16 +211,10 @@ class Server {
     std::atomic<bool> running_{false};
     std::atomic<bool> stop_requested_{false};
 
-    // Heterogeneous-lookup map (C++20): hot-path lookups by string_view
-    // avoid a std::string allocation per key.
-    struct SvHash {
-        using is_transparent = void;
-        size_t operator()(std::string_view s) const noexcept {
-            return std::hash<std::string_view>{}(s);
-        }
-    };
+    // Open-addressing key index with arena-stored keys (csrc/server/kvmap.h)
+    // — node-based maps measured ~350 µs of insert cost per 2048-key write.
     std::mutex kv_mu_;
-    std::unordered_map<std::string, Ref<BlockEntry>, SvHash, std::equal_to<>> kv_;
+    KvMap kv_;
 
     // stats
     std::atomic<uint64_t> n_writes_{0}, n_reads_{0}, n_put_{0}, n_get_{0};
