@@ -23,6 +23,9 @@ struct EntrySlab {
 
     void* take(size_t n) {
         std::lock_guard<std::mutex> lk(mu);
+        return take_locked(n);
+    }
+    void* take_locked(size_t n) {
         if (free_list.empty()) {
             chunks.emplace_back(new char[n * kPerChunk]);
             char* base = chunks.back().get();
@@ -32,6 +35,15 @@ struct EntrySlab {
         void* p = free_list.back();
         free_list.pop_back();
         return p;
+    }
+    // Pre-size the freelist for a burst (one lock for the whole batch).
+    void prepare(size_t n, size_t count) {
+        std::lock_guard<std::mutex> lk(mu);
+        while (free_list.size() < count) {
+            chunks.emplace_back(new char[n * kPerChunk]);
+            char* base = chunks.back().get();
+            for (size_t i = 0; i < kPerChunk; i++) free_list.push_back(base + i * n);
+        }
     }
     void give(void* p) {
         std::lock_guard<std::mutex> lk(mu);
@@ -528,6 +540,19 @@ Shard* Server::shard_least_used() {
     return best;
 }
 
+void Server::erase_entries(const std::vector<Ref<BlockEntry>>& entries) {
+    // Rare path (allocation failure / failed copy): remove entries from the
+    // index by identity. O(map) scan, but keys are not kept around on the
+    // hot path.
+    std::lock_guard<std::mutex> lk(kv_mu_);
+    std::vector<std::string> victims;
+    kv_.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
+        for (auto& e : entries)
+            if (val.get() == e.get()) victims.emplace_back(key);
+    });
+    for (auto& k : victims) kv_.erase(k);
+}
+
 size_t Server::evict_lru_locked(Shard* shard, size_t bytes) {
     // Candidates: committed, idle (only the map holds a ref), on this shard.
     // Key views stay valid across erase (the arena is append-only).
@@ -596,7 +621,6 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
     Shard::CopyJob job;
     job.bytes_per_block = page;
     std::vector<Ref<BlockEntry>> new_entries;
-    std::vector<std::string> new_keys;
     std::vector<uint64_t> fresh_offs;
     {
         // Short critical section: dedup (first write wins) + placeholder
@@ -606,7 +630,6 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
         uint64_t t = tick();
         std::lock_guard<std::mutex> lk(kv_mu_);
         new_entries.reserve(msg.blocks.size());
-        new_keys.reserve(msg.blocks.size());
         fresh_offs.reserve(msg.blocks.size());
         for (auto& b : msg.blocks) {
             auto* e = new BlockEntry();
@@ -619,7 +642,6 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
             kv_.emplace(b.first, ref, &inserted);
             if (!inserted) continue;  // dedup: first write wins (e freed via ref)
             new_entries.push_back(std::move(ref));
-            new_keys.emplace_back(b.first);
             fresh_offs.push_back(b.second);
         }
     }
@@ -638,11 +660,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
         if (evict_lru_locked(shard, page * n_fresh) > 0) alloc_ok = try_alloc();
     }
     if (!alloc_ok) {
-        std::lock_guard<std::mutex> lk(kv_mu_);
-        for (size_t i = 0; i < n_fresh; i++) {
-            Ref<BlockEntry>* v = kv_.find(new_keys[i]);
-            if (v && v->get() == new_entries[i].get()) kv_.erase(new_keys[i]);
-        }
+        erase_entries(new_entries);  // rollback (rare path: scans the index)
         return send_status(c, OUT_OF_MEMORY);
     }
     job.src.reserve(n_fresh);
@@ -670,8 +688,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
     static const bool sdbg = getenv("IFS_SERVER_DEBUG") != nullptr;
     auto t_start = std::chrono::steady_clock::now();
     auto entries = std::make_shared<std::vector<Ref<BlockEntry>>>(std::move(new_entries));
-    auto keys = std::make_shared<std::vector<std::string>>(std::move(new_keys));
-    job.done = [this, c, entries, keys, sync_resp, t_start](bool ok) {
+    job.done = [this, c, entries, sync_resp, t_start](bool ok) {
         if (sdbg && entries->size() > 64) {
             auto us = std::chrono::duration<double, std::micro>(
                           std::chrono::steady_clock::now() - t_start)
@@ -679,15 +696,11 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
             fprintf(stderr, "[sdbg] write n=%zu submit->complete=%.0fus\n", entries->size(),
                     us);
         }
-        c->owner->post([this, c, entries, keys, ok, sync_resp] {
+        c->owner->post([this, c, entries, ok, sync_resp] {
             if (ok) {
                 for (auto& e : *entries) e->committed = true;
             } else {
-                std::lock_guard<std::mutex> lk(kv_mu_);
-                for (size_t i = 0; i < keys->size(); i++) {
-                    Ref<BlockEntry>* v = kv_.find((*keys)[i]);
-                    if (v && v->get() == (*entries)[i].get()) kv_.erase((*keys)[i]);
-                }
+                erase_entries(*entries);  // copy failed: drop the keys
             }
             if (sync_resp) send_status(c, ok ? FINISH : INTERNAL_ERROR);
             finish_task(c);
