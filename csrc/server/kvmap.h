@@ -31,9 +31,18 @@ class KvMap {
 
     size_t size() const { return size_; }
 
-    Ref<BlockEntry>* find(std::string_view key) {
+    // Batched-probe support: precompute hashes, prefetch the slot lines a
+    // few iterations ahead, then probe — hides the one cold cache miss per
+    // key that dominates bulk inserts into a multi-MB table.
+    static uint64_t hash_of(std::string_view key) { return hash_key(key); }
+    void prefetch(uint64_t h) const {
+        __builtin_prefetch(&slots_[h & (slots_.size() - 1)]);
+    }
+
+    Ref<BlockEntry>* find(std::string_view key) { return find_hashed(key, hash_key(key)); }
+
+    Ref<BlockEntry>* find_hashed(std::string_view key, uint64_t h) {
         size_t mask = slots_.size() - 1;
-        uint64_t h = hash_key(key);
         for (size_t i = h & mask;; i = (i + 1) & mask) {
             Slot& s = slots_[i];
             if (s.state == kEmpty) return nullptr;
@@ -43,9 +52,13 @@ class KvMap {
 
     // Insert if absent. Returns the value slot and sets *inserted.
     Ref<BlockEntry>* emplace(std::string_view key, Ref<BlockEntry> val, bool* inserted) {
+        return emplace_hashed(key, hash_key(key), std::move(val), inserted);
+    }
+
+    Ref<BlockEntry>* emplace_hashed(std::string_view key, uint64_t h, Ref<BlockEntry> val,
+                                    bool* inserted) {
         maybe_grow();
         size_t mask = slots_.size() - 1;
-        uint64_t h = hash_key(key);
         size_t first_tomb = SIZE_MAX;
         for (size_t i = h & mask;; i = (i + 1) & mask) {
             Slot& s = slots_[i];

@@ -608,6 +608,8 @@ void* resolve_client_base(Server::Conn* c, const Server::LocalView& msg) {
 }  // namespace
 
 void Server::op_local_write(Conn* c, const LocalView& msg) {
+    static const bool pdbg = getenv("IFS_SERVER_DEBUG") != nullptr;
+    auto p0 = std::chrono::steady_clock::now();
     if (!gpu::available()) return send_status(c, SYSTEM_ERROR);
     if (msg.ipc_len != gpu::kIpcHandleSize || msg.block_size <= 0)
         return send_status(c, INVALID_REQ);
@@ -628,10 +630,19 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
         // concurrent IO threads don't serialize on the index for the whole
         // request (uncommitted entries are invisible to reads).
         uint64_t t = tick();
+        size_t nb = msg.blocks.size();
+        // Hash pass + prefetch pipeline: the emplace probe into the multi-MB
+        // slot array is one cold cache miss per key; prefetching ~16 keys
+        // ahead overlaps those misses.
+        std::vector<uint64_t> hashes(nb);
+        for (size_t i = 0; i < nb; i++) hashes[i] = KvMap::hash_of(msg.blocks[i].first);
+        constexpr size_t kPf = 16;
         std::lock_guard<std::mutex> lk(kv_mu_);
-        new_entries.reserve(msg.blocks.size());
-        fresh_offs.reserve(msg.blocks.size());
-        for (auto& b : msg.blocks) {
+        for (size_t i = 0; i < std::min(kPf, nb); i++) kv_.prefetch(hashes[i]);
+        new_entries.reserve(nb);
+        fresh_offs.reserve(nb);
+        for (size_t i = 0; i < nb; i++) {
+            if (i + kPf < nb) kv_.prefetch(hashes[i + kPf]);
             auto* e = new BlockEntry();
             e->size = page;
             e->shard = shard;
@@ -639,12 +650,13 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
             e->last_access = t;
             Ref<BlockEntry> ref(e);
             bool inserted = false;
-            kv_.emplace(b.first, ref, &inserted);
+            kv_.emplace_hashed(msg.blocks[i].first, hashes[i], ref, &inserted);
             if (!inserted) continue;  // dedup: first write wins (e freed via ref)
             new_entries.push_back(std::move(ref));
-            fresh_offs.push_back(b.second);
+            fresh_offs.push_back(msg.blocks[i].second);
         }
     }
+    auto p1 = std::chrono::steady_clock::now();
     // Batched allocation (shard allocator lock only).
     size_t n_fresh = new_entries.size();
     std::vector<std::pair<void*, int>> slots;
@@ -663,6 +675,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
         erase_entries(new_entries);  // rollback (rare path: scans the index)
         return send_status(c, OUT_OF_MEMORY);
     }
+    auto p2 = std::chrono::steady_clock::now();
     job.src.reserve(n_fresh);
     job.dst.reserve(n_fresh);
     for (size_t i = 0; i < n_fresh; i++) {
@@ -671,6 +684,14 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
         e->pool_idx = slots[i].second;
         job.src.push_back(reinterpret_cast<uint64_t>(client_ptr + fresh_offs[i]));
         job.dst.push_back(reinterpret_cast<uint64_t>(slots[i].first));
+    }
+    auto p3 = std::chrono::steady_clock::now();
+    if (pdbg && n_fresh > 64) {
+        auto us = [](auto a, auto b) {
+            return std::chrono::duration<double, std::micro>(b - a).count();
+        };
+        fprintf(stderr, "[pdbg] insert=%.0f alloc=%.0f build=%.0f\n", us(p0, p1), us(p1, p2),
+                us(p2, p3));
     }
 
     n_writes_.fetch_add(1);
@@ -728,9 +749,17 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
     held->reserve(msg.blocks.size());
     uint64_t read_tick = tick();
     {
+        size_t nb2 = msg.blocks.size();
+        std::vector<uint64_t> hashes(nb2);
+        for (size_t i = 0; i < nb2; i++) hashes[i] = KvMap::hash_of(msg.blocks[i].first);
+        constexpr size_t kPf = 16;
         std::lock_guard<std::mutex> lk(kv_mu_);
+        for (size_t i = 0; i < std::min(kPf, nb2); i++) kv_.prefetch(hashes[i]);
+        size_t bi = 0;
         for (auto& b : msg.blocks) {
-            Ref<BlockEntry>* v = kv_.find(b.first);
+            if (bi + kPf < nb2) kv_.prefetch(hashes[bi + kPf]);
+            Ref<BlockEntry>* v = kv_.find_hashed(b.first, hashes[bi]);
+            bi++;
             if (!v || !(*v)->committed) {
                 return send_status(c, KEY_NOT_FOUND);
             }
