@@ -62,6 +62,27 @@ void* BlockEntry::operator new(size_t n) { return entry_slab().take(n); }
 void BlockEntry::operator delete(void* p) { entry_slab().give(p); }
 
 namespace {
+// One lock for a whole burst of entry allocations (hot write path).
+struct SlabBatch {
+    std::vector<void*> slots;
+    size_t next = 0;
+    explicit SlabBatch(size_t count) {
+        slots.reserve(count);
+        auto& slab = entry_slab();
+        std::lock_guard<std::mutex> lk(slab.mu);
+        for (size_t i = 0; i < count; i++) slots.push_back(slab.take_locked(sizeof(BlockEntry)));
+    }
+    ~SlabBatch() {
+        if (next >= slots.size()) return;
+        auto& slab = entry_slab();
+        std::lock_guard<std::mutex> lk(slab.mu);
+        for (size_t i = next; i < slots.size(); i++) slab.free_list.push_back(slots[i]);
+    }
+    BlockEntry* make() { return new (slots[next++]) BlockEntry(); }
+};
+}  // namespace
+
+namespace {
 
 struct WriteReq {
     uv_write_t req;
@@ -637,13 +658,14 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
         std::vector<uint64_t> hashes(nb);
         for (size_t i = 0; i < nb; i++) hashes[i] = KvMap::hash_of(msg.blocks[i].first);
         constexpr size_t kPf = 16;
+        SlabBatch slab_batch(nb);  // one lock for the whole entry burst
         std::lock_guard<std::mutex> lk(kv_mu_);
         for (size_t i = 0; i < std::min(kPf, nb); i++) kv_.prefetch(hashes[i]);
         new_entries.reserve(nb);
         fresh_offs.reserve(nb);
         for (size_t i = 0; i < nb; i++) {
             if (i + kPf < nb) kv_.prefetch(hashes[i + kPf]);
-            auto* e = new BlockEntry();
+            auto* e = slab_batch.make();
             e->size = page;
             e->shard = shard;
             e->committed = false;

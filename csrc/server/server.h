@@ -89,6 +89,9 @@ struct BlockEntry : RefCounted {
     }
     static void* operator new(size_t n);
     static void operator delete(void* p);
+    // placement forms (class-level operator new hides the global ones)
+    static void* operator new(size_t, void* p) { return p; }
+    static void operator delete(void*, void*) {}
 };
 
 class Server {
