@@ -286,7 +286,19 @@ void Shard::completion_loop() {
         }
         bool ok = t.slot != nullptr;
         if (t.slot) {
-            ok = gpu::event_sync(t.slot->event);
+            // Spin briefly before blocking: hipEventSynchronize's interrupt
+            // wake costs ~20 us; most copies finish within the spin window.
+            bool done = false;
+            for (int spin = 0; spin < 4000; spin++) {
+                if (gpu::event_query(t.slot->event)) {
+                    done = true;
+                    break;
+                }
+#if defined(__x86_64__)
+                __builtin_ia32_pause();
+#endif
+            }
+            ok = done || gpu::event_sync(t.slot->event);
             std::lock_guard<std::mutex> lk(task_mu_);
             t.slot->busy = false;
         }
