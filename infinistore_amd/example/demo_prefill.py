@@ -23,10 +23,10 @@ PAGE_ELEMS = 16384  # 32 KB fp16 pages
 PAGES_PER_LAYER = 16
 
 
-def main():
+def main(port: int = 22345):
     assert torch.cuda.is_available(), "demo needs a GPU"
     cfg = ifs.ClientConfig(
-        host_addr="127.0.0.1", service_port=22345,
+        host_addr="127.0.0.1", service_port=port,
         connection_type=ifs.TYPE_LOCAL_GPU,
     )
     conn = ifs.InfinityConnection(cfg)
@@ -92,4 +92,6 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    import sys
+
+    main(int(sys.argv[1]) if len(sys.argv) > 1 else 22345)

@@ -331,3 +331,10 @@ def test_example_client_run(gpu_server):
 
     ex.run(conn, "cuda:0", "cuda:0")
     conn.close()
+
+
+def test_demo_prefill_example(gpu_server):
+    """The layer-overlap prefill pattern example end to end."""
+    from infinistore_amd.example import demo_prefill
+
+    demo_prefill.main(gpu_server)
