@@ -640,98 +640,85 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
 
     Shard* shard = shard_for_device(msg.device);
     size_t page = static_cast<size_t>(msg.block_size);
+    size_t nb = msg.blocks.size();
+    constexpr size_t kPf = 16;
 
-    Shard::CopyJob job;
-    job.bytes_per_block = page;
-    std::vector<Ref<BlockEntry>> new_entries;
-    std::vector<uint64_t> fresh_offs;
+    // Phase A — dedup check only (short kv_mu_ hold, prefetch-pipelined).
+    // The authoritative first-write-wins decision happens at the insert pass
+    // (phase D); a key that appears between A and D just wastes one block
+    // copy whose memory is released after the kernel completes.
+    std::vector<uint64_t> hashes(nb);
+    for (size_t i = 0; i < nb; i++) hashes[i] = KvMap::hash_of(msg.blocks[i].first);
+    std::vector<uint32_t> fresh;
+    fresh.reserve(nb);
     {
-        // Short critical section: dedup (first write wins) + placeholder
-        // inserts. Allocation and job building run outside kv_mu_ so
-        // concurrent IO threads don't serialize on the index for the whole
-        // request (uncommitted entries are invisible to reads).
-        uint64_t t = tick();
-        size_t nb = msg.blocks.size();
-        // Hash pass + prefetch pipeline: the emplace probe into the multi-MB
-        // slot array is one cold cache miss per key; prefetching ~16 keys
-        // ahead overlaps those misses.
-        std::vector<uint64_t> hashes(nb);
-        for (size_t i = 0; i < nb; i++) hashes[i] = KvMap::hash_of(msg.blocks[i].first);
-        constexpr size_t kPf = 16;
-        SlabBatch slab_batch(nb);  // one lock for the whole entry burst
         std::lock_guard<std::mutex> lk(kv_mu_);
         for (size_t i = 0; i < std::min(kPf, nb); i++) kv_.prefetch(hashes[i]);
-        new_entries.reserve(nb);
-        fresh_offs.reserve(nb);
         for (size_t i = 0; i < nb; i++) {
             if (i + kPf < nb) kv_.prefetch(hashes[i + kPf]);
-            auto* e = slab_batch.make();
-            e->size = page;
-            e->shard = shard;
-            e->committed = false;
-            e->last_access = t;
-            Ref<BlockEntry> ref(e);
-            bool inserted = false;
-            kv_.emplace_hashed(msg.blocks[i].first, hashes[i], ref, &inserted);
-            if (!inserted) continue;  // dedup: first write wins (e freed via ref)
-            new_entries.push_back(std::move(ref));
-            fresh_offs.push_back(msg.blocks[i].second);
+            if (!kv_.find_hashed(msg.blocks[i].first, hashes[i]))
+                fresh.push_back(static_cast<uint32_t>(i));
         }
     }
     auto p1 = std::chrono::steady_clock::now();
-    // Batched allocation (shard allocator lock only).
-    size_t n_fresh = new_entries.size();
+
+    n_writes_.fetch_add(1);
+    size_t n_fresh = fresh.size();
+    bytes_in_.fetch_add(n_fresh * page);
+    bool sync_resp = (msg.flags & kLocalFlagSyncResponse) != 0;
+    if (n_fresh == 0) {
+        // everything was a duplicate — nothing to copy
+        return send_status(c, sync_resp ? FINISH : TASK_ACCEPTED);
+    }
+
+    // Phase B — batched allocation (shard allocator lock only).
     std::vector<std::pair<void*, int>> slots;
     slots.reserve(n_fresh);
     auto try_alloc = [&] {
-        return n_fresh == 0 ||
-               shard->allocate(page, n_fresh,
+        return shard->allocate(page, n_fresh,
                                [&](void* p, int idx) { slots.push_back({p, idx}); });
     };
     bool alloc_ok = try_alloc();
     if (!alloc_ok && opt_.auto_evict) {
         std::lock_guard<std::mutex> lk(kv_mu_);
-        if (evict_lru_locked(shard, page * n_fresh) > 0) alloc_ok = try_alloc();
+        if (evict_lru_locked(shard, page * n_fresh * 2) > 0) alloc_ok = try_alloc();
     }
-    if (!alloc_ok) {
-        erase_entries(new_entries);  // rollback (rare path: scans the index)
-        return send_status(c, OUT_OF_MEMORY);
-    }
+    if (!alloc_ok) return send_status(c, OUT_OF_MEMORY);
     auto p2 = std::chrono::steady_clock::now();
+
+    // Phase C — create entries, build the copy job and LAUNCH it; the index
+    // insert pass (D) then runs while the kernel is in flight.
+    uint64_t t = tick();
+    SlabBatch slab_batch(n_fresh);
+    auto entries = std::make_shared<std::vector<Ref<BlockEntry>>>();
+    entries->reserve(n_fresh);
+    Shard::CopyJob job;
+    job.bytes_per_block = page;
     job.src.reserve(n_fresh);
     job.dst.reserve(n_fresh);
     for (size_t i = 0; i < n_fresh; i++) {
-        BlockEntry* e = new_entries[i].get();
+        auto* e = slab_batch.make();
         e->ptr = slots[i].first;
+        e->size = page;
         e->pool_idx = slots[i].second;
-        job.src.push_back(reinterpret_cast<uint64_t>(client_ptr + fresh_offs[i]));
+        e->shard = shard;
+        e->committed = false;
+        e->last_access = t;
+        entries->emplace_back(e);
+        job.src.push_back(
+            reinterpret_cast<uint64_t>(client_ptr + msg.blocks[fresh[i]].second));
         job.dst.push_back(reinterpret_cast<uint64_t>(slots[i].first));
     }
-    auto p3 = std::chrono::steady_clock::now();
-    if (pdbg && n_fresh > 64) {
-        auto us = [](auto a, auto b) {
-            return std::chrono::duration<double, std::micro>(b - a).count();
-        };
-        fprintf(stderr, "[pdbg] insert=%.0f alloc=%.0f build=%.0f\n", us(p0, p1), us(p1, p2),
-                us(p2, p3));
-    }
-
-    n_writes_.fetch_add(1);
-    bytes_in_.fetch_add(job.src.size() * page);
     maybe_extend(shard);
-
-    if (job.src.empty()) {
-        // everything was a duplicate — nothing to copy
-        return send_status(c, (msg.flags & kLocalFlagSyncResponse) ? FINISH : TASK_ACCEPTED);
-    }
 
     c->remain.fetch_add(1);
     c->ref();
-    bool sync_resp = (msg.flags & kLocalFlagSyncResponse) != 0;
     static const bool sdbg = getenv("IFS_SERVER_DEBUG") != nullptr;
     auto t_start = std::chrono::steady_clock::now();
-    auto entries = std::make_shared<std::vector<Ref<BlockEntry>>>(std::move(new_entries));
-    job.done = [this, c, entries, sync_resp, t_start](bool ok) {
+    // won[i] is written in phase D on this thread BEFORE the posted commit
+    // lambda can run on the same (owner-loop) thread.
+    auto won = std::make_shared<std::vector<uint8_t>>(n_fresh, 0);
+    job.done = [this, c, entries, won, sync_resp, t_start](bool ok) {
         if (sdbg && entries->size() > 64) {
             auto us = std::chrono::duration<double, std::micro>(
                           std::chrono::steady_clock::now() - t_start)
@@ -739,21 +726,49 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
             fprintf(stderr, "[sdbg] write n=%zu submit->complete=%.0fus\n", entries->size(),
                     us);
         }
-        c->owner->post([this, c, entries, ok, sync_resp] {
+        c->owner->post([this, c, entries, won, ok, sync_resp] {
             if (ok) {
-                for (auto& e : *entries) e->committed = true;
+                for (size_t i = 0; i < entries->size(); i++)
+                    if ((*won)[i]) (*entries)[i]->committed = true;
             } else {
-                erase_entries(*entries);  // copy failed: drop the keys
+                std::vector<Ref<BlockEntry>> winners;
+                for (size_t i = 0; i < entries->size(); i++)
+                    if ((*won)[i]) winners.push_back((*entries)[i]);
+                erase_entries(winners);  // copy failed: drop the keys
             }
             if (sync_resp) send_status(c, ok ? FINISH : INTERNAL_ERROR);
             finish_task(c);
         });
     };
     // Async write: respond before submitting so the client's next request
-    // overlaps the kernel (the prefill overlap pattern). Sync-response write
-    // (flags&1): one round trip, response sent on completion instead.
+    // overlaps the kernel. Sync-response write (flags&1): one round trip,
+    // response sent on completion instead.
     if (!sync_resp) send_status(c, TASK_ACCEPTED);
-    if (!shard->submit_copy(std::move(job))) finish_task(c);
+    bool submitted = shard->submit_copy(std::move(job));
+
+    // Phase D — insert pass, overlapped with the in-flight kernel. Losers
+    // (a racing writer inserted the key first) keep their block alive until
+    // the copy completes, then release it; they are never committed.
+    {
+        std::lock_guard<std::mutex> lk(kv_mu_);
+        for (size_t i = 0; i < std::min(kPf, n_fresh); i++) kv_.prefetch(hashes[fresh[i]]);
+        for (size_t i = 0; i < n_fresh; i++) {
+            if (i + kPf < n_fresh) kv_.prefetch(hashes[fresh[i + kPf]]);
+            bool inserted = false;
+            kv_.emplace_hashed(msg.blocks[fresh[i]].first, hashes[fresh[i]], (*entries)[i],
+                               &inserted);
+            (*won)[i] = inserted ? 1 : 0;
+        }
+    }
+    auto p3 = std::chrono::steady_clock::now();
+    if (pdbg && n_fresh > 64) {
+        auto us = [](auto a, auto b) {
+            return std::chrono::duration<double, std::micro>(b - a).count();
+        };
+        fprintf(stderr, "[pdbg] dedup=%.0f alloc=%.0f build+insert=%.0f\n", us(p0, p1),
+                us(p1, p2), us(p2, p3));
+    }
+    if (!submitted) finish_task(c);
 }
 
 void Server::op_local_read(Conn* c, const LocalView& msg) {
