@@ -224,7 +224,12 @@ bool Server::start() {
         w->stop_async.data = w.get();
         workers_.push_back(std::move(w));
     }
-    kv_.reserve(1u << 20);  // avoid rehash storms during bulk prefill writes
+    // Pre-size the index to avoid rehash storms during bulk prefill writes.
+    // IFS_KV_INITIAL (tests only) shrinks the initial capacity so rehash and
+    // tombstone-compaction paths are exercised by small workloads.
+    size_t kv_initial = 1u << 20;
+    if (const char* env = getenv("IFS_KV_INITIAL")) kv_initial = strtoull(env, nullptr, 10);
+    kv_.reserve(kv_initial);
     running_.store(true);
     stop_requested_.store(false);
     main_io_.start();

@@ -117,3 +117,63 @@ def test_garbage_flatbuffer_body(cpu_server):
     s.sendall(struct.pack("<IcI", 0xDEADBEEF, b"S", 0))
     assert struct.unpack("<i", s.recv(4))[0] == 0
     s.close()
+
+
+def test_kv_index_rehash_churn(ports):
+    """Force the kv index through growth rehashes and tombstone compaction:
+    tiny initial capacity + insert/delete churn, verifying contents
+    throughout."""
+    import os
+    import subprocess
+    import sys
+    import time as _t
+
+    # The knob is read at server start; run in a subprocess with it set.
+    code = r"""
+import os, torch, uuid
+import infinistore_amd as ifs
+port = int(os.environ["PORT"])
+ifs.register_server(ifs.ServerConfig(service_port=port, manage_port=port+1,
+                                     prealloc_size=1, minimal_allocate_size=16,
+                                     cpu_only=True))
+cfg = ifs.ClientConfig(host_addr="127.0.0.1", service_port=port,
+                       connection_type=ifs.TYPE_RDMA, link_type="TCP")
+conn = ifs.InfinityConnection(cfg)
+conn.connect()
+src = torch.arange(4096, dtype=torch.float32)
+dst = torch.zeros(4096, dtype=torch.float32)
+conn.register_mr(src); conn.register_mr(dst)
+live = {}
+import random
+rng = random.Random(3)
+for gen in range(40):
+    keys = [f"g{gen}-{i}-{uuid.uuid4().hex[:6]}" for i in range(64)]
+    blocks = conn.allocate_rdma(keys, 1024)
+    conn.rdma_write_cache(src, [ (i % 16) * 256 for i in range(64)], 256, blocks)
+    conn.sync()
+    for i, k in enumerate(keys):
+        live[k] = (i % 16) * 256
+    # delete a random half of an old generation
+    victims = rng.sample(list(live.keys()), min(40, len(live)//2))
+    conn.delete_keys(victims)
+    for v in victims: live.pop(v)
+    # verify a random sample reads back right
+    sample = rng.sample(list(live.items()), min(8, len(live)))
+    for k, off in sample:
+        conn.read_cache(dst, [(k, 0)], 256)
+        conn.sync()
+        assert torch.equal(dst[:256], src[off:off+256]), k
+assert ifs.get_kvmap_len() == len(live)
+conn.delete_keys(list(live.keys()))
+assert ifs.get_kvmap_len() == 0
+conn.close(); ifs.unregister_server()
+print("CHURN_OK")
+"""
+    service_port, _ = ports
+    env = dict(os.environ)
+    env["IFS_KV_INITIAL"] = "256"
+    env["PORT"] = str(service_port)
+    r = subprocess.run([sys.executable, "-c", code], env=env, capture_output=True,
+                       text=True, timeout=300, cwd=os.path.dirname(os.path.dirname(
+                           os.path.abspath(__file__))))
+    assert r.returncode == 0 and "CHURN_OK" in r.stdout, r.stdout + r.stderr
