@@ -110,6 +110,25 @@ class KvMap {
         }
     }
 
+    // Bounded scan from a persistent cursor (clock-hand eviction): visits up
+    // to max_slots slots, calling fn for full ones; fn returning false stops
+    // early. The cursor wraps and is updated for the next call.
+    template <typename Fn>  // fn(string_view, Ref<BlockEntry>&) -> bool
+    void scan_from(size_t* cursor, size_t max_slots, Fn&& fn) {
+        size_t n = slots_.size();
+        size_t step = 0;
+        for (; step < max_slots && step < n; step++) {
+            Slot& s = slots_[(*cursor + step) & (n - 1)];
+            if (s.state == kFull && !fn(key_of(s), s.val)) {
+                step++;
+                break;
+            }
+        }
+        *cursor = (*cursor + step) & (n - 1);
+    }
+
+    size_t capacity() const { return slots_.size(); }
+
     void clear() {
         for (auto& s : slots_) {
             if (s.state == kFull) s.val.~Ref<BlockEntry>();

@@ -580,24 +580,37 @@ void Server::erase_entries(const std::vector<Ref<BlockEntry>>& entries) {
 }
 
 size_t Server::evict_lru_locked(Shard* shard, size_t bytes) {
+    // Sampled clock-hand eviction: scan bounded slot windows from a
+    // persistent cursor, evict the least-recently-accessed half of each
+    // sample — O(evicted) amortized instead of a full index scan per
+    // eviction event (which capped sustained full-pool churn).
     // Candidates: committed, idle (only the map holds a ref), on this shard.
     // Key views stay valid across erase (the arena is append-only).
-    std::vector<std::pair<uint64_t, std::string_view>> cands;
-    kv_.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
-        BlockEntry* e = val.get();
-        if (e->shard == shard && e->committed && e->ref_count() == 1)
-            cands.push_back({e->last_access, key});
-    });
-    std::sort(cands.begin(), cands.end(),
-              [](const auto& a, const auto& b) { return a.first < b.first; });
     size_t freed = 0;
-    for (auto& [tick_v, key] : cands) {
-        if (freed >= bytes) break;
-        Ref<BlockEntry>* v = kv_.find(key);
-        if (!v) continue;
-        freed += (*v)->size;
-        kv_.erase(key);
-        n_evicted_.fetch_add(1);
+    size_t scanned = 0;
+    const size_t scan_cap = kv_.capacity();  // at most one full revolution
+    while (freed < bytes && scanned < scan_cap) {
+        std::vector<std::pair<uint64_t, std::string_view>> sample;
+        sample.reserve(128);
+        size_t window = 4096;
+        kv_.scan_from(&evict_hand_, window, [&](std::string_view key, Ref<BlockEntry>& val) {
+            BlockEntry* e = val.get();
+            if (e->shard == shard && e->committed && e->ref_count() == 1)
+                sample.push_back({e->last_access, key});
+            return sample.size() < 128;
+        });
+        scanned += window;
+        if (sample.empty()) continue;
+        std::sort(sample.begin(), sample.end(),
+                  [](const auto& a, const auto& b) { return a.first < b.first; });
+        size_t take = std::max<size_t>(1, sample.size() / 2);
+        for (size_t i = 0; i < take && freed < bytes; i++) {
+            Ref<BlockEntry>* v = kv_.find(sample[i].second);
+            if (!v) continue;
+            freed += (*v)->size;
+            kv_.erase(sample[i].second);
+            n_evicted_.fetch_add(1);
+        }
     }
     if (freed) DEBUG("auto-evicted %zu bytes from shard dev=%d", freed, shard->device());
     return freed;

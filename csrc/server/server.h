@@ -199,6 +199,7 @@ class Server {
     // Evict >= `bytes` of LRU committed idle entries on `shard`.
     // Caller must hold kv_mu_. Returns bytes freed.
     size_t evict_lru_locked(Shard* shard, size_t bytes);
+    size_t evict_hand_ = 0;  // clock cursor (kv_mu_)
     void erase_entries(const std::vector<Ref<BlockEntry>>& entries);
     uint64_t tick() { return access_tick_.fetch_add(1, std::memory_order_relaxed); }
     std::atomic<uint64_t> access_tick_{1};
