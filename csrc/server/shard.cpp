@@ -192,6 +192,36 @@ bool Shard::submit_copy(CopyJob&& job) {
         return true;
     }
 
+    // Fan very large jobs out across the stream pool: one 1 GB request as a
+    // single kernel leaves other streams idle and interacts badly with
+    // concurrent large requests; ~4096-block (512 MB at 128 KB) sub-jobs
+    // spread the work and complete independently.
+    constexpr size_t kFanChunk = 4096;
+    if (on_gpu() && n > kFanChunk && streams_.size() > 1) {
+        size_t n_sub = (n + kFanChunk - 1) / kFanChunk;
+        auto pending = std::make_shared<std::atomic<int>>(static_cast<int>(n_sub));
+        auto all_ok = std::make_shared<std::atomic<bool>>(true);
+        auto done = std::make_shared<std::function<void(bool)>>(std::move(job.done));
+        for (size_t off = 0; off < n; off += kFanChunk) {
+            size_t take = std::min(kFanChunk, n - off);
+            CopyJob sub;
+            sub.bytes_per_block = job.bytes_per_block;
+            sub.src.assign(job.src.begin() + static_cast<long>(off),
+                           job.src.begin() + static_cast<long>(off + take));
+            sub.dst.assign(job.dst.begin() + static_cast<long>(off),
+                           job.dst.begin() + static_cast<long>(off + take));
+            sub.done = [pending, all_ok, done](bool ok) {
+                if (!ok) all_ok->store(false);
+                if (pending->fetch_sub(1) == 1 && *done) (*done)(all_ok->load());
+            };
+            if (!submit_copy(std::move(sub))) {
+                all_ok->store(false);
+                if (pending->fetch_sub(1) == 1 && *done) (*done)(all_ok->load());
+            }
+        }
+        return true;
+    }
+
     if (!on_gpu()) {
         // CPU shard: copies run inline on the caller (loop) thread.
         for (size_t i = 0; i < n; i++)
