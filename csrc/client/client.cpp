@@ -478,6 +478,17 @@ int ClientConn::delete_keys(const std::vector<std::string>& keys) {
     return n;
 }
 
+std::string ClientConn::get_stats() {
+    if (!connected_) return "{}";
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!send_req(OP_STATS, nullptr, 0)) return "{}";
+    int code = 0;
+    if (!recv_status(&code) || code != FINISH) return "{}";
+    std::vector<uint8_t> payload;
+    if (!recv_payload(&payload)) return "{}";
+    return std::string(payload.begin(), payload.end());
+}
+
 int ClientConn::get_match_last_index(const std::vector<std::string>& keys) {
     if (!connected_) return -1;
     auto body = build_match_request(keys);

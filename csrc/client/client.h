@@ -89,6 +89,8 @@ class ClientConn {
     // Delete keys; returns the number removed (extension: engine-driven
     // eviction — the reference only offers wholesale purge).
     int delete_keys(const std::vector<std::string>& keys);
+    // Server stats JSON over the wire (extension).
+    std::string get_stats();
 
     bool rdma_connected() const { return rdma_connected_; }
 

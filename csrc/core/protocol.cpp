@@ -16,6 +16,7 @@ std::string op_name(char op) {
         case OP_TCP_PUT: return "tcp_put";
         case OP_TCP_GET: return "tcp_get";
         case OP_DELETE: return "delete";
+        case OP_STATS: return "stats";
         case OP_W_FAST: return "local_write_fast";
         case OP_R_FAST: return "local_read_fast";
         default: return "unknown";

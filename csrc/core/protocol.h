@@ -49,6 +49,7 @@ constexpr char OP_GET_MATCH_LAST_IDX = 'M';
 constexpr char OP_TCP_PUT = 'P';  // inline block data put (emulated RDMA_WRITE)
 constexpr char OP_TCP_GET = 'G';  // inline block data get (emulated server push)
 constexpr char OP_DELETE = 'X';   // delete keys (extension: engine-driven eviction)
+constexpr char OP_STATS = 'Q';    // server stats JSON over the wire (extension)
 // Packed fast-path local ops (extension): same semantics as OP_W/OP_R but a
 // flat binary layout the hot path can build/parse at memcpy speed — the
 // flatbuffers LocalMetaRequest ops remain accepted for wire compatibility.

@@ -272,7 +272,8 @@ PYBIND11_MODULE(_native, m) {
         .def("check_exist", &ClientConn::check_exist, py::call_guard<py::gil_scoped_release>())
         .def("get_match_last_index", &ClientConn::get_match_last_index,
              py::call_guard<py::gil_scoped_release>())
-        .def("delete_keys", &ClientConn::delete_keys, py::call_guard<py::gil_scoped_release>());
+        .def("delete_keys", &ClientConn::delete_keys, py::call_guard<py::gil_scoped_release>())
+        .def("get_stats", &ClientConn::get_stats, py::call_guard<py::gil_scoped_release>());
 
     // ---- server ----
     m.def("start_server", &start_server, py::call_guard<py::gil_scoped_release>());

@@ -538,6 +538,11 @@ void Server::handle_request(Conn* c, char op, std::vector<uint8_t> body) {
             return op_match_index(c, body);
         case OP_DELETE:
             return op_delete(c, body);
+        case OP_STATS: {
+            std::string js = stats_json();
+            return send_status_payload(c, FINISH,
+                                       reinterpret_cast<const uint8_t*>(js.data()), js.size());
+        }
         default:
             WARN("unknown op '%c'", op);
             return send_status(c, INVALID_REQ);

@@ -231,6 +231,16 @@ def main():
         print(f"saturation: {args.clients} clients, aggregate {agg:.2f} MB/s "
               f"(per-client write {statistics.mean(r[0] for r in results):.2f} MB/s, "
               f"read {statistics.mean(r[1] for r in results):.2f} MB/s)")
+        # Server-side truth (the client-side walls include Python/CPU
+        # contention in the worker processes):
+        try:
+            probe = make_conn(args, False)
+            stats = json.loads(probe.conn.get_stats())
+            print(f"server counters: bytes_in={stats['bytes_in']/1e9:.1f} GB "
+                  f"bytes_out={stats['bytes_out']/1e9:.1f} GB op_us={stats.get('op_us')}")
+            probe.close()
+        except Exception:
+            pass
         if server_proc is not None:
             server_proc.terminate()
             server_proc.wait(timeout=20)

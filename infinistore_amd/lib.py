@@ -519,6 +519,11 @@ class InfinityConnection:
             raise Exception("can't find a match")
         return ret
 
+    def get_server_stats_remote(self) -> str:
+        """Server stats JSON fetched over the wire (extension) — works from
+        any client, no management port needed."""
+        return self.conn.get_stats()
+
     def delete_keys(self, keys: List[str]) -> int:
         """Delete keys from the store; returns the number removed.
         Extension over the reference (which only offers wholesale purge) —
