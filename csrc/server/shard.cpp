@@ -238,9 +238,12 @@ bool Shard::submit_copy(CopyJob&& job) {
             aligned = (job.src[i] % 16 == 0) && (job.dst[i] % 16 == 0);
     }
 
-    std::lock_guard<std::mutex> submit_lk(submit_mu_);
-    StreamCtx& sc = streams_[next_stream_];
-    next_stream_ = (next_stream_ + 1) % static_cast<int>(streams_.size());
+    // HIP stream enqueues are thread-safe; each job's ops precede its event
+    // in stream order regardless of interleaving with other submitters, so
+    // no submit-wide lock is needed (a lock here convoyed all clients behind
+    // slot waits).
+    StreamCtx& sc =
+        streams_[next_stream_.fetch_add(1) % static_cast<uint32_t>(streams_.size())];
 
     // Small aligned batches: descriptors ride in the kernel arguments.
     if (aligned && n <= 16) {

@@ -29,7 +29,7 @@ struct ShardOptions {
     size_t pool_bytes = 0;    // initial arena size
     size_t block_granule = 64 << 10;  // bitmap granule (minimal_allocate_size)
     int n_streams = 4;
-    int slots_per_stream = 4;
+    int slots_per_stream = 16;
     size_t max_descs_per_slot = 65536;  // 64K blocks -> 1 MiB of u64 ptrs/side
     bool auto_extend = false;
     size_t extend_bytes = 10ull << 30;
@@ -131,8 +131,7 @@ class Shard {
     std::mutex alloc_mu_;
 
     std::vector<StreamCtx> streams_;
-    int next_stream_ = 0;
-    std::mutex submit_mu_;
+    std::atomic<uint32_t> next_stream_{0};
 
     std::deque<PendingTask> tasks_;
     std::mutex task_mu_;
