@@ -152,11 +152,15 @@ def _worker(args_dict, q, barrier, wid):
     device = f"cuda:{(ns.src_gpu + wid) % n_dev}" if local else "cpu"
     conn = make_conn(ns, local)
     try:
+        run_once(ns, conn, local, device)  # warm: IPC opens, allocator, caches
         barrier.wait(timeout=300)  # start all clients together (steady state)
         t0 = time.perf_counter()
-        w, r = run_once(ns, conn, local, device)
+        w = r = 0.0
+        iters = max(1, ns.iteration)
+        for _ in range(iters):
+            w, r = run_once(ns, conn, local, device)
         wall = time.perf_counter() - t0
-        q.put((w, r, wall))
+        q.put((w, r, wall, iters))
     finally:
         conn.close()
 
@@ -227,7 +231,8 @@ def main():
         for p in procs:
             p.join()
         wall = max(r[2] for r in results)  # steady-state window (post-barrier)
-        agg = args.clients * (args.size << 20) * 2 / wall / 1e6
+        iters = results[0][3]
+        agg = args.clients * (args.size << 20) * 2 * iters / wall / 1e6
         print(f"saturation: {args.clients} clients, aggregate {agg:.2f} MB/s "
               f"(per-client write {statistics.mean(r[0] for r in results):.2f} MB/s, "
               f"read {statistics.mean(r[1] for r in results):.2f} MB/s)")
