@@ -257,7 +257,7 @@ bool Shard::submit_copy(CopyJob&& job) {
         {
             std::lock_guard<std::mutex> lk(task_mu_);
             if (!ok) slot->busy = false;
-            tasks_.push_back({&sc, ok ? slot : nullptr, std::move(job.done)});
+            sc.pending.push_back({ok ? slot : nullptr, std::move(job.done)});
         }
         task_cv_.notify_one();
         return true;
@@ -290,15 +290,15 @@ bool Shard::submit_copy(CopyJob&& job) {
             {
                 std::lock_guard<std::mutex> lk(task_mu_);
                 slot->busy = false;
-                tasks_.push_back({&sc, nullptr, std::move(job.done)});
+                sc.pending.push_back({nullptr, std::move(job.done)});
             }
             task_cv_.notify_one();
             return true;
         }
         {
             std::lock_guard<std::mutex> lk(task_mu_);
-            tasks_.push_back(
-                {&sc, slot, last ? std::move(job.done) : std::function<void(bool)>()});
+            sc.pending.push_back(
+                {slot, last ? std::move(job.done) : std::function<void(bool)>()});
         }
         task_cv_.notify_one();
         off += take;
@@ -308,30 +308,54 @@ bool Shard::submit_copy(CopyJob&& job) {
 
 void Shard::completion_loop() {
     gpu::set_device(opt_.device);
+    // Per-stream FIFO queues, completed out of order ACROSS streams: one
+    // long kernel on stream A must not delay the response of a finished
+    // copy on stream B (cross-client latency coupling).
     for (;;) {
-        PendingTask t;
+        PendingTask t{};
+        bool have = false;
         {
             std::unique_lock<std::mutex> lk(task_mu_);
-            task_cv_.wait(lk, [this] { return stopping_ || !tasks_.empty(); });
-            if (stopping_ && tasks_.empty()) return;
-            t = std::move(tasks_.front());
-            tasks_.pop_front();
+            for (int pass = 0; !have; pass++) {
+                bool any_pending = false;
+                for (auto& sc : streams_) {
+                    if (sc.pending.empty()) continue;
+                    any_pending = true;
+                    PendingTask& front = sc.pending.front();
+                    if (!front.slot || gpu::event_query(front.slot->event)) {
+                        t = std::move(front);
+                        sc.pending.pop_front();
+                        have = true;
+                        break;
+                    }
+                }
+                if (have) break;
+                if (stopping_ && !any_pending) return;
+                if (!any_pending) {
+                    task_cv_.wait(lk, [&] {
+                        if (stopping_) return true;
+                        for (auto& sc : streams_)
+                            if (!sc.pending.empty()) return true;
+                        return false;
+                    });
+                    if (stopping_) {
+                        bool empty = true;
+                        for (auto& sc : streams_) empty &= sc.pending.empty();
+                        if (empty) return;
+                    }
+                } else {
+                    // Events pending but none complete yet: spin politely.
+                    lk.unlock();
+#if defined(__x86_64__)
+                    for (int i = 0; i < 64; i++) __builtin_ia32_pause();
+#endif
+                    lk.lock();
+                }
+            }
         }
         bool ok = t.slot != nullptr;
         if (t.slot) {
-            // Spin briefly before blocking: hipEventSynchronize's interrupt
-            // wake costs ~20 us; most copies finish within the spin window.
-            bool done = false;
-            for (int spin = 0; spin < 4000; spin++) {
-                if (gpu::event_query(t.slot->event)) {
-                    done = true;
-                    break;
-                }
-#if defined(__x86_64__)
-                __builtin_ia32_pause();
-#endif
-            }
-            ok = done || gpu::event_sync(t.slot->event);
+            ok = gpu::event_query(t.slot->event) || gpu::event_sync(t.slot->event);
             std::lock_guard<std::mutex> lk(task_mu_);
             t.slot->busy = false;
         }

@@ -111,15 +111,15 @@ class Shard {
         gpu::Event event = nullptr;
         bool busy = false;
     };
+    struct PendingTask {
+        Slot* slot;
+        std::function<void(bool)> done;  // may be empty for chunked sub-jobs
+    };
     struct StreamCtx {
         gpu::Stream stream = nullptr;
         std::vector<Slot> slots;
+        std::deque<PendingTask> pending;  // FIFO per stream (task_mu_)
         int next_slot = 0;
-    };
-    struct PendingTask {
-        StreamCtx* sc;
-        Slot* slot;
-        std::function<void(bool)> done;  // may be empty for chunked sub-jobs
     };
 
     Slot* acquire_slot(StreamCtx& sc);
@@ -133,7 +133,6 @@ class Shard {
     std::vector<StreamCtx> streams_;
     std::atomic<uint32_t> next_stream_{0};
 
-    std::deque<PendingTask> tasks_;
     std::mutex task_mu_;
     std::condition_variable task_cv_;
     std::condition_variable slot_cv_;
