@@ -643,6 +643,14 @@ void* resolve_client_base(Server::Conn* c, const Server::LocalView& msg) {
     std::vector<uint8_t> key(msg.ipc, msg.ipc + msg.ipc_len);
     auto it = c->ipc_cache.find(key);
     if (it != c->ipc_cache.end()) return it->second.first;
+    // Bound the cache: a client that churns tensors would otherwise pin
+    // every old allocation via its stale mapping. Only flush when nothing
+    // is in flight (a mapping may be read by a queued kernel).
+    if (c->ipc_cache.size() >= 64 && c->remain.load() == 0 &&
+        c->fabric_inflight.load() == 0) {
+        for (auto& kv2 : c->ipc_cache) gpu::ipc_close(kv2.second.first);
+        c->ipc_cache.clear();
+    }
     gpu::IpcHandle h;
     memcpy(h.bytes, msg.ipc, gpu::kIpcHandleSize);
     void* base = gpu::ipc_open(h, msg.device);
