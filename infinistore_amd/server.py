@@ -164,6 +164,10 @@ def main():
     if args.prevent_oom:
         prevent_oom()
 
+    if torch.cuda.is_available() and torch.cuda.device_count() > 1:
+        from .warmup import check_p2p_access
+
+        check_p2p_access()
     register_server(config)
     print(f"infinistore-amd serving on :{config.service_port} "
           f"(manage :{config.manage_port})", flush=True)

@@ -9,6 +9,25 @@ import torch
 from . import lib
 
 
+def check_p2p_access() -> bool:
+    """Report peer-access availability between every GPU pair (role of the
+    reference's server.py:99-109 helper). Returns True when every pair can
+    peer (xGMI on an MI355X node)."""
+    n = torch.cuda.device_count()
+    ok = True
+    for i in range(n):
+        for j in range(n):
+            if i == j:
+                continue
+            can = torch.cuda.can_device_access_peer(i, j)
+            if not can:
+                lib.Logger.warn(f"no peer access {i} -> {j}")
+                ok = False
+    if ok and n > 1:
+        lib.Logger.info(f"p2p access OK across {n} GPUs")
+    return ok
+
+
 def warmup(service_port: int) -> bool:
     if not torch.cuda.is_available():
         lib.Logger.warn("warmup skipped: no GPU")
