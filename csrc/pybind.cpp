@@ -53,6 +53,7 @@ struct ServerConfigPy {
     int cpu_shards = 1;               // CPU-mode shard count
     bool auto_evict = false;          // LRU-evict on allocation failure
     int io_threads = 3;               // worker IO loops (0 = single loop)
+    int extend_size = 10;             // GB per auto-extend arena
 };
 
 bool start_server(const ServerConfigPy& cfg) {
@@ -73,6 +74,7 @@ bool start_server(const ServerConfigPy& cfg) {
     opt.link_type = cfg.link_type;
     opt.auto_evict = cfg.auto_evict;
     opt.io_threads = cfg.io_threads;
+    opt.extend_bytes = static_cast<size_t>(cfg.extend_size) << 30;
     if (!cfg.cpu_only && gpu::available()) {
         if (!cfg.devices.empty()) {
             opt.devices = cfg.devices;
@@ -155,7 +157,8 @@ PYBIND11_MODULE(_native, m) {
         .def_readwrite("cpu_only", &ServerConfigPy::cpu_only)
         .def_readwrite("cpu_shards", &ServerConfigPy::cpu_shards)
         .def_readwrite("auto_evict", &ServerConfigPy::auto_evict)
-        .def_readwrite("io_threads", &ServerConfigPy::io_threads);
+        .def_readwrite("io_threads", &ServerConfigPy::io_threads)
+        .def_readwrite("extend_size", &ServerConfigPy::extend_size);
 
     // ---- client connection ----
     py::class_<ClientConn>(m, "Connection")

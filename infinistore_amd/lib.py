@@ -108,6 +108,7 @@ class ServerConfig(_native.ServerConfig):
         self.cpu_shards = kwargs.get("cpu_shards", 1)
         self.auto_evict = kwargs.get("auto_evict", False)
         self.io_threads = kwargs.get("io_threads", 3)
+        self.extend_size = kwargs.get("extend_size", 10)
 
     def __repr__(self):
         return (

@@ -401,3 +401,77 @@ def test_auto_evict_lru(ports):
         conn.close()
     finally:
         ifs.unregister_server()
+
+
+def test_auto_extend_grows_pool(ports):
+    """auto_increase: the pool extends in the background when the last arena
+    crosses the usage threshold (role of the reference's add_mempool flow)."""
+    import json
+    import time
+
+    service_port, manage_port = ports
+    cfg = ifs.ServerConfig(
+        service_port=service_port,
+        manage_port=manage_port,
+        prealloc_size=1,           # 1 GB initial
+        extend_size=1,             # +1 GB per extension
+        minimal_allocate_size=1024,
+        cpu_only=True,
+        auto_increase=True,
+    )
+    ifs.register_server(cfg)
+    try:
+        conn = make_client(service_port)
+        page = 1 << 20
+        src = torch.zeros(page // 4)
+        conn.register_mr(src)
+        total0 = json.loads(ifs.get_server_stats())["total_blocks"]
+        # push usage past 80% to trigger background extension
+        keys = [f"x-{i}" for i in range(850)]
+        blocks = conn.allocate_rdma(keys, page)
+        assert len(blocks) == 850
+        deadline = time.time() + 20
+        while time.time() < deadline:
+            if json.loads(ifs.get_server_stats())["total_blocks"] > total0:
+                break
+            time.sleep(0.2)
+        assert json.loads(ifs.get_server_stats())["total_blocks"] > total0
+        # and the extra capacity is usable
+        more = conn.allocate_rdma([f"y-{i}" for i in range(400)], page)
+        assert len(more) == 400
+        conn.close()
+    finally:
+        ifs.unregister_server()
+
+
+def test_fabric_via_container_ip(cpu_server):
+    """Connect via the container's non-loopback address (the cross-host
+    code path, minus the physical network)."""
+    import socket as _socket
+
+    try:
+        ip = _socket.gethostbyname(_socket.gethostname())
+    except OSError:
+        pytest.skip("no resolvable host address")
+    if ip.startswith("127."):
+        pytest.skip("hostname resolves to loopback")
+    cfg = ifs.ClientConfig(
+        host_addr=ip, service_port=cpu_server,
+        connection_type=ifs.TYPE_RDMA, link_type="TCP",
+    )
+    conn = ifs.InfinityConnection(cfg)
+    conn.connect()
+    try:
+        src = torch.arange(1024, dtype=torch.float32)
+        dst = torch.zeros_like(src)
+        conn.register_mr(src)
+        conn.register_mr(dst)
+        key = f"ip-{uuid.uuid4()}"
+        blocks = conn.allocate_rdma([key], 4096)
+        conn.rdma_write_cache(src, [0], 1024, blocks)
+        conn.sync()
+        conn.read_cache(dst, [(key, 0)], 1024)
+        conn.sync()
+        assert torch.equal(src, dst)
+    finally:
+        conn.close()
