@@ -5,6 +5,7 @@
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/socket.h>
+#include <sys/un.h>
 #include <unistd.h>
 
 #include <algorithm>
@@ -41,6 +42,28 @@ void ClientConn::close_conn() {
 int ClientConn::init_connection(const ClientConfigC& cfg) {
     set_log_level(cfg.log_level.c_str());
     if (connected_) return -1;
+    // Same-host fast path: the server exposes a Unix-domain socket next to
+    // its TCP port (lower latency than TCP loopback; same wire protocol).
+    if (cfg.host_addr == "127.0.0.1" || cfg.host_addr == "localhost") {
+        std::string path = "/tmp/infinistore-amd-" + std::to_string(cfg.service_port) + ".sock";
+        struct sockaddr_un ua{};
+        if (path.size() < sizeof(ua.sun_path)) {
+            int ufd = ::socket(AF_UNIX, SOCK_STREAM, 0);
+            if (ufd >= 0) {
+                ua.sun_family = AF_UNIX;
+                strncpy(ua.sun_path, path.c_str(), sizeof(ua.sun_path) - 1);
+                if (::connect(ufd, reinterpret_cast<struct sockaddr*>(&ua), sizeof(ua)) == 0) {
+                    struct timeval tv{60, 0};
+                    setsockopt(ufd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+                    fd_ = ufd;
+                    connected_ = true;
+                    DEBUG("connected via UDS %s", path.c_str());
+                    return 0;
+                }
+                ::close(ufd);
+            }
+        }
+    }
     struct addrinfo hints{}, *res = nullptr;
     hints.ai_family = AF_INET;
     hints.ai_socktype = SOCK_STREAM;

@@ -6,6 +6,7 @@
 #include <chrono>
 #include <cstring>
 #include <future>
+#include <unistd.h>
 
 #include "../core/log.h"
 
@@ -94,7 +95,7 @@ void send_buf(Server::Conn* c, std::vector<uint8_t> data);
 void conn_close(Server::Conn* c) {
     if (c->closed) return;
     c->closed = true;
-    uv_close(reinterpret_cast<uv_handle_t*>(&c->tcp), [](uv_handle_t* h) {
+    uv_close(c->handle(), [](uv_handle_t* h) {
         auto* c = static_cast<Server::Conn*>(h->data);
         c->unref();
     });
@@ -122,7 +123,7 @@ void send_buf(Server::Conn* c, std::vector<uint8_t> data) {
     wr->req.data = wr;
     uv_buf_t b = uv_buf_init(reinterpret_cast<char*>(wr->data.data()),
                              static_cast<unsigned>(wr->data.size()));
-    int r = uv_write(&wr->req, reinterpret_cast<uv_stream_t*>(&c->tcp), &b, 1,
+    int r = uv_write(&wr->req, c->stream(), &b, 1,
                      [](uv_write_t* req, int status) {
                          auto* wr = static_cast<WriteReq*>(req->data);
                          delete wr;
@@ -214,6 +215,21 @@ bool Server::start() {
         uv_loop_close(&main_io_.loop);
         return false;
     }
+    // Same-host fast transport: a Unix-domain listener next to the TCP one.
+    pipe_path_ = "/tmp/infinistore-amd-" + std::to_string(opt_.service_port) + ".sock";
+    unlink(pipe_path_.c_str());
+    uv_pipe_init(&main_io_.loop, &pipe_listener_, 0);
+    pipe_listener_.data = this;
+    if (uv_pipe_bind(&pipe_listener_, pipe_path_.c_str()) == 0 &&
+        uv_listen(reinterpret_cast<uv_stream_t*>(&pipe_listener_), 512,
+                  &Server::on_new_pipe_connection) == 0) {
+        pipe_listening_ = true;
+    } else {
+        WARN("UDS listener at %s unavailable; same-host clients use TCP",
+             pipe_path_.c_str());
+        uv_close(reinterpret_cast<uv_handle_t*>(&pipe_listener_), nullptr);
+    }
+
     for (int i = 0; i < std::max(0, opt_.io_threads); i++) {
         auto w = std::make_unique<IoLoop>();
         w->srv = this;
@@ -269,7 +285,14 @@ void Server::IoLoop::on_stop(uv_async_t* h) {
         conn_close(c);
     }
     io->conns.clear();
-    if (io->is_main) uv_close(reinterpret_cast<uv_handle_t*>(&io->srv->listener_), nullptr);
+    if (io->is_main) {
+        uv_close(reinterpret_cast<uv_handle_t*>(&io->srv->listener_), nullptr);
+        if (io->srv->pipe_listening_) {
+            uv_close(reinterpret_cast<uv_handle_t*>(&io->srv->pipe_listener_), nullptr);
+            unlink(io->srv->pipe_path_.c_str());
+            io->srv->pipe_listening_ = false;
+        }
+    }
     uv_close(reinterpret_cast<uv_handle_t*>(&io->post_async), nullptr);
     uv_close(reinterpret_cast<uv_handle_t*>(&io->stop_async), nullptr);
 }
@@ -331,12 +354,42 @@ void Server::on_new_connection(uv_stream_t* server, int status) {
         io = srv->workers_[i].get();
     }
     if (io == &srv->main_io_)
-        srv->adopt_fd(io, fd2);
+        srv->adopt_fd(io, fd2, false);
     else
-        io->post([srv, io, fd2] { srv->adopt_fd(io, fd2); });
+        io->post([srv, io, fd2] { srv->adopt_fd(io, fd2, false); });
 }
 
-void Server::adopt_fd(IoLoop* io, int fd) {
+void Server::on_new_pipe_connection(uv_stream_t* server, int status) {
+    auto* srv = static_cast<Server*>(server->data);
+    if (status < 0) return;
+    uv_pipe_t* tmp = new uv_pipe_t();
+    uv_pipe_init(&srv->main_io_.loop, tmp, 0);
+    tmp->data = nullptr;
+    auto close_tmp = [](uv_handle_t* h) { delete reinterpret_cast<uv_pipe_t*>(h); };
+    if (uv_accept(server, reinterpret_cast<uv_stream_t*>(tmp)) != 0) {
+        uv_close(reinterpret_cast<uv_handle_t*>(tmp), close_tmp);
+        return;
+    }
+    uv_os_fd_t fd;
+    if (uv_fileno(reinterpret_cast<uv_handle_t*>(tmp), &fd) != 0) {
+        uv_close(reinterpret_cast<uv_handle_t*>(tmp), close_tmp);
+        return;
+    }
+    int fd2 = dup(fd);
+    uv_close(reinterpret_cast<uv_handle_t*>(tmp), close_tmp);
+    IoLoop* io = &srv->main_io_;
+    if (!srv->workers_.empty()) {
+        uint32_t i = srv->next_worker_.fetch_add(1) %
+                     static_cast<uint32_t>(srv->workers_.size());
+        io = srv->workers_[i].get();
+    }
+    if (io == &srv->main_io_)
+        srv->adopt_fd(io, fd2, true);
+    else
+        io->post([srv, io, fd2] { srv->adopt_fd(io, fd2, true); });
+}
+
+void Server::adopt_fd(IoLoop* io, int fd, bool is_pipe) {
     if (stop_requested_.load()) {
         ::close(fd);
         return;
@@ -344,16 +397,26 @@ void Server::adopt_fd(IoLoop* io, int fd) {
     auto* c = new Conn();
     c->srv = this;
     c->owner = io;
-    uv_tcp_init(&io->loop, &c->tcp);
-    c->tcp.data = c;
-    if (uv_tcp_open(&c->tcp, fd) != 0) {
-        conn_close(c);
-        return;
+    c->is_pipe = is_pipe;
+    if (is_pipe) {
+        uv_pipe_init(&io->loop, &c->pipe, 0);
+        c->pipe.data = c;
+        if (uv_pipe_open(&c->pipe, fd) != 0) {
+            conn_close(c);
+            return;
+        }
+    } else {
+        uv_tcp_init(&io->loop, &c->tcp);
+        c->tcp.data = c;
+        if (uv_tcp_open(&c->tcp, fd) != 0) {
+            conn_close(c);
+            return;
+        }
+        uv_tcp_nodelay(&c->tcp, 1);
     }
-    uv_tcp_nodelay(&c->tcp, 1);
     io->conns.push_back(c);
     uv_read_start(
-        reinterpret_cast<uv_stream_t*>(&c->tcp),
+        c->stream(),
         [](uv_handle_t*, size_t suggested, uv_buf_t* buf) {
             buf->base = static_cast<char*>(malloc(suggested));
             buf->len = suggested;

@@ -175,7 +175,8 @@ class Server {
     friend struct VerbsPeer;
 
     static void on_new_connection(uv_stream_t* server, int status);
-    void adopt_fd(IoLoop* io, int fd);  // runs on io's thread
+    void adopt_fd(IoLoop* io, int fd, bool is_pipe);  // runs on io's thread
+    static void on_new_pipe_connection(uv_stream_t* server, int status);
     void post(std::function<void()> fn) { main_io_.post(std::move(fn)); }
 
     // ---- request handling (loop thread) ----
@@ -213,6 +214,9 @@ class Server {
     std::vector<std::unique_ptr<IoLoop>> workers_;
     std::atomic<uint32_t> next_worker_{0};
     uv_tcp_t listener_;
+    uv_pipe_t pipe_listener_;
+    bool pipe_listening_ = false;
+    std::string pipe_path_;
     std::atomic<bool> running_{false};
     std::atomic<bool> stop_requested_{false};
 
@@ -242,7 +246,16 @@ class Server {
 struct Server::Conn : RefCounted {
     Server* srv = nullptr;
     Server::IoLoop* owner = nullptr;  // the loop thread serving this conn
-    uv_tcp_t tcp;
+    // Transport handle: TCP socket or Unix-domain pipe (same-host clients
+    // connect via the UDS listener — lower latency than TCP loopback).
+    union {
+        uv_tcp_t tcp;
+        uv_pipe_t pipe;
+    };
+    bool is_pipe = false;
+    uv_stream_t* stream() { return reinterpret_cast<uv_stream_t*>(&tcp); }
+    uv_handle_t* handle() { return reinterpret_cast<uv_handle_t*>(&tcp); }
+    Conn() : tcp{} {}
     bool closed = false;
 
     // read state machine

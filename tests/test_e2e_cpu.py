@@ -475,3 +475,18 @@ def test_fabric_via_container_ip(cpu_server):
         assert torch.equal(src, dst)
     finally:
         conn.close()
+
+
+def test_tcp_transport_explicit(cpu_server):
+    """Force the TCP transport (127.0.0.2 avoids the UDS fast path) so both
+    transports stay covered."""
+    cfg = ifs.ClientConfig(
+        host_addr="127.0.0.2", service_port=cpu_server,
+        connection_type=ifs.TYPE_RDMA, link_type="TCP",
+    )
+    conn = ifs.InfinityConnection(cfg)
+    conn.connect()
+    try:
+        put_get_roundtrip(conn, 2048, 512)
+    finally:
+        conn.close()
