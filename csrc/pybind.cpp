@@ -196,6 +196,38 @@ PYBIND11_MODULE(_native, m) {
             py::arg("op"), py::arg("keys_blob"), py::arg("offsets"), py::arg("n"),
             py::arg("block_size"), py::arg("ptr"), py::arg("device"),
             py::arg("sync_response") = false)
+        .def(
+            "rw_local_keys",
+            [](ClientConn& c, const std::string& op, py::list keys, py::buffer offsets,
+               uint64_t element_size, int block_size, uintptr_t ptr, int device,
+               bool sync_response) {
+                // Build the NUL-joined key blob and byte offsets in C++ —
+                // no Python-side join/numpy work on the hot path.
+                py::buffer_info ob = offsets.request();
+                size_t n = static_cast<size_t>(py::len(keys));
+                if (ob.itemsize != 8 || static_cast<size_t>(ob.size) < n)
+                    throw std::runtime_error("offsets must be uint64[n]");
+                const uint64_t* offs = static_cast<const uint64_t*>(ob.ptr);
+                std::string blob;
+                blob.reserve(n * 48);
+                std::vector<uint64_t> byte_offs(n);
+                for (size_t i = 0; i < n; i++) {
+                    Py_ssize_t klen = 0;
+                    const char* ks =
+                        PyUnicode_AsUTF8AndSize(keys[i].ptr(), &klen);
+                    if (!ks) throw std::runtime_error("keys must be str");
+                    if (i) blob.push_back('\0');
+                    blob.append(ks, static_cast<size_t>(klen));
+                    byte_offs[i] = offs[i] * element_size;
+                }
+                py::gil_scoped_release rel;
+                return c.rw_local_packed(op.empty() ? 'W' : op[0], blob.data(), blob.size(),
+                                         byte_offs.data(), n, block_size, ptr, device,
+                                         sync_response);
+            },
+            py::arg("op"), py::arg("keys"), py::arg("offsets"), py::arg("element_size"),
+            py::arg("block_size"), py::arg("ptr"), py::arg("device"),
+            py::arg("sync_response") = false)
         .def("sync_local", &ClientConn::sync_local, py::call_guard<py::gil_scoped_release>())
         .def("register_mr", &ClientConn::register_mr, py::call_guard<py::gil_scoped_release>())
         .def(

@@ -341,10 +341,9 @@ class InfinityConnection:
         self._verify(cache)
         assert self.local_connected, "write_pages uses the local GPU path"
         es = cache.element_size()
-        blob = "\x00".join(keys).encode()
-        offs = np.asarray(offsets, dtype=np.uint64) * np.uint64(es)
-        ret = self.conn.rw_local_fast(
-            self.OP_W, blob, offs.tobytes(), len(keys), page_size * es,
+        offs = np.asarray(offsets, dtype=np.uint64)
+        ret = self.conn.rw_local_keys(
+            self.OP_W, keys, offs, es, page_size * es,
             cache.data_ptr(), _remap_device_id(cache), sync,
         )
         if ret < 0:
@@ -355,10 +354,9 @@ class InfinityConnection:
         self._verify(cache)
         es = cache.element_size()
         if self.local_connected:
-            blob = "\x00".join(keys).encode()
-            offs = np.asarray(offsets, dtype=np.uint64) * np.uint64(es)
-            ret = self.conn.rw_local_fast(
-                self.OP_R, blob, offs.tobytes(), len(keys), page_size * es,
+            offs = np.asarray(offsets, dtype=np.uint64)
+            ret = self.conn.rw_local_keys(
+                self.OP_R, keys, offs, es, page_size * es,
                 cache.data_ptr(), _remap_device_id(cache),
             )
         elif self.rdma_connected:
