@@ -9,9 +9,9 @@ import torch
 import infinistore_amd as ifs
 
 
-async def main():
+async def main(port: int = 22345):
     cfg = ifs.ClientConfig(
-        host_addr="127.0.0.1", service_port=22345,
+        host_addr="127.0.0.1", service_port=port,
         connection_type=ifs.TYPE_RDMA, link_type="TCP",
     )
     conn = ifs.InfinityConnection(cfg)
@@ -38,4 +38,6 @@ async def main():
 
 
 if __name__ == "__main__":
-    asyncio.run(main())
+    import sys
+
+    asyncio.run(main(int(sys.argv[1]) if len(sys.argv) > 1 else 22345))

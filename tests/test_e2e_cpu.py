@@ -490,3 +490,9 @@ def test_tcp_transport_explicit(cpu_server):
         put_get_roundtrip(conn, 2048, 512)
     finally:
         conn.close()
+
+
+def test_example_client_async(cpu_server):
+    from infinistore_amd.example import client_async
+
+    asyncio.run(client_async.main(cpu_server))

@@ -338,3 +338,18 @@ def test_demo_prefill_example(gpu_server):
     from infinistore_amd.example import demo_prefill
 
     demo_prefill.main(gpu_server)
+
+
+def test_gpu_page_hashes_wrapper():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from infinistore_amd.kv_connector import gpu_page_hashes
+
+    t = torch.randn(4 * 4096, device="cuda:0")
+    h1 = gpu_page_hashes(t, [0, 4096, 8192, 12288], 4096)
+    h2 = gpu_page_hashes(t, [0, 4096, 8192, 12288], 4096)
+    assert h1 == h2 and len(set(h1)) == 4  # deterministic chain, distinct
+    t2 = t.clone()
+    t2[5000] += 1.0  # page 1 changes -> pages 1..3 chain digests change
+    h3 = gpu_page_hashes(t2, [0, 4096, 8192, 12288], 4096)
+    assert h3[0] == h1[0] and h3[1] != h1[1] and h3[3] != h1[3]
