@@ -622,10 +622,7 @@ Shard* Server::shard_least_used() {
     Shard* best = shards_[0].get();
     size_t best_used = SIZE_MAX;
     for (auto& s : shards_) {
-        size_t total = s->total_blocks();
         size_t used = s->used_blocks();
-        size_t free_blocks = total - used;
-        (void)free_blocks;
         if (used < best_used) {
             best_used = used;
             best = s.get();

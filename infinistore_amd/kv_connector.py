@@ -108,9 +108,8 @@ class PagedKVConnector:
                 self.conn.read_pages(kv_out, keys, page_offsets, page_elems)
             else:
                 self._ensure_mr(kv_out)
-                es = kv_out.element_size()
-                blocks = [(k, int(o) * es) for k, o in zip(keys, page_offsets)]
-                self.conn.conn.r_rdma(blocks, page_elems * es, kv_out.data_ptr())
+                blocks = [(k, int(o)) for k, o in zip(keys, page_offsets)]
+                self.conn.read_cache(kv_out, blocks, page_elems)
             self.conn.sync()
             return True
         except Exception:
