@@ -149,19 +149,6 @@ void* MemoryPool::allocate_below(size_t size, size_t limit_block) {
     return static_cast<uint8_t*>(base_) + start * block_size_;
 }
 
-size_t MemoryPool::high_water() const {
-    for (size_t w = n_words_; w-- > 0;) {
-        uint64_t bits = bits_[w];
-        if (w == n_words_ - 1) {
-            // Mask out the synthetic tail-used bits of a partial last word.
-            size_t tail = n_words_ * 64 - n_blocks_;
-            if (tail) bits &= ~(~0ull << (64 - tail));
-        }
-        if (bits) return w * 64 + (64 - static_cast<size_t>(__builtin_clzll(bits)));
-    }
-    return 0;
-}
-
 size_t MemoryPool::largest_free_run() const {
     size_t best = 0, run = 0;
     for (size_t i = 0; i < n_blocks_; i++) {

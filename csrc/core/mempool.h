@@ -45,8 +45,6 @@ class MemoryPool {
         return (reinterpret_cast<uintptr_t>(ptr) - reinterpret_cast<uintptr_t>(base_)) /
                block_size_;
     }
-    // Highest used block index + 1 (0 if empty) — the pool's "high-water mark".
-    size_t high_water() const;
     // Largest contiguous free run, in blocks.
     size_t largest_free_run() const;
 
