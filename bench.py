@@ -47,6 +47,8 @@ def parse_args():
     p.add_argument("--cross", action="store_true",
                    help="each rank reads keys written by the next rank "
                         "(forces xGMI cross-shard traffic)")
+    p.add_argument("--no-pipeline", action="store_true",
+                   help="disable put/get step pipelining (sequential loop)")
     return p.parse_args()
 
 
@@ -189,36 +191,64 @@ def main():
                "get_sync": 0.0}
     debug = os.environ.get("IFS_BENCH_DEBUG")
 
+    # Pipelined mode (default on the local path): the put of step s+1 is
+    # issued as a plain async request (server responds before submitting the
+    # copy) right before the blocking get of step s, so the put's gather
+    # kernel + index insert run on their own HIP stream underneath the get's
+    # scatter kernel; the trailing sync only drains whatever is left of the
+    # put commit. Every step still moves the full put+get payload — only the
+    # host-side request latency is hidden. Sequential when --cross (needs a
+    # barrier between a step's put commit and the peer's get) or debugging.
+    pipeline = use_local_path and not cross and not debug and not args.no_pipeline
+
     sync_all()
     t0 = time.perf_counter()
     put_time = 0.0
     get_time = 0.0
-    for s in range(args.steps):
+    if pipeline:
         tp = time.perf_counter()
-        if debug and use_local_path:
-            conn.write_pages(src, put_keys[s], offsets_np, elems_per_block, sync=True)
-            tb = time.perf_counter()
-            conn.sync()
-            tc = time.perf_counter()
-            debug_t["put_req"] += tb - tp
-            debug_t["put_sync"] += tc - tb
-        else:
-            do_put(put_keys[s])
+        conn.write_pages(src, put_keys[0], offsets_np, elems_per_block, sync=True)
         put_time += time.perf_counter() - tp
-        if cross:
-            dist.barrier()  # readers wait for the writer of their keys
-        tg = time.perf_counter()
-        if debug and use_local_path:
+        for s in range(args.steps):
+            tp = time.perf_counter()
+            if s + 1 < args.steps:
+                conn.write_pages(src, put_keys[s + 1], offsets_np,
+                                 elems_per_block, sync=False)
+            tg = time.perf_counter()
             conn.read_pages(dst, get_keys[s], offsets_np, elems_per_block)
-            tb = time.perf_counter()
-            conn.sync()
-            debug_t["get_req"] += tb - tg
-            debug_t["get_sync"] += time.perf_counter() - tb
-        else:
-            do_get(get_keys[s])
-        get_time += time.perf_counter() - tg
+            conn.sync()  # get is sync-response; this drains put(s+1)'s commit
+            put_time += tg - tp
+            get_time += time.perf_counter() - tg
+    else:
+        for s in range(args.steps):
+            tp = time.perf_counter()
+            if debug and use_local_path:
+                conn.write_pages(src, put_keys[s], offsets_np, elems_per_block, sync=True)
+                tb = time.perf_counter()
+                conn.sync()
+                tc = time.perf_counter()
+                debug_t["put_req"] += tb - tp
+                debug_t["put_sync"] += tc - tb
+            else:
+                do_put(put_keys[s])
+            put_time += time.perf_counter() - tp
+            if cross:
+                dist.barrier()  # readers wait for the writer of their keys
+            tg = time.perf_counter()
+            if debug and use_local_path:
+                conn.read_pages(dst, get_keys[s], offsets_np, elems_per_block)
+                tb = time.perf_counter()
+                conn.sync()
+                debug_t["get_req"] += tb - tg
+                debug_t["get_sync"] += time.perf_counter() - tb
+            else:
+                do_get(get_keys[s])
+            get_time += time.perf_counter() - tg
     sync_all()
     elapsed = time.perf_counter() - t0
+    if pipeline and not torch.equal(src.cpu(), dst.cpu()):
+        print(json.dumps({"error": "data mismatch in pipelined loop"}))
+        sys.exit(1)
     if debug:
         per = {k: round(v / args.steps * 1e6, 1) for k, v in debug_t.items()}
         print(f"rank {rank} per-step us: {per}", file=sys.stderr)
