@@ -49,6 +49,10 @@ def parse_args():
                         "(forces xGMI cross-shard traffic)")
     p.add_argument("--no-pipeline", action="store_true",
                    help="disable put/get step pipelining (sequential loop)")
+    p.add_argument("--conns", type=int, default=2,
+                   help="parallel client connections per rank (local path): "
+                        "each runs its own pipelined step loop over 1/conns "
+                        "of the blocks in its own thread")
     return p.parse_args()
 
 
@@ -122,6 +126,17 @@ def main():
         conn.register_mr(src)
         conn.register_mr(dst)
 
+    # Extra connections for intra-rank parallelism (local path only): the
+    # blocking waits release the GIL and the server spreads connections over
+    # its IO worker loops and HIP streams, so two half-size pipelined loops
+    # overlap each other's host-side request latency.
+    n_conns = args.conns if (use_local_path and not args.cross) else 1
+    conns = [conn]
+    for _ in range(n_conns - 1):
+        c = ifs.InfinityConnection(ccfg)
+        c.connect()
+        conns.append(c)
+
     import numpy as np
 
     run_id = uuid.uuid4().hex[:8]
@@ -175,12 +190,26 @@ def main():
     put_keys = [step_keys(s) for s in range(args.steps)]
     get_keys = [step_keys(s, read_rank) for s in range(args.steps)]
 
+    # Per-connection block slices (contiguous ranges; the absolute element
+    # offsets stay valid because both tensors are shared by all conns).
+    bounds = [(c * args.blocks) // n_conns for c in range(n_conns + 1)]
+    csl = [slice(bounds[c], bounds[c + 1]) for c in range(n_conns)]
+    coff = [offsets_np[s] for s in csl]
+
     # ---- correctness spot-check + warmup ----
     for w in range(args.warmup):
-        do_put([f"warm-{k}" for k in step_keys(w)])
-        if cross:
-            dist.barrier()
-        do_get([f"warm-{k}" for k in step_keys(w, read_rank)])
+        if n_conns > 1:
+            wk = [f"warm-{k}" for k in step_keys(w)]
+            for c in range(n_conns):  # warm every conn's IPC export + slab
+                conns[c].write_pages(src, wk[csl[c]], coff[c],
+                                     elems_per_block, sync=True)
+                conns[c].read_pages(dst, wk[csl[c]], coff[c], elems_per_block)
+                conns[c].sync()
+        else:
+            do_put([f"warm-{k}" for k in step_keys(w)])
+            if cross:
+                dist.barrier()
+            do_get([f"warm-{k}" for k in step_keys(w, read_rank)])
     if not torch.equal(src.cpu(), dst.cpu()):
         print(json.dumps({"error": "data mismatch in warmup"}))
         sys.exit(1)
@@ -206,19 +235,33 @@ def main():
     put_time = 0.0
     get_time = 0.0
     if pipeline:
-        tp = time.perf_counter()
-        conn.write_pages(src, put_keys[0], offsets_np, elems_per_block, sync=True)
-        put_time += time.perf_counter() - tp
-        for s in range(args.steps):
-            tp = time.perf_counter()
-            if s + 1 < args.steps:
-                conn.write_pages(src, put_keys[s + 1], offsets_np,
-                                 elems_per_block, sync=False)
-            tg = time.perf_counter()
-            conn.read_pages(dst, get_keys[s], offsets_np, elems_per_block)
-            conn.sync()  # get is sync-response; this drains put(s+1)'s commit
-            put_time += tg - tp
-            get_time += time.perf_counter() - tg
+        def run_conn(c):
+            cn, o, sl_ = conns[c], coff[c], csl[c]
+            pt = gt = 0.0
+            t = time.perf_counter()
+            cn.write_pages(src, put_keys[0][sl_], o, elems_per_block, sync=True)
+            pt += time.perf_counter() - t
+            for s in range(args.steps):
+                t = time.perf_counter()
+                if s + 1 < args.steps:
+                    cn.write_pages(src, put_keys[s + 1][sl_], o,
+                                   elems_per_block, sync=False)
+                tg = time.perf_counter()
+                cn.read_pages(dst, get_keys[s][sl_], o, elems_per_block)
+                cn.sync()  # get is sync-response; this drains put(s+1)'s commit
+                pt += tg - t
+                gt += time.perf_counter() - tg
+            return pt, gt
+
+        if n_conns == 1:
+            put_time, get_time = run_conn(0)
+        else:
+            import concurrent.futures as cf
+
+            with cf.ThreadPoolExecutor(n_conns) as ex:
+                res = list(ex.map(run_conn, range(n_conns)))
+            put_time = max(r[0] for r in res)
+            get_time = max(r[1] for r in res)
     else:
         for s in range(args.steps):
             tp = time.perf_counter()
@@ -324,7 +367,8 @@ def main():
         }
         print(json.dumps(result))
 
-    conn.close()
+    for c in conns:
+        c.close()
     if dist:
         dist.barrier()
     if rank == 0 and not external:
