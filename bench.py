@@ -95,6 +95,7 @@ def main():
             minimal_allocate_size=args.block_kb,
             cpu_only=not have_gpu,
             devices=list(range(n_shards)) if have_gpu else [],
+            io_threads=max(3, args.conns + 1),
         )
         ifs.register_server(scfg)
     if dist:
