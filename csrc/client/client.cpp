@@ -31,6 +31,10 @@ void ClientConn::close_conn() {
     if (worker_.joinable()) worker_.join();
     worker_started_ = false;
     verbs_.reset();
+    if (shm_active_) {
+        shm_active_ = false;
+        shm_.unmap();  // server side is torn down by the socket close below
+    }
     if (fd_ >= 0) {
         ::close(fd_);
         fd_ = -1;
@@ -58,6 +62,8 @@ int ClientConn::init_connection(const ClientConfigC& cfg) {
                     fd_ = ufd;
                     connected_ = true;
                     DEBUG("connected via UDS %s", path.c_str());
+                    if (cfg.connection_type == "LOCAL_GPU" && !getenv("IFS_NO_SHM"))
+                        try_shm_setup(cfg.service_port);
                     return 0;
                 }
                 ::close(ufd);
@@ -95,8 +101,127 @@ int ClientConn::init_connection(const ClientConfigC& cfg) {
     setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
     fd_ = fd;
     connected_ = true;
+    if ((cfg.host_addr == "127.0.0.1" || cfg.host_addr == "localhost") &&
+        cfg.connection_type == "LOCAL_GPU" && !getenv("IFS_NO_SHM"))
+        try_shm_setup(cfg.service_port);
     return 0;
 }
+
+// ---------------------------------------------------------------------------
+// shared-memory ring transport (same-host fast path for the packed ops)
+// ---------------------------------------------------------------------------
+bool ClientConn::try_shm_setup(int port) {
+    static std::atomic<uint32_t> counter{0};
+    std::string name = "/ifs-" + std::to_string(getpid()) + "-" + std::to_string(port) + "-" +
+                       std::to_string(counter.fetch_add(1));
+    if (!shmring::create_segment(name, 1u << 20, 64u << 10, &shm_)) return false;
+    bool ok = false;
+    if (send_req(OP_SHM_SETUP, reinterpret_cast<const uint8_t*>(name.data()), name.size())) {
+        int code = 0;
+        if (recv_status(&code) && code == FINISH) ok = true;
+    }
+    shm_unlink(name.c_str());  // both sides hold mappings; drop the name
+    if (!ok) {
+        shm_.unmap();
+        DEBUG("shm ring setup declined; staying on the socket");
+        return false;
+    }
+    shm_active_ = true;
+    DEBUG("shm ring transport active");
+    return true;
+}
+
+// Non-blocking: collect any responses already in the ring. Responses whose
+// seq nobody is waiting for are async-write errors; remember the first.
+void ClientConn::shm_drain_responses() {
+    for (;;) {
+        uint32_t len = 0;
+        uint64_t skip = 0;
+        const uint8_t* rec = shm_.resp->peek(&len, &skip);
+        if (!rec) return;
+        if (len >= sizeof(shmring::RespRec)) {
+            shmring::RespRec r;
+            memcpy(&r, rec, sizeof(r));
+            if (r.status != 0 && r.status != TASK_ACCEPTED && r.status != FINISH &&
+                shm_async_err_ == 0)
+                shm_async_err_ = r.status;
+        }
+        shm_.resp->consume(skip);
+    }
+}
+
+int ClientConn::shm_wait(uint64_t seq) {
+    auto t0 = std::chrono::steady_clock::now();
+    int spins = 0;
+    for (;;) {
+        uint32_t len = 0;
+        uint64_t skip = 0;
+        const uint8_t* rec = shm_.resp->peek(&len, &skip);
+        if (rec) {
+            if (len == 0) {  // wrap marker
+                shm_.resp->consume(skip);
+                continue;
+            }
+            shmring::RespRec r{};
+            memcpy(&r, rec, std::min(sizeof(r), size_t(len)));
+            shm_.resp->consume(skip);
+            if (r.h.seq == seq) return r.status;
+            if (r.status != 0 && r.status != TASK_ACCEPTED && r.status != FINISH &&
+                shm_async_err_ == 0)
+                shm_async_err_ = r.status;
+            continue;
+        }
+        // Spin briefly (responses usually land in tens of µs), then back off;
+        // give up after the same 60 s the socket path uses.
+        if (++spins < 2000) {
+#if defined(__x86_64__)
+            __builtin_ia32_pause();
+#endif
+        } else {
+            usleep(spins < 4000 ? 20 : 200);
+            if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(60)) {
+                ERROR("shm ring response timeout (seq %llu)",
+                      static_cast<unsigned long long>(seq));
+                return -1;
+            }
+        }
+    }
+}
+
+int ClientConn::shm_request(char op, const uint8_t* body, size_t n, bool want_resp) {
+    uint32_t need = shmring::rec_len(n);
+    if (need > shm_.req->cap / 2) return kShmNoFit;
+    shm_drain_responses();
+    uint64_t seq = ++shm_seq_;
+    uint64_t adv = 0;
+    uint8_t* dst = shm_.req->claim(need, &adv);
+    auto t0 = std::chrono::steady_clock::now();
+    while (!dst) {  // ring full: the server is behind; wait for space
+        shm_drain_responses();
+#if defined(__x86_64__)
+        __builtin_ia32_pause();
+#endif
+        if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(60)) return -1;
+        dst = shm_.req->claim(need, &adv);
+    }
+    shmring::RecHdr h{};
+    h.len = need;
+    h.op = static_cast<uint8_t>(op);
+    h.body_len = static_cast<uint32_t>(n);
+    h.seq = seq;
+    memcpy(dst, &h, sizeof(h));
+    if (n) memcpy(dst + sizeof(h), body, n);
+    shm_.req->publish(adv);
+    if (!want_resp) {
+        shm_unacked_++;
+        return 0;
+    }
+    int code = shm_wait(seq);
+    shm_unacked_ = 0;  // a sync-response op drains the ring ordering-wise
+    return code;
+}
+
+int ClientConn::shm_ring_sync() { return shm_request(OP_SYNC, nullptr, 0, /*want_resp=*/true); }
 
 int ClientConn::setup_rdma(const ClientConfigC& cfg) {
     if (!connected_) return -1;
@@ -131,6 +256,10 @@ int ClientConn::setup_rdma(const ClientConfigC& cfg) {
 // framing helpers (io_mu_ must be held)
 // ---------------------------------------------------------------------------
 bool ClientConn::send_req(char op, const uint8_t* body, size_t n) {
+    // Ring requests and socket requests are handled by different server
+    // threads; a ring sync round trip here restores the total order the
+    // socket alone used to give (only needed when ring ops are unacked).
+    if (shm_active_ && shm_unacked_ > 0) shm_ring_sync();
     Header h{kMagic, op, static_cast<uint32_t>(n)};
     if (!send_exact(fd_, &h, sizeof(h))) return false;
     if (n && !send_exact(fd_, body, n)) return false;
@@ -216,6 +345,25 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
     static const bool dbg = getenv("IFS_CLIENT_DEBUG") != nullptr;
     auto t0 = std::chrono::steady_clock::now();
     std::lock_guard<std::mutex> lk(io_mu_);
+    if (shm_active_) {
+        bool want_resp = (h.flags & kLocalFlagSyncResponse) != 0;
+        int code = shm_request(wire_op, body.data(), body.size(), want_resp);
+        if (code != kShmNoFit) {
+            if (dbg && n > 64) {
+                auto t2 = std::chrono::steady_clock::now();
+                fprintf(stderr, "[cdbg] op=%c n=%zu shm total=%.0fus\n", op, n,
+                        std::chrono::duration<double, std::micro>(t2 - t0).count());
+            }
+            if (want_resp && code != TASK_ACCEPTED && code != FINISH) {
+                WARN("rw_local(shm) op=%c -> %d", op, code);
+                return code < 0 ? code : -code;
+            }
+            if (op == 'W' && !sync_response) local_dirty_ = true;
+            return 0;
+        }
+        // falls through to the socket (record too large for the ring);
+        // send_req's ring sync keeps ordering.
+    }
     if (!send_req(wire_op, body.data(), body.size())) return -1;
     auto t1 = std::chrono::steady_clock::now();
     int code = 0;
@@ -239,7 +387,18 @@ int ClientConn::sync_local() {
     std::lock_guard<std::mutex> lk(io_mu_);
     // Reads complete before their response (kLocalFlagSyncResponse), so the
     // sync round trip is only needed after writes.
-    if (!local_dirty_) return 0;
+    if (!local_dirty_ && shm_unacked_ == 0) return 0;
+    if (shm_active_) {
+        int code = shm_ring_sync();
+        if (code != 0) return code < 0 ? code : -code;
+        local_dirty_ = false;
+        if (shm_async_err_) {  // surface an earlier async ring write failure
+            int e = shm_async_err_;
+            shm_async_err_ = 0;
+            return -e;
+        }
+        return 0;
+    }
     if (!send_req(OP_SYNC, nullptr, 0)) return -1;
     int remain = -1;
     if (!recv_status(&remain)) return -1;

@@ -24,6 +24,7 @@
 #include <vector>
 
 #include "../core/protocol.h"
+#include "../core/shm_ring.h"
 
 namespace ifs {
 
@@ -93,6 +94,9 @@ class ClientConn {
     std::string get_stats();
 
     bool rdma_connected() const { return rdma_connected_; }
+    // Shared-memory ring transport active (same-host fast path for the
+    // packed local ops; csrc/core/shm_ring.h).
+    bool shm_active() const { return shm_active_; }
 
    private:
     // synchronous framed request/response (io_mu_ held)
@@ -112,6 +116,22 @@ class ClientConn {
 
     void worker_main();
     void enqueue(std::function<void()> fn);
+
+    // ---- shm ring transport ----
+    bool try_shm_setup(int port);  // after connect; silent fallback on failure
+    // Push one record; want_resp: wait for and return the server status
+    // (mapped like the socket path). Returns kShmNoFit when the record does
+    // not fit the ring (caller falls back to the socket after a ring sync).
+    static constexpr int kShmNoFit = INT32_MIN;
+    int shm_request(char op, const uint8_t* body, size_t n, bool want_resp);
+    int shm_wait(uint64_t seq);           // drain responses until seq; io_mu_
+    int shm_ring_sync();                  // OP_SYNC over the ring; io_mu_
+    void shm_drain_responses();           // non-blocking pop; io_mu_
+    shmring::Segment shm_;
+    bool shm_active_ = false;
+    uint64_t shm_seq_ = 0;
+    uint64_t shm_unacked_ = 0;  // async ring writes since last ring sync
+    int shm_async_err_ = 0;     // first error for an unawaited seq (io_mu_)
 
     int fd_ = -1;
     bool connected_ = false;

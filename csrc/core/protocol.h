@@ -56,6 +56,11 @@ constexpr char OP_STATS = 'Q';    // server stats JSON over the wire (extension)
 // Body: PackedLocalHdr, u64 offsets[n], NUL-separated key bytes (n keys).
 constexpr char OP_W_FAST = 'w';
 constexpr char OP_R_FAST = 'r';
+// Shared-memory ring handshake (extension, same-host clients): body is the
+// shm_open name of a client-created segment (csrc/core/shm_ring.h); after a
+// FINISH response the client sends OP_W_FAST/OP_R_FAST/OP_SYNC through the
+// ring instead of the socket.
+constexpr char OP_SHM_SETUP = 'h';
 
 #pragma pack(push, 1)
 struct PackedLocalHdr {

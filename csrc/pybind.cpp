@@ -308,7 +308,8 @@ PYBIND11_MODULE(_native, m) {
         .def("get_match_last_index", &ClientConn::get_match_last_index,
              py::call_guard<py::gil_scoped_release>())
         .def("delete_keys", &ClientConn::delete_keys, py::call_guard<py::gil_scoped_release>())
-        .def("get_stats", &ClientConn::get_stats, py::call_guard<py::gil_scoped_release>());
+        .def("get_stats", &ClientConn::get_stats, py::call_guard<py::gil_scoped_release>())
+        .def("shm_active", &ClientConn::shm_active);
 
     // ---- server ----
     m.def("start_server", &start_server, py::call_guard<py::gil_scoped_release>());
@@ -401,6 +402,57 @@ PYBIND11_MODULE(_native, m) {
     });
 
     // WR flow-control simulation (chain split + outstanding cap + overflow).
+    // shm ring framing/wrap test harness: push `msgs` through a tiny ring
+    // with an interleaved consumer, return the bodies read back in order.
+    m.def("_dbg_shmring_echo", [](const std::vector<py::bytes>& msgs, uint32_t cap,
+                                  int drain_every) {
+        std::vector<uint8_t> mem(ifs::shmring::Ring::footprint(cap));
+        auto* ring = reinterpret_cast<ifs::shmring::Ring*>(mem.data());
+        new (&ring->head) std::atomic<uint64_t>(0);
+        new (&ring->tail) std::atomic<uint64_t>(0);
+        ring->cap = cap;
+        std::vector<py::bytes> out;
+        uint64_t seq = 0;
+        auto drain = [&] {
+            for (;;) {
+                uint32_t len = 0;
+                uint64_t skip = 0;
+                const uint8_t* rec = ring->peek(&len, &skip);
+                if (!rec) return;
+                if (len) {
+                    ifs::shmring::RecHdr h;
+                    memcpy(&h, rec, sizeof(h));
+                    if (h.seq != ++seq) throw std::runtime_error("seq out of order");
+                    out.emplace_back(reinterpret_cast<const char*>(rec + sizeof(h)),
+                                     h.body_len);
+                }
+                ring->consume(skip);
+            }
+        };
+        int pushed = 0;
+        for (auto& m2 : msgs) {
+            std::string body = m2;
+            uint32_t need = ifs::shmring::rec_len(body.size());
+            uint64_t adv = 0;
+            uint8_t* dst = ring->claim(need, &adv);
+            for (long spin = 0; !dst; spin++) {  // full: drain like the consumer would
+                drain();
+                dst = ring->claim(need, &adv);
+                if (spin > 1000000) throw std::runtime_error("ring stuck");
+            }
+            ifs::shmring::RecHdr h{};
+            h.len = need;
+            h.op = 'T';
+            h.body_len = static_cast<uint32_t>(body.size());
+            h.seq = static_cast<uint64_t>(pushed + 1);
+            memcpy(dst, &h, sizeof(h));
+            memcpy(dst + sizeof(h), body.data(), body.size());
+            ring->publish(adv);
+            if (++pushed % std::max(1, drain_every) == 0) drain();
+        }
+        drain();
+        return out;
+    });
     m.def("_dbg_wrflow_sim", [](int n_wrs, int batch, int cap, int complete_after) {
         std::deque<size_t> inflight_chains;
         std::vector<size_t> posted_sizes;

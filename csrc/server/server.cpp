@@ -282,6 +282,7 @@ void Server::IoLoop::on_stop(uv_async_t* h) {
     // remain.
     for (auto* c : io->conns) {
         io->srv->verbs_teardown(c);
+        io->srv->shm_teardown(c);
         conn_close(c);
     }
     io->conns.clear();
@@ -429,6 +430,7 @@ void Server::adopt_fd(IoLoop* io, int fd, bool is_pipe) {
                 auto& v = c->owner->conns;
                 v.erase(std::remove(v.begin(), v.end(), c), v.end());
                 srv->verbs_teardown(c);
+                srv->shm_teardown(c);
                 conn_close(c);
                 return;
             }
@@ -451,6 +453,7 @@ void Server::adopt_fd(IoLoop* io, int fd, bool is_pipe) {
                         auto& v = c->owner->conns;
                         v.erase(std::remove(v.begin(), v.end(), c), v.end());
                         srv->verbs_teardown(c);
+                        srv->shm_teardown(c);
                         conn_close(c);
                         return;
                     }
@@ -460,6 +463,7 @@ void Server::adopt_fd(IoLoop* io, int fd, bool is_pipe) {
                         auto& v = c->owner->conns;
                         v.erase(std::remove(v.begin(), v.end(), c), v.end());
                         srv->verbs_teardown(c);
+                        srv->shm_teardown(c);
                         conn_close(c);
                         return;
                     }
@@ -500,24 +504,24 @@ Server::LocalView to_view(const LocalMetaMsg& msg) {
 }
 
 // Packed fast-path body: PackedLocalHdr, u64 offsets[n], NUL-separated keys.
-bool parse_packed_local(const std::vector<uint8_t>& body, Server::LocalView* v) {
-    if (body.size() < sizeof(PackedLocalHdr)) return false;
+bool parse_packed_local(const uint8_t* body, size_t body_len, Server::LocalView* v) {
+    if (body_len < sizeof(PackedLocalHdr)) return false;
     PackedLocalHdr h;
-    memcpy(&h, body.data(), sizeof(h));
+    memcpy(&h, body, sizeof(h));
     size_t n = h.n_blocks;
     size_t off_end = sizeof(h) + n * 8;
-    if (body.size() < off_end) return false;
+    if (body_len < off_end) return false;
     v->device = h.device;
     v->pid = h.pid;
     v->base_ptr = h.base_ptr;
     v->base_offset = h.base_offset;
     v->block_size = h.block_size;
     v->flags = h.flags;
-    v->ipc = body.data() + offsetof(PackedLocalHdr, ipc);
+    v->ipc = body + offsetof(PackedLocalHdr, ipc);
     v->ipc_len = 64;
-    const uint64_t* offs = reinterpret_cast<const uint64_t*>(body.data() + sizeof(h));
-    const char* kp = reinterpret_cast<const char*>(body.data() + off_end);
-    const char* kend = reinterpret_cast<const char*>(body.data() + body.size());
+    const uint64_t* offs = reinterpret_cast<const uint64_t*>(body + sizeof(h));
+    const char* kp = reinterpret_cast<const char*>(body + off_end);
+    const char* kend = reinterpret_cast<const char*>(body + body_len);
     v->blocks.reserve(n);
     for (size_t i = 0; i < n; i++) {
         const char* nul = static_cast<const char*>(memchr(kp, 0, kend - kp));
@@ -554,25 +558,29 @@ void Server::handle_request(Conn* c, char op, std::vector<uint8_t> body) {
         case OP_W: {
             LocalMetaMsg msg;
             if (!parse_local_meta(body.data(), body.size(), &msg)) return send_status(c, INVALID_REQ);
-            return op_local_write(c, to_view(msg));
+            return op_local_write(c, to_view(msg), ReqCtx{});
         }
         case OP_R: {
             LocalMetaMsg msg;
             if (!parse_local_meta(body.data(), body.size(), &msg)) return send_status(c, INVALID_REQ);
-            return op_local_read(c, to_view(msg));
+            return op_local_read(c, to_view(msg), ReqCtx{});
         }
         case OP_W_FAST: {
             LocalView v;
-            if (!parse_packed_local(body, &v)) return send_status(c, INVALID_REQ);
-            return op_local_write(c, v);
+            if (!parse_packed_local(body.data(), body.size(), &v))
+                return send_status(c, INVALID_REQ);
+            return op_local_write(c, v, ReqCtx{});
         }
         case OP_R_FAST: {
             LocalView v;
-            if (!parse_packed_local(body, &v)) return send_status(c, INVALID_REQ);
-            return op_local_read(c, v);
+            if (!parse_packed_local(body.data(), body.size(), &v))
+                return send_status(c, INVALID_REQ);
+            return op_local_read(c, v, ReqCtx{});
         }
         case OP_SYNC:
-            return op_sync(c);
+            return op_sync(c, ReqCtx{});
+        case OP_SHM_SETUP:
+            return op_shm_setup(c, body);
         case OP_RDMA_EXCHANGE:
             return op_exchange(c, body);
         case OP_RDMA_ALLOCATE: {
@@ -719,14 +727,14 @@ void* resolve_client_base(Server::Conn* c, const Server::LocalView& msg) {
 }
 }  // namespace
 
-void Server::op_local_write(Conn* c, const LocalView& msg) {
+void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     static const bool pdbg = getenv("IFS_SERVER_DEBUG") != nullptr;
     auto p0 = std::chrono::steady_clock::now();
-    if (!gpu::available()) return send_status(c, SYSTEM_ERROR);
+    if (!gpu::available()) return reply_local(c, ctx, SYSTEM_ERROR);
     if (msg.ipc_len != gpu::kIpcHandleSize || msg.block_size <= 0)
-        return send_status(c, INVALID_REQ);
+        return reply_local(c, ctx, INVALID_REQ);
     void* base = resolve_client_base(c, msg);
-    if (!base) return send_status(c, INTERNAL_ERROR);
+    if (!base) return reply_local(c, ctx, INTERNAL_ERROR);
     uint8_t* client_ptr = static_cast<uint8_t*>(base) + msg.base_offset;
 
     Shard* shard = shard_for_device(msg.device);
@@ -759,7 +767,8 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
     bool sync_resp = (msg.flags & kLocalFlagSyncResponse) != 0;
     if (n_fresh == 0) {
         // everything was a duplicate — nothing to copy
-        return send_status(c, sync_resp ? FINISH : TASK_ACCEPTED);
+        if (ctx.shm && !sync_resp) return;  // ring async writes are unacked
+        return reply_local(c, ctx, sync_resp ? FINISH : TASK_ACCEPTED);
     }
 
     // Phase B — batched allocation (shard allocator lock only).
@@ -774,7 +783,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
         std::lock_guard<std::mutex> lk(kv_mu_);
         if (evict_lru_locked(shard, page * n_fresh * 2) > 0) alloc_ok = try_alloc();
     }
-    if (!alloc_ok) return send_status(c, OUT_OF_MEMORY);
+    if (!alloc_ok) return reply_local(c, ctx, OUT_OF_MEMORY);
     auto p2 = std::chrono::steady_clock::now();
 
     // Phase C — create entries, build the copy job and LAUNCH it; the index
@@ -806,10 +815,14 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
     c->ref();
     static const bool sdbg = getenv("IFS_SERVER_DEBUG") != nullptr;
     auto t_start = std::chrono::steady_clock::now();
-    // won[i] is written in phase D on this thread BEFORE the posted commit
-    // lambda can run on the same (owner-loop) thread.
+    // won[i] is written in phase D after the kernel is submitted. On the
+    // socket path the commit lambda is posted to the owner loop — the same
+    // thread running this handler — so it cannot run before phase D ends.
+    // On the shm path the commit runs on a completion thread; commit_mu
+    // (held here across submit + phase D) provides that ordering instead.
     auto won = std::make_shared<std::vector<uint8_t>>(n_fresh, 0);
-    job.done = [this, c, entries, won, sync_resp, t_start](bool ok) {
+    auto commit_mu = std::make_shared<std::mutex>();
+    job.done = [this, c, entries, won, sync_resp, ctx, commit_mu, t_start](bool ok) {
         if (sdbg && entries->size() > 64) {
             auto us = std::chrono::duration<double, std::micro>(
                           std::chrono::steady_clock::now() - t_start)
@@ -817,7 +830,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
             fprintf(stderr, "[sdbg] write n=%zu submit->complete=%.0fus\n", entries->size(),
                     us);
         }
-        c->owner->post([this, c, entries, won, ok, sync_resp] {
+        auto fin = [this, c, entries, won, ok, sync_resp, ctx] {
             if (ok) {
                 for (size_t i = 0; i < entries->size(); i++)
                     if ((*won)[i]) (*entries)[i]->committed = true;
@@ -827,14 +840,24 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
                     if ((*won)[i]) winners.push_back((*entries)[i]);
                 erase_entries(winners);  // copy failed: drop the keys
             }
-            if (sync_resp) send_status(c, ok ? FINISH : INTERNAL_ERROR);
-            finish_task(c);
-        });
+            if (sync_resp) reply_local(c, ctx, ok ? FINISH : INTERNAL_ERROR);
+            finish_task(c, /*on_owner=*/!ctx.shm);
+        };
+        if (ctx.shm) {
+            std::lock_guard<std::mutex> lk(*commit_mu);  // wait out phase D
+            fin();
+        } else {
+            c->owner->post([fin, commit_mu] {
+                std::lock_guard<std::mutex> lk(*commit_mu);
+                fin();
+            });
+        }
     };
     // Async write: respond before submitting so the client's next request
     // overlaps the kernel. Sync-response write (flags&1): one round trip,
-    // response sent on completion instead.
-    if (!sync_resp) send_status(c, TASK_ACCEPTED);
+    // response sent on completion instead. Ring async writes are unacked.
+    if (!sync_resp && !ctx.shm) send_status(c, TASK_ACCEPTED);
+    std::unique_lock<std::mutex> commit_lk(*commit_mu);
     bool submitted = shard->submit_copy(std::move(job));
 
     // Phase D — insert pass, overlapped with the in-flight kernel. Losers
@@ -851,6 +874,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
             (*won)[i] = inserted ? 1 : 0;
         }
     }
+    commit_lk.unlock();  // phase D done: completion may commit
     auto p3 = std::chrono::steady_clock::now();
     if (pdbg && n_fresh > 64) {
         auto us = [](auto a, auto b) {
@@ -859,15 +883,15 @@ void Server::op_local_write(Conn* c, const LocalView& msg) {
         fprintf(stderr, "[pdbg] dedup=%.0f alloc=%.0f build+insert=%.0f\n", us(p0, p1),
                 us(p1, p2), us(p2, p3));
     }
-    if (!submitted) finish_task(c);
+    if (!submitted) finish_task(c, /*on_owner=*/!ctx.shm);
 }
 
-void Server::op_local_read(Conn* c, const LocalView& msg) {
-    if (!gpu::available()) return send_status(c, SYSTEM_ERROR);
+void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
+    if (!gpu::available()) return reply_local(c, ctx, SYSTEM_ERROR);
     if (msg.ipc_len != gpu::kIpcHandleSize || msg.block_size <= 0)
-        return send_status(c, INVALID_REQ);
+        return reply_local(c, ctx, INVALID_REQ);
     void* base = resolve_client_base(c, msg);
-    if (!base) return send_status(c, INTERNAL_ERROR);
+    if (!base) return reply_local(c, ctx, INTERNAL_ERROR);
     uint8_t* client_ptr = static_cast<uint8_t*>(base) + msg.base_offset;
     size_t page = static_cast<size_t>(msg.block_size);
 
@@ -889,7 +913,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
             Ref<BlockEntry>* v = kv_.find_hashed(b.first, hashes[bi]);
             bi++;
             if (!v || !(*v)->committed) {
-                return send_status(c, KEY_NOT_FOUND);
+                return reply_local(c, ctx, KEY_NOT_FOUND);
             }
             BlockEntry* e = v->get();
             e->last_access = read_tick;
@@ -903,7 +927,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
     n_reads_.fetch_add(1);
     bytes_out_.fetch_add(msg.blocks.size() * page);
     bool sync_resp = (msg.flags & kLocalFlagSyncResponse) != 0;
-    if (jobs.empty()) return send_status(c, sync_resp ? FINISH : TASK_ACCEPTED);
+    if (jobs.empty()) return reply_local(c, ctx, sync_resp ? FINISH : TASK_ACCEPTED);
 
     c->remain.fetch_add(1);
     c->ref();
@@ -913,7 +937,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
         Shard::CopyJob j = std::move(job);
         static const bool sdbg2 = getenv("IFS_SERVER_DEBUG") != nullptr;
         auto t_start2 = std::chrono::steady_clock::now();
-        j.done = [this, c, held, pending, all_ok, sync_resp, t_start2](bool ok) {
+        j.done = [this, c, held, pending, all_ok, sync_resp, ctx, t_start2](bool ok) {
             if (sdbg2 && held->size() > 64) {
                 auto us = std::chrono::duration<double, std::micro>(
                               std::chrono::steady_clock::now() - t_start2)
@@ -922,37 +946,199 @@ void Server::op_local_read(Conn* c, const LocalView& msg) {
             }
             if (!ok) all_ok->store(false);
             if (pending->fetch_sub(1) == 1) {
-                c->owner->post([this, c, held, all_ok, sync_resp] {
-                    if (sync_resp)
-                        send_status(c, all_ok->load() ? FINISH : INTERNAL_ERROR);
-                    finish_task(c);
-                });
+                if (ctx.shm) {
+                    // reply straight from the completion thread: the ring
+                    // write needs no loop-thread affinity.
+                    if (sync_resp) reply_local(c, ctx, all_ok->load() ? FINISH : INTERNAL_ERROR);
+                    finish_task(c, /*on_owner=*/false);
+                } else {
+                    c->owner->post([this, c, held, all_ok, sync_resp] {
+                        if (sync_resp)
+                            send_status(c, all_ok->load() ? FINISH : INTERNAL_ERROR);
+                        finish_task(c, /*on_owner=*/true);
+                    });
+                }
             }
         };
         if (!shard->submit_copy(std::move(j))) {
             all_ok->store(false);
             if (pending->fetch_sub(1) == 1) {
-                finish_task(c);
-                return send_status(c, INTERNAL_ERROR);
+                finish_task(c, /*on_owner=*/!ctx.shm);
+                return reply_local(c, ctx, INTERNAL_ERROR);
             }
         }
     }
-    if (!sync_resp) send_status(c, TASK_ACCEPTED);
+    if (!sync_resp && !ctx.shm) send_status(c, TASK_ACCEPTED);
 }
 
-void Server::op_sync(Conn* c) {
+void Server::op_sync(Conn* c, const ReqCtx& ctx) {
+    std::lock_guard<std::mutex> lk(c->sync_mu);
     if (c->remain.load() == 0) {
-        send_status(c, 0);
+        reply_local(c, ctx, 0);
     } else {
         c->sync_waiting = true;  // answered by finish_task when remain drains
+        c->sync_ctx = ctx;
     }
 }
 
-// Runs on the loop thread after a local-path copy completes.
-void Server::finish_task(Conn* c) {
-    if (c->remain.fetch_sub(1) == 1 && c->sync_waiting) {
-        c->sync_waiting = false;
-        send_status(c, 0);
+// Owner loop thread (socket completions) or a completion thread (shm).
+void Server::finish_task(Conn* c, bool on_owner) {
+    if (c->remain.fetch_sub(1) == 1) {
+        ReqCtx ctx;
+        bool respond = false;
+        {
+            std::lock_guard<std::mutex> lk(c->sync_mu);
+            if (c->sync_waiting) {
+                c->sync_waiting = false;
+                respond = true;
+                ctx = c->sync_ctx;
+            }
+        }
+        if (respond) {
+            if (ctx.shm || on_owner) {
+                reply_local(c, ctx, 0);
+            } else {
+                // socket reply from a completion thread: hop to the owner
+                // loop (uv_write is not thread-safe).
+                c->ref();
+                c->owner->post([this, c, ctx] {
+                    reply_local(c, ctx, 0);
+                    c->unref();
+                });
+            }
+        }
+    }
+    c->unref();
+}
+
+void Server::reply_local(Conn* c, const ReqCtx& ctx, int code) {
+    if (ctx.shm && c->shm) return c->shm->push_resp(ctx.seq, code);
+    send_status(c, code);
+}
+
+// ---- shared-memory ring transport -----------------------------------------
+void Server::ShmPeer::push_resp(uint64_t seq, int status) {
+    std::lock_guard<std::mutex> lk(resp_mu);
+    shmring::RespRec r{};
+    r.h.len = sizeof(r);
+    r.h.op = 0;
+    r.h.body_len = sizeof(r) - sizeof(r.h);
+    r.h.seq = seq;
+    r.status = status;
+    uint64_t adv = 0;
+    uint8_t* dst = seg.resp->claim(sizeof(r), &adv);
+    if (!dst) {
+        // Client stopped draining (crashed / gone): drop the response; the
+        // socket EOF tears the conn down.
+        WARN("shm resp ring full; dropping response seq=%llu",
+             static_cast<unsigned long long>(seq));
+        return;
+    }
+    memcpy(dst, &r, sizeof(r));
+    seg.resp->publish(adv);
+}
+
+void Server::op_shm_setup(Conn* c, const std::vector<uint8_t>& body) {
+    // Body: shm_open name of a client-created segment. Refuse when already
+    // set up, name is implausible, or too many pollers are running.
+    if (c->shm || body.empty() || body.size() > 100 || body[0] != '/')
+        return send_status(c, INVALID_REQ);
+    if (shm_peers_.load() >= 32) return send_status(c, SYSTEM_ERROR);
+    std::string name(reinterpret_cast<const char*>(body.data()), body.size());
+    auto* p = new ShmPeer();
+    p->srv = this;
+    p->c = c;
+    if (!shmring::open_segment(name, &p->seg)) {
+        delete p;
+        return send_status(c, INTERNAL_ERROR);
+    }
+    c->shm = p;
+    shm_peers_.fetch_add(1);
+    c->ref();  // held by the poller thread; released at poller exit
+    p->th = std::thread([this, p] { shm_poll_main(p); });
+    INFO("shm ring attached (%zu bytes)", p->seg.len);
+    send_status(c, FINISH);
+}
+
+void Server::shm_teardown(Conn* c) {
+    if (!c->shm) return;
+    c->shm->stop.store(true, std::memory_order_release);
+    if (c->shm->th.joinable()) c->shm->th.join();
+    shm_peers_.fetch_sub(1);
+    // The peer + mapping are freed in ~Conn: in-flight completion threads
+    // may still write responses into the (now client-less) ring.
+}
+
+void Server::shm_poll_main(ShmPeer* p) {
+    Conn* c = p->c;
+    auto last_work = std::chrono::steady_clock::now();
+    while (!p->stop.load(std::memory_order_acquire)) {
+        uint32_t len = 0;
+        uint64_t skip = 0;
+        const uint8_t* rec = p->seg.req->peek(&len, &skip);
+        if (!rec) {
+            // Adaptive idle: spin while recently hot (sub-µs pickup), back
+            // off to 50 µs sleeps when the ring has been quiet.
+            if (std::chrono::steady_clock::now() - last_work > std::chrono::milliseconds(2)) {
+                usleep(50);
+            } else {
+#if defined(__x86_64__)
+                for (int i = 0; i < 64; i++) __builtin_ia32_pause();
+#else
+                std::this_thread::yield();
+#endif
+            }
+            continue;
+        }
+        if (len == 0) {  // wrap marker
+            p->seg.req->consume(skip);
+            continue;
+        }
+        shmring::RecHdr h;
+        memcpy(&h, rec, sizeof(h));
+        const uint8_t* body = rec + sizeof(h);
+        ReqCtx ctx{h.seq, true};
+        auto t0 = std::chrono::steady_clock::now();
+        char op = static_cast<char>(h.op);
+        if (h.body_len + sizeof(h) > len) {
+            reply_local(c, ctx, INVALID_REQ);
+        } else {
+            // The record stays in the ring while the handler runs (views
+            // point into it); consume() below frees the space.
+            switch (op) {
+                case OP_W_FAST: {
+                    LocalView v;
+                    if (!parse_packed_local(body, h.body_len, &v))
+                        reply_local(c, ctx, INVALID_REQ);
+                    else
+                        op_local_write(c, v, ctx);
+                    break;
+                }
+                case OP_R_FAST: {
+                    LocalView v;
+                    if (!parse_packed_local(body, h.body_len, &v))
+                        reply_local(c, ctx, INVALID_REQ);
+                    else
+                        op_local_read(c, v, ctx);
+                    break;
+                }
+                case OP_SYNC:
+                    op_sync(c, ctx);
+                    break;
+                default:
+                    reply_local(c, ctx, INVALID_REQ);
+            }
+        }
+        p->seg.req->consume(skip);
+        last_work = std::chrono::steady_clock::now();
+        auto us = std::chrono::duration_cast<std::chrono::microseconds>(last_work - t0).count();
+        auto& st = op_stats_[static_cast<uint8_t>(op) & 127];
+        st.count.fetch_add(1, std::memory_order_relaxed);
+        st.total_us.fetch_add(static_cast<uint64_t>(us), std::memory_order_relaxed);
+        uint64_t prev = st.max_us.load(std::memory_order_relaxed);
+        while (static_cast<uint64_t>(us) > prev &&
+               !st.max_us.compare_exchange_weak(prev, static_cast<uint64_t>(us))) {
+        }
     }
     c->unref();
 }

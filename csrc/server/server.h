@@ -38,6 +38,7 @@
 
 #include "../core/mempool.h"
 #include "../core/protocol.h"
+#include "../core/shm_ring.h"
 #include "../core/utils.h"
 #include "../fabric/verbs_fabric.h"
 #include "../gpu/gpu.h"
@@ -116,6 +117,30 @@ class Server {
 
     struct Conn;  // defined below (file-local helpers + server_verbs use it)
 
+    // How a request arrived: over the socket (default) or via the conn's
+    // shared-memory ring. Replies for shm requests are written straight into
+    // the response ring — from ANY thread — instead of uv_write on the owner
+    // loop, which is what lets completion threads answer without the
+    // uv_async hop.
+    struct ReqCtx {
+        uint64_t seq = 0;
+        bool shm = false;
+    };
+
+    // Shared-memory ring peer (csrc/core/shm_ring.h): one per conn that
+    // completed OP_SHM_SETUP. The poller thread holds a conn ref; the
+    // segment stays mapped until ~Conn so late completion-thread replies
+    // never touch unmapped memory.
+    struct ShmPeer {
+        Server* srv = nullptr;
+        Conn* c = nullptr;
+        shmring::Segment seg;
+        std::thread th;
+        std::atomic<bool> stop{false};
+        std::mutex resp_mu;  // serializes response writers (poller + completions)
+        void push_resp(uint64_t seq, int status);
+    };
+
     // Uniform view of a local-path request (flatbuffers or packed fast-path
     // format). Key views point into the request body / parsed message and
     // are only valid during the handling call.
@@ -181,10 +206,19 @@ class Server {
 
     // ---- request handling (loop thread) ----
     void handle_request(Conn* c, char op, std::vector<uint8_t> body);
-    void op_local_write(Conn* c, const LocalView& msg);
-    void op_local_read(Conn* c, const LocalView& msg);
-    void op_sync(Conn* c);
-    void finish_task(Conn* c);  // remain-- (+ deferred sync reply)
+    void op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx);
+    void op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx);
+    void op_sync(Conn* c, const ReqCtx& ctx);
+    // remain-- (+ deferred sync reply). on_owner: caller runs on the conn's
+    // owner loop thread (socket replies must; shm replies need not).
+    void finish_task(Conn* c, bool on_owner);
+    // Route a status to the request's transport (ring: any thread; socket:
+    // owner loop thread only).
+    void reply_local(Conn* c, const ReqCtx& ctx, int code);
+    void op_shm_setup(Conn* c, const std::vector<uint8_t>& body);
+    void shm_teardown(Conn* c);       // owner loop thread; joins the poller
+    void shm_poll_main(ShmPeer* p);   // poller thread body
+    std::atomic<int> shm_peers_{0};
     void op_exchange(Conn* c, const std::vector<uint8_t>& body);
     void op_allocate(Conn* c, const RemoteMetaMsg& msg);
     void op_tcp_put(Conn* c, std::vector<uint8_t> body);
@@ -282,9 +316,21 @@ struct Server::Conn : RefCounted {
     // verbs fabric peer (owned; torn down on the loop thread).
     Server::VerbsPeer* verbs = nullptr;
 
+    // shared-memory ring peer (poller joined at teardown; freed in ~Conn).
+    Server::ShmPeer* shm = nullptr;
+    // Guards the {remain==0, sync_waiting, sync_ctx} decision — with the shm
+    // transport, op_sync (poller thread) races finish_task (completion
+    // threads); on the socket path both ran on the owner loop.
+    std::mutex sync_mu;
+    Server::ReqCtx sync_ctx;
+
     ~Conn() override {
         for (auto& kv : ipc_cache) {
             if (gpu::available()) gpu::ipc_close(kv.second.first);
+        }
+        if (shm) {
+            shm->seg.unmap();
+            delete shm;
         }
     }
 };

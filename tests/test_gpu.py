@@ -353,3 +353,54 @@ def test_gpu_page_hashes_wrapper():
     t2[5000] += 1.0  # page 1 changes -> pages 1..3 chain digests change
     h3 = gpu_page_hashes(t2, [0, 4096, 8192, 12288], 4096)
     assert h3[0] == h1[0] and h3[1] != h1[1] and h3[3] != h1[3]
+
+
+def test_shm_transport_roundtrip(gpu_server):
+    """Same-host packed ops ride the shared-memory ring (csrc/core/shm_ring.h):
+    the transport must be active on a local conn, round-trip correctly
+    (including requests big enough to wrap the 1 MB ring several times), and
+    fall back to the socket when disabled via IFS_NO_SHM."""
+    conn = local_conn(gpu_server)
+    try:
+        assert conn.conn.shm_active(), "shm ring should be active on a local conn"
+        n_blocks, page = 600, 4096  # ~20 KB of keys/offsets per request
+        src = torch.randn(n_blocks * page // 4, device="cuda:0")
+        dst = torch.zeros_like(src)
+        offs = [i * page // 4 for i in range(n_blocks)]
+        for rep in range(8):  # > 8 MB of request records => ring wraps
+            keys = [f"shm-{uuid.uuid4()}-{rep}-{i}" for i in range(n_blocks)]
+            conn.local_gpu_write_cache(src, list(zip(keys, offs)), page // 4)
+            assert conn.sync() == 0
+            dst.zero_()
+            conn.read_cache(dst, list(zip(keys, offs)), page // 4)
+            conn.sync()
+            assert torch.equal(src, dst)
+    finally:
+        conn.close()
+
+
+def test_shm_disabled_fallback(gpu_server):
+    env = os.environ.copy()
+    env["IFS_NO_SHM"] = "1"
+    code = subprocess.run(
+        [sys.executable, "-c", f"""
+import torch, uuid
+import infinistore_amd as ifs
+cfg = ifs.ClientConfig(host_addr="127.0.0.1", service_port={gpu_server},
+                       connection_type=ifs.TYPE_LOCAL_GPU)
+c = ifs.InfinityConnection(cfg)
+c.connect()
+assert not c.conn.shm_active()
+src = torch.randn(8192, device="cuda:0")
+dst = torch.zeros_like(src)
+key = f"nshm-{{uuid.uuid4()}}"
+c.local_gpu_write_cache(src, [(key, 0)], 8192)
+c.sync()
+c.read_cache(dst, [(key, 0)], 8192)
+c.sync()
+assert torch.equal(src, dst)
+c.close()
+"""],
+        env=env, cwd=REPO, timeout=120,
+    ).returncode
+    assert code == 0
