@@ -129,14 +129,24 @@ def main():
 
     # Extra connections for intra-rank parallelism (local path only): the
     # blocking waits release the GIL and the server spreads connections over
-    # its IO worker loops and HIP streams, so two half-size pipelined loops
-    # overlap each other's host-side request latency.
+    # its IO worker loops / ring pollers and HIP streams, so two half-size
+    # pipelined loops overlap each other's host-side request latency.
+    # Each loop uses a write-conn + read-conn PAIR so the async write handler
+    # is never queued in front of the blocking read on one connection — the
+    # natural shape for disaggregated serving (prefill writes, decode reads).
     n_conns = args.conns if (use_local_path and not args.cross) else 1
-    conns = [conn]
+    conns = [conn]   # read conns (conns[0] also serves the latency phase)
+    wconns = []      # write conns (writes + syncs)
     for _ in range(n_conns - 1):
         c = ifs.InfinityConnection(ccfg)
         c.connect()
         conns.append(c)
+    for _ in range(n_conns if use_local_path else 0):
+        c = ifs.InfinityConnection(ccfg)
+        c.connect()
+        wconns.append(c)
+    if not wconns:
+        wconns = conns
 
     import numpy as np
 
@@ -199,11 +209,11 @@ def main():
 
     # ---- correctness spot-check + warmup ----
     for w in range(args.warmup):
-        if n_conns > 1:
+        if use_local_path and not cross:
             wk = [f"warm-{k}" for k in step_keys(w)]
             for c in range(n_conns):  # warm every conn's IPC export + slab
-                conns[c].write_pages(src, wk[csl[c]], coff[c],
-                                     elems_per_block, sync=True)
+                wconns[c].write_pages(src, wk[csl[c]], coff[c],
+                                      elems_per_block, sync=True)
                 conns[c].read_pages(dst, wk[csl[c]], coff[c], elems_per_block)
                 conns[c].sync()
         else:
@@ -237,19 +247,20 @@ def main():
     get_time = 0.0
     if pipeline:
         def run_conn(c):
-            cn, o, sl_ = conns[c], coff[c], csl[c]
+            rc_, wc_, o, sl_ = conns[c], wconns[c], coff[c], csl[c]
             pt = gt = 0.0
             t = time.perf_counter()
-            cn.write_pages(src, put_keys[0][sl_], o, elems_per_block, sync=True)
+            wc_.write_pages(src, put_keys[0][sl_], o, elems_per_block, sync=True)
             pt += time.perf_counter() - t
             for s in range(args.steps):
                 t = time.perf_counter()
                 if s + 1 < args.steps:
-                    cn.write_pages(src, put_keys[s + 1][sl_], o,
-                                   elems_per_block, sync=False)
+                    wc_.write_pages(src, put_keys[s + 1][sl_], o,
+                                    elems_per_block, sync=False)
                 tg = time.perf_counter()
-                cn.read_pages(dst, get_keys[s][sl_], o, elems_per_block)
-                cn.sync()  # get is sync-response; this drains put(s+1)'s commit
+                rc_.read_pages(dst, get_keys[s][sl_], o, elems_per_block)
+                rc_.sync()
+                wc_.sync()  # drain put(s+1)'s commit before the next read
                 pt += tg - t
                 gt += time.perf_counter() - tg
             return pt, gt
@@ -370,6 +381,9 @@ def main():
 
     for c in conns:
         c.close()
+    for c in wconns:
+        if c not in conns:
+            c.close()
     if dist:
         dist.barrier()
     if rank == 0 and not external:

@@ -370,7 +370,7 @@ def test_shm_transport_roundtrip(gpu_server):
         for rep in range(8):  # > 8 MB of request records => ring wraps
             keys = [f"shm-{uuid.uuid4()}-{rep}-{i}" for i in range(n_blocks)]
             conn.local_gpu_write_cache(src, list(zip(keys, offs)), page // 4)
-            assert conn.sync() == 0
+            conn.sync()
             dst.zero_()
             conn.read_cache(dst, list(zip(keys, offs)), page // 4)
             conn.sync()
