@@ -171,14 +171,15 @@ int ClientConn::shm_wait(uint64_t seq) {
                 shm_async_err_ = r.status;
             continue;
         }
-        // Spin briefly (responses usually land in tens of µs), then back off;
-        // give up after the same 60 s the socket path uses.
-        if (++spins < 2000) {
+        // Spin through the whole expected wait (a batched op completes in
+        // 100-300 µs; usleep granularity is ~50 µs and would dominate), then
+        // back off for ms-scale waits; give up after the socket path's 60 s.
+        if (++spins < 50000) {
 #if defined(__x86_64__)
             __builtin_ia32_pause();
 #endif
         } else {
-            usleep(spins < 4000 ? 20 : 200);
+            usleep(100);
             if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(60)) {
                 ERROR("shm ring response timeout (seq %llu)",
                       static_cast<unsigned long long>(seq));
