@@ -750,11 +750,16 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     for (size_t i = 0; i < nb; i++) hashes[i] = KvMap::hash_of(msg.blocks[i].first);
     std::vector<uint32_t> fresh;
     fresh.reserve(nb);
-    {
+    // Chunked critical sections: concurrent readers (other conns' pollers /
+    // loop threads) interleave between chunks instead of waiting out the
+    // whole pass.
+    constexpr size_t kChunk = 256;
+    for (size_t c0 = 0; c0 < nb; c0 += kChunk) {
+        size_t c1 = std::min(nb, c0 + kChunk);
         std::lock_guard<std::mutex> lk(kv_mu_);
-        for (size_t i = 0; i < std::min(kPf, nb); i++) kv_.prefetch(hashes[i]);
-        for (size_t i = 0; i < nb; i++) {
-            if (i + kPf < nb) kv_.prefetch(hashes[i + kPf]);
+        for (size_t i = c0; i < std::min(c0 + kPf, c1); i++) kv_.prefetch(hashes[i]);
+        for (size_t i = c0; i < c1; i++) {
+            if (i + kPf < c1) kv_.prefetch(hashes[i + kPf]);
             if (!kv_.find_hashed(msg.blocks[i].first, hashes[i]))
                 fresh.push_back(static_cast<uint32_t>(i));
         }
@@ -863,11 +868,12 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     // Phase D — insert pass, overlapped with the in-flight kernel. Losers
     // (a racing writer inserted the key first) keep their block alive until
     // the copy completes, then release it; they are never committed.
-    {
+    for (size_t c0 = 0; c0 < n_fresh; c0 += kChunk) {
+        size_t c1 = std::min(n_fresh, c0 + kChunk);
         std::lock_guard<std::mutex> lk(kv_mu_);
-        for (size_t i = 0; i < std::min(kPf, n_fresh); i++) kv_.prefetch(hashes[fresh[i]]);
-        for (size_t i = 0; i < n_fresh; i++) {
-            if (i + kPf < n_fresh) kv_.prefetch(hashes[fresh[i + kPf]]);
+        for (size_t i = c0; i < std::min(c0 + kPf, c1); i++) kv_.prefetch(hashes[fresh[i]]);
+        for (size_t i = c0; i < c1; i++) {
+            if (i + kPf < c1) kv_.prefetch(hashes[fresh[i + kPf]]);
             bool inserted = false;
             kv_.emplace_hashed(msg.blocks[fresh[i]].first, hashes[fresh[i]], (*entries)[i],
                                &inserted);
@@ -905,23 +911,26 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         std::vector<uint64_t> hashes(nb2);
         for (size_t i = 0; i < nb2; i++) hashes[i] = KvMap::hash_of(msg.blocks[i].first);
         constexpr size_t kPf = 16;
-        std::lock_guard<std::mutex> lk(kv_mu_);
-        for (size_t i = 0; i < std::min(kPf, nb2); i++) kv_.prefetch(hashes[i]);
-        size_t bi = 0;
-        for (auto& b : msg.blocks) {
-            if (bi + kPf < nb2) kv_.prefetch(hashes[bi + kPf]);
-            Ref<BlockEntry>* v = kv_.find_hashed(b.first, hashes[bi]);
-            bi++;
-            if (!v || !(*v)->committed) {
-                return reply_local(c, ctx, KEY_NOT_FOUND);
+        constexpr size_t kChunk = 256;  // interleave with writers' passes
+        for (size_t c0 = 0; c0 < nb2; c0 += kChunk) {
+            size_t c1 = std::min(nb2, c0 + kChunk);
+            std::lock_guard<std::mutex> lk(kv_mu_);
+            for (size_t i = c0; i < std::min(c0 + kPf, c1); i++) kv_.prefetch(hashes[i]);
+            for (size_t bi = c0; bi < c1; bi++) {
+                auto& b = msg.blocks[bi];
+                if (bi + kPf < c1) kv_.prefetch(hashes[bi + kPf]);
+                Ref<BlockEntry>* v = kv_.find_hashed(b.first, hashes[bi]);
+                if (!v || !(*v)->committed) {
+                    return reply_local(c, ctx, KEY_NOT_FOUND);
+                }
+                BlockEntry* e = v->get();
+                e->last_access = read_tick;
+                auto& job = jobs[e->shard];
+                job.bytes_per_block = page;
+                job.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
+                job.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.second));
+                held->push_back(*v);
             }
-            BlockEntry* e = v->get();
-            e->last_access = read_tick;
-            auto& job = jobs[e->shard];
-            job.bytes_per_block = page;
-            job.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
-            job.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.second));
-            held->push_back(*v);
         }
     }
     n_reads_.fetch_add(1);
