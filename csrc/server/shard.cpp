@@ -21,7 +21,8 @@ Shard::~Shard() {
     }
     fabric_cv_.notify_all();
     if (fabric_thread_.joinable()) fabric_thread_.join();
-    if (completion_thread_.joinable()) completion_thread_.join();
+    for (auto& t : completion_threads_)
+        if (t.joinable()) t.join();
     for (auto& sc : streams_) {
         for (auto& sl : sc.slots) {
             if (sl.event) gpu::event_destroy(sl.event);
@@ -63,7 +64,12 @@ bool Shard::init() {
                 if (!sl.h_src || !sl.h_dst || !sl.d_src || !sl.d_dst || !sl.event) return false;
             }
         }
-        completion_thread_ = std::thread([this] { completion_loop(); });
+        // One completion thread PER STREAM: done-callbacks can block (the
+        // shm write commit waits for the insert pass) and a single thread
+        // serializing all streams would convoy an already-finished read's
+        // reply behind a blocked write commit.
+        for (size_t si = 0; si < streams_.size(); si++)
+            completion_threads_.emplace_back([this, si] { completion_loop(si); });
     } else {
         void* arena = nullptr;
         if (posix_memalign(&arena, 4096, opt_.pool_bytes) != 0) {
@@ -259,7 +265,7 @@ bool Shard::submit_copy(CopyJob&& job) {
             if (!ok) slot->busy = false;
             sc.pending.push_back({ok ? slot : nullptr, std::move(job.done)});
         }
-        task_cv_.notify_one();
+        task_cv_.notify_all();
         return true;
     }
 
@@ -292,7 +298,7 @@ bool Shard::submit_copy(CopyJob&& job) {
                 slot->busy = false;
                 sc.pending.push_back({nullptr, std::move(job.done)});
             }
-            task_cv_.notify_one();
+            task_cv_.notify_all();
             return true;
         }
         {
@@ -300,27 +306,25 @@ bool Shard::submit_copy(CopyJob&& job) {
             sc.pending.push_back(
                 {slot, last ? std::move(job.done) : std::function<void(bool)>()});
         }
-        task_cv_.notify_one();
+        task_cv_.notify_all();
         off += take;
     }
     return true;
 }
 
-void Shard::completion_loop() {
+void Shard::completion_loop(size_t stream_idx) {
     gpu::set_device(opt_.device);
-    // Per-stream FIFO queues, completed out of order ACROSS streams: one
-    // long kernel on stream A must not delay the response of a finished
-    // copy on stream B (cross-client latency coupling).
+    // One thread per stream, so streams complete fully independently: a
+    // done-callback that blocks (e.g. the shm write commit waiting for the
+    // insert pass) stalls only its own stream's FIFO.
+    auto& sc = streams_[stream_idx];
     for (;;) {
         PendingTask t{};
         bool have = false;
         {
             std::unique_lock<std::mutex> lk(task_mu_);
-            for (int pass = 0; !have; pass++) {
-                bool any_pending = false;
-                for (auto& sc : streams_) {
-                    if (sc.pending.empty()) continue;
-                    any_pending = true;
+            while (!have) {
+                if (!sc.pending.empty()) {
                     PendingTask& front = sc.pending.front();
                     if (!front.slot || gpu::event_query(front.slot->event)) {
                         t = std::move(front);
@@ -328,29 +332,17 @@ void Shard::completion_loop() {
                         have = true;
                         break;
                     }
-                }
-                if (have) break;
-                if (stopping_ && !any_pending) return;
-                if (!any_pending) {
-                    task_cv_.wait(lk, [&] {
-                        if (stopping_) return true;
-                        for (auto& sc : streams_)
-                            if (!sc.pending.empty()) return true;
-                        return false;
-                    });
-                    if (stopping_) {
-                        bool empty = true;
-                        for (auto& sc : streams_) empty &= sc.pending.empty();
-                        if (empty) return;
-                    }
-                } else {
-                    // Events pending but none complete yet: spin politely.
+                    // Event pending but not complete yet: spin politely.
                     lk.unlock();
 #if defined(__x86_64__)
                     for (int i = 0; i < 64; i++) __builtin_ia32_pause();
 #endif
                     lk.lock();
+                    continue;
                 }
+                if (stopping_) return;
+                task_cv_.wait(lk, [&] { return stopping_ || !sc.pending.empty(); });
+                if (stopping_ && sc.pending.empty()) return;
             }
         }
         bool ok = t.slot != nullptr;

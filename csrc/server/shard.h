@@ -123,7 +123,7 @@ class Shard {
     };
 
     Slot* acquire_slot(StreamCtx& sc);
-    void completion_loop();
+    void completion_loop(size_t stream_idx);
     void fabric_loop();
 
     ShardOptions opt_;
@@ -136,7 +136,7 @@ class Shard {
     std::mutex task_mu_;
     std::condition_variable task_cv_;
     std::condition_variable slot_cv_;
-    std::thread completion_thread_;
+    std::vector<std::thread> completion_threads_;  // one per stream
     std::atomic<bool> stopping_{false};
     bool inited_ = false;
 
