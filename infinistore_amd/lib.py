@@ -332,8 +332,15 @@ class InfinityConnection:
     # Same semantics as local_gpu_write_cache/read_cache but takes keys and
     # offsets as parallel sequences; `offsets` may be a reusable
     # np.ndarray(dtype=uint64) of ELEMENT offsets, skipping per-call tuple
-    # marshalling on the hot path.
-    def write_pages(self, cache: torch.Tensor, keys: List[str], offsets, page_size: int,
+    # marshalling on the hot path. `keys` may also be a pre-serialized bytes
+    # blob (the NUL-joined key list, e.g. via pack_keys) — engines that hold
+    # a page hash chain can serialize it once and skip the per-request join.
+    @staticmethod
+    def pack_keys(keys: List[str]) -> bytes:
+        """Serialize a key list for the write_pages/read_pages blob form."""
+        return "\0".join(keys).encode()
+
+    def write_pages(self, cache: torch.Tensor, keys, offsets, page_size: int,
                     sync: bool = False):
         """sync=True completes the write in ONE round trip (the response is
         sent when the copy finishes); sync=False (default) returns after the
@@ -342,23 +349,35 @@ class InfinityConnection:
         assert self.local_connected, "write_pages uses the local GPU path"
         es = cache.element_size()
         offs = np.asarray(offsets, dtype=np.uint64)
-        ret = self.conn.rw_local_keys(
-            self.OP_W, keys, offs, es, page_size * es,
-            cache.data_ptr(), _remap_device_id(cache), sync,
-        )
+        if isinstance(keys, (bytes, bytearray, memoryview)):
+            ret = self.conn.rw_local_blob(
+                self.OP_W, keys, offs, es, page_size * es,
+                cache.data_ptr(), _remap_device_id(cache), sync,
+            )
+        else:
+            ret = self.conn.rw_local_keys(
+                self.OP_W, keys, offs, es, page_size * es,
+                cache.data_ptr(), _remap_device_id(cache), sync,
+            )
         if ret < 0:
             raise Exception(f"Failed to write to infinistore, ret = {ret}")
         return 0
 
-    def read_pages(self, cache: torch.Tensor, keys: List[str], offsets, page_size: int):
+    def read_pages(self, cache: torch.Tensor, keys, offsets, page_size: int):
         self._verify(cache)
         es = cache.element_size()
         if self.local_connected:
             offs = np.asarray(offsets, dtype=np.uint64)
-            ret = self.conn.rw_local_keys(
-                self.OP_R, keys, offs, es, page_size * es,
-                cache.data_ptr(), _remap_device_id(cache),
-            )
+            if isinstance(keys, (bytes, bytearray, memoryview)):
+                ret = self.conn.rw_local_blob(
+                    self.OP_R, keys, offs, es, page_size * es,
+                    cache.data_ptr(), _remap_device_id(cache),
+                )
+            else:
+                ret = self.conn.rw_local_keys(
+                    self.OP_R, keys, offs, es, page_size * es,
+                    cache.data_ptr(), _remap_device_id(cache),
+                )
         elif self.rdma_connected:
             blocks = [(k, int(o) * es) for k, o in zip(keys, offsets)]
             ret = self.conn.r_rdma(blocks, page_size * es, cache.data_ptr())

@@ -369,10 +369,16 @@ def test_shm_transport_roundtrip(gpu_server):
         offs = [i * page // 4 for i in range(n_blocks)]
         for rep in range(8):  # > 8 MB of request records => ring wraps
             keys = [f"shm-{uuid.uuid4()}-{rep}-{i}" for i in range(n_blocks)]
-            conn.local_gpu_write_cache(src, list(zip(keys, offs)), page // 4)
-            conn.sync()
-            dst.zero_()
-            conn.read_cache(dst, list(zip(keys, offs)), page // 4)
+            if rep % 2:  # pre-serialized key-blob form
+                blob = ifs.InfinityConnection.pack_keys(keys)
+                conn.write_pages(src, blob, offs, page // 4, sync=True)
+                dst.zero_()
+                conn.read_pages(dst, blob, offs, page // 4)
+            else:
+                conn.local_gpu_write_cache(src, list(zip(keys, offs)), page // 4)
+                conn.sync()
+                dst.zero_()
+                conn.read_cache(dst, list(zip(keys, offs)), page // 4)
             conn.sync()
             assert torch.equal(src, dst)
     finally:
