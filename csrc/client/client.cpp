@@ -209,6 +209,7 @@ int ClientConn::shm_request(char op, const uint8_t* body, size_t n, bool want_re
     h.len = need;
     h.op = static_cast<uint8_t>(op);
     h.body_len = static_cast<uint32_t>(n);
+    h.t_push_us = shmring::mono_us();
     h.seq = seq;
     memcpy(dst, &h, sizeof(h));
     if (n) memcpy(dst + sizeof(h), body, n);

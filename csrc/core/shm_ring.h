@@ -51,9 +51,16 @@ struct RecHdr {
     uint8_t op;
     uint8_t _pad[3];
     uint32_t body_len;
-    uint32_t _pad2;
+    uint32_t t_push_us;  // producer CLOCK_MONOTONIC µs (mod 2^32): same clock
+                         // domain on one host, used for queue-delay tracing
     uint64_t seq;
 };
+
+inline uint32_t mono_us() {
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return static_cast<uint32_t>(ts.tv_sec * 1000000ull + ts.tv_nsec / 1000);
+}
 static_assert(sizeof(RecHdr) == 24, "RecHdr layout");
 
 inline uint32_t rec_len(size_t body) {

@@ -901,6 +901,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     uint8_t* client_ptr = static_cast<uint8_t*>(base) + msg.base_offset;
     size_t page = static_cast<size_t>(msg.block_size);
 
+    auto tr0 = std::chrono::steady_clock::now();
     // Group blocks by owning shard (keys may live on different GPUs).
     std::map<Shard*, Shard::CopyJob> jobs;
     auto held = std::make_shared<std::vector<Ref<BlockEntry>>>();
@@ -933,6 +934,8 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
             }
         }
     }
+    static const bool rdbg2 = getenv("IFS_SERVER_DEBUG") != nullptr;
+    auto tcol = std::chrono::steady_clock::now();
     n_reads_.fetch_add(1);
     bytes_out_.fetch_add(msg.blocks.size() * page);
     bool sync_resp = (msg.flags & kLocalFlagSyncResponse) != 0;
@@ -976,6 +979,14 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
                 return reply_local(c, ctx, INTERNAL_ERROR);
             }
         }
+    }
+    if (rdbg2 && msg.blocks.size() > 256) {
+        auto us = [](auto a, auto b) {
+            return std::chrono::duration<double, std::micro>(b - a).count();
+        };
+        auto tsub = std::chrono::steady_clock::now();
+        fprintf(stderr, "[rdbg] read n=%zu collect=%.0f submit=%.0f\n", msg.blocks.size(),
+                us(tr0, tcol), us(tcol, tsub));
     }
     if (!sync_resp && !ctx.shm) send_status(c, TASK_ACCEPTED);
 }
@@ -1109,6 +1120,10 @@ void Server::shm_poll_main(ShmPeer* p) {
         ReqCtx ctx{h.seq, true};
         auto t0 = std::chrono::steady_clock::now();
         char op = static_cast<char>(h.op);
+        static const bool rdbg = getenv("IFS_SERVER_DEBUG") != nullptr;
+        if (rdbg && h.body_len > 4096)
+            fprintf(stderr, "[rdbg] op=%c pickup_wait=%uus\n", op,
+                    shmring::mono_us() - h.t_push_us);
         if (h.body_len + sizeof(h) > len) {
             reply_local(c, ctx, INVALID_REQ);
         } else {
