@@ -643,7 +643,7 @@ void Server::erase_entries(const std::vector<Ref<BlockEntry>>& entries) {
     // Rare path (allocation failure / failed copy): remove entries from the
     // index by identity. O(map) scan, but keys are not kept around on the
     // hot path.
-    std::lock_guard<std::mutex> lk(kv_mu_);
+    std::lock_guard<std::shared_mutex> lk(kv_mu_);
     std::vector<std::string> victims;
     kv_.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
         for (auto& e : entries)
@@ -669,7 +669,7 @@ size_t Server::evict_lru_locked(Shard* shard, size_t bytes) {
         kv_.scan_from(&evict_hand_, window, [&](std::string_view key, Ref<BlockEntry>& val) {
             BlockEntry* e = val.get();
             if (e->shard == shard && e->committed && e->ref_count() == 1)
-                sample.push_back({e->last_access, key});
+                sample.push_back({e->last_access.load(std::memory_order_relaxed), key});
             return sample.size() < 128;
         });
         scanned += window;
@@ -756,7 +756,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     constexpr size_t kChunk = 256;
     for (size_t c0 = 0; c0 < nb; c0 += kChunk) {
         size_t c1 = std::min(nb, c0 + kChunk);
-        std::lock_guard<std::mutex> lk(kv_mu_);
+        std::shared_lock<std::shared_mutex> lk(kv_mu_);
         for (size_t i = c0; i < std::min(c0 + kPf, c1); i++) kv_.prefetch(hashes[i]);
         for (size_t i = c0; i < c1; i++) {
             if (i + kPf < c1) kv_.prefetch(hashes[i + kPf]);
@@ -785,7 +785,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     };
     bool alloc_ok = try_alloc();
     if (!alloc_ok && opt_.auto_evict) {
-        std::lock_guard<std::mutex> lk(kv_mu_);
+        std::lock_guard<std::shared_mutex> lk(kv_mu_);
         if (evict_lru_locked(shard, page * n_fresh * 2) > 0) alloc_ok = try_alloc();
     }
     if (!alloc_ok) return reply_local(c, ctx, OUT_OF_MEMORY);
@@ -808,7 +808,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         e->pool_idx = slots[i].second;
         e->shard = shard;
         e->committed = false;
-        e->last_access = t;
+        e->last_access.store(t, std::memory_order_relaxed);
         entries->emplace_back(e);
         job.src.push_back(
             reinterpret_cast<uint64_t>(client_ptr + msg.blocks[fresh[i]].second));
@@ -870,7 +870,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     // the copy completes, then release it; they are never committed.
     for (size_t c0 = 0; c0 < n_fresh; c0 += kChunk) {
         size_t c1 = std::min(n_fresh, c0 + kChunk);
-        std::lock_guard<std::mutex> lk(kv_mu_);
+        std::lock_guard<std::shared_mutex> lk(kv_mu_);
         for (size_t i = c0; i < std::min(c0 + kPf, c1); i++) kv_.prefetch(hashes[fresh[i]]);
         for (size_t i = c0; i < c1; i++) {
             if (i + kPf < c1) kv_.prefetch(hashes[fresh[i + kPf]]);
@@ -915,7 +915,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         constexpr size_t kChunk = 256;  // interleave with writers' passes
         for (size_t c0 = 0; c0 < nb2; c0 += kChunk) {
             size_t c1 = std::min(nb2, c0 + kChunk);
-            std::lock_guard<std::mutex> lk(kv_mu_);
+            std::shared_lock<std::shared_mutex> lk(kv_mu_);
             for (size_t i = c0; i < std::min(c0 + kPf, c1); i++) kv_.prefetch(hashes[i]);
             for (size_t bi = c0; bi < c1; bi++) {
                 auto& b = msg.blocks[bi];
@@ -925,7 +925,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
                     return reply_local(c, ctx, KEY_NOT_FOUND);
                 }
                 BlockEntry* e = v->get();
-                e->last_access = read_tick;
+                e->last_access.store(read_tick, std::memory_order_relaxed);
                 auto& job = jobs[e->shard];
                 job.bytes_per_block = page;
                 job.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
@@ -1191,7 +1191,7 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
     blocks.reserve(keys.size());
     std::vector<std::string> created;
     {
-        std::lock_guard<std::mutex> lk(kv_mu_);
+        std::lock_guard<std::shared_mutex> lk(kv_mu_);
         for (auto& key : keys) {
             if (kv_.find(key) != nullptr) {
                 blocks.push_back({0, 0, 0});  // FAKE block: dup key, client skips
@@ -1217,7 +1217,7 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
             e->size = page;
             e->pool_idx = pool_idx;
             e->shard = shard;
-            e->last_access = tick();
+            e->last_access.store(tick(), std::memory_order_relaxed);
             Ref<BlockEntry> ref(e);
             bool ins = false;
             kv_.emplace(key, ref, &ins);
@@ -1242,13 +1242,13 @@ void Server::commit_addrs(Conn* c, const std::vector<uint64_t>& addrs) {
 
 bool Server::collect_read_entries(const std::vector<std::string>& keys,
                                   std::vector<Ref<BlockEntry>>* out) {
-    std::lock_guard<std::mutex> lk(kv_mu_);
+    std::shared_lock<std::shared_mutex> lk(kv_mu_);
     out->reserve(keys.size());
     uint64_t t = tick();
     for (auto& key : keys) {
         Ref<BlockEntry>* v = kv_.find(key);
         if (!v || !(*v)->committed) return false;
-        (*v)->last_access = t;
+        (*v)->last_access.store(t, std::memory_order_relaxed);
         out->push_back(*v);
     }
     return true;
@@ -1395,7 +1395,7 @@ void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body) {
     std::string key(reinterpret_cast<const char*>(body.data() + 4), len);
     bool exists;
     {
-        std::lock_guard<std::mutex> lk(kv_mu_);
+        std::shared_lock<std::shared_mutex> lk(kv_mu_);
         Ref<BlockEntry>* v = kv_.find(key);
         exists = v && (*v)->committed;
     }
@@ -1411,7 +1411,7 @@ void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body) {
     // divergence from the reference, which counts uncommitted keys as
     // present (infinistore.cpp:1097); an uncommitted key cannot be read, so
     // reporting it as a hit would make the subsequent read_cache fail.
-    std::lock_guard<std::mutex> lk(kv_mu_);
+    std::shared_lock<std::shared_mutex> lk(kv_mu_);
     auto present = [&](size_t i) {
         Ref<BlockEntry>* v = kv_.find(keys[i]);
         return v && (*v)->committed;
@@ -1433,7 +1433,7 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
         return send_status(c, INVALID_REQ);
     int n = 0;
     {
-        std::lock_guard<std::mutex> lk(kv_mu_);
+        std::lock_guard<std::shared_mutex> lk(kv_mu_);
         for (auto& k : keys) n += kv_.erase(k) ? 1 : 0;
     }
     send_status(c, n);
@@ -1448,7 +1448,7 @@ std::pair<size_t, size_t> Server::compact() {
     // request thread can start a read against a block while it moves
     // (requests block on the mutex for the few ms this takes).
     size_t moved = 0, bytes = 0;
-    std::lock_guard<std::mutex> lk(kv_mu_);
+    std::lock_guard<std::shared_mutex> lk(kv_mu_);
     for (auto& shard_up : shards_) {
         Shard* shard = shard_up.get();
         std::vector<std::pair<void*, size_t>> movable;
@@ -1497,12 +1497,12 @@ std::pair<size_t, size_t> Server::compact() {
 }
 
 size_t Server::kvmap_len() {
-    std::lock_guard<std::mutex> lk(kv_mu_);
+    std::shared_lock<std::shared_mutex> lk(kv_mu_);
     return kv_.size();
 }
 
 size_t Server::purge() {
-    std::lock_guard<std::mutex> lk(kv_mu_);
+    std::lock_guard<std::shared_mutex> lk(kv_mu_);
     size_t n = kv_.size();
     kv_.clear();
     return n;

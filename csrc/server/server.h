@@ -30,6 +30,7 @@
 #include <map>
 #include <memory>
 #include <mutex>
+#include <shared_mutex>
 #include <string>
 #include <string_view>
 #include <thread>
@@ -84,7 +85,9 @@ struct BlockEntry : RefCounted {
     int pool_idx = -1;
     Shard* shard = nullptr;
     bool committed = false;
-    uint64_t last_access = 0;  // LRU tick (auto_evict)
+    // LRU tick (auto_evict); atomic: bumped under the SHARED kv lock by
+    // concurrent readers.
+    std::atomic<uint64_t> last_access{0};
     ~BlockEntry() override {
         if (shard && ptr) shard->deallocate(ptr, size, pool_idx);
     }
@@ -256,7 +259,11 @@ class Server {
 
     // Open-addressing key index with arena-stored keys (csrc/server/kvmap.h)
     // — node-based maps measured ~350 µs of insert cost per 2048-key write.
-    std::mutex kv_mu_;
+    // Reader-writer lock: lookups (read collect, dedup pre-check, queries)
+    // take it shared and run concurrently; inserts/erases/rehash take it
+    // exclusive. Measured: with 4 conns active, exclusive-only kv_mu_ cost
+    // reads 60-450 µs of lock wait per 1024-key collect.
+    std::shared_mutex kv_mu_;
     KvMap kv_;
 
     // stats
