@@ -254,22 +254,37 @@ def main():
     get_time = 0.0
     if pipeline:
         def run_conn(c):
+            # Two-deep pipeline per pair: writes lead two steps, reads one.
+            # Invariant before pushing R(k): W(k) committed (wc_.sync() at
+            # the end of iteration k-1 drains the W(k+1) pushed there, and
+            # the prime below covers W(0)/W(1)). While the client waits on
+            # R(s)'s ticket, W(s+2) and R(s+1) are already queued server-
+            # side, so the GPU always has copy work in flight.
             rc_, wc_, o = conns[c], wconns[c], coff[c]
+            K = args.steps
             pt = gt = 0.0
             t = time.perf_counter()
             wc_.write_pages(src, put_blobs[0][c], o, elems_per_block, sync=True)
+            if K > 1:
+                wc_.write_pages(src, put_blobs[1][c], o, elems_per_block,
+                                sync=False)
+                wc_.sync()
             pt += time.perf_counter() - t
-            for s in range(args.steps):
+            tk = rc_.read_pages_async(dst, get_blobs[0][c], o, elems_per_block)
+            for s in range(K):
                 t = time.perf_counter()
-                if s + 1 < args.steps:
-                    wc_.write_pages(src, put_blobs[s + 1][c], o,
+                if s + 2 < K:
+                    wc_.write_pages(src, put_blobs[s + 2][c], o,
                                     elems_per_block, sync=False)
+                tk_next = (rc_.read_pages_async(dst, get_blobs[s + 1][c], o,
+                                                elems_per_block)
+                           if s + 1 < K else None)
                 tg = time.perf_counter()
-                rc_.read_pages(dst, get_blobs[s][c], o, elems_per_block)
-                rc_.sync()
-                wc_.sync()  # drain put(s+1)'s commit before the next read
+                rc_.wait_read(tk)
+                wc_.sync()  # commits W(s+2) before R(s+2) is pushed next iter
                 pt += tg - t
                 gt += time.perf_counter() - tg
+                tk = tk_next
             return pt, gt
 
         if n_conns == 1:

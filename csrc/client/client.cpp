@@ -142,6 +142,7 @@ void ClientConn::shm_drain_responses() {
         if (len >= sizeof(shmring::RespRec)) {
             shmring::RespRec r;
             memcpy(&r, rec, sizeof(r));
+            if (shm_results_.size() < 4096) shm_results_[r.h.seq] = r.status;
             if (r.status != 0 && r.status != TASK_ACCEPTED && r.status != FINISH &&
                 shm_async_err_ == 0)
                 shm_async_err_ = r.status;
@@ -151,6 +152,12 @@ void ClientConn::shm_drain_responses() {
 }
 
 int ClientConn::shm_wait(uint64_t seq) {
+    auto it0 = shm_results_.find(seq);
+    if (it0 != shm_results_.end()) {  // already drained while waiting elsewhere
+        int st = it0->second;
+        shm_results_.erase(it0);
+        return st;
+    }
     auto t0 = std::chrono::steady_clock::now();
     int spins = 0;
     for (;;) {
@@ -166,6 +173,9 @@ int ClientConn::shm_wait(uint64_t seq) {
             memcpy(&r, rec, std::min(sizeof(r), size_t(len)));
             shm_.resp->consume(skip);
             if (r.h.seq == seq) return r.status;
+            // Out-of-order response: stash for a later wait (bounded — a
+            // runaway map means tickets are being dropped by the caller).
+            if (shm_results_.size() < 4096) shm_results_[r.h.seq] = r.status;
             if (r.status != 0 && r.status != TASK_ACCEPTED && r.status != FINISH &&
                 shm_async_err_ == 0)
                 shm_async_err_ = r.status;
@@ -189,7 +199,8 @@ int ClientConn::shm_wait(uint64_t seq) {
     }
 }
 
-int ClientConn::shm_request(char op, const uint8_t* body, size_t n, bool want_resp) {
+int ClientConn::shm_request(char op, const uint8_t* body, size_t n, bool want_resp,
+                            uint64_t* out_ticket) {
     uint32_t need = shmring::rec_len(n);
     if (need > shm_.req->cap / 2) return kShmNoFit;
     shm_drain_responses();
@@ -218,9 +229,23 @@ int ClientConn::shm_request(char op, const uint8_t* body, size_t n, bool want_re
         shm_unacked_++;
         return 0;
     }
+    if (out_ticket) {  // ticketed: the caller waits later (wait_local_ticket)
+        *out_ticket = seq;
+        shm_unacked_++;
+        return 0;
+    }
     int code = shm_wait(seq);
     shm_unacked_ = 0;  // a sync-response op drains the ring ordering-wise
     return code;
+}
+
+int ClientConn::wait_local_ticket(uint64_t ticket) {
+    if (ticket == 0) return 0;
+    std::lock_guard<std::mutex> lk(io_mu_);
+    if (!shm_active_) return -1;
+    int code = shm_wait(ticket);
+    if (code != TASK_ACCEPTED && code != FINISH && code != 0) return code < 0 ? code : -code;
+    return 0;
 }
 
 int ClientConn::shm_ring_sync() { return shm_request(OP_SYNC, nullptr, 0, /*want_resp=*/true); }
@@ -297,7 +322,9 @@ int ClientConn::rw_local(char op, const std::vector<std::pair<std::string, uint6
 
 int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
                                 const uint64_t* offsets, size_t n, int block_size,
-                                uintptr_t ptr, int device_id, bool sync_response) {
+                                uintptr_t ptr, int device_id, bool sync_response,
+                                uint64_t* out_ticket) {
+    if (out_ticket) *out_ticket = 0;  // 0 = completed synchronously
     if (!connected_) return -1;
     if (!gpu::available()) {
         ERROR("local path requires a GPU");
@@ -349,8 +376,10 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
     std::lock_guard<std::mutex> lk(io_mu_);
     if (shm_active_) {
         bool want_resp = (h.flags & kLocalFlagSyncResponse) != 0;
-        int code = shm_request(wire_op, body.data(), body.size(), want_resp);
+        int code = shm_request(wire_op, body.data(), body.size(), want_resp,
+                               want_resp ? out_ticket : nullptr);
         if (code != kShmNoFit) {
+            if (out_ticket && *out_ticket) return 0;  // ticketed: wait later
             if (dbg && n > 64) {
                 auto t2 = std::chrono::steady_clock::now();
                 fprintf(stderr, "[cdbg] op=%c n=%zu shm total=%.0fus\n", op, n,

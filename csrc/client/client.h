@@ -64,7 +64,10 @@ class ClientConn {
     // straight into the packed wire format (OP_W_FAST / OP_R_FAST).
     int rw_local_packed(char op, const char* keys_blob, size_t blob_len,
                         const uint64_t* offsets, size_t n, int block_size, uintptr_t ptr,
-                        int device_id, bool sync_response = false);
+                        int device_id, bool sync_response = false,
+                        uint64_t* out_ticket = nullptr);
+    // Wait for a ticketed (async shm) op; ticket 0 = already complete.
+    int wait_local_ticket(uint64_t ticket);
     int sync_local();
 
     // ---- RDMA-semantics path ----
@@ -123,7 +126,8 @@ class ClientConn {
     // (mapped like the socket path). Returns kShmNoFit when the record does
     // not fit the ring (caller falls back to the socket after a ring sync).
     static constexpr int kShmNoFit = INT32_MIN;
-    int shm_request(char op, const uint8_t* body, size_t n, bool want_resp);
+    int shm_request(char op, const uint8_t* body, size_t n, bool want_resp,
+                    uint64_t* out_ticket = nullptr);
     int shm_wait(uint64_t seq);           // drain responses until seq; io_mu_
     int shm_ring_sync();                  // OP_SYNC over the ring; io_mu_
     void shm_drain_responses();           // non-blocking pop; io_mu_
@@ -132,6 +136,7 @@ class ClientConn {
     uint64_t shm_seq_ = 0;
     uint64_t shm_unacked_ = 0;  // async ring writes since last ring sync
     int shm_async_err_ = 0;     // first error for an unawaited seq (io_mu_)
+    std::unordered_map<uint64_t, int> shm_results_;  // ticketed responses (io_mu_)
 
     int fd_ = -1;
     bool connected_ = false;

@@ -252,6 +252,37 @@ PYBIND11_MODULE(_native, m) {
             py::arg("op"), py::arg("blob"), py::arg("offsets"), py::arg("element_size"),
             py::arg("block_size"), py::arg("ptr"), py::arg("device"),
             py::arg("sync_response") = false)
+        .def(
+            "rw_local_blob_async",
+            [](ClientConn& c, const std::string& op, py::buffer blob, py::buffer offsets,
+               uint64_t element_size, int block_size, uintptr_t ptr, int device) {
+                // Ticketed form: push the (sync-response) op and return a
+                // ticket to pass to wait_local_ticket — the engine overlaps
+                // its own work with the copy. Ticket 0 = already complete
+                // (op fell back to the blocking socket path).
+                py::buffer_info bb = blob.request();
+                py::buffer_info ob = offsets.request();
+                if (ob.itemsize != 8) throw std::runtime_error("offsets must be uint64[n]");
+                size_t n = static_cast<size_t>(ob.size);
+                const uint64_t* offs = static_cast<const uint64_t*>(ob.ptr);
+                std::vector<uint64_t> byte_offs(n);
+                for (size_t i = 0; i < n; i++) byte_offs[i] = offs[i] * element_size;
+                const char* bp = static_cast<const char*>(bb.ptr);
+                size_t blen = static_cast<size_t>(bb.size) * bb.itemsize;
+                uint64_t ticket = 0;
+                int ret;
+                {
+                    py::gil_scoped_release rel;
+                    ret = c.rw_local_packed(op.empty() ? 'R' : op[0], bp, blen,
+                                            byte_offs.data(), n, block_size, ptr, device,
+                                            /*sync_response=*/true, &ticket);
+                }
+                return py::make_tuple(ret, ticket);
+            },
+            py::arg("op"), py::arg("blob"), py::arg("offsets"), py::arg("element_size"),
+            py::arg("block_size"), py::arg("ptr"), py::arg("device"))
+        .def("wait_local_ticket", &ClientConn::wait_local_ticket,
+             py::call_guard<py::gil_scoped_release>())
         .def("sync_local", &ClientConn::sync_local, py::call_guard<py::gil_scoped_release>())
         .def("register_mr", &ClientConn::register_mr, py::call_guard<py::gil_scoped_release>())
         .def(

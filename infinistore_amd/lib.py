@@ -386,6 +386,31 @@ class InfinityConnection:
         if ret < 0:
             raise Exception(f"Failed to read from infinistore, ret = {ret}")
 
+    def read_pages_async(self, cache: torch.Tensor, keys, offsets, page_size: int):
+        """Ticketed read (local path, shm ring): pushes the request and
+        returns a ticket; call wait_read(ticket) before using the data. Lets
+        an engine overlap its own work (or further requests) with the copy —
+        e.g. prefetching the next sequence's KV pages while decoding. Falls
+        back to a blocking read (ticket 0) when the ring is unavailable."""
+        self._verify(cache)
+        assert self.local_connected, "read_pages_async uses the local GPU path"
+        es = cache.element_size()
+        offs = np.asarray(offsets, dtype=np.uint64)
+        if not isinstance(keys, (bytes, bytearray, memoryview)):
+            keys = self.pack_keys(keys)
+        ret, ticket = self.conn.rw_local_blob_async(
+            self.OP_R, keys, offs, es, page_size * es,
+            cache.data_ptr(), _remap_device_id(cache),
+        )
+        if ret < 0:
+            raise Exception(f"Failed to read from infinistore, ret = {ret}")
+        return ticket
+
+    def wait_read(self, ticket: int):
+        ret = self.conn.wait_local_ticket(ticket)
+        if ret < 0:
+            raise Exception(f"Async read failed, ret = {ret}")
+
     def read_cache(self, cache: torch.Tensor, blocks: List[Tuple[str, int]], page_size: int):
         """Read pages into `cache` (offsets in elements)."""
         self._verify(cache)
