@@ -840,10 +840,18 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
                 for (size_t i = 0; i < entries->size(); i++)
                     if ((*won)[i]) (*entries)[i]->committed = true;
             } else {
-                std::vector<Ref<BlockEntry>> winners;
-                for (size_t i = 0; i < entries->size(); i++)
-                    if ((*won)[i]) winners.push_back((*entries)[i]);
-                erase_entries(winners);  // copy failed: drop the keys
+                // Copy failed: drop the keys. Always from the owner loop —
+                // erase_entries takes the kv lock exclusively, and compact()
+                // waits on completion-thread futures WHILE holding that
+                // lock, so taking it on a completion thread could deadlock.
+                c->ref();
+                c->owner->post([this, c, entries, won] {
+                    std::vector<Ref<BlockEntry>> winners;
+                    for (size_t i = 0; i < entries->size(); i++)
+                        if ((*won)[i]) winners.push_back((*entries)[i]);
+                    erase_entries(winners);
+                    c->unref();
+                });
             }
             if (sync_resp) reply_local(c, ctx, ok ? FINISH : INTERNAL_ERROR);
             finish_task(c, /*on_owner=*/!ctx.shm);
