@@ -1071,7 +1071,7 @@ void Server::op_shm_setup(Conn* c, const std::vector<uint8_t>& body) {
     // set up, name is implausible, or too many pollers are running.
     if (c->shm || body.empty() || body.size() > 100 || body[0] != '/')
         return send_status(c, INVALID_REQ);
-    if (shm_peers_.load() >= 32) return send_status(c, SYSTEM_ERROR);
+    if (shm_peers_.load() >= 64) return send_status(c, SYSTEM_ERROR);
     std::string name(reinterpret_cast<const char*>(body.data()), body.size());
     auto* p = new ShmPeer();
     p->srv = this;

@@ -15,6 +15,7 @@ Examples:
 import argparse
 import json
 import statistics
+import sys
 import time
 import uuid
 
@@ -108,6 +109,14 @@ def run_once(args, conn, local, device):
     if args.verify:
         assert torch.equal(src.cpu(), dst.cpu()), "verification failed"
 
+    # Steady-state footprint: drop this iteration's keys (each iteration
+    # writes a fresh key set; an engine similarly evicts finished
+    # sequences). Runs after the timed windows but inside the wall clock.
+    try:
+        conn.delete_keys(keys)
+    except Exception:
+        pass
+
     return total_bytes / w_time / 1e6, total_bytes / r_time / 1e6
 
 
@@ -150,8 +159,9 @@ def _worker(args_dict, q, barrier, wid):
     local = ns.local_gpu and torch.cuda.is_available()
     n_dev = torch.cuda.device_count() if torch.cuda.is_available() else 1
     device = f"cuda:{(ns.src_gpu + wid) % n_dev}" if local else "cpu"
-    conn = make_conn(ns, local)
+    conn = None
     try:
+        conn = make_conn(ns, local)
         run_once(ns, conn, local, device)  # warm: IPC opens, allocator, caches
         barrier.wait(timeout=300)  # start all clients together (steady state)
         t0 = time.perf_counter()
@@ -161,8 +171,11 @@ def _worker(args_dict, q, barrier, wid):
             w, r = run_once(ns, conn, local, device)
         wall = time.perf_counter() - t0
         q.put((w, r, wall, iters))
+    except Exception as e:  # report instead of leaving the parent hanging
+        q.put(("error", f"worker {wid}: {e}", 0.0, 0))
     finally:
-        conn.close()
+        if conn is not None:
+            conn.close()
 
 
 def _spawn_server(args):
@@ -211,6 +224,11 @@ def main():
     args = parse_args()
     if args.shape:
         apply_shape(args)
+    if args.spawn_server and args.clients > 1:
+        # Keys live until each iteration's trailing delete; size the pool
+        # for all clients' live iterations plus slack.
+        need = args.clients * args.size * 2 * 1.3 / 1024 + 1
+        args.prealloc_size = max(args.prealloc_size, int(need))
     server_proc = _spawn_server(args) if args.spawn_server else None
     local = args.local_gpu and torch.cuda.is_available()
     device = f"cuda:{args.src_gpu}" if local else "cpu"
@@ -229,7 +247,16 @@ def main():
             p.start()
         results = [q.get(timeout=600) for _ in procs]
         for p in procs:
-            p.join()
+            p.join(timeout=60)
+            if p.is_alive():
+                p.terminate()
+        errors = [r for r in results if r[0] == "error"]
+        if errors:
+            for e in errors:
+                print(e[1], file=sys.stderr)
+            if server_proc:
+                server_proc.terminate()
+            raise SystemExit(1)
         wall = max(r[2] for r in results)  # steady-state window (post-barrier)
         iters = results[0][3]
         agg = args.clients * (args.size << 20) * 2 * iters / wall / 1e6
