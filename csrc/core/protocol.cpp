@@ -19,6 +19,7 @@ std::string op_name(char op) {
         case OP_STATS: return "stats";
         case OP_W_FAST: return "local_write_fast";
         case OP_R_FAST: return "local_read_fast";
+        case OP_SHM_SETUP: return "shm_setup";
         default: return "unknown";
     }
 }
