@@ -245,7 +245,7 @@ bool Server::start() {
     // tombstone-compaction paths are exercised by small workloads.
     size_t kv_initial = 1u << 20;
     if (const char* env = getenv("IFS_KV_INITIAL")) kv_initial = strtoull(env, nullptr, 10);
-    kv_.reserve(kv_initial);
+    for (auto& st : kv_) st.map.reserve(kv_initial / kStripes);
     running_.store(true);
     stop_requested_.store(false);
     main_io_.start();
@@ -642,47 +642,63 @@ Shard* Server::shard_least_used() {
 void Server::erase_entries(const std::vector<Ref<BlockEntry>>& entries) {
     // Rare path (allocation failure / failed copy): remove entries from the
     // index by identity. O(map) scan, but keys are not kept around on the
-    // hot path.
-    std::lock_guard<std::shared_mutex> lk(kv_mu_);
-    std::vector<std::string> victims;
-    kv_.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
-        for (auto& e : entries)
-            if (val.get() == e.get()) victims.emplace_back(key);
-    });
-    for (auto& k : victims) kv_.erase(k);
+    // hot path. One stripe at a time.
+    for (auto& st : kv_) {
+        std::lock_guard<std::shared_mutex> lk(st.mu);
+        std::vector<std::string> victims;
+        st.map.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
+            for (auto& e : entries)
+                if (val.get() == e.get()) victims.emplace_back(key);
+        });
+        for (auto& k : victims) st.map.erase(k);
+    }
 }
 
-size_t Server::evict_lru_locked(Shard* shard, size_t bytes) {
+size_t Server::evict_lru(Shard* shard, size_t bytes) {
     // Sampled clock-hand eviction: scan bounded slot windows from a
-    // persistent cursor, evict the least-recently-accessed half of each
-    // sample — O(evicted) amortized instead of a full index scan per
-    // eviction event (which capped sustained full-pool churn).
+    // persistent per-stripe cursor, evict the least-recently-accessed half
+    // of each sample — O(evicted) amortized instead of a full index scan
+    // per eviction event (which capped sustained full-pool churn).
     // Candidates: committed, idle (only the map holds a ref), on this shard.
     // Key views stay valid across erase (the arena is append-only).
+    // Stripes are visited round-robin (shared cursor) holding one stripe
+    // lock at a time.
     size_t freed = 0;
-    size_t scanned = 0;
-    const size_t scan_cap = kv_.capacity();  // at most one full revolution
-    while (freed < bytes && scanned < scan_cap) {
+    for (size_t visited = 0; freed < bytes && visited < kStripes; visited++) {
+        auto& st = kv_[evict_stripe_rr_.fetch_add(1) % kStripes];
+        std::lock_guard<std::shared_mutex> lk(st.mu);
+        size_t scanned = 0;
+        const size_t scan_cap = st.map.capacity();  // one revolution max
         std::vector<std::pair<uint64_t, std::string_view>> sample;
         sample.reserve(128);
-        size_t window = 4096;
-        kv_.scan_from(&evict_hand_, window, [&](std::string_view key, Ref<BlockEntry>& val) {
-            BlockEntry* e = val.get();
-            if (e->shard == shard && e->committed && e->ref_count() == 1)
-                sample.push_back({e->last_access.load(std::memory_order_relaxed), key});
-            return sample.size() < 128;
-        });
-        scanned += window;
-        if (sample.empty()) continue;
-        std::sort(sample.begin(), sample.end(),
-                  [](const auto& a, const auto& b) { return a.first < b.first; });
-        size_t take = std::max<size_t>(1, sample.size() / 2);
-        for (size_t i = 0; i < take && freed < bytes; i++) {
-            Ref<BlockEntry>* v = kv_.find(sample[i].second);
-            if (!v) continue;
-            freed += (*v)->size;
-            kv_.erase(sample[i].second);
-            n_evicted_.fetch_add(1);
+        while (freed < bytes && scanned < scan_cap) {
+            size_t window = 4096;
+            st.map.scan_from(&st.evict_hand, window,
+                             [&](std::string_view key, Ref<BlockEntry>& val) {
+                                 BlockEntry* e = val.get();
+                                 if (e->shard == shard && e->committed && e->ref_count() == 1)
+                                     sample.push_back(
+                                         {e->last_access.load(std::memory_order_relaxed), key});
+                                 return sample.size() < 128;
+                             });
+            scanned += window;
+            // In a sparse table one window yields few candidates; evicting
+            // from a 1-2 entry "sample" degrades LRU to "next key after the
+            // cursor". Accumulate across windows until the sample is big
+            // enough to rank (or the stripe is fully scanned).
+            if (sample.size() < 128 && scanned < scan_cap) continue;
+            if (sample.empty()) break;
+            std::sort(sample.begin(), sample.end(),
+                      [](const auto& a, const auto& b) { return a.first < b.first; });
+            size_t take = std::max<size_t>(1, sample.size() / 2);
+            for (size_t i = 0; i < take && freed < bytes; i++) {
+                Ref<BlockEntry>* v = st.map.find(sample[i].second);
+                if (!v) continue;
+                freed += (*v)->size;
+                st.map.erase(sample[i].second);
+                n_evicted_.fetch_add(1);
+            }
+            sample.clear();
         }
     }
     if (freed) DEBUG("auto-evicted %zu bytes from shard dev=%d", freed, shard->device());
@@ -750,18 +766,23 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     for (size_t i = 0; i < nb; i++) hashes[i] = KvMap::hash_of(msg.blocks[i].first);
     std::vector<uint32_t> fresh;
     fresh.reserve(nb);
-    // Chunked critical sections: concurrent readers (other conns' pollers /
-    // loop threads) interleave between chunks instead of waiting out the
-    // whole pass.
-    constexpr size_t kChunk = 256;
-    for (size_t c0 = 0; c0 < nb; c0 += kChunk) {
-        size_t c1 = std::min(nb, c0 + kChunk);
-        std::shared_lock<std::shared_mutex> lk(kv_mu_);
-        for (size_t i = c0; i < std::min(c0 + kPf, c1); i++) kv_.prefetch(hashes[i]);
-        for (size_t i = c0; i < c1; i++) {
-            if (i + kPf < c1) kv_.prefetch(hashes[i + kPf]);
-            if (!kv_.find_hashed(msg.blocks[i].first, hashes[i]))
-                fresh.push_back(static_cast<uint32_t>(i));
+    // Bucket keys by stripe once; each stripe's lookups run under that
+    // stripe's SHARED lock (~nb/16 keys per hold), so concurrent requests
+    // on other stripes never wait.
+    std::array<std::vector<uint32_t>, kStripes> by_stripe;
+    for (size_t i = 0; i < nb; i++)
+        by_stripe[stripe_of(hashes[i])].push_back(static_cast<uint32_t>(i));
+    for (size_t si = 0; si < kStripes; si++) {
+        auto& list = by_stripe[si];
+        if (list.empty()) continue;
+        std::shared_lock<std::shared_mutex> lk(kv_[si].mu);
+        auto& m = kv_[si].map;
+        size_t ln = list.size();
+        for (size_t i = 0; i < std::min(kPf, ln); i++) m.prefetch(hashes[list[i]]);
+        for (size_t i = 0; i < ln; i++) {
+            if (i + kPf < ln) m.prefetch(hashes[list[i + kPf]]);
+            uint32_t gi = list[i];
+            if (!m.find_hashed(msg.blocks[gi].first, hashes[gi])) fresh.push_back(gi);
         }
     }
     auto p1 = std::chrono::steady_clock::now();
@@ -785,8 +806,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     };
     bool alloc_ok = try_alloc();
     if (!alloc_ok && opt_.auto_evict) {
-        std::lock_guard<std::shared_mutex> lk(kv_mu_);
-        if (evict_lru_locked(shard, page * n_fresh * 2) > 0) alloc_ok = try_alloc();
+        if (evict_lru(shard, page * n_fresh * 2) > 0) alloc_ok = try_alloc();
     }
     if (!alloc_ok) return reply_local(c, ctx, OUT_OF_MEMORY);
     auto p2 = std::chrono::steady_clock::now();
@@ -876,16 +896,25 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     // Phase D — insert pass, overlapped with the in-flight kernel. Losers
     // (a racing writer inserted the key first) keep their block alive until
     // the copy completes, then release it; they are never committed.
-    for (size_t c0 = 0; c0 < n_fresh; c0 += kChunk) {
-        size_t c1 = std::min(n_fresh, c0 + kChunk);
-        std::lock_guard<std::shared_mutex> lk(kv_mu_);
-        for (size_t i = c0; i < std::min(c0 + kPf, c1); i++) kv_.prefetch(hashes[fresh[i]]);
-        for (size_t i = c0; i < c1; i++) {
-            if (i + kPf < c1) kv_.prefetch(hashes[fresh[i + kPf]]);
-            bool inserted = false;
-            kv_.emplace_hashed(msg.blocks[fresh[i]].first, hashes[fresh[i]], (*entries)[i],
-                               &inserted);
-            (*won)[i] = inserted ? 1 : 0;
+    {
+        std::array<std::vector<uint32_t>, kStripes> ins_by_stripe;  // pos in fresh
+        for (uint32_t p = 0; p < n_fresh; p++)
+            ins_by_stripe[stripe_of(hashes[fresh[p]])].push_back(p);
+        for (size_t si = 0; si < kStripes; si++) {
+            auto& list = ins_by_stripe[si];
+            if (list.empty()) continue;
+            std::lock_guard<std::shared_mutex> lk(kv_[si].mu);
+            auto& m = kv_[si].map;
+            size_t ln = list.size();
+            for (size_t i = 0; i < std::min(kPf, ln); i++) m.prefetch(hashes[fresh[list[i]]]);
+            for (size_t i = 0; i < ln; i++) {
+                if (i + kPf < ln) m.prefetch(hashes[fresh[list[i + kPf]]]);
+                uint32_t p = list[i];
+                bool inserted = false;
+                m.emplace_hashed(msg.blocks[fresh[p]].first, hashes[fresh[p]], (*entries)[p],
+                                 &inserted);
+                (*won)[p] = inserted ? 1 : 0;
+            }
         }
     }
     commit_lk.unlock();  // phase D done: completion may commit
@@ -920,15 +949,20 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         std::vector<uint64_t> hashes(nb2);
         for (size_t i = 0; i < nb2; i++) hashes[i] = KvMap::hash_of(msg.blocks[i].first);
         constexpr size_t kPf = 16;
-        constexpr size_t kChunk = 256;  // interleave with writers' passes
-        for (size_t c0 = 0; c0 < nb2; c0 += kChunk) {
-            size_t c1 = std::min(nb2, c0 + kChunk);
-            std::shared_lock<std::shared_mutex> lk(kv_mu_);
-            for (size_t i = c0; i < std::min(c0 + kPf, c1); i++) kv_.prefetch(hashes[i]);
-            for (size_t bi = c0; bi < c1; bi++) {
-                auto& b = msg.blocks[bi];
-                if (bi + kPf < c1) kv_.prefetch(hashes[bi + kPf]);
-                Ref<BlockEntry>* v = kv_.find_hashed(b.first, hashes[bi]);
+        std::array<std::vector<uint32_t>, kStripes> by_stripe;
+        for (size_t i = 0; i < nb2; i++)
+            by_stripe[stripe_of(hashes[i])].push_back(static_cast<uint32_t>(i));
+        for (size_t si = 0; si < kStripes; si++) {
+            auto& list = by_stripe[si];
+            if (list.empty()) continue;
+            std::shared_lock<std::shared_mutex> lk(kv_[si].mu);
+            auto& m = kv_[si].map;
+            size_t ln = list.size();
+            for (size_t i = 0; i < std::min(kPf, ln); i++) m.prefetch(hashes[list[i]]);
+            for (size_t i = 0; i < ln; i++) {
+                if (i + kPf < ln) m.prefetch(hashes[list[i + kPf]]);
+                auto& b = msg.blocks[list[i]];
+                Ref<BlockEntry>* v = m.find_hashed(b.first, hashes[list[i]]);
                 if (!v || !(*v)->committed) {
                     return reply_local(c, ctx, KEY_NOT_FOUND);
                 }
@@ -1197,43 +1231,59 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
     Shard* shard = shard_least_used();
     std::vector<RemoteBlockWire> blocks;
     blocks.reserve(keys.size());
-    std::vector<std::string> created;
-    {
-        std::lock_guard<std::shared_mutex> lk(kv_mu_);
-        for (auto& key : keys) {
-            if (kv_.find(key) != nullptr) {
+    std::vector<std::pair<std::string, uint64_t>> created;  // key, hash
+    auto rollback = [&] {
+        for (auto& [k, h] : created) {
+            auto& st = kv_[stripe_of(h)];
+            std::lock_guard<std::shared_mutex> lk(st.mu);
+            st.map.erase(k);
+        }
+    };
+    for (auto& key : keys) {
+        uint64_t h = KvMap::hash_of(key);
+        auto& st = kv_[stripe_of(h)];
+        {
+            std::shared_lock<std::shared_mutex> sl(st.mu);
+            if (st.map.find_hashed(key, h) != nullptr) {
                 blocks.push_back({0, 0, 0});  // FAKE block: dup key, client skips
                 continue;
             }
-            void* ptr = nullptr;
-            int pool_idx = -1;
-            auto try_alloc = [&] {
-                return shard->allocate(page, 1, [&](void* p, int idx) {
-                    ptr = p;
-                    pool_idx = idx;
-                });
-            };
-            bool ok = try_alloc();
-            if (!ok && opt_.auto_evict && evict_lru_locked(shard, page) > 0) ok = try_alloc();
-            if (!ok) {
-                for (auto& k : created) kv_.erase(k);
-                *status = OUT_OF_MEMORY;
-                return {};
-            }
-            auto* e = new BlockEntry();
-            e->ptr = ptr;
-            e->size = page;
-            e->pool_idx = pool_idx;
-            e->shard = shard;
-            e->last_access.store(tick(), std::memory_order_relaxed);
-            Ref<BlockEntry> ref(e);
-            bool ins = false;
-            kv_.emplace(key, ref, &ins);
-            created.push_back(key);
-            c->pending_rdma.emplace(reinterpret_cast<uint64_t>(ptr), ref);
-            blocks.push_back({static_cast<uint32_t>(shard->device() + 1), 0,
-                              reinterpret_cast<uint64_t>(ptr)});
         }
+        void* ptr = nullptr;
+        int pool_idx = -1;
+        auto try_alloc = [&] {
+            return shard->allocate(page, 1, [&](void* p, int idx) {
+                ptr = p;
+                pool_idx = idx;
+            });
+        };
+        bool ok = try_alloc();
+        if (!ok && opt_.auto_evict && evict_lru(shard, page) > 0) ok = try_alloc();
+        if (!ok) {
+            rollback();
+            *status = OUT_OF_MEMORY;
+            return {};
+        }
+        auto* e = new BlockEntry();
+        e->ptr = ptr;
+        e->size = page;
+        e->pool_idx = pool_idx;
+        e->shard = shard;
+        e->last_access.store(tick(), std::memory_order_relaxed);
+        Ref<BlockEntry> ref(e);
+        bool ins = false;
+        {
+            std::lock_guard<std::shared_mutex> lk(st.mu);
+            st.map.emplace_hashed(key, h, ref, &ins);
+        }
+        if (!ins) {  // raced with another writer: first write wins
+            blocks.push_back({0, 0, 0});
+            continue;  // `ref` releases and frees the block via ~BlockEntry
+        }
+        created.push_back({key, h});
+        c->pending_rdma.emplace(reinterpret_cast<uint64_t>(ptr), ref);
+        blocks.push_back({static_cast<uint32_t>(shard->device() + 1), 0,
+                          reinterpret_cast<uint64_t>(ptr)});
     }
     maybe_extend(shard);
     return blocks;
@@ -1250,11 +1300,13 @@ void Server::commit_addrs(Conn* c, const std::vector<uint64_t>& addrs) {
 
 bool Server::collect_read_entries(const std::vector<std::string>& keys,
                                   std::vector<Ref<BlockEntry>>* out) {
-    std::shared_lock<std::shared_mutex> lk(kv_mu_);
     out->reserve(keys.size());
     uint64_t t = tick();
     for (auto& key : keys) {
-        Ref<BlockEntry>* v = kv_.find(key);
+        uint64_t h = KvMap::hash_of(key);
+        auto& st = kv_[stripe_of(h)];
+        std::shared_lock<std::shared_mutex> lk(st.mu);
+        Ref<BlockEntry>* v = st.map.find_hashed(key, h);
         if (!v || !(*v)->committed) return false;
         (*v)->last_access.store(t, std::memory_order_relaxed);
         out->push_back(*v);
@@ -1403,8 +1455,10 @@ void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body) {
     std::string key(reinterpret_cast<const char*>(body.data() + 4), len);
     bool exists;
     {
-        std::shared_lock<std::shared_mutex> lk(kv_mu_);
-        Ref<BlockEntry>* v = kv_.find(key);
+        uint64_t h = KvMap::hash_of(key);
+        auto& st = kv_[stripe_of(h)];
+        std::shared_lock<std::shared_mutex> lk(st.mu);
+        Ref<BlockEntry>* v = st.map.find_hashed(key, h);
         exists = v && (*v)->committed;
     }
     send_status(c, exists ? 0 : 1);
@@ -1419,9 +1473,11 @@ void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body) {
     // divergence from the reference, which counts uncommitted keys as
     // present (infinistore.cpp:1097); an uncommitted key cannot be read, so
     // reporting it as a hit would make the subsequent read_cache fail.
-    std::shared_lock<std::shared_mutex> lk(kv_mu_);
     auto present = [&](size_t i) {
-        Ref<BlockEntry>* v = kv_.find(keys[i]);
+        uint64_t h = KvMap::hash_of(keys[i]);
+        auto& st = kv_[stripe_of(h)];
+        std::shared_lock<std::shared_mutex> lk(st.mu);
+        Ref<BlockEntry>* v = st.map.find_hashed(keys[i], h);
         return v && (*v)->committed;
     };
     long left = 0, right = static_cast<long>(keys.size());
@@ -1440,9 +1496,11 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
     if (!parse_match_request(body.data(), body.size(), &keys))
         return send_status(c, INVALID_REQ);
     int n = 0;
-    {
-        std::lock_guard<std::shared_mutex> lk(kv_mu_);
-        for (auto& k : keys) n += kv_.erase(k) ? 1 : 0;
+    for (auto& k : keys) {
+        uint64_t h = KvMap::hash_of(k);
+        auto& st = kv_[stripe_of(h)];
+        std::lock_guard<std::shared_mutex> lk(st.mu);
+        n += st.map.erase(k) ? 1 : 0;
     }
     send_status(c, n);
 }
@@ -1452,19 +1510,23 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
 // ---------------------------------------------------------------------------
 std::pair<size_t, size_t> Server::compact() {
     if (!running_.load()) return {0, 0};
-    // Maintenance op: kv_mu_ is held for the WHOLE plan+copy+swap so no
-    // request thread can start a read against a block while it moves
-    // (requests block on the mutex for the few ms this takes).
+    // Maintenance op: ALL stripe locks are held (taken in index order) for
+    // the WHOLE plan+copy+swap so no request thread can start a read
+    // against a block while it moves (requests block for the few ms this
+    // takes). This is the only path holding more than one stripe lock.
     size_t moved = 0, bytes = 0;
-    std::lock_guard<std::shared_mutex> lk(kv_mu_);
+    std::vector<std::unique_lock<std::shared_mutex>> locks;
+    locks.reserve(kStripes);
+    for (auto& st : kv_) locks.emplace_back(st.mu);
     for (auto& shard_up : shards_) {
         Shard* shard = shard_up.get();
         std::vector<std::pair<void*, size_t>> movable;
-        kv_.for_each([&](std::string_view, Ref<BlockEntry>& val) {
-            BlockEntry* e = val.get();
-            if (e->shard == shard && e->committed && e->ref_count() == 1)
-                movable.push_back({e->ptr, e->size});
-        });
+        for (auto& st : kv_)
+            st.map.for_each([&](std::string_view, Ref<BlockEntry>& val) {
+                BlockEntry* e = val.get();
+                if (e->shard == shard && e->committed && e->ref_count() == 1)
+                    movable.push_back({e->ptr, e->size});
+            });
         if (movable.empty()) continue;
         auto moves = shard->plan_compaction(movable);
         if (moves.empty()) continue;
@@ -1489,30 +1551,38 @@ std::pair<size_t, size_t> Server::compact() {
         // Swap pointers in the index and free the old slots.
         std::map<void*, Shard::Move*> by_old;
         for (auto& m : moves) by_old[m.old_ptr] = &m;
-        kv_.for_each([&](std::string_view, Ref<BlockEntry>& val) {
-            BlockEntry* e = val.get();
-            auto it = by_old.find(e->ptr);
-            if (it == by_old.end() || e->shard != shard) return;
-            Shard::Move* m = it->second;
-            e->ptr = m->new_ptr;
-            e->pool_idx = m->pool_idx;
-            shard->deallocate(m->old_ptr, m->size, m->pool_idx);
-            moved++;
-            bytes += m->size;
-        });
+        for (auto& st : kv_)
+            st.map.for_each([&](std::string_view, Ref<BlockEntry>& val) {
+                BlockEntry* e = val.get();
+                auto it = by_old.find(e->ptr);
+                if (it == by_old.end() || e->shard != shard) return;
+                Shard::Move* m = it->second;
+                e->ptr = m->new_ptr;
+                e->pool_idx = m->pool_idx;
+                shard->deallocate(m->old_ptr, m->size, m->pool_idx);
+                moved++;
+                bytes += m->size;
+            });
     }
     return {moved, bytes};
 }
 
 size_t Server::kvmap_len() {
-    std::shared_lock<std::shared_mutex> lk(kv_mu_);
-    return kv_.size();
+    size_t n = 0;
+    for (auto& st : kv_) {
+        std::shared_lock<std::shared_mutex> lk(st.mu);
+        n += st.map.size();
+    }
+    return n;
 }
 
 size_t Server::purge() {
-    std::lock_guard<std::shared_mutex> lk(kv_mu_);
-    size_t n = kv_.size();
-    kv_.clear();
+    size_t n = 0;
+    for (auto& st : kv_) {
+        std::lock_guard<std::shared_mutex> lk(st.mu);
+        n += st.map.size();
+        st.map.clear();
+    }
     return n;
 }
 
