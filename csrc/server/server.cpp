@@ -805,8 +805,12 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
                                [&](void* p, int idx) { slots.push_back({p, idx}); });
     };
     bool alloc_ok = try_alloc();
-    if (!alloc_ok && opt_.auto_evict) {
-        if (evict_lru(shard, page * n_fresh * 2) > 0) alloc_ok = try_alloc();
+    // Retry loop: with concurrent writers a single evict+retry can lose its
+    // freed blocks to a racing allocation; keep evicting until the
+    // allocation lands or eviction runs dry.
+    for (int attempt = 0; !alloc_ok && opt_.auto_evict && attempt < 4; attempt++) {
+        if (evict_lru(shard, page * n_fresh * 2) == 0) break;
+        alloc_ok = try_alloc();
     }
     if (!alloc_ok) return reply_local(c, ctx, OUT_OF_MEMORY);
     auto p2 = std::chrono::steady_clock::now();
@@ -1258,7 +1262,10 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
             });
         };
         bool ok = try_alloc();
-        if (!ok && opt_.auto_evict && evict_lru(shard, page) > 0) ok = try_alloc();
+        for (int attempt = 0; !ok && opt_.auto_evict && attempt < 4; attempt++) {
+            if (evict_lru(shard, page * 4) == 0) break;
+            ok = try_alloc();
+        }
         if (!ok) {
             rollback();
             *status = OUT_OF_MEMORY;
@@ -1496,11 +1503,18 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
     if (!parse_match_request(body.data(), body.size(), &keys))
         return send_status(c, INVALID_REQ);
     int n = 0;
-    for (auto& k : keys) {
-        uint64_t h = KvMap::hash_of(k);
-        auto& st = kv_[stripe_of(h)];
-        std::lock_guard<std::shared_mutex> lk(st.mu);
-        n += st.map.erase(k) ? 1 : 0;
+    {
+        std::array<std::vector<uint32_t>, kStripes> by_stripe;
+        std::vector<uint64_t> hashes(keys.size());
+        for (size_t i = 0; i < keys.size(); i++) {
+            hashes[i] = KvMap::hash_of(keys[i]);
+            by_stripe[stripe_of(hashes[i])].push_back(static_cast<uint32_t>(i));
+        }
+        for (size_t si = 0; si < kStripes; si++) {
+            if (by_stripe[si].empty()) continue;
+            std::lock_guard<std::shared_mutex> lk(kv_[si].mu);
+            for (uint32_t i : by_stripe[si]) n += kv_[si].map.erase(keys[i]) ? 1 : 0;
+        }
     }
     send_status(c, n);
 }
