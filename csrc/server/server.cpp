@@ -1510,10 +1510,17 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
             hashes[i] = KvMap::hash_of(keys[i]);
             by_stripe[stripe_of(hashes[i])].push_back(static_cast<uint32_t>(i));
         }
+        constexpr size_t kPf = 16;
         for (size_t si = 0; si < kStripes; si++) {
-            if (by_stripe[si].empty()) continue;
+            auto& list = by_stripe[si];
+            if (list.empty()) continue;
             std::lock_guard<std::shared_mutex> lk(kv_[si].mu);
-            for (uint32_t i : by_stripe[si]) n += kv_[si].map.erase(keys[i]) ? 1 : 0;
+            auto& m = kv_[si].map;
+            for (size_t i = 0; i < std::min(kPf, list.size()); i++) m.prefetch(hashes[list[i]]);
+            for (size_t i = 0; i < list.size(); i++) {
+                if (i + kPf < list.size()) m.prefetch(hashes[list[i + kPf]]);
+                n += m.erase(keys[list[i]]) ? 1 : 0;
+            }
         }
     }
     send_status(c, n);
