@@ -115,6 +115,29 @@ class PagedKVConnector:
         except Exception:
             return False
 
+    def load_layer_async(self, layer: int, kv_out: torch.Tensor,
+                         page_keys: List[str], page_offsets, page_elems: int):
+        """Ticketed prefetch (local path): push the gather and return a
+        ticket; call wait_load(ticket) before touching kv_out. Lets decode
+        overlap the next request's KV fetch with current compute. Returns
+        None if any page was missing or the path is unavailable (caller
+        falls back to load_layer / recompute)."""
+        if not self.local:
+            return None
+        keys = [self._key(layer, k) for k in page_keys]
+        try:
+            return self.conn.read_pages_async(kv_out, keys, page_offsets,
+                                              page_elems)
+        except Exception:
+            return None
+
+    def wait_load(self, ticket) -> bool:
+        try:
+            self.conn.wait_read(ticket)
+            return True
+        except Exception:
+            return False
+
     def evict(self, page_keys: List[str]) -> int:
         """Drop a sequence's pages (all layers)."""
         keys = [self._key(layer, k) for layer in range(self.n_layers)

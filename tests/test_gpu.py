@@ -410,3 +410,34 @@ c.close()
         env=env, cwd=REPO, timeout=120,
     ).returncode
     assert code == 0
+
+
+def test_ticketed_async_reads(gpu_server):
+    """read_pages_async/wait_read: multiple reads in flight on one conn,
+    responses consumed out of submission order."""
+    conn = local_conn(gpu_server)
+    try:
+        page = 32768
+        n = 64
+        src = torch.randn(page * n // 4, device="cuda:0")
+        dsts = [torch.zeros(page * n // 4, device="cuda:0") for _ in range(3)]
+        offs = [i * page // 4 for i in range(n)]
+        keysets = []
+        for rep in range(3):
+            keys = [f"tk-{uuid.uuid4()}-{rep}-{i}" for i in range(n)]
+            conn.write_pages(src, keys, offs, page // 4, sync=True)
+            keysets.append(keys)
+        tickets = [conn.read_pages_async(dsts[r], keysets[r], offs, page // 4)
+                   for r in range(3)]
+        for t in reversed(tickets):  # wait out of order
+            conn.wait_read(t)
+        for d in dsts:
+            assert torch.equal(src, d)
+        # missing key surfaces as an error at wait time (or at push for the
+        # blocking fallback)
+        import pytest as _pytest
+        with _pytest.raises(Exception):
+            t = conn.read_pages_async(dsts[0], ["tk-missing-key"], [0], page // 4)
+            conn.wait_read(t)
+    finally:
+        conn.close()
