@@ -55,6 +55,36 @@ def make_app(config: ServerConfig):
         moved, moved_bytes = await loop.run_in_executor(None, lib.compact_pool)
         return {"moved_blocks": moved, "moved_bytes": moved_bytes}
 
+    @app.get("/metrics")
+    async def metrics():
+        # Prometheus exposition of the server counters (scrape target for
+        # production fleets). Gauges are re-set per scrape from the native
+        # stats snapshot; per-op handler timings become *_count / *_us_total.
+        import json
+
+        from fastapi.responses import PlainTextResponse
+
+        s = json.loads(lib.get_server_stats())
+        lines = []
+
+        def g(name, val, help_=""):
+            if help_:
+                lines.append(f"# HELP infinistore_{name} {help_}")
+            lines.append(f"# TYPE infinistore_{name} gauge")
+            lines.append(f"infinistore_{name} {val}")
+
+        g("kv_len", s.get("kv_len", 0), "Stored keys")
+        g("used_blocks", s.get("used_blocks", 0))
+        g("total_blocks", s.get("total_blocks", 0))
+        g("bytes_in_total", s.get("bytes_in", 0), "Payload bytes written")
+        g("bytes_out_total", s.get("bytes_out", 0), "Payload bytes read")
+        g("evicted_total", s.get("evicted", 0))
+        for op, st in (s.get("op_us") or {}).items():
+            lines.append(f'infinistore_op_count{{op="{op}"}} {st.get("count", 0)}')
+            lines.append(f'infinistore_op_avg_us{{op="{op}"}} {st.get("avg_us", 0)}')
+            lines.append(f'infinistore_op_max_us{{op="{op}"}} {st.get("max_us", 0)}')
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     @app.post("/selftest/{port}")
     async def selftest(port: int):
         # Loopback roundtrip through the full client stack (CPU tensor).

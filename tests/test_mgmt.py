@@ -55,6 +55,23 @@ def test_selftest_endpoint(app_client):
     assert r["status"] == "ok"
 
 
+def test_metrics_endpoint(app_client):
+    tc, cpu_server = app_client
+    conn = make_client(cpu_server)
+    try:
+        src = torch.zeros(256)
+        conn.register_mr(src)
+        blocks = conn.allocate_rdma(["metrics-key"], 1024)
+        conn.rdma_write_cache(src, [0], 256, blocks)
+        conn.sync()
+    finally:
+        conn.close()
+    text = tc.get("/metrics").text
+    assert "infinistore_kv_len 1" in text
+    assert 'infinistore_op_count{op="allocate"}' in text
+    assert "# TYPE infinistore_bytes_in_total gauge" in text
+
+
 def test_stats_json_parses():
     s = ifs.get_server_stats()
     json.loads(s)
