@@ -441,3 +441,12 @@ def test_ticketed_async_reads(gpu_server):
             conn.wait_read(t)
     finally:
         conn.close()
+
+
+def test_disaggregated_decode_gpu(gpu_server):
+    """Prefill→store→decode demo over the local IPC + HIP gather path;
+    decode logits rebuilt from cached pages must match the monolithic
+    forward (infinistore_amd/example/disaggregated.py)."""
+    from infinistore_amd.example.disaggregated import main as disagg_main
+
+    disagg_main(port=gpu_server, device="cuda:0")
