@@ -323,7 +323,7 @@ int ClientConn::rw_local(char op, const std::vector<std::pair<std::string, uint6
 int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
                                 const uint64_t* offsets, size_t n, int block_size,
                                 uintptr_t ptr, int device_id, bool sync_response,
-                                uint64_t* out_ticket) {
+                                uint64_t* out_ticket, uint32_t extra_flags) {
     if (out_ticket) *out_ticket = 0;  // 0 = completed synchronously
     if (!connected_) return -1;
     if (!gpu::available()) {
@@ -361,7 +361,7 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
     // Reads complete in one round trip (response deferred to completion);
     // writes default to async so uploads overlap compute (prefill pattern),
     // with an opt-in single-round-trip mode (write_pages(sync=True)).
-    h.flags = (op == 'R' || sync_response) ? kLocalFlagSyncResponse : 0;
+    h.flags = ((op == 'R' || sync_response) ? kLocalFlagSyncResponse : 0) | extra_flags;
     h.rsvd = 0;
     memcpy(h.ipc, handle.bytes, gpu::kIpcHandleSize);
 

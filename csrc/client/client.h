@@ -65,7 +65,7 @@ class ClientConn {
     int rw_local_packed(char op, const char* keys_blob, size_t blob_len,
                         const uint64_t* offsets, size_t n, int block_size, uintptr_t ptr,
                         int device_id, bool sync_response = false,
-                        uint64_t* out_ticket = nullptr);
+                        uint64_t* out_ticket = nullptr, uint32_t extra_flags = 0);
     // Wait for a ticketed (async shm) op; ticket 0 = already complete.
     int wait_local_ticket(uint64_t ticket);
     int sync_local();

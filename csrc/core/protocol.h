@@ -80,6 +80,10 @@ static_assert(sizeof(PackedLocalHdr) == 104, "packed local header size");
 // PackedLocalHdr.flags: defer the response until the copy completes (one
 // round trip instead of request-ack + sync).
 constexpr uint32_t kLocalFlagSyncResponse = 1;
+// Store the (bf16) payload quantized to fp8 e4m3 with one scale per block —
+// half the HBM per cached page; reads dequantize transparently (extension,
+// csrc/gpu/gpu.hip quant kernels).
+constexpr uint32_t kLocalFlagQuantFp8 = 2;
 
 std::string op_name(char op);
 

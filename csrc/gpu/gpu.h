@@ -90,6 +90,17 @@ bool launch_copy_blocks(int dev, Stream s, const uint64_t* dev_src_ptrs,
 bool launch_copy_blocks_inline(int dev, Stream s, const uint64_t* src_ptrs,
                                const uint64_t* dst_ptrs, int n_blocks, size_t bytes_per_block);
 
+// --- fp8 KV compression ------------------------------------------------------
+// Quantize n blocks of bf16 (elems_per_block each) into fp8 e4m3 blocks of
+// half the byte size; one scale (absmax/448) per block -> dev_scales[i].
+bool launch_quant_blocks(int dev, Stream s, const uint64_t* dev_src_ptrs,
+                         const uint64_t* dev_dst_ptrs, float* dev_scales, int n_blocks,
+                         size_t elems_per_block);
+// Inverse: fp8 blocks + scales -> bf16 blocks.
+bool launch_dequant_blocks(int dev, Stream s, const uint64_t* dev_src_ptrs,
+                           const uint64_t* dev_dst_ptrs, const float* dev_scales, int n_blocks,
+                           size_t elems_per_block);
+
 // --- block fingerprint ------------------------------------------------------
 // 64-bit position-salted fingerprint per block -> out_hashes[i] (device mem).
 bool launch_hash_blocks(int dev, Stream s, const uint64_t* dev_ptrs, int n_blocks,

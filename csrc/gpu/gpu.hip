@@ -340,6 +340,147 @@ bool launch_copy_blocks(int dev, Stream stream, const uint64_t* dev_src_ptrs,
     return true;
 }
 
+// --- fp8 KV-cache compression ----------------------------------------------
+// MI355X-native extension (no reference equivalent): pages enter the pool
+// quantized bf16 -> OCP fp8 e4m3 with ONE power-agnostic scale per page
+// (absmax/448), halving HBM footprint per cached page; reads dequantize
+// back to bf16 on the way out. Both kernels are memory-bound streaming
+// passes: one 512-thread workgroup per page, vectorized 8-elems-per-lane.
+// Encoding is done in integer bit math (round-to-nearest-even on the
+// normal path) so results match torch.float8_e4m3fn casts to 1 ulp.
+
+__device__ __forceinline__ uint8_t f32_to_e4m3(float f) {
+    uint32_t bits = __float_as_uint(f);
+    uint8_t sign = static_cast<uint8_t>((bits >> 24) & 0x80);
+    uint32_t u = bits & 0x7fffffffu;
+    if (u >= 0x43e00000u) return sign | 0x7e;  // |x| >= 448 -> clamp to max
+    int e = static_cast<int>(u >> 23) - 127;
+    if (e < -6) {  // subnormal region: unit 2^-9
+        if (e < -10) return sign;  // rounds to zero
+        uint32_t full = 0x800000u | (u & 0x7fffffu);  // 1.m (24-bit)
+        int shift = 23 - (e + 9);                     // 2^-9 units
+        uint32_t half = 1u << (shift - 1);
+        uint32_t q = (full + half) >> shift;          // round half away
+        if (q > 8) q = 8;                             // 8 == min normal code
+        return sign | static_cast<uint8_t>(q);
+    }
+    // RNE at mantissa bit 20.
+    uint32_t rounded = u + 0x0007ffffu + ((u >> 20) & 1u);
+    if (rounded >= 0x43e00000u) return sign | 0x7e;
+    e = static_cast<int>(rounded >> 23) - 127;
+    uint32_t m = (rounded >> 20) & 7u;
+    return sign | static_cast<uint8_t>(((e + 7) << 3) | m);
+}
+
+__device__ __forceinline__ float e4m3_to_f32(uint8_t v) {
+    uint32_t e = (v >> 3) & 0xfu;
+    uint32_t m = v & 7u;
+    float mag;
+    if (e == 0)
+        mag = static_cast<float>(m) * 0.001953125f;  // m * 2^-9
+    else
+        mag = __uint_as_float(((e - 7 + 127) << 23) | (m << 20));
+    return (v & 0x80) ? -mag : mag;
+}
+
+// One workgroup per page. Pass 1: workgroup absmax (wavefront shuffle
+// reduce + LDS). Pass 2: scale + convert, 8 bf16 in (one uint4) -> 8 fp8
+// out (one uint2) per lane per iteration.
+__global__ void quant_blocks_bf16_fp8_kernel(const uint64_t* __restrict__ src_ptrs,
+                                             const uint64_t* __restrict__ dst_ptrs,
+                                             float* __restrict__ scales,
+                                             uint64_t elems_per_block) {
+    const uint16_t* src = reinterpret_cast<const uint16_t*>(src_ptrs[blockIdx.x]);
+    uint8_t* dst = reinterpret_cast<uint8_t*>(dst_ptrs[blockIdx.x]);
+    __shared__ float red[8];  // 512 threads / 64-wide wavefronts
+    float m = 0.f;
+    uint64_t n8 = elems_per_block / 8;
+    const uint4* src4 = reinterpret_cast<const uint4*>(src);
+    for (uint64_t u = threadIdx.x; u < n8; u += blockDim.x) {
+        uint4 pk = src4[u];
+        const uint16_t* h = reinterpret_cast<const uint16_t*>(&pk);
+#pragma unroll
+        for (int i = 0; i < 8; i++) {
+            // bf16 absmax == f32 absmax of the widened bits.
+            float f = __uint_as_float(static_cast<uint32_t>(h[i]) << 16);
+            m = fmaxf(m, fabsf(f));
+        }
+    }
+    for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_down(m, off, 64));
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = m;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float mm = red[0];
+        for (int w = 1; w < static_cast<int>(blockDim.x) / 64; w++) mm = fmaxf(mm, red[w]);
+        red[0] = mm > 0.f ? mm / 448.f : 1.f;  // scale
+        scales[blockIdx.x] = red[0];
+    }
+    __syncthreads();
+    float inv = 1.f / red[0];
+    for (uint64_t u = threadIdx.x; u < n8; u += blockDim.x) {
+        uint4 pk = src4[u];
+        const uint16_t* h = reinterpret_cast<const uint16_t*>(&pk);
+        uint8_t out[8];
+#pragma unroll
+        for (int i = 0; i < 8; i++) {
+            float f = __uint_as_float(static_cast<uint32_t>(h[i]) << 16);
+            out[i] = f32_to_e4m3(f * inv);
+        }
+        reinterpret_cast<uint2*>(dst)[u] = *reinterpret_cast<const uint2*>(out);
+    }
+}
+
+// Dequantize fp8 pages back into bf16 client memory (round-to-nearest-even
+// bf16 truncation): 8 fp8 in (uint2) -> 8 bf16 out (uint4) per lane.
+__global__ void dequant_blocks_fp8_bf16_kernel(const uint64_t* __restrict__ src_ptrs,
+                                               const uint64_t* __restrict__ dst_ptrs,
+                                               const float* __restrict__ scales,
+                                               uint64_t elems_per_block) {
+    const uint8_t* src = reinterpret_cast<const uint8_t*>(src_ptrs[blockIdx.x]);
+    uint16_t* dst = reinterpret_cast<uint16_t*>(dst_ptrs[blockIdx.x]);
+    float scale = scales[blockIdx.x];
+    uint64_t n8 = elems_per_block / 8;
+    for (uint64_t u = threadIdx.x; u < n8; u += blockDim.x) {
+        uint2 pk = reinterpret_cast<const uint2*>(src)[u];
+        const uint8_t* b = reinterpret_cast<const uint8_t*>(&pk);
+        uint16_t out[8];
+#pragma unroll
+        for (int i = 0; i < 8; i++) {
+            float f = e4m3_to_f32(b[i]) * scale;
+            uint32_t fb = __float_as_uint(f);
+            fb += 0x7fffu + ((fb >> 16) & 1u);  // RNE to bf16
+            out[i] = static_cast<uint16_t>(fb >> 16);
+        }
+        reinterpret_cast<uint4*>(dst)[u] = *reinterpret_cast<const uint4*>(out);
+    }
+}
+
+bool launch_quant_blocks(int dev, Stream stream, const uint64_t* dev_src_ptrs,
+                         const uint64_t* dev_dst_ptrs, float* dev_scales, int n_blocks,
+                         size_t elems_per_block) {
+    if (n_blocks <= 0) return true;
+    if (elems_per_block % 8 != 0) return false;
+    HIP_OK(hipSetDevice(dev));
+    hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+    hipLaunchKernelGGL(quant_blocks_bf16_fp8_kernel, dim3(n_blocks), dim3(512), 0, s,
+                       dev_src_ptrs, dev_dst_ptrs, dev_scales, elems_per_block);
+    HIP_OK(hipGetLastError());
+    return true;
+}
+
+bool launch_dequant_blocks(int dev, Stream stream, const uint64_t* dev_src_ptrs,
+                           const uint64_t* dev_dst_ptrs, const float* dev_scales, int n_blocks,
+                           size_t elems_per_block) {
+    if (n_blocks <= 0) return true;
+    if (elems_per_block % 8 != 0) return false;
+    HIP_OK(hipSetDevice(dev));
+    hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+    hipLaunchKernelGGL(dequant_blocks_fp8_bf16_kernel, dim3(n_blocks), dim3(512), 0, s,
+                       dev_src_ptrs, dev_dst_ptrs, dev_scales, elems_per_block);
+    HIP_OK(hipGetLastError());
+    return true;
+}
+
 // --- fingerprint -----------------------------------------------------------
 // Position-salted 64-bit mix (splitmix64 finalizer); XOR-combined across a
 // block so the reduction is order-free, deterministic, and parallel.

@@ -200,7 +200,7 @@ PYBIND11_MODULE(_native, m) {
             "rw_local_keys",
             [](ClientConn& c, const std::string& op, py::list keys, py::buffer offsets,
                uint64_t element_size, int block_size, uintptr_t ptr, int device,
-               bool sync_response) {
+               bool sync_response, uint32_t extra_flags) {
                 // Build the NUL-joined key blob and byte offsets in C++ —
                 // no Python-side join/numpy work on the hot path.
                 py::buffer_info ob = offsets.request();
@@ -223,16 +223,16 @@ PYBIND11_MODULE(_native, m) {
                 py::gil_scoped_release rel;
                 return c.rw_local_packed(op.empty() ? 'W' : op[0], blob.data(), blob.size(),
                                          byte_offs.data(), n, block_size, ptr, device,
-                                         sync_response);
+                                         sync_response, nullptr, extra_flags);
             },
             py::arg("op"), py::arg("keys"), py::arg("offsets"), py::arg("element_size"),
             py::arg("block_size"), py::arg("ptr"), py::arg("device"),
-            py::arg("sync_response") = false)
+            py::arg("sync_response") = false, py::arg("extra_flags") = 0)
         .def(
             "rw_local_blob",
             [](ClientConn& c, const std::string& op, py::buffer blob, py::buffer offsets,
                uint64_t element_size, int block_size, uintptr_t ptr, int device,
-               bool sync_response) {
+               bool sync_response, uint32_t extra_flags) {
                 // Zero-pack fast path: keys as one NUL-separated bytes blob
                 // (engines cache the serialized page-key chain; re-joining
                 // 2k Python strings costs ~30 µs per request otherwise).
@@ -247,11 +247,12 @@ PYBIND11_MODULE(_native, m) {
                 size_t blen = static_cast<size_t>(bb.size) * bb.itemsize;
                 py::gil_scoped_release rel;
                 return c.rw_local_packed(op.empty() ? 'W' : op[0], bp, blen, byte_offs.data(),
-                                         n, block_size, ptr, device, sync_response);
+                                         n, block_size, ptr, device, sync_response, nullptr,
+                                         extra_flags);
             },
             py::arg("op"), py::arg("blob"), py::arg("offsets"), py::arg("element_size"),
             py::arg("block_size"), py::arg("ptr"), py::arg("device"),
-            py::arg("sync_response") = false)
+            py::arg("sync_response") = false, py::arg("extra_flags") = 0)
         .def(
             "rw_local_blob_async",
             [](ClientConn& c, const std::string& op, py::buffer blob, py::buffer offsets,

@@ -757,6 +757,11 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     size_t page = static_cast<size_t>(msg.block_size);
     size_t nb = msg.blocks.size();
     constexpr size_t kPf = 16;
+    // fp8 ingest compression (extension): bf16 pages stored at half size
+    // with one scale per page; requires a 16-byte-divisible page.
+    const bool quant = (msg.flags & kLocalFlagQuantFp8) != 0;
+    if (quant && (page % 16 != 0)) return reply_local(c, ctx, INVALID_REQ);
+    const size_t stored = quant ? page / 2 : page;
 
     // Phase A — dedup check only (short kv_mu_ hold, prefetch-pipelined).
     // The authoritative first-write-wins decision happens at the insert pass
@@ -801,7 +806,8 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     std::vector<std::pair<void*, int>> slots;
     slots.reserve(n_fresh);
     auto try_alloc = [&] {
-        return shard->allocate(page, n_fresh,
+        slots.clear();
+        return shard->allocate(stored, n_fresh,
                                [&](void* p, int idx) { slots.push_back({p, idx}); });
     };
     bool alloc_ok = try_alloc();
@@ -825,13 +831,20 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     job.bytes_per_block = page;
     job.src.reserve(n_fresh);
     job.dst.reserve(n_fresh);
+    auto scales_out =
+        quant ? std::make_shared<std::vector<float>>(n_fresh, 1.f) : nullptr;
+    if (quant) {
+        job.xform = Shard::CopyJob::Xform::kQuantBf16Fp8;
+        job.scales_out = scales_out;
+    }
     for (size_t i = 0; i < n_fresh; i++) {
         auto* e = slab_batch.make();
         e->ptr = slots[i].first;
-        e->size = page;
+        e->size = stored;
         e->pool_idx = slots[i].second;
         e->shard = shard;
         e->committed = false;
+        e->fp8 = quant;
         e->last_access.store(t, std::memory_order_relaxed);
         entries->emplace_back(e);
         job.src.push_back(
@@ -851,7 +864,8 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     // (held here across submit + phase D) provides that ordering instead.
     auto won = std::make_shared<std::vector<uint8_t>>(n_fresh, 0);
     auto commit_mu = std::make_shared<std::mutex>();
-    job.done = [this, c, entries, won, sync_resp, ctx, commit_mu, t_start](bool ok) {
+    job.done = [this, c, entries, won, sync_resp, ctx, commit_mu, scales_out,
+                t_start](bool ok) {
         if (sdbg && entries->size() > 64) {
             auto us = std::chrono::duration<double, std::micro>(
                           std::chrono::steady_clock::now() - t_start)
@@ -859,10 +873,13 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
             fprintf(stderr, "[sdbg] write n=%zu submit->complete=%.0fus\n", entries->size(),
                     us);
         }
-        auto fin = [this, c, entries, won, ok, sync_resp, ctx] {
+        auto fin = [this, c, entries, won, ok, sync_resp, ctx, scales_out] {
             if (ok) {
                 for (size_t i = 0; i < entries->size(); i++)
-                    if ((*won)[i]) (*entries)[i]->committed = true;
+                    if ((*won)[i]) {
+                        if (scales_out) (*entries)[i]->scale = (*scales_out)[i];
+                        (*entries)[i]->committed = true;
+                    }
             } else {
                 // Copy failed: drop the keys. Always from the owner loop —
                 // erase_entries takes the kv lock exclusively, and compact()
@@ -943,8 +960,10 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     size_t page = static_cast<size_t>(msg.block_size);
 
     auto tr0 = std::chrono::steady_clock::now();
-    // Group blocks by owning shard (keys may live on different GPUs).
+    // Group blocks by owning shard (keys may live on different GPUs); fp8-
+    // compressed entries go into separate dequantizing jobs.
     std::map<Shard*, Shard::CopyJob> jobs;
+    std::map<Shard*, Shard::CopyJob> qjobs;
     auto held = std::make_shared<std::vector<Ref<BlockEntry>>>();
     held->reserve(msg.blocks.size());
     uint64_t read_tick = tick();
@@ -972,10 +991,20 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
                 }
                 BlockEntry* e = v->get();
                 e->last_access.store(read_tick, std::memory_order_relaxed);
-                auto& job = jobs[e->shard];
-                job.bytes_per_block = page;
-                job.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
-                job.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.second));
+                if (e->fp8) {
+                    if (e->size * 2 != page) return reply_local(c, ctx, INVALID_REQ);
+                    auto& qj = qjobs[e->shard];
+                    qj.bytes_per_block = page;
+                    qj.xform = Shard::CopyJob::Xform::kDequantFp8Bf16;
+                    qj.scales_in.push_back(e->scale);
+                    qj.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
+                    qj.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.second));
+                } else {
+                    auto& job = jobs[e->shard];
+                    job.bytes_per_block = page;
+                    job.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
+                    job.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.second));
+                }
                 held->push_back(*v);
             }
         }
@@ -985,13 +1014,15 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     n_reads_.fetch_add(1);
     bytes_out_.fetch_add(msg.blocks.size() * page);
     bool sync_resp = (msg.flags & kLocalFlagSyncResponse) != 0;
-    if (jobs.empty()) return reply_local(c, ctx, sync_resp ? FINISH : TASK_ACCEPTED);
+    if (jobs.empty() && qjobs.empty())
+        return reply_local(c, ctx, sync_resp ? FINISH : TASK_ACCEPTED);
 
     c->remain.fetch_add(1);
     c->ref();
-    auto pending = std::make_shared<std::atomic<int>>(static_cast<int>(jobs.size()));
+    auto pending =
+        std::make_shared<std::atomic<int>>(static_cast<int>(jobs.size() + qjobs.size()));
     auto all_ok = std::make_shared<std::atomic<bool>>(true);
-    for (auto& [shard, job] : jobs) {
+    auto submit_one = [&](Shard* shard, Shard::CopyJob& job) -> bool {
         Shard::CopyJob j = std::move(job);
         static const bool sdbg2 = getenv("IFS_SERVER_DEBUG") != nullptr;
         auto t_start2 = std::chrono::steady_clock::now();
@@ -1022,10 +1053,16 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
             all_ok->store(false);
             if (pending->fetch_sub(1) == 1) {
                 finish_task(c, /*on_owner=*/!ctx.shm);
-                return reply_local(c, ctx, INTERNAL_ERROR);
+                reply_local(c, ctx, INTERNAL_ERROR);
+                return false;  // error reply sent; stop submitting
             }
         }
-    }
+        return true;
+    };
+    for (auto& [shard, job] : jobs)
+        if (!submit_one(shard, job)) return;
+    for (auto& [shard, job] : qjobs)
+        if (!submit_one(shard, job)) return;
     if (rdbg2 && msg.blocks.size() > 256) {
         auto us = [](auto a, auto b) {
             return std::chrono::duration<double, std::micro>(b - a).count();
@@ -1315,6 +1352,9 @@ bool Server::collect_read_entries(const std::vector<std::string>& keys,
         std::shared_lock<std::shared_mutex> lk(st.mu);
         Ref<BlockEntry>* v = st.map.find_hashed(key, h);
         if (!v || !(*v)->committed) return false;
+        // fp8-compressed entries are a local-GPU-path feature: the TCP/verbs
+        // fabric moves raw bytes and cannot dequantize on the way out.
+        if ((*v)->fp8) return false;
         (*v)->last_access.store(t, std::memory_order_relaxed);
         out->push_back(*v);
     }

@@ -86,6 +86,10 @@ struct BlockEntry : RefCounted {
     int pool_idx = -1;
     Shard* shard = nullptr;
     bool committed = false;
+    // fp8-compressed page (extension): `size` is the STORED byte size
+    // (half the logical bf16 page); reads dequantize with `scale`.
+    bool fp8 = false;
+    float scale = 1.f;
     // LRU tick (auto_evict); atomic: bumped under the SHARED kv lock by
     // concurrent readers.
     std::atomic<uint64_t> last_access{0};

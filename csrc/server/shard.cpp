@@ -30,6 +30,8 @@ Shard::~Shard() {
             if (sl.h_dst) gpu::free_host_pinned(sl.h_dst);
             if (sl.d_src) gpu::free_device(sl.d_src);
             if (sl.d_dst) gpu::free_device(sl.d_dst);
+            if (sl.h_scale) gpu::free_host_pinned(sl.h_scale);
+            if (sl.d_scale) gpu::free_device(sl.d_scale);
         }
         if (sc.stream) gpu::stream_destroy(sc.stream);
     }
@@ -56,12 +58,17 @@ bool Shard::init() {
             if (!sc.stream) return false;
             sc.slots.resize(opt_.slots_per_stream);
             for (auto& sl : sc.slots) {
+                size_t scale_bytes = opt_.max_descs_per_slot * sizeof(float);
                 sl.h_src = static_cast<uint64_t*>(gpu::alloc_host_pinned(desc_bytes));
                 sl.h_dst = static_cast<uint64_t*>(gpu::alloc_host_pinned(desc_bytes));
                 sl.d_src = static_cast<uint64_t*>(gpu::alloc_device(opt_.device, desc_bytes));
                 sl.d_dst = static_cast<uint64_t*>(gpu::alloc_device(opt_.device, desc_bytes));
+                sl.h_scale = static_cast<float*>(gpu::alloc_host_pinned(scale_bytes));
+                sl.d_scale = static_cast<float*>(gpu::alloc_device(opt_.device, scale_bytes));
                 sl.event = gpu::event_create(opt_.device);
-                if (!sl.h_src || !sl.h_dst || !sl.d_src || !sl.d_dst || !sl.event) return false;
+                if (!sl.h_src || !sl.h_dst || !sl.d_src || !sl.d_dst || !sl.h_scale ||
+                    !sl.d_scale || !sl.event)
+                    return false;
             }
         }
         // One completion thread PER STREAM: done-callbacks can block (the
@@ -212,6 +219,12 @@ bool Shard::submit_copy(CopyJob&& job) {
             size_t take = std::min(kFanChunk, n - off);
             CopyJob sub;
             sub.bytes_per_block = job.bytes_per_block;
+            sub.xform = job.xform;
+            sub.scales_out = job.scales_out;
+            sub.scales_off = job.scales_off + off;
+            if (!job.scales_in.empty())
+                sub.scales_in.assign(job.scales_in.begin() + static_cast<long>(off),
+                                     job.scales_in.begin() + static_cast<long>(off + take));
             sub.src.assign(job.src.begin() + static_cast<long>(off),
                            job.src.begin() + static_cast<long>(off + take));
             sub.dst.assign(job.dst.begin() + static_cast<long>(off),
@@ -229,6 +242,7 @@ bool Shard::submit_copy(CopyJob&& job) {
     }
 
     if (!on_gpu()) {
+        if (job.xform != CopyJob::Xform::kCopy) return false;  // GPU-only
         // CPU shard: copies run inline on the caller (loop) thread.
         for (size_t i = 0; i < n; i++)
             memcpy(reinterpret_cast<void*>(job.dst[i]), reinterpret_cast<const void*>(job.src[i]),
@@ -252,7 +266,7 @@ bool Shard::submit_copy(CopyJob&& job) {
         streams_[next_stream_.fetch_add(1) % static_cast<uint32_t>(streams_.size())];
 
     // Small aligned batches: descriptors ride in the kernel arguments.
-    if (aligned && n <= 16) {
+    if (aligned && n <= 16 && job.xform == CopyJob::Xform::kCopy) {
         Slot* slot = acquire_slot(sc);
         if (!slot) return false;
         bool ok = gpu::set_device(opt_.device) &&
@@ -263,7 +277,7 @@ bool Shard::submit_copy(CopyJob&& job) {
         {
             std::lock_guard<std::mutex> lk(task_mu_);
             if (!ok) slot->busy = false;
-            sc.pending.push_back({ok ? slot : nullptr, std::move(job.done)});
+            sc.pending.push_back({ok ? slot : nullptr, std::move(job.done), nullptr, 0, 0});
         }
         task_cv_.notify_all();
         return true;
@@ -286,8 +300,26 @@ bool Shard::submit_copy(CopyJob&& job) {
                                          sc.stream);
         ok = ok && gpu::memcpy_h2d_async(slot->d_dst, slot->h_dst, take * sizeof(uint64_t),
                                          sc.stream);
-        ok = ok && gpu::launch_copy_blocks(opt_.device, sc.stream, slot->d_src, slot->d_dst,
-                                           static_cast<int>(take), job.bytes_per_block, aligned);
+        if (job.xform == CopyJob::Xform::kCopy) {
+            ok = ok && gpu::launch_copy_blocks(opt_.device, sc.stream, slot->d_src, slot->d_dst,
+                                               static_cast<int>(take), job.bytes_per_block,
+                                               aligned);
+        } else if (job.xform == CopyJob::Xform::kQuantBf16Fp8) {
+            ok = ok && gpu::launch_quant_blocks(opt_.device, sc.stream, slot->d_src, slot->d_dst,
+                                                slot->d_scale, static_cast<int>(take),
+                                                job.bytes_per_block / 2);
+            // Bring the per-block scales back before the completion event.
+            ok = ok && gpu::memcpy_d2h_async(slot->h_scale, slot->d_scale, take * sizeof(float),
+                                             sc.stream);
+        } else {  // kDequantFp8Bf16
+            memcpy(slot->h_scale, job.scales_in.data() + off, take * sizeof(float));
+            ok = ok && gpu::memcpy_h2d_async(slot->d_scale, slot->h_scale, take * sizeof(float),
+                                             sc.stream);
+            ok = ok && gpu::launch_dequant_blocks(opt_.device, sc.stream, slot->d_src,
+                                                  slot->d_dst, slot->d_scale,
+                                                  static_cast<int>(take),
+                                                  job.bytes_per_block / 2);
+        }
         ok = ok && gpu::event_record(slot->event, sc.stream);
         if (!ok) {
             // Report the failure through the done callback (even if earlier
@@ -296,15 +328,17 @@ bool Shard::submit_copy(CopyJob&& job) {
             {
                 std::lock_guard<std::mutex> lk(task_mu_);
                 slot->busy = false;
-                sc.pending.push_back({nullptr, std::move(job.done)});
+                sc.pending.push_back({nullptr, std::move(job.done), nullptr, 0, 0});
             }
             task_cv_.notify_all();
             return true;
         }
         {
             std::lock_guard<std::mutex> lk(task_mu_);
-            sc.pending.push_back(
-                {slot, last ? std::move(job.done) : std::function<void(bool)>()});
+            bool q = job.xform == CopyJob::Xform::kQuantBf16Fp8;
+            sc.pending.push_back({slot,
+                                  last ? std::move(job.done) : std::function<void(bool)>(),
+                                  q ? job.scales_out : nullptr, job.scales_off + off, take});
         }
         task_cv_.notify_all();
         off += take;
@@ -348,6 +382,12 @@ void Shard::completion_loop(size_t stream_idx) {
         bool ok = t.slot != nullptr;
         if (t.slot) {
             ok = gpu::event_query(t.slot->event) || gpu::event_sync(t.slot->event);
+            if (ok && t.scales_out && t.scales_n) {
+                // Must precede the slot release: the next submitter reuses
+                // the pinned h_scale buffer.
+                memcpy(t.scales_out->data() + t.scales_off, t.slot->h_scale,
+                       t.scales_n * sizeof(float));
+            }
             std::lock_guard<std::mutex> lk(task_mu_);
             t.slot->busy = false;
         }

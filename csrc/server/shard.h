@@ -39,9 +39,19 @@ class Shard {
    public:
     // Descriptor list: uniform-size block copies src[i] -> dst[i].
     struct CopyJob {
+        // kCopy moves bytes; the fp8 transforms convert while moving
+        // (bytes_per_block is the LOGICAL bf16 size in all modes — the fp8
+        // side of a transform is half that).
+        enum class Xform { kCopy = 0, kQuantBf16Fp8, kDequantFp8Bf16 };
         std::vector<uint64_t> src;
         std::vector<uint64_t> dst;
         size_t bytes_per_block = 0;
+        Xform xform = Xform::kCopy;
+        // kQuant: per-block scales, written at [scales_off..) before done().
+        std::shared_ptr<std::vector<float>> scales_out;
+        size_t scales_off = 0;
+        // kDequant: per-block scale inputs.
+        std::vector<float> scales_in;
         // Completion callback; invoked exactly once from the shard completion
         // thread (GPU) or inline (CPU shard). ok=false means the copy failed.
         std::function<void(bool ok)> done;
@@ -108,12 +118,19 @@ class Shard {
         uint64_t* h_dst = nullptr;
         uint64_t* d_src = nullptr;  // device descriptor buffers
         uint64_t* d_dst = nullptr;
+        float* h_scale = nullptr;   // fp8 transform scale staging
+        float* d_scale = nullptr;
         gpu::Event event = nullptr;
         bool busy = false;
     };
     struct PendingTask {
         Slot* slot;
         std::function<void(bool)> done;  // may be empty for chunked sub-jobs
+        // quant jobs: copy slot->h_scale[0..n) into (*scales_out)[off..)
+        // BEFORE the slot is released (the next submitter reuses h_scale).
+        std::shared_ptr<std::vector<float>> scales_out;
+        size_t scales_off = 0;
+        size_t scales_n = 0;
     };
     struct StreamCtx {
         gpu::Stream stream = nullptr;

@@ -20,7 +20,7 @@ Differences (MI355X-native build):
 import os
 import time
 import asyncio
-from typing import List, Tuple
+from typing import List, Optional, Tuple
 
 import numpy as np
 import torch
@@ -340,24 +340,38 @@ class InfinityConnection:
         """Serialize a key list for the write_pages/read_pages blob form."""
         return "\0".join(keys).encode()
 
+    FLAG_QUANT_FP8 = 2  # wire flag: store bf16 pages fp8-compressed
+
     def write_pages(self, cache: torch.Tensor, keys, offsets, page_size: int,
-                    sync: bool = False):
+                    sync: bool = False, quant: Optional[str] = None):
         """sync=True completes the write in ONE round trip (the response is
         sent when the copy finishes); sync=False (default) returns after the
-        server accepts, so uploads overlap compute — call sync() later."""
+        server accepts, so uploads overlap compute — call sync() later.
+
+        quant="fp8": store the pages quantized to fp8 e4m3 (one absmax/448
+        scale per page) at HALF the HBM footprint; requires a bf16 cache
+        tensor, and reads of these keys dequantize back to bf16
+        transparently. Doubles effective cache capacity at ~3-bit mantissa
+        precision — the usual KV-cache quantization trade."""
         self._verify(cache)
         assert self.local_connected, "write_pages uses the local GPU path"
+        flags = 0
+        if quant is not None:
+            assert quant == "fp8", f"unsupported quant mode {quant!r}"
+            assert cache.dtype == torch.bfloat16, "fp8 quant requires a bf16 cache"
+            assert page_size % 8 == 0, "fp8 quant needs page_size % 8 == 0"
+            flags = self.FLAG_QUANT_FP8
         es = cache.element_size()
         offs = np.asarray(offsets, dtype=np.uint64)
         if isinstance(keys, (bytes, bytearray, memoryview)):
             ret = self.conn.rw_local_blob(
                 self.OP_W, keys, offs, es, page_size * es,
-                cache.data_ptr(), _remap_device_id(cache), sync,
+                cache.data_ptr(), _remap_device_id(cache), sync, flags,
             )
         else:
             ret = self.conn.rw_local_keys(
                 self.OP_W, keys, offs, es, page_size * es,
-                cache.data_ptr(), _remap_device_id(cache), sync,
+                cache.data_ptr(), _remap_device_id(cache), sync, flags,
             )
         if ret < 0:
             raise Exception(f"Failed to write to infinistore, ret = {ret}")

@@ -450,3 +450,57 @@ def test_disaggregated_decode_gpu(gpu_server):
     from infinistore_amd.example.disaggregated import main as disagg_main
 
     disagg_main(port=gpu_server, device="cuda:0")
+
+
+def test_fp8_quantized_pages(gpu_server):
+    """fp8 e4m3 ingest compression: pages stored at half size with a per-page
+    scale; reads dequantize to bf16. Numerics vs torch.float8_e4m3fn
+    roundtrip with the same scaling; capacity vs plain storage."""
+    conn = local_conn(gpu_server)
+    try:
+        page_elems = 65536  # 128 KB bf16 -> 64 KB stored
+        n = 16
+        torch.manual_seed(3)
+        src = (torch.randn(n * page_elems, device="cuda:0") * 3).to(torch.bfloat16)
+        dst = torch.zeros_like(src)
+        offs = [i * page_elems for i in range(n)]
+        keys = [f"fp8-{uuid.uuid4()}-{i}" for i in range(n)]
+
+        stats0 = json.loads(conn.get_server_stats())
+        conn.write_pages(src, keys, offs, page_elems, sync=True, quant="fp8")
+        stats1 = json.loads(conn.get_server_stats())
+        conn.read_pages(dst, keys, offs, page_elems)
+        conn.sync()
+
+        # Half the blocks of a plain write (64 KB granule on this server).
+        pkeys = [f"plain-{k}" for k in keys]
+        conn.write_pages(src, pkeys, offs, page_elems, sync=True)
+        stats2 = json.loads(conn.get_server_stats())
+        q_blocks = stats1["used_blocks"] - stats0["used_blocks"]
+        p_blocks = stats2["used_blocks"] - stats1["used_blocks"]
+        assert q_blocks * 2 == p_blocks, (q_blocks, p_blocks)
+
+        for i in range(n):
+            page = src[offs[i] : offs[i] + page_elems].float()
+            scale = page.abs().max().item() / 448.0 or 1.0
+            ref = ((page / scale).to(torch.float8_e4m3fn).float() * scale
+                   ).to(torch.bfloat16).float()
+            got = dst[offs[i] : offs[i] + page_elems].float()
+            # normal-path RNE matches torch exactly; allow one subnormal ulp
+            # plus bf16 rounding at the edges
+            diff = (got - ref).abs().max().item()
+            assert diff <= scale * 2 ** -6, (i, diff, scale)
+            # end-to-end error vs the original bf16 page: fp8 mantissa step
+            rel = (got - page).abs().max().item() / page.abs().max().item()
+            assert rel <= 0.07, (i, rel)
+
+        # fabric reads of compressed entries are refused, not corrupted
+        rdma = make_client(gpu_server)
+        host = torch.zeros(page_elems, dtype=torch.bfloat16)
+        rdma.register_mr(host)
+        import pytest as _pytest
+        with _pytest.raises(Exception):
+            rdma.read_cache(host, [(keys[0], 0)], page_elems)
+        rdma.close()
+    finally:
+        conn.close()
