@@ -466,16 +466,16 @@ def test_fp8_quantized_pages(gpu_server):
         offs = [i * page_elems for i in range(n)]
         keys = [f"fp8-{uuid.uuid4()}-{i}" for i in range(n)]
 
-        stats0 = json.loads(conn.get_server_stats())
+        stats0 = json.loads(conn.get_server_stats_remote())
         conn.write_pages(src, keys, offs, page_elems, sync=True, quant="fp8")
-        stats1 = json.loads(conn.get_server_stats())
+        stats1 = json.loads(conn.get_server_stats_remote())
         conn.read_pages(dst, keys, offs, page_elems)
         conn.sync()
 
         # Half the blocks of a plain write (64 KB granule on this server).
         pkeys = [f"plain-{k}" for k in keys]
         conn.write_pages(src, pkeys, offs, page_elems, sync=True)
-        stats2 = json.loads(conn.get_server_stats())
+        stats2 = json.loads(conn.get_server_stats_remote())
         q_blocks = stats1["used_blocks"] - stats0["used_blocks"]
         p_blocks = stats2["used_blocks"] - stats1["used_blocks"]
         assert q_blocks * 2 == p_blocks, (q_blocks, p_blocks)
