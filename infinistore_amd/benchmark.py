@@ -48,6 +48,8 @@ def parse_args():
                         "--size is derived from --seq-len x --batch")
     p.add_argument("--seq-len", type=int, default=8192)
     p.add_argument("--batch", type=int, default=1)
+    p.add_argument("--quant", choices=["fp8"], default=None,
+                   help="store pages fp8-compressed (local-GPU path, bf16)")
     p.add_argument("--verify", action="store_true")
     p.add_argument("--json", action="store_true", help="print a JSON summary")
     p.add_argument("--spawn-server", action="store_true",
@@ -73,12 +75,14 @@ def run_once(args, conn, local, device):
     block_bytes = args.block_size << 10
     total_bytes = args.size << 20
     n_blocks = total_bytes // block_bytes
-    elems = total_bytes // 4
-    page_elems = block_bytes // 4
+    dt = torch.bfloat16 if args.quant else torch.float32
+    es = 2 if args.quant else 4
+    elems = total_bytes // es
+    page_elems = block_bytes // es
 
-    src = torch.rand(elems, dtype=torch.float32, device=device)
+    src = torch.rand(elems, dtype=dt, device=device)
     dst_dev = f"cuda:{args.dst_gpu}" if local else device
-    dst = torch.zeros(elems, dtype=torch.float32, device=dst_dev)
+    dst = torch.zeros(elems, dtype=dt, device=dst_dev)
     if not local:
         conn.register_mr(src)
         conn.register_mr(dst)
@@ -93,7 +97,12 @@ def run_once(args, conn, local, device):
     for s0 in range(0, n_blocks, per_step):
         sl = slice(s0, min(s0 + per_step, n_blocks))
         if local:
-            conn.local_gpu_write_cache(src, list(zip(keys[sl], offsets[sl])), page_elems)
+            if args.quant:
+                conn.write_pages(src, keys[sl], offsets[sl], page_elems,
+                                 quant=args.quant)
+            else:
+                conn.local_gpu_write_cache(src, list(zip(keys[sl], offsets[sl])),
+                                           page_elems)
         else:
             blocks = conn.allocate_rdma(keys[sl], block_bytes)
             conn.rdma_write_cache(src, offsets[sl], page_elems, blocks)

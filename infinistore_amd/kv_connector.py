@@ -50,9 +50,12 @@ class PagedKVConnector:
     """
 
     def __init__(self, host: str, port: int, model_tag: str, n_layers: int,
-                 local: bool = True):
+                 local: bool = True, quant: Optional[str] = None):
+        """quant="fp8": store KV pages fp8-compressed (half HBM per page,
+        ~3-bit mantissa; reads return bf16). Local path + bf16 KV only."""
         self.model_tag = model_tag
         self.n_layers = n_layers
+        self.quant = quant
         cfg = lib.ClientConfig(
             host_addr=host,
             service_port=port,
@@ -87,7 +90,8 @@ class PagedKVConnector:
         compute (writes are async until sync())."""
         keys = [self._key(layer, k) for k in page_keys]
         if self.local:
-            self.conn.write_pages(kv, keys, page_offsets, page_elems)
+            self.conn.write_pages(kv, keys, page_offsets, page_elems,
+                                  quant=self.quant)
         else:
             self._ensure_mr(kv)
             es = kv.element_size()
