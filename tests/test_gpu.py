@@ -450,6 +450,7 @@ def test_disaggregated_decode_gpu(gpu_server):
     from infinistore_amd.example.disaggregated import main as disagg_main
 
     disagg_main(port=gpu_server, device="cuda:0")
+    disagg_main(port=gpu_server, device="cuda:0", quant="fp8")
 
 
 def test_fp8_quantized_pages(gpu_server):
