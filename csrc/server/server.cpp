@@ -763,7 +763,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     if (quant && (page % 16 != 0)) return reply_local(c, ctx, INVALID_REQ);
     const size_t stored = quant ? page / 2 : page;
 
-    // Phase A — dedup check only (short kv_mu_ hold, prefetch-pipelined).
+    // Phase A — dedup check only (shared stripe locks, prefetch-pipelined).
     // The authoritative first-write-wins decision happens at the insert pass
     // (phase D); a key that appears between A and D just wastes one block
     // copy whose memory is released after the kernel completes.
@@ -1607,7 +1607,7 @@ std::pair<size_t, size_t> Server::compact() {
             if (!shard->submit_copy(std::move(jj))) {
                 cp.set_value(false);
             }
-            cf.wait();  // shard completion thread fulfills (never takes kv_mu_)
+            cf.wait();  // completion thread fulfills (never takes stripe locks)
         }
         // Swap pointers in the index and free the old slots.
         std::map<void*, Shard::Move*> by_old;
