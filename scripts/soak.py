@@ -28,6 +28,10 @@ def main():
     p.add_argument("--port", type=int, default=23751)
     p.add_argument("--pool-gb", type=int, default=6)
     p.add_argument("--block-kb", type=int, default=128)
+    p.add_argument("--quant-frac", type=float, default=0.0,
+                   help="fraction of write generations stored fp8-compressed "
+                        "(GPU path): exercises mixed plain/fp8 entries under "
+                        "eviction + compaction churn")
     args = p.parse_args()
 
     have_gpu = torch.cuda.is_available()
@@ -67,8 +71,10 @@ def main():
             while time.time() < stop:
                 gen += 1
                 keys = [f"t{tid}-g{gen}-{uuid.uuid4().hex[:8]}-{i}" for i in range(nb)]
+                q = "fp8" if (have_gpu and rng.random() < args.quant_frac) else None
                 if have_gpu:
-                    conn.write_pages(src, keys, offs, page_elems, sync=True)
+                    conn.write_pages(src, keys, offs, page_elems, sync=True,
+                                     quant=q)
                 else:
                     blocks = conn.allocate_rdma(keys, page_elems * 2)
                     conn.rdma_write_cache(src, [int(o) for o in offs], page_elems, blocks)
