@@ -1571,23 +1571,38 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
 // ---------------------------------------------------------------------------
 std::pair<size_t, size_t> Server::compact() {
     if (!running_.load()) return {0, 0};
-    // Maintenance op: ALL stripe locks are held (taken in index order) for
-    // the WHOLE plan+copy+swap so no request thread can start a read
-    // against a block while it moves (requests block for the few ms this
-    // takes). This is the only path holding more than one stripe lock.
+    // Defragmentation in three phases per shard. The stripe locks are NOT
+    // held across the copy wait: a write handler holds its job's commit
+    // mutex while its insert pass takes stripe locks, and the completion
+    // thread that fulfills our copy futures can be blocked on that same
+    // commit mutex — holding the stripes here closed that into a deadlock
+    // (found by the mixed fp8/eviction soak).
+    //  1. snapshot (all stripe locks, in order): pick committed idle
+    //     entries, take a Ref on each so neither delete nor eviction can
+    //     free the old block mid-copy.
+    //  2. copy (no locks): batched kernel moves into the planned slots.
+    //  3. swap (all stripe locks): re-check each entry is still present and
+    //     still idle (a reader that started during the copy captured the
+    //     OLD pointer and holds a ref — skip those), then flip ptr/pool and
+    //     free the old slot. Unapplied moves free their new slot instead.
     size_t moved = 0, bytes = 0;
-    std::vector<std::unique_lock<std::shared_mutex>> locks;
-    locks.reserve(kStripes);
-    for (auto& st : kv_) locks.emplace_back(st.mu);
     for (auto& shard_up : shards_) {
         Shard* shard = shard_up.get();
+        std::vector<Ref<BlockEntry>> held;
         std::vector<std::pair<void*, size_t>> movable;
-        for (auto& st : kv_)
-            st.map.for_each([&](std::string_view, Ref<BlockEntry>& val) {
-                BlockEntry* e = val.get();
-                if (e->shard == shard && e->committed && e->ref_count() == 1)
-                    movable.push_back({e->ptr, e->size});
-            });
+        {
+            std::vector<std::unique_lock<std::shared_mutex>> locks;
+            locks.reserve(kStripes);
+            for (auto& st : kv_) locks.emplace_back(st.mu);
+            for (auto& st : kv_)
+                st.map.for_each([&](std::string_view, Ref<BlockEntry>& val) {
+                    BlockEntry* e = val.get();
+                    if (e->shard == shard && e->committed && e->ref_count() == 1) {
+                        held.push_back(val);
+                        movable.push_back({e->ptr, e->size});
+                    }
+                });
+        }
         if (movable.empty()) continue;
         auto moves = shard->plan_compaction(movable);
         if (moves.empty()) continue;
@@ -1599,6 +1614,7 @@ std::pair<size_t, size_t> Server::compact() {
             j.src.push_back(reinterpret_cast<uint64_t>(m.old_ptr));
             j.dst.push_back(reinterpret_cast<uint64_t>(m.new_ptr));
         }
+        bool copies_ok = true;
         for (auto& [sz, j2] : by_size) {
             std::promise<bool> cp;
             auto cf = cp.get_future();
@@ -1607,23 +1623,37 @@ std::pair<size_t, size_t> Server::compact() {
             if (!shard->submit_copy(std::move(jj))) {
                 cp.set_value(false);
             }
-            cf.wait();  // completion thread fulfills (never takes stripe locks)
+            if (!cf.get()) copies_ok = false;
         }
-        // Swap pointers in the index and free the old slots.
+
         std::map<void*, Shard::Move*> by_old;
         for (auto& m : moves) by_old[m.old_ptr] = &m;
-        for (auto& st : kv_)
-            st.map.for_each([&](std::string_view, Ref<BlockEntry>& val) {
-                BlockEntry* e = val.get();
-                auto it = by_old.find(e->ptr);
-                if (it == by_old.end() || e->shard != shard) return;
-                Shard::Move* m = it->second;
-                e->ptr = m->new_ptr;
-                e->pool_idx = m->pool_idx;
-                shard->deallocate(m->old_ptr, m->size, m->pool_idx);
-                moved++;
-                bytes += m->size;
-            });
+        {
+            std::vector<std::unique_lock<std::shared_mutex>> locks;
+            locks.reserve(kStripes);
+            for (auto& st : kv_) locks.emplace_back(st.mu);
+            for (auto& st : kv_)
+                st.map.for_each([&](std::string_view, Ref<BlockEntry>& val) {
+                    BlockEntry* e = val.get();
+                    auto it = by_old.find(e->ptr);
+                    if (it == by_old.end() || e->shard != shard) return;
+                    // map ref + our `held` ref = 2; more means an in-flight
+                    // read captured the old pointer during the copy.
+                    if (!copies_ok || e->ref_count() > 2) return;
+                    Shard::Move* m = it->second;
+                    e->ptr = m->new_ptr;
+                    e->pool_idx = m->pool_idx;
+                    shard->deallocate(m->old_ptr, m->size, m->pool_idx);
+                    m->old_ptr = nullptr;  // applied
+                    moved++;
+                    bytes += m->size;
+                });
+        }
+        for (auto& m : moves)
+            if (m.old_ptr != nullptr)  // unapplied: release the planned slot
+                shard->deallocate(m.new_ptr, m.size, m.pool_idx);
+        // held refs drop here; entries erased during the copy free their
+        // (old) blocks now.
     }
     return {moved, bytes};
 }
