@@ -52,6 +52,7 @@ struct ServerConfigPy {
     bool cpu_only = false;            // force CPU pool even if GPUs exist
     int cpu_shards = 1;               // CPU-mode shard count
     bool auto_evict = false;          // LRU-evict on allocation failure
+    int ttl_seconds = 0;              // key time-to-live (0 = forever)
     int io_threads = 3;               // worker IO loops (0 = single loop)
     int extend_size = 10;             // GB per auto-extend arena
 };
@@ -73,6 +74,7 @@ bool start_server(const ServerConfigPy& cfg) {
     opt.ib_port = cfg.ib_port;
     opt.link_type = cfg.link_type;
     opt.auto_evict = cfg.auto_evict;
+    opt.ttl_seconds = cfg.ttl_seconds;
     opt.io_threads = cfg.io_threads;
     opt.extend_bytes = static_cast<size_t>(cfg.extend_size) << 30;
     if (!cfg.cpu_only && gpu::available()) {
@@ -157,6 +159,7 @@ PYBIND11_MODULE(_native, m) {
         .def_readwrite("cpu_only", &ServerConfigPy::cpu_only)
         .def_readwrite("cpu_shards", &ServerConfigPy::cpu_shards)
         .def_readwrite("auto_evict", &ServerConfigPy::auto_evict)
+        .def_readwrite("ttl_seconds", &ServerConfigPy::ttl_seconds)
         .def_readwrite("io_threads", &ServerConfigPy::io_threads)
         .def_readwrite("extend_size", &ServerConfigPy::extend_size);
 

@@ -639,6 +639,12 @@ Shard* Server::shard_least_used() {
     return best;
 }
 
+uint32_t Server::now_sec() {
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return static_cast<uint32_t>(ts.tv_sec);
+}
+
 void Server::erase_entries(const std::vector<Ref<BlockEntry>>& entries) {
     // Rare path (allocation failure / failed copy): remove entries from the
     // index by identity. O(map) scan, but keys are not kept around on the
@@ -678,7 +684,10 @@ size_t Server::evict_lru(Shard* shard, size_t bytes) {
                                  BlockEntry* e = val.get();
                                  if (e->shard == shard && e->committed && e->ref_count() == 1)
                                      sample.push_back(
-                                         {e->last_access.load(std::memory_order_relaxed), key});
+                                         {expired(e) ? 0  // expired: evict first
+                                                     : e->last_access.load(
+                                                           std::memory_order_relaxed),
+                                          key});
                                  return sample.size() < 128;
                              });
             scanned += window;
@@ -787,7 +796,8 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         for (size_t i = 0; i < ln; i++) {
             if (i + kPf < ln) m.prefetch(hashes[list[i + kPf]]);
             uint32_t gi = list[i];
-            if (!m.find_hashed(msg.blocks[gi].first, hashes[gi])) fresh.push_back(gi);
+            Ref<BlockEntry>* v = m.find_hashed(msg.blocks[gi].first, hashes[gi]);
+            if (!v || expired(v->get())) fresh.push_back(gi);
         }
     }
     auto p1 = std::chrono::steady_clock::now();
@@ -845,6 +855,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         e->shard = shard;
         e->committed = false;
         e->fp8 = quant;
+        e->born_sec = now_sec();
         e->last_access.store(t, std::memory_order_relaxed);
         entries->emplace_back(e);
         job.src.push_back(
@@ -932,8 +943,13 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
                 if (i + kPf < ln) m.prefetch(hashes[fresh[list[i + kPf]]]);
                 uint32_t p = list[i];
                 bool inserted = false;
-                m.emplace_hashed(msg.blocks[fresh[p]].first, hashes[fresh[p]], (*entries)[p],
-                                 &inserted);
+                Ref<BlockEntry>* slot2 = m.emplace_hashed(msg.blocks[fresh[p]].first,
+                                                          hashes[fresh[p]], (*entries)[p],
+                                                          &inserted);
+                if (!inserted && slot2 && expired(slot2->get())) {
+                    *slot2 = (*entries)[p];  // expired loser: replace in place
+                    inserted = true;
+                }
                 (*won)[p] = inserted ? 1 : 0;
             }
         }
@@ -986,7 +1002,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
                 if (i + kPf < ln) m.prefetch(hashes[list[i + kPf]]);
                 auto& b = msg.blocks[list[i]];
                 Ref<BlockEntry>* v = m.find_hashed(b.first, hashes[list[i]]);
-                if (!v || !(*v)->committed) {
+                if (!v || !(*v)->committed || expired(v->get())) {
                     return reply_local(c, ctx, KEY_NOT_FOUND);
                 }
                 BlockEntry* e = v->get();
@@ -1285,7 +1301,8 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
         auto& st = kv_[stripe_of(h)];
         {
             std::shared_lock<std::shared_mutex> sl(st.mu);
-            if (st.map.find_hashed(key, h) != nullptr) {
+            Ref<BlockEntry>* v0 = st.map.find_hashed(key, h);
+            if (v0 != nullptr && !expired(v0->get())) {
                 blocks.push_back({0, 0, 0});  // FAKE block: dup key, client skips
                 continue;
             }
@@ -1313,12 +1330,17 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
         e->size = page;
         e->pool_idx = pool_idx;
         e->shard = shard;
+        e->born_sec = now_sec();
         e->last_access.store(tick(), std::memory_order_relaxed);
         Ref<BlockEntry> ref(e);
         bool ins = false;
         {
             std::lock_guard<std::shared_mutex> lk(st.mu);
-            st.map.emplace_hashed(key, h, ref, &ins);
+            Ref<BlockEntry>* slot2 = st.map.emplace_hashed(key, h, ref, &ins);
+            if (!ins && slot2 && expired(slot2->get())) {
+                *slot2 = ref;  // expired entry: replace
+                ins = true;
+            }
         }
         if (!ins) {  // raced with another writer: first write wins
             blocks.push_back({0, 0, 0});
@@ -1351,7 +1373,7 @@ bool Server::collect_read_entries(const std::vector<std::string>& keys,
         auto& st = kv_[stripe_of(h)];
         std::shared_lock<std::shared_mutex> lk(st.mu);
         Ref<BlockEntry>* v = st.map.find_hashed(key, h);
-        if (!v || !(*v)->committed) return false;
+        if (!v || !(*v)->committed || expired(v->get())) return false;
         // fp8-compressed entries are a local-GPU-path feature: the TCP/verbs
         // fabric moves raw bytes and cannot dequantize on the way out.
         if ((*v)->fp8) return false;
@@ -1506,7 +1528,7 @@ void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body) {
         auto& st = kv_[stripe_of(h)];
         std::shared_lock<std::shared_mutex> lk(st.mu);
         Ref<BlockEntry>* v = st.map.find_hashed(key, h);
-        exists = v && (*v)->committed;
+        exists = v && (*v)->committed && !expired(v->get());
     }
     send_status(c, exists ? 0 : 1);
 }
@@ -1525,7 +1547,7 @@ void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body) {
         auto& st = kv_[stripe_of(h)];
         std::shared_lock<std::shared_mutex> lk(st.mu);
         Ref<BlockEntry>* v = st.map.find_hashed(keys[i], h);
-        return v && (*v)->committed;
+        return v && (*v)->committed && !expired(v->get());
     };
     long left = 0, right = static_cast<long>(keys.size());
     while (left < right) {

@@ -72,6 +72,10 @@ struct ServerOptions {
     // reference only fails the write; engines can also evict explicitly via
     // delete_keys).
     bool auto_evict = false;
+    // Time-to-live for stored keys, seconds; 0 = keys live until
+    // deleted/evicted/purged (extension — production prefix caches expire
+    // stale prompts; the reference has no expiry at all).
+    int ttl_seconds = 0;
 };
 
 class Server;
@@ -90,6 +94,7 @@ struct BlockEntry : RefCounted {
     // (half the logical bf16 page); reads dequantize with `scale`.
     bool fp8 = false;
     float scale = 1.f;
+    uint32_t born_sec = 0;  // CLOCK_MONOTONIC seconds at insert (TTL)
     // LRU tick (auto_evict); atomic: bumped under the SHARED kv lock by
     // concurrent readers.
     std::atomic<uint64_t> last_access{0};
@@ -244,6 +249,13 @@ class Server {
     size_t evict_lru(Shard* shard, size_t bytes);
     void erase_entries(const std::vector<Ref<BlockEntry>>& entries);
     uint64_t tick() { return access_tick_.fetch_add(1, std::memory_order_relaxed); }
+    static uint32_t now_sec();
+    // TTL check (lazy expiry): true when the entry is past its lifetime —
+    // lookups treat it as absent; the evictor reclaims it first.
+    bool expired(const BlockEntry* e) const {
+        return opt_.ttl_seconds > 0 &&
+               now_sec() - e->born_sec > static_cast<uint32_t>(opt_.ttl_seconds);
+    }
     std::atomic<uint64_t> access_tick_{1};
     std::atomic<uint64_t> n_evicted_{0};
 

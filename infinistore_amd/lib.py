@@ -107,6 +107,7 @@ class ServerConfig(_native.ServerConfig):
         self.cpu_only = kwargs.get("cpu_only", False)
         self.cpu_shards = kwargs.get("cpu_shards", 1)
         self.auto_evict = kwargs.get("auto_evict", False)
+        self.ttl_seconds = kwargs.get("ttl_seconds", 0)
         self.io_threads = kwargs.get("io_threads", 3)
         self.extend_size = kwargs.get("extend_size", 10)
 

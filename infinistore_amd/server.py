@@ -138,6 +138,9 @@ def parse_args():
                    help="worker IO loop threads (0 = single loop)")
     p.add_argument("--auto-increase", action="store_true",
                    help="extend the pool automatically when nearly full")
+    p.add_argument("--ttl-seconds", type=int, default=0,
+                   help="expire keys this many seconds after insert "
+                        "(0 = keys live until deleted/evicted/purged)")
     p.add_argument("--auto-evict", action="store_true",
                    help="LRU-evict committed idle keys when the pool is full")
     p.add_argument("--devices", default="",
@@ -183,6 +186,7 @@ def main():
         io_threads=args.io_threads,
         auto_increase=args.auto_increase,
         auto_evict=args.auto_evict,
+        ttl_seconds=args.ttl_seconds,
         devices=devices,
         cpu_only=args.cpu_only,
         dev_name=args.dev_name,

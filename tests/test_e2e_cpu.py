@@ -496,3 +496,51 @@ def test_example_client_async(cpu_server):
     from infinistore_amd.example import client_async
 
     asyncio.run(client_async.main(cpu_server))
+
+
+def test_ttl_expiry(ports):
+    """TTL (extension): keys expire `ttl_seconds` after insert — lookups
+    treat them as absent, an expired key can be overwritten with fresh
+    data, and the evictor reclaims expired entries first."""
+    import time
+
+    service_port, manage_port = ports
+    cfg = ifs.ServerConfig(
+        service_port=service_port, manage_port=manage_port,
+        prealloc_size=1, minimal_allocate_size=16, cpu_only=True,
+        ttl_seconds=2, auto_evict=True,
+    )
+    ifs.register_server(cfg)
+    try:
+        conn = make_client(service_port)
+        src = torch.arange(4096, dtype=torch.float32)
+        dst = torch.zeros_like(src)
+        conn.register_mr(src)
+        conn.register_mr(dst)
+        blocks = conn.allocate_rdma(["ttl-key"], 4096 * 4)
+        conn.rdma_write_cache(src, [0], 4096, blocks)
+        conn.sync()
+        assert conn.check_exist("ttl-key")
+        conn.read_cache(dst, [("ttl-key", 0)], 4096)
+        conn.sync()
+        assert torch.equal(src, dst)
+
+        time.sleep(3)  # past the 2 s TTL
+        assert not conn.check_exist("ttl-key")
+        with pytest.raises(Exception):
+            conn.read_cache(dst, [("ttl-key", 0)], 4096)
+
+        # the expired key is overwritable (allocate returns a REAL block,
+        # not the dup-key FAKE) and fresh data round-trips
+        src2 = src + 1
+        conn.register_mr(src2)
+        blocks2 = conn.allocate_rdma(["ttl-key"], 4096 * 4)
+        assert blocks2[0] != (0, 0) and tuple(blocks2[0])[1] != 0
+        conn.rdma_write_cache(src2, [0], 4096, blocks2)
+        conn.sync()
+        conn.read_cache(dst, [("ttl-key", 0)], 4096)
+        conn.sync()
+        assert torch.equal(src2, dst)
+        conn.close()
+    finally:
+        ifs.unregister_server()
