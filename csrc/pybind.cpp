@@ -377,6 +377,24 @@ PYBIND11_MODULE(_native, m) {
     m.def("get_kvmap_len", &get_kvmap_len_py);
     m.def("server_stats", &server_stats_py);
     m.def("server_compact", &server_compact_py, py::call_guard<py::gil_scoped_release>());
+    m.def("server_snapshot", [](const std::string& path) {
+        std::pair<size_t, size_t> r{0, 0};
+        bool ok;
+        {
+            py::gil_scoped_release rel;
+            ok = g_server && g_server->snapshot(path, &r);
+        }
+        return py::make_tuple(ok, r.first, r.second);
+    });
+    m.def("server_restore", [](const std::string& path) {
+        std::pair<size_t, size_t> r{0, 0};
+        bool ok;
+        {
+            py::gil_scoped_release rel;
+            ok = g_server && g_server->restore(path, &r);
+        }
+        return py::make_tuple(ok, r.first, r.second);
+    });
 
     // ---- logging ----
     m.def("set_log_level", [](const std::string& lvl) { set_log_level(lvl.c_str()); });

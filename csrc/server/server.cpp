@@ -1699,6 +1699,137 @@ size_t Server::purge() {
     return n;
 }
 
+// ---- snapshot / restore ----------------------------------------------------
+namespace {
+constexpr uint64_t kSnapMagic = 0x53494653504e3150ull;  // "SIFSPN1P"
+struct SnapEntryHdr {
+    uint32_t key_len;
+    uint32_t fp8;
+    uint64_t size;  // stored bytes
+    float scale;
+    uint32_t _pad;
+};
+}  // namespace
+
+bool Server::snapshot(const std::string& path, std::pair<size_t, size_t>* out) {
+    *out = {0, 0};
+    // Pin a consistent set of committed entries (refs keep blocks alive and
+    // make compaction skip them — it requires ref_count()==1), then stream
+    // them out without holding any lock.
+    std::vector<std::pair<std::string, Ref<BlockEntry>>> pinned;
+    for (auto& st : kv_) {
+        std::shared_lock<std::shared_mutex> lk(st.mu);
+        st.map.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
+            if (val->committed && !expired(val.get()))
+                pinned.emplace_back(std::string(key), val);
+        });
+    }
+    FILE* f = fopen(path.c_str(), "wb");
+    if (!f) {
+        ERROR("snapshot: cannot open %s", path.c_str());
+        return false;
+    }
+    uint64_t magic = kSnapMagic, count = pinned.size();
+    bool ok = fwrite(&magic, 8, 1, f) == 1 && fwrite(&count, 8, 1, f) == 1;
+    std::vector<uint8_t> buf;
+    for (auto& [key, ref] : pinned) {
+        if (!ok) break;
+        BlockEntry* e = ref.get();
+        SnapEntryHdr h{static_cast<uint32_t>(key.size()), e->fp8 ? 1u : 0u, e->size,
+                       e->scale, 0};
+        buf.resize(e->size);
+        if (e->shard->on_gpu()) {
+            if (!gpu::memcpy_d2h(buf.data(), e->ptr, e->size)) {
+                ok = false;
+                break;
+            }
+        } else {
+            memcpy(buf.data(), e->ptr, e->size);
+        }
+        ok = fwrite(&h, sizeof(h), 1, f) == 1 && fwrite(key.data(), 1, key.size(), f) == key.size() &&
+             fwrite(buf.data(), 1, e->size, f) == e->size;
+        out->first++;
+        out->second += e->size;
+    }
+    fclose(f);
+    if (!ok) ERROR("snapshot: write failed at entry %zu", out->first);
+    return ok;
+}
+
+bool Server::restore(const std::string& path, std::pair<size_t, size_t>* out) {
+    *out = {0, 0};
+    FILE* f = fopen(path.c_str(), "rb");
+    if (!f) {
+        ERROR("restore: cannot open %s", path.c_str());
+        return false;
+    }
+    uint64_t magic = 0, count = 0;
+    if (fread(&magic, 8, 1, f) != 1 || magic != kSnapMagic || fread(&count, 8, 1, f) != 1) {
+        ERROR("restore: bad snapshot header");
+        fclose(f);
+        return false;
+    }
+    std::vector<uint8_t> buf;
+    std::string key;
+    bool ok = true;
+    for (uint64_t i = 0; i < count && ok; i++) {
+        SnapEntryHdr h{};
+        if (fread(&h, sizeof(h), 1, f) != 1 || h.key_len > 4096 || h.size > (1u << 30)) {
+            ok = false;
+            break;
+        }
+        key.resize(h.key_len);
+        buf.resize(h.size);
+        if (fread(key.data(), 1, h.key_len, f) != h.key_len ||
+            fread(buf.data(), 1, h.size, f) != h.size) {
+            ok = false;
+            break;
+        }
+        Shard* shard = shard_least_used();
+        void* ptr = nullptr;
+        int pool_idx = -1;
+        if (!shard->allocate(h.size, 1, [&](void* p, int idx) {
+                ptr = p;
+                pool_idx = idx;
+            })) {
+            WARN("restore: pool full after %zu entries", out->first);
+            break;  // partial restore is still useful
+        }
+        bool copied = shard->on_gpu() ? gpu::memcpy_h2d(ptr, buf.data(), h.size)
+                                      : (memcpy(ptr, buf.data(), h.size), true);
+        if (!copied) {
+            shard->deallocate(ptr, h.size, pool_idx);
+            ok = false;
+            break;
+        }
+        auto* e = new BlockEntry();
+        e->ptr = ptr;
+        e->size = h.size;
+        e->pool_idx = pool_idx;
+        e->shard = shard;
+        e->fp8 = h.fp8 != 0;
+        e->scale = h.scale;
+        e->born_sec = now_sec();
+        e->committed = true;
+        e->last_access.store(tick(), std::memory_order_relaxed);
+        Ref<BlockEntry> ref(e);
+        uint64_t hh = KvMap::hash_of(key);
+        auto& st = kv_[stripe_of(hh)];
+        bool ins = false;
+        {
+            std::lock_guard<std::shared_mutex> lk(st.mu);
+            st.map.emplace_hashed(key, hh, ref, &ins);
+        }
+        if (ins) {
+            out->first++;
+            out->second += h.size;
+        }
+        // !ins: live key wins over the snapshot; `ref` frees the block.
+    }
+    fclose(f);
+    return ok;
+}
+
 std::string Server::stats_json() {
     char buf[1024];
     size_t used = 0, total = 0, frag = 0;

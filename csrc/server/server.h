@@ -126,6 +126,12 @@ class Server {
     // Defragment the pools: move committed, idle blocks to lower addresses
     // using the batched copy kernel. Returns (moved_blocks, moved_bytes).
     std::pair<size_t, size_t> compact();
+    // Warm-restart persistence (extension; the reference cache is purely
+    // volatile): dump every committed, unexpired entry (key, metadata,
+    // page bytes) to `path` / load such a dump back into the pool.
+    // Returns (entries, payload_bytes) or (0,0) + false on IO failure.
+    bool snapshot(const std::string& path, std::pair<size_t, size_t>* out);
+    bool restore(const std::string& path, std::pair<size_t, size_t>* out);
     int num_shards() const { return static_cast<int>(shards_.size()); }
 
     struct Conn;  // defined below (file-local helpers + server_verbs use it)

@@ -23,6 +23,8 @@ from .lib import (
     get_kvmap_len,
     get_server_stats,
     compact_pool,
+    snapshot_pool,
+    restore_pool,
     fingerprint_blocks,
 )
 
@@ -43,6 +45,8 @@ __all__ = [
     "purge_kv_map",
     "get_kvmap_len",
     "get_server_stats",
+    "snapshot_pool",
+    "restore_pool",
     "compact_pool",
     "fingerprint_blocks",
 ]

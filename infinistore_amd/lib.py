@@ -167,6 +167,24 @@ def purge_kv_map():
     return _native.purge_kv_map()
 
 
+def snapshot_pool(path: str):
+    """Dump every committed page to `path` for a warm restart (extension —
+    the reference cache is volatile). Returns (entries, payload_bytes)."""
+    ok, n, b = _native.server_snapshot(path)
+    if not ok:
+        raise Exception(f"snapshot to {path} failed")
+    return n, b
+
+
+def restore_pool(path: str):
+    """Load a snapshot back into the pool (existing keys win; stops cleanly
+    when the pool fills). Returns (entries, payload_bytes)."""
+    ok, n, b = _native.server_restore(path)
+    if not ok:
+        raise Exception(f"restore from {path} failed")
+    return n, b
+
+
 def get_server_stats():
     """JSON string with server counters (extension over the reference)."""
     return _native.server_stats()

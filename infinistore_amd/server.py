@@ -49,6 +49,18 @@ def make_app(config: ServerConfig):
 
         return json.loads(lib.get_server_stats())
 
+    @app.post("/snapshot")
+    async def snapshot(path: str):
+        loop = asyncio.get_running_loop()
+        n, b = await loop.run_in_executor(None, lib.snapshot_pool, path)
+        return {"entries": n, "bytes": b}
+
+    @app.post("/restore")
+    async def restore(path: str):
+        loop = asyncio.get_running_loop()
+        n, b = await loop.run_in_executor(None, lib.restore_pool, path)
+        return {"entries": n, "bytes": b}
+
     @app.post("/compact")
     async def compact():
         loop = asyncio.get_running_loop()
@@ -138,6 +150,8 @@ def parse_args():
                    help="worker IO loop threads (0 = single loop)")
     p.add_argument("--auto-increase", action="store_true",
                    help="extend the pool automatically when nearly full")
+    p.add_argument("--restore-from", default="",
+                   help="load a snapshot file into the pool at startup")
     p.add_argument("--ttl-seconds", type=int, default=0,
                    help="expire keys this many seconds after insert "
                         "(0 = keys live until deleted/evicted/purged)")
@@ -195,6 +209,7 @@ def main():
     )
     config.verify()
     lib.check_supported()
+    _restore_path = args.restore_from
     if args.prevent_oom:
         prevent_oom()
 
@@ -203,6 +218,10 @@ def main():
 
         check_p2p_access()
     register_server(config)
+    if _restore_path:
+        n, b = lib.restore_pool(_restore_path)
+        print(f"restored {n} entries ({b >> 20} MB) from {_restore_path}",
+              flush=True)
     print(f"infinistore-amd serving on :{config.service_port} "
           f"(manage :{config.manage_port})", flush=True)
 

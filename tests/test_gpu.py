@@ -505,3 +505,49 @@ def test_fp8_quantized_pages(gpu_server):
         rdma.close()
     finally:
         conn.close()
+
+
+def test_snapshot_restore_hbm(tmp_path):
+    """Snapshot/restore of HBM-resident pages (D2H stream out, H2D back),
+    including fp8-compressed entries whose scales must survive — run in a
+    subprocess that hosts the server in-process so the python snapshot API
+    is reachable."""
+    code = subprocess.run(
+        [sys.executable, "-c", f"""
+import torch, uuid
+import infinistore_amd as ifs
+from conftest import free_port
+import sys
+sys.path.insert(0, {str(REPO + '/tests')!r})
+port = free_port()
+ifs.register_server(ifs.ServerConfig(service_port=port, manage_port=port+1,
+                                     prealloc_size=2, minimal_allocate_size=64))
+cfg = ifs.ClientConfig(host_addr="127.0.0.1", service_port=port,
+                       connection_type=ifs.TYPE_LOCAL_GPU)
+c = ifs.InfinityConnection(cfg); c.connect()
+n, pe = 8, 65536
+src = torch.randn(n*pe, dtype=torch.bfloat16, device="cuda:0")
+dst = torch.zeros_like(src)
+offs = [i*pe for i in range(n)]
+keys = [f"s-{{i}}" for i in range(n)]
+qkeys = [f"q-{{i}}" for i in range(n)]
+c.write_pages(src, keys, offs, pe, sync=True)
+c.write_pages(src, qkeys, offs, pe, sync=True, quant="fp8")
+snap = {str(tmp_path / 'hbm.snap')!r}
+sn, sb = ifs.snapshot_pool(snap)
+assert sn == 2*n, sn
+ifs.purge_kv_map()
+rn, rb = ifs.restore_pool(snap)
+assert rn == 2*n, rn
+c.read_pages(dst, keys, offs, pe); c.sync()
+assert torch.equal(src, dst)
+dst.zero_()
+c.read_pages(dst, qkeys, offs, pe); c.sync()
+err = (dst.float()-src.float()).abs().max() / src.abs().max()
+assert err < 0.08, err.item()
+c.close(); ifs.unregister_server()
+print("SNAP OK")
+"""],
+        cwd=os.path.join(REPO, "tests"), timeout=180,
+    ).returncode
+    assert code == 0
