@@ -514,11 +514,9 @@ def test_snapshot_restore_hbm(tmp_path):
     is reachable."""
     code = subprocess.run(
         [sys.executable, "-c", f"""
-import torch, uuid
+import torch
 import infinistore_amd as ifs
 from conftest import free_port
-import sys
-sys.path.insert(0, {str(REPO + '/tests')!r})
 port = free_port()
 ifs.register_server(ifs.ServerConfig(service_port=port, manage_port=port+1,
                                      prealloc_size=2, minimal_allocate_size=64))
@@ -548,6 +546,7 @@ assert err < 0.08, err.item()
 c.close(); ifs.unregister_server()
 print("SNAP OK")
 """],
-        cwd=os.path.join(REPO, "tests"), timeout=180,
+        cwd=REPO, env={**os.environ, "PYTHONPATH": REPO + ":" + os.path.join(REPO, "tests")},
+        timeout=180,
     ).returncode
     assert code == 0
