@@ -963,7 +963,12 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         fprintf(stderr, "[pdbg] dedup=%.0f alloc=%.0f build+insert=%.0f\n", us(p0, p1),
                 us(p1, p2), us(p2, p3));
     }
-    if (!submitted) finish_task(c, /*on_owner=*/!ctx.shm);
+    if (!submitted) {
+        // The job never launched (e.g. a transform on a CPU shard): the done
+        // callback will not fire, so answer sync-response writers here.
+        if (sync_resp) reply_local(c, ctx, INTERNAL_ERROR);
+        finish_task(c, /*on_owner=*/!ctx.shm);
+    }
 }
 
 void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
