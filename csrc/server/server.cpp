@@ -250,6 +250,32 @@ bool Server::start() {
     stop_requested_.store(false);
     main_io_.start();
     for (auto& w : workers_) w->start();
+    if (opt_.ttl_seconds > 0) {
+        ttl_thread_ = std::thread([this] {
+            // Sweep every ttl/4 (>=1 s): one bounded pass over each stripe,
+            // erasing expired idle entries. Lookups already treat expired
+            // keys as absent; this just returns their memory early.
+            auto period = std::chrono::seconds(std::max(1, opt_.ttl_seconds / 4));
+            std::unique_lock<std::mutex> lk(ttl_mu_);
+            while (!ttl_cv_.wait_for(lk, period,
+                                     [this] { return stop_requested_.load(); })) {
+                lk.unlock();
+                size_t swept = 0;
+                for (auto& st : kv_) {
+                    std::lock_guard<std::shared_mutex> ex(st.mu);
+                    std::vector<std::string> victims;
+                    st.map.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
+                        if (expired(val.get()) && val->ref_count() == 1)
+                            victims.emplace_back(key);
+                    });
+                    for (auto& k : victims) st.map.erase(k);
+                    swept += victims.size();
+                }
+                if (swept) DEBUG("ttl sweep: %zu expired entries erased", swept);
+                lk.lock();
+            }
+        });
+    }
     INFO("server listening on 0.0.0.0:%d with %zu shard(s), %zu IO loops",
          opt_.service_port, shards_.size(), workers_.size() + 1);
     return true;
@@ -265,6 +291,8 @@ void Server::IoLoop::start() {
 void Server::stop() {
     if (!running_.load()) return;
     stop_requested_.store(true);
+    ttl_cv_.notify_all();
+    if (ttl_thread_.joinable()) ttl_thread_.join();
     for (auto& w : workers_) w->request_stop();
     main_io_.request_stop();
     for (auto& w : workers_)

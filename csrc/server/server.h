@@ -26,6 +26,7 @@
 
 #include <array>
 #include <atomic>
+#include <condition_variable>
 #include <cstdint>
 #include <functional>
 #include <map>
@@ -278,6 +279,11 @@ class Server {
     std::string pipe_path_;
     std::atomic<bool> running_{false};
     std::atomic<bool> stop_requested_{false};
+    // TTL sweeper (only when ttl_seconds > 0): periodically erases expired
+    // entries so memory frees proactively, not just on lookup/eviction.
+    std::thread ttl_thread_;
+    std::mutex ttl_mu_;
+    std::condition_variable ttl_cv_;
 
     // Open-addressing key index with arena-stored keys (csrc/server/kvmap.h)
     // — node-based maps measured ~350 µs of insert cost per 2048-key write.

@@ -529,6 +529,9 @@ def test_ttl_expiry(ports):
         assert not conn.check_exist("ttl-key")
         with pytest.raises(Exception):
             conn.read_cache(dst, [("ttl-key", 0)], 4096)
+        # the background sweeper reclaims the memory proactively (period
+        # ttl/4, min 1 s — by now it has run)
+        assert ifs.get_kvmap_len() == 0
 
         # the expired key is overwritable (allocate returns a REAL block,
         # not the dup-key FAKE) and fresh data round-trips
