@@ -530,7 +530,11 @@ def test_ttl_expiry(ports):
         with pytest.raises(Exception):
             conn.read_cache(dst, [("ttl-key", 0)], 4096)
         # the background sweeper reclaims the memory proactively (period
-        # ttl/4, min 1 s — by now it has run)
+        # ttl/4, min 1 s; poll past the second-granularity boundary)
+        for _ in range(60):
+            if ifs.get_kvmap_len() == 0:
+                break
+            time.sleep(0.2)
         assert ifs.get_kvmap_len() == 0
 
         # the expired key is overwritable (allocate returns a REAL block,
