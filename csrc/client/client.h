@@ -97,6 +97,8 @@ class ClientConn {
     std::string get_stats();
 
     bool rdma_connected() const { return rdma_connected_; }
+    // Which data plane won the OP_RDMA_EXCHANGE negotiation.
+    bool using_verbs() const { return verbs_ != nullptr; }
     // Shared-memory ring transport active (same-host fast path for the
     // packed local ops; csrc/core/shm_ring.h).
     bool shm_active() const { return shm_active_; }

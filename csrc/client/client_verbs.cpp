@@ -136,7 +136,11 @@ VerbsClient::~VerbsClient() = default;
 std::unique_ptr<VerbsClient> VerbsClient::establish(int fd, const ClientConfigC& cfg,
                                                     bool* attempted) {
     *attempted = false;
-    if (!vf::compiled_in() || cfg.link_type == "TCP") return nullptr;
+    if (!vf::compiled_in() || cfg.link_type == "TCP") {
+        DEBUG("verbs skipped (compiled_in=%d link=%s)", vf::compiled_in() ? 1 : 0,
+              cfg.link_type.c_str());
+        return nullptr;
+    }
     vf::Options o;
     o.dev_name = cfg.dev_name;
     o.ib_port = cfg.ib_port;
@@ -144,7 +148,10 @@ std::unique_ptr<VerbsClient> VerbsClient::establish(int fd, const ClientConfigC&
     {
         std::lock_guard<std::mutex> lk(g_drv_mu);
         if (!client_driver().ready()) {
-            if (!vf::device_available(o)) return nullptr;
+            if (!vf::device_available(o)) {
+                DEBUG("verbs skipped: no usable RDMA device");
+                return nullptr;
+            }
             std::string err;
             if (!client_driver().init(o, &err)) {
                 WARN("client verbs driver init failed: %s", err.c_str());

@@ -368,7 +368,8 @@ PYBIND11_MODULE(_native, m) {
              py::call_guard<py::gil_scoped_release>())
         .def("delete_keys", &ClientConn::delete_keys, py::call_guard<py::gil_scoped_release>())
         .def("get_stats", &ClientConn::get_stats, py::call_guard<py::gil_scoped_release>())
-        .def("shm_active", &ClientConn::shm_active);
+        .def("shm_active", &ClientConn::shm_active)
+        .def("using_verbs", &ClientConn::using_verbs);
 
     // ---- server ----
     m.def("start_server", &start_server, py::call_guard<py::gil_scoped_release>());

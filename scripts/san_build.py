@@ -3,6 +3,12 @@
 
 Usage:  python scripts/san_build.py asan [pytest args...]
         python scripts/san_build.py tsan [pytest args...]
+        python scripts/san_build.py mockverbs [pytest args...]
+
+mockverbs builds against tests/mock_verbs (an in-process loopback ibverbs
+provider) so the verbs fabric — unreachable in this NIC-less environment —
+executes for real: QP bring-up, SEND metadata, RDMA_WRITE chains,
+WRITE_WITH_IMM reads, WrFlow. No sanitizer in that mode.
 
 Host C++ TUs are rebuilt with g++ -fsanitize={address,thread}; the HIP TU
 (csrc/gpu/gpu.hip) is compiled uninstrumented (device code cannot carry the
@@ -35,25 +41,32 @@ DEFAULT_TESTS = {
     "tsan": ["tests/test_stress.py::test_many_threads_one_server",
              "tests/test_stress.py::test_async_ops_interleaved",
              "tests/test_shm_ring.py"],
+    "mockverbs": ["tests/test_verbs_loopback.py"],
 }
 
 
 def main():
     mode = sys.argv[1] if len(sys.argv) > 1 else "asan"
-    assert mode in ("asan", "tsan"), "mode must be asan|tsan"
-    san = "address" if mode == "asan" else "thread"
+    assert mode in ("asan", "tsan", "mockverbs"), "mode must be asan|tsan|mockverbs"
+    san = {"asan": "address", "tsan": "thread", "mockverbs": None}[mode]
     tests = sys.argv[2:] or DEFAULT_TESTS[mode]
 
     import pybind11
 
     build = f"/tmp/{mode}_build"
     os.makedirs(build + "/infinistore_amd", exist_ok=True)
-    flags = ["-O1", "-g", "-std=c++20", "-fPIC", f"-fsanitize={san}",
+    flags = ["-O1", "-g", "-std=c++20", "-fPIC",
              "-fno-omit-frame-pointer", "-D__HIP_PLATFORM_AMD__", "-Icsrc",
              f"-I{pybind11.get_include()}",
              f"-I{sysconfig.get_paths()['include']}", "-I/opt/rocm/include"]
+    if san:
+        flags.insert(4, f"-fsanitize={san}")
+    srcs = list(SRCS)
+    if mode == "mockverbs":
+        flags.insert(0, "-Itests/mock_verbs")  # <infiniband/verbs.h> -> mock
+        srcs.append("tests/mock_verbs/mock_verbs.cpp")
     objs = []
-    for s in SRCS:
+    for s in srcs:
         o = f"{build}/{s.replace('/', '_')}.o"
         src = os.path.join(REPO, s)
         if not os.path.exists(o) or os.path.getmtime(o) < os.path.getmtime(src):
@@ -65,7 +78,8 @@ def main():
             print(f"  [{mode}] {s}")
         objs.append(o)
     gpu_o = f"{build}/gpu.o"
-    if not os.path.exists(gpu_o):
+    gpu_src = os.path.join(REPO, "csrc/gpu/gpu.hip")
+    if not os.path.exists(gpu_o) or os.path.getmtime(gpu_o) < os.path.getmtime(gpu_src):
         nosan = [f for f in flags if "sanitize" not in f and f != "-fno-omit-frame-pointer"]
         r = subprocess.run(["/opt/rocm/bin/hipcc", "-c", "csrc/gpu/gpu.hip", "-o",
                             gpu_o, "--offload-arch=gfx950"] + nosan,
@@ -77,28 +91,33 @@ def main():
 
     suffix = sysconfig.get_config_var("EXT_SUFFIX")
     out = f"{build}/infinistore_amd/_native{suffix}"
-    r = subprocess.run(["g++", "-shared", f"-fsanitize={san}", "-o", out] + objs +
-                       ["-luv", "-L/opt/rocm/lib", "-lamdhip64"],
-                       capture_output=True, text=True, cwd=REPO)
+    link = ["g++", "-shared"] + ([f"-fsanitize={san}"] if san else []) + ["-o", out] + objs +            ["-luv", "-L/opt/rocm/lib", "-lamdhip64"]
+    r = subprocess.run(link, capture_output=True, text=True, cwd=REPO)
     if r.returncode:
         print(r.stderr[:4000])
         return 1
 
-    # Shadow package: real python sources + instrumented _native.
-    for f in os.listdir(os.path.join(REPO, "infinistore_amd")):
-        if f.endswith(".py"):
-            src = os.path.join(REPO, "infinistore_amd", f)
-            dst = f"{build}/infinistore_amd/{f}"
-            with open(src) as a, open(dst, "w") as b:
-                b.write(a.read())
+    # Shadow package: real python sources (incl. subpackages) + the
+    # instrumented _native.
+    pkg = os.path.join(REPO, "infinistore_amd")
+    for root, _dirs, files in os.walk(pkg):
+        rel = os.path.relpath(root, pkg)
+        os.makedirs(os.path.join(build, "infinistore_amd", rel), exist_ok=True)
+        for f in files:
+            if f.endswith(".py"):
+                src = os.path.join(root, f)
+                dst = os.path.join(build, "infinistore_amd", rel, f)
+                with open(src) as a, open(dst, "w") as b:
+                    b.write(a.read())
 
-    librt = subprocess.run(
-        ["gcc", f"-print-file-name=lib{'asan' if mode == 'asan' else 'tsan'}.so"],
-        capture_output=True, text=True).stdout.strip()
     env = dict(os.environ)
     env["PYTHONPATH"] = build
-    env["LD_PRELOAD"] = librt
     env["IFS_SKIP_BUILD"] = "1"
+    if san:
+        librt = subprocess.run(
+            ["gcc", f"-print-file-name=lib{'asan' if mode == 'asan' else 'tsan'}.so"],
+            capture_output=True, text=True).stdout.strip()
+        env["LD_PRELOAD"] = librt
     if mode == "asan":
         env["ASAN_OPTIONS"] = "detect_leaks=0:abort_on_error=1"
     else:
@@ -107,9 +126,16 @@ def main():
             open(supp, "w").write("# TSAN suppressions (python runtime noise)\n"
                                   "race:_Py\nrace:Py\n")
         env["TSAN_OPTIONS"] = "report_bugs=1:halt_on_error=0:suppressions=" + supp
+    # Run from the BUILD dir: `python -m pytest` puts the cwd at the front
+    # of sys.path, so running from the repo would import the repo's own
+    # infinistore_amd and silently test the NORMAL build instead of the
+    # instrumented one (import the module in-test and check __file__ if in
+    # doubt). Test paths become absolute for the same reason.
+    abs_tests = [t if os.path.isabs(t.split("::")[0]) else os.path.join(REPO, t)
+                 for t in tests]
     print(f"  [{mode}] running: pytest {' '.join(tests)}")
-    return subprocess.call([sys.executable, "-m", "pytest", "-x", "-q"] + tests,
-                           env=env, cwd=REPO)
+    return subprocess.call([sys.executable, "-m", "pytest", "-x", "-q"] + abs_tests,
+                           env=env, cwd=build)
 
 
 if __name__ == "__main__":
