@@ -2,6 +2,7 @@
 
 #include <poll.h>
 
+#include <atomic>
 #include <condition_variable>
 #include <cstring>
 #include <deque>
@@ -31,7 +32,7 @@ struct VerbsClient::Impl {
     std::unique_ptr<vf::Endpoint> ep;
     std::unique_ptr<WrFlow> flow;
     std::thread poller;
-    bool stop = false;
+    std::atomic<bool> stop{false};
     int wake_pipe[2] = {-1, -1};
 
     std::mutex mu;
@@ -47,10 +48,7 @@ struct VerbsClient::Impl {
     std::mutex op_mu;  // serializes allocate/write/read ops
 
     ~Impl() {
-        {
-            std::lock_guard<std::mutex> lk(mu);
-            stop = true;
-        }
+        stop.store(true, std::memory_order_release);
         if (wake_pipe[1] >= 0) {
             char b = 1;
             ssize_t r = ::write(wake_pipe[1], &b, 1);
@@ -98,10 +96,7 @@ struct VerbsClient::Impl {
         fds[1].fd = wake_pipe[0];
         fds[1].events = POLLIN;
         for (;;) {
-            {
-                std::lock_guard<std::mutex> lk(mu);
-                if (stop) return;
-            }
+            if (stop.load(std::memory_order_acquire)) return;
             int r = ::poll(fds, 2, 1000);
             if (r < 0 && errno != EINTR) return;
             if (fds[0].revents & POLLIN) {

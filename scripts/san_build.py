@@ -42,13 +42,17 @@ DEFAULT_TESTS = {
              "tests/test_stress.py::test_async_ops_interleaved",
              "tests/test_shm_ring.py"],
     "mockverbs": ["tests/test_verbs_loopback.py"],
+    "mockverbs-tsan": ["tests/test_verbs_loopback.py"],
+    "mockverbs-asan": ["tests/test_verbs_loopback.py"],
 }
 
 
 def main():
     mode = sys.argv[1] if len(sys.argv) > 1 else "asan"
-    assert mode in ("asan", "tsan", "mockverbs"), "mode must be asan|tsan|mockverbs"
-    san = {"asan": "address", "tsan": "thread", "mockverbs": None}[mode]
+    assert mode in ("asan", "tsan", "mockverbs", "mockverbs-tsan", "mockverbs-asan"), \
+        "mode must be asan|tsan|mockverbs|mockverbs-tsan|mockverbs-asan"
+    san = {"asan": "address", "tsan": "thread", "mockverbs": None,
+           "mockverbs-tsan": "thread", "mockverbs-asan": "address"}[mode]
     tests = sys.argv[2:] or DEFAULT_TESTS[mode]
 
     import pybind11
@@ -62,7 +66,7 @@ def main():
     if san:
         flags.insert(4, f"-fsanitize={san}")
     srcs = list(SRCS)
-    if mode == "mockverbs":
+    if mode.startswith("mockverbs"):
         flags.insert(0, "-Itests/mock_verbs")  # <infiniband/verbs.h> -> mock
         srcs.append("tests/mock_verbs/mock_verbs.cpp")
     objs = []
@@ -115,12 +119,12 @@ def main():
     env["IFS_SKIP_BUILD"] = "1"
     if san:
         librt = subprocess.run(
-            ["gcc", f"-print-file-name=lib{'asan' if mode == 'asan' else 'tsan'}.so"],
+            ["gcc", f"-print-file-name=lib{'asan' if san == 'address' else 'tsan'}.so"],
             capture_output=True, text=True).stdout.strip()
         env["LD_PRELOAD"] = librt
-    if mode == "asan":
+    if san == "address":
         env["ASAN_OPTIONS"] = "detect_leaks=0:abort_on_error=1"
-    else:
+    elif san:
         supp = os.path.join(REPO, "scripts/tsan.supp")
         if not os.path.exists(supp):
             open(supp, "w").write("# TSAN suppressions (python runtime noise)\n"

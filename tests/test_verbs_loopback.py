@@ -101,3 +101,43 @@ def test_verbs_many_blocks_wr_chaining(cpu_server):
         assert torch.equal(src, dst)
     finally:
         conn.close()
+
+
+def test_verbs_async_api(cpu_server):
+    """allocate_rdma_async / rdma_write_cache_async / read_cache_async over
+    the verbs data plane (client CQ thread + callback marshalling)."""
+    import asyncio
+
+    conn = verbs_client(cpu_server)
+    try:
+        src = torch.randn(8192, dtype=torch.float32)
+        dst = torch.zeros_like(src)
+        conn.register_mr(src)
+        conn.register_mr(dst)
+        run = uuid.uuid4().hex[:8]
+        keys = [f"va-{run}-{i}" for i in range(8)]
+        offsets = [i * 1024 for i in range(8)]
+
+        async def go():
+            blocks = await conn.allocate_rdma_async(keys, 1024 * 4)
+            await conn.rdma_write_cache_async(src, offsets, 1024, blocks)
+            conn.sync()
+            await conn.read_cache_async(dst, list(zip(keys, offsets)), 1024)
+            conn.sync()
+
+        asyncio.run(go())
+        assert torch.equal(src, dst)
+    finally:
+        conn.close()
+
+
+def test_verbs_missing_key(cpu_server):
+    conn = verbs_client(cpu_server)
+    try:
+        dst = torch.zeros(1024, dtype=torch.float32)
+        conn.register_mr(dst)
+        with pytest.raises(Exception):
+            conn.read_cache(dst, [(f"vm-{uuid.uuid4().hex}", 0)], 1024)
+            conn.sync()
+    finally:
+        conn.close()
