@@ -1,7 +1,9 @@
 #include "client.h"
 
 #include <arpa/inet.h>
+#include <limits.h>
 #include <netdb.h>
+#include <sys/uio.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/socket.h>
@@ -59,6 +61,9 @@ int ClientConn::init_connection(const ClientConfigC& cfg) {
                 if (::connect(ufd, reinterpret_cast<struct sockaddr*>(&ua), sizeof(ua)) == 0) {
                     struct timeval tv{60, 0};
                     setsockopt(ufd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+                    int bufsz = 4 << 20;  // bulk fabric payloads
+                    setsockopt(ufd, SOL_SOCKET, SO_SNDBUF, &bufsz, sizeof(bufsz));
+                    setsockopt(ufd, SOL_SOCKET, SO_RCVBUF, &bufsz, sizeof(bufsz));
                     fd_ = ufd;
                     connected_ = true;
                     DEBUG("connected via UDS %s", path.c_str());
@@ -99,6 +104,9 @@ int ClientConn::init_connection(const ClientConfigC& cfg) {
     // blocks indefinitely on its sockets).
     struct timeval tv{60, 0};
     setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+    int bufsz = 4 << 20;  // bulk fabric payloads
+    setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &bufsz, sizeof(bufsz));
+    setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &bufsz, sizeof(bufsz));
     fd_ = fd;
     connected_ = true;
     if ((cfg.host_addr == "127.0.0.1" || cfg.host_addr == "localhost") &&
@@ -282,6 +290,52 @@ int ClientConn::setup_rdma(const ClientConfigC& cfg) {
 // ---------------------------------------------------------------------------
 // framing helpers (io_mu_ must be held)
 // ---------------------------------------------------------------------------
+namespace {
+// Gathered send: avoids assembling multi-MB bodies just to copy them into
+// the socket (the TCP fabric's put path sends tensor pages directly).
+bool recv_iov(int fd, struct iovec* iov, int cnt) {
+    while (cnt > 0) {
+        ssize_t r = ::readv(fd, iov, std::min(cnt, IOV_MAX));
+        if (r <= 0) {
+            if (r < 0 && errno == EINTR) continue;
+            return false;
+        }
+        size_t left = static_cast<size_t>(r);
+        while (cnt > 0 && left >= iov->iov_len) {
+            left -= iov->iov_len;
+            iov++;
+            cnt--;
+        }
+        if (cnt > 0 && left) {
+            iov->iov_base = static_cast<uint8_t*>(iov->iov_base) + left;
+            iov->iov_len -= left;
+        }
+    }
+    return true;
+}
+
+bool send_iov(int fd, struct iovec* iov, int cnt) {
+    while (cnt > 0) {
+        ssize_t w = ::writev(fd, iov, std::min(cnt, IOV_MAX));
+        if (w < 0) {
+            if (errno == EINTR) continue;
+            return false;
+        }
+        size_t left = static_cast<size_t>(w);
+        while (cnt > 0 && left >= iov->iov_len) {
+            left -= iov->iov_len;
+            iov++;
+            cnt--;
+        }
+        if (cnt > 0 && left) {
+            iov->iov_base = static_cast<uint8_t*>(iov->iov_base) + left;
+            iov->iov_len -= left;
+        }
+    }
+    return true;
+}
+}  // namespace
+
 bool ClientConn::send_req(char op, const uint8_t* body, size_t n) {
     // Ring requests and socket requests are handled by different server
     // threads; a ring sync round trip here restores the total order the
@@ -541,26 +595,54 @@ int ClientConn::do_w_rdma(const uint64_t* offsets, size_t n_offsets, int block_s
     std::vector<uint64_t> all_addrs;
     {
         std::unique_lock<std::mutex> lk(io_mu_);
+        std::vector<uint8_t> body;       // assembly fallback (device tensors)
+        std::vector<uint8_t> prefix;     // header+addrs for the gathered path
+        std::vector<struct iovec> iov;
         for (size_t start = 0; start < idx.size(); start += per) {
             size_t take = std::min(per, idx.size() - start);
-            std::vector<uint8_t> body(8 + take * 8 + take * bs);
-            uint32_t n32 = static_cast<uint32_t>(take), bs32 = static_cast<uint32_t>(bs);
-            memcpy(body.data(), &n32, 4);
-            memcpy(body.data() + 4, &bs32, 4);
-            uint64_t* addrs = reinterpret_cast<uint64_t*>(body.data() + 8);
-            uint8_t* payload = body.data() + 8 + take * 8;
-            for (size_t j = 0; j < take; j++) {
-                size_t i = idx[start + j];
-                addrs[j] = blocks[i].remote_addr;
-                all_addrs.push_back(blocks[i].remote_addr);
-                const void* src = reinterpret_cast<const void*>(base_ptr + offsets[i]);
-                if (dev) {
-                    if (!gpu::memcpy_d2h(payload + j * bs, src, bs)) return -1;
-                } else {
-                    memcpy(payload + j * bs, src, bs);
+            // Host tensors: gathered send straight from the tensor pages —
+            // no multi-MB body assembly. (Device tensors still stage D2H;
+            // iovec count is bounded by IOV_MAX.)
+            bool gather = !dev && take + 2 <= static_cast<size_t>(IOV_MAX);
+            if (gather) {
+                prefix.resize(sizeof(Header) + 8 + take * 8);
+                auto* h = reinterpret_cast<Header*>(prefix.data());
+                *h = Header{kMagic, OP_TCP_PUT,
+                            static_cast<uint32_t>(8 + take * 8 + take * bs)};
+                uint32_t n32 = static_cast<uint32_t>(take), bs32 = static_cast<uint32_t>(bs);
+                memcpy(prefix.data() + sizeof(Header), &n32, 4);
+                memcpy(prefix.data() + sizeof(Header) + 4, &bs32, 4);
+                uint64_t* addrs =
+                    reinterpret_cast<uint64_t*>(prefix.data() + sizeof(Header) + 8);
+                iov.clear();
+                iov.push_back({prefix.data(), prefix.size()});
+                for (size_t j = 0; j < take; j++) {
+                    size_t i = idx[start + j];
+                    addrs[j] = blocks[i].remote_addr;
+                    all_addrs.push_back(blocks[i].remote_addr);
+                    iov.push_back({reinterpret_cast<void*>(base_ptr + offsets[i]), bs});
                 }
+                if (!send_iov(fd_, iov.data(), static_cast<int>(iov.size()))) return -1;
+            } else {
+                body.resize(8 + take * 8 + take * bs);
+                uint32_t n32 = static_cast<uint32_t>(take), bs32 = static_cast<uint32_t>(bs);
+                memcpy(body.data(), &n32, 4);
+                memcpy(body.data() + 4, &bs32, 4);
+                uint64_t* addrs = reinterpret_cast<uint64_t*>(body.data() + 8);
+                uint8_t* payload = body.data() + 8 + take * 8;
+                for (size_t j = 0; j < take; j++) {
+                    size_t i = idx[start + j];
+                    addrs[j] = blocks[i].remote_addr;
+                    all_addrs.push_back(blocks[i].remote_addr);
+                    const void* src = reinterpret_cast<const void*>(base_ptr + offsets[i]);
+                    if (dev) {
+                        if (!gpu::memcpy_d2h(payload + j * bs, src, bs)) return -1;
+                    } else {
+                        memcpy(payload + j * bs, src, bs);
+                    }
+                }
+                if (!send_req(OP_TCP_PUT, body.data(), body.size())) return -1;
             }
-            if (!send_req(OP_TCP_PUT, body.data(), body.size())) return -1;
             n_chunks++;
         }
         for (size_t ci = 0; ci < n_chunks; ci++) {
@@ -626,7 +708,21 @@ int ClientConn::do_r_rdma(const std::vector<std::pair<std::string, uint64_t>>& b
             WARN("read -> %d", code);
             return -code;
         }
-        if (!recv_payload(&payload)) return -1;
+        uint32_t len = 0;
+        if (!recv_exact(fd_, &len, 4)) return -1;
+        // Host tensors: scatter the payload straight into the destination
+        // pages with readv (the payload is the request's blocks in order) —
+        // no intermediate buffer or per-block memcpy.
+        if (!dev && len == blocks.size() * bs) {
+            std::vector<struct iovec> iov;
+            iov.reserve(blocks.size());
+            for (auto& b : blocks)
+                iov.push_back({reinterpret_cast<void*>(base_ptr + b.second), bs});
+            if (!recv_iov(fd_, iov.data(), static_cast<int>(iov.size()))) return -1;
+            return 0;
+        }
+        payload.resize(len);
+        if (len && !recv_exact(fd_, payload.data(), len)) return -1;
     }
     if (payload.size() < blocks.size() * bs) return -1;
     for (size_t i = 0; i < blocks.size(); i++) {

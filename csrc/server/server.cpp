@@ -88,6 +88,8 @@ namespace {
 struct WriteReq {
     uv_write_t req;
     std::vector<uint8_t> data;
+    std::shared_ptr<uint8_t[]> raw;  // alternative payload (uninitialized buf)
+    size_t raw_len = 0;
 };
 
 void send_buf(Server::Conn* c, std::vector<uint8_t> data);
@@ -114,6 +116,26 @@ void send_status_payload(Server::Conn* c, int code, const uint8_t* payload, size
     memcpy(v.data() + 4, &len, 4);
     if (n) memcpy(v.data() + 8, payload, n);
     send_buf(c, std::move(v));
+}
+
+void send_raw(Server::Conn* c, std::shared_ptr<uint8_t[]> buf, size_t len) {
+    if (c->closed) return;
+    auto* wr = new WriteReq();
+    wr->raw = std::move(buf);
+    wr->raw_len = len;
+    wr->req.data = wr;
+    uv_buf_t b = uv_buf_init(reinterpret_cast<char*>(wr->raw.get()),
+                             static_cast<unsigned>(len));
+    int r = uv_write(&wr->req, c->stream(), &b, 1,
+                     [](uv_write_t* req, int status) {
+                         auto* wr2 = static_cast<WriteReq*>(req->data);
+                         delete wr2;
+                         if (status < 0) DEBUG("uv_write status %d", status);
+                     });
+    if (r != 0) {
+        delete wr;
+        conn_close(c);
+    }
 }
 
 void send_buf(Server::Conn* c, std::vector<uint8_t> data) {
@@ -423,6 +445,9 @@ void Server::adopt_fd(IoLoop* io, int fd, bool is_pipe) {
         ::close(fd);
         return;
     }
+    int bufsz = 4 << 20;  // bulk fabric payloads ride this socket
+    setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &bufsz, sizeof(bufsz));
+    setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &bufsz, sizeof(bufsz));
     auto* c = new Conn();
     c->srv = this;
     c->owner = io;
@@ -1507,12 +1532,14 @@ void Server::op_tcp_get(Conn* c, const RemoteMetaMsg& msg) {
     if (!collect_read_entries(msg.keys, &entries)) return send_status(c, KEY_NOT_FOUND);
     // The framed response (status + len + blocks) is ONE buffer, filled by
     // the shards' fabric workers; the response is sent on completion.
+    // Allocated UNINITIALIZED: value-initializing a 256 MB vector costs
+    // ~100 ms before any data moves.
     size_t total = entries.size() * page;
-    auto resp = std::make_shared<std::vector<uint8_t>>(8 + total);
+    std::shared_ptr<uint8_t[]> resp(new uint8_t[8 + total]);
     int code = FINISH;
     uint32_t len32 = static_cast<uint32_t>(total);
-    memcpy(resp->data(), &code, 4);
-    memcpy(resp->data() + 4, &len32, 4);
+    memcpy(resp.get(), &code, 4);
+    memcpy(resp.get() + 4, &len32, 4);
     std::map<Shard*, Shard::FabricJob> jobs;
     for (size_t i = 0; i < entries.size(); i++) {
         BlockEntry* e = entries[i].get();
@@ -1528,15 +1555,16 @@ void Server::op_tcp_get(Conn* c, const RemoteMetaMsg& msg) {
     auto all_ok = std::make_shared<std::atomic<bool>>(true);
     auto held_sp = std::make_shared<std::vector<Ref<BlockEntry>>>(std::move(entries));
     c->ref();
+    size_t resp_len = 8 + total;
     for (auto& [shard, job] : jobs) {
         Shard::FabricJob j = std::move(job);
-        j.host = resp;
-        j.done = [this, c, resp, pending, all_ok, held_sp](bool ok) {
+        j.raw_host = resp;
+        j.done = [this, c, resp, resp_len, pending, all_ok, held_sp](bool ok) {
             if (!ok) all_ok->store(false);
             if (pending->fetch_sub(1) == 1) {
-                c->owner->post([this, c, resp, all_ok] {
+                c->owner->post([this, c, resp, resp_len, all_ok] {
                     if (all_ok->load())
-                        send_buf(c, std::move(*resp));
+                        send_raw(c, resp, resp_len);
                     else
                         send_status(c, INTERNAL_ERROR);
                     c->unref();

@@ -417,11 +417,13 @@ void Shard::fabric_loop() {
             job = std::move(fabric_q_.front());
             fabric_q_.pop_front();
         }
-        bool ok = job.host != nullptr;
+        uint8_t* base = job.raw_host ? job.raw_host.get()
+                                     : (job.host ? job.host->data() : nullptr);
+        bool ok = base != nullptr;
         size_t bs = job.bytes_per_block;
         if (job.is_put) {
             for (size_t i = 0; i < job.block_ptrs.size() && ok; i++) {
-                const uint8_t* src = job.host->data() + job.host_offsets[i];
+                const uint8_t* src = base + job.host_offsets[i];
                 void* dst = reinterpret_cast<void*>(job.block_ptrs[i]);
                 if (on_gpu())
                     ok = gpu::memcpy_h2d(dst, src, bs);
@@ -430,7 +432,7 @@ void Shard::fabric_loop() {
             }
         } else {
             for (size_t i = 0; i < job.block_ptrs.size() && ok; i++) {
-                uint8_t* dst = job.host->data() + job.host_offsets[i];
+                uint8_t* dst = base + job.host_offsets[i];
                 const void* src = reinterpret_cast<const void*>(job.block_ptrs[i]);
                 if (on_gpu())
                     ok = gpu::memcpy_d2h(dst, src, bs);

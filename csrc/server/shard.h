@@ -83,6 +83,7 @@ class Shard {
     struct FabricJob {
         bool is_put = false;
         std::shared_ptr<std::vector<uint8_t>> host;  // source (put) / dest (get)
+        std::shared_ptr<uint8_t[]> raw_host;         // alternative uninit dest
         std::vector<uint64_t> block_ptrs;            // pool block addresses
         std::vector<size_t> host_offsets;            // absolute offsets into *host
         size_t bytes_per_block = 0;
