@@ -1556,6 +1556,57 @@ void Server::op_tcp_get(Conn* c, const RemoteMetaMsg& msg) {
     auto held_sp = std::make_shared<std::vector<Ref<BlockEntry>>>(std::move(entries));
     c->ref();
     size_t resp_len = 8 + total;
+
+    // Single-shard big responses STREAM: the job is split into segments and
+    // each segment is written to the socket as its copy lands, overlapping
+    // the pool→host copy with the send (they were strictly sequential —
+    // 256 MB paid copy + send back to back). Payload offsets are ascending
+    // within the single job and the fabric worker runs segments FIFO, so
+    // in-order streaming holds. Multi-shard responses interleave offsets
+    // and keep the single-shot path.
+    constexpr size_t kSegBytes = 16u << 20;
+    if (jobs.size() == 1 && total > 2 * kSegBytes) {
+        auto& [shard, whole] = *jobs.begin();
+        size_t nb = whole.block_ptrs.size();
+        size_t per_seg = std::max<size_t>(1, kSegBytes / page);
+        size_t n_segs = (nb + per_seg - 1) / per_seg;
+        auto segs_left = std::make_shared<std::atomic<int>>(static_cast<int>(n_segs));
+        for (size_t off = 0; off < nb; off += per_seg) {
+            size_t take = std::min(per_seg, nb - off);
+            Shard::FabricJob j;
+            j.is_put = false;
+            j.bytes_per_block = page;
+            j.raw_host = resp;
+            j.block_ptrs.assign(whole.block_ptrs.begin() + static_cast<long>(off),
+                                whole.block_ptrs.begin() + static_cast<long>(off + take));
+            j.host_offsets.assign(whole.host_offsets.begin() + static_cast<long>(off),
+                                  whole.host_offsets.begin() + static_cast<long>(off + take));
+            bool first = off == 0;
+            size_t seg_start = first ? 0 : 8 + off * page;  // frame rides seg 0
+            size_t seg_len = (first ? 8 : 0) + take * page;
+            j.done = [this, c, resp, segs_left, all_ok, held_sp, seg_start,
+                      seg_len](bool ok) {
+                if (!ok) all_ok->store(false);
+                bool last = segs_left->fetch_sub(1) == 1;
+                c->ref();
+                c->owner->post([this, c, resp, all_ok, seg_start, seg_len, last] {
+                    if (all_ok->load()) {
+                        std::shared_ptr<uint8_t[]> view(resp, resp.get() + seg_start);
+                        send_raw(c, std::move(view), seg_len);
+                    } else if (last) {
+                        // some earlier segments may already be on the wire;
+                        // the client's length-checked read will fail cleanly
+                        conn_close(c);
+                    }
+                    c->unref();
+                    if (last) c->unref();  // the op's own ref
+                });
+            };
+            shard->submit_fabric(std::move(j));
+        }
+        return;
+    }
+
     for (auto& [shard, job] : jobs) {
         Shard::FabricJob j = std::move(job);
         j.raw_host = resp;
