@@ -551,3 +551,42 @@ def test_ttl_expiry(ports):
         conn.close()
     finally:
         ifs.unregister_server()
+
+
+def test_large_streamed_read(ports):
+    """Single-shard reads >32 MB take the segment-streaming path (the
+    response is written to the socket while later segments still copy);
+    verify byte-exactness across the segment boundaries."""
+    service_port, manage_port = ports
+    cfg = ifs.ServerConfig(
+        service_port=service_port, manage_port=manage_port,
+        prealloc_size=1, minimal_allocate_size=128, cpu_only=True,
+    )
+    ifs.register_server(cfg)
+    try:
+        conn = make_client(service_port)
+        n, pe = 384, 32768  # 384 x 128 KB = 48 MB > 2 x 16 MB segments
+        src = torch.rand(n * pe)
+        dst = torch.zeros_like(src)
+        conn.register_mr(src)
+        conn.register_mr(dst)
+        keys = [f"big-{i}" for i in range(n)]
+        offs = [i * pe for i in range(n)]
+        blocks = conn.allocate_rdma(keys, pe * 4)
+        conn.rdma_write_cache(src, offs, pe, blocks)
+        conn.sync()
+        conn.read_cache(dst, list(zip(keys, offs)), pe)
+        conn.sync()
+        assert torch.equal(src, dst)
+        # and a shuffled-offset read (host offsets still ascend per request
+        # order on the wire; destination scatter is the client's readv)
+        import random
+        order = list(range(n))
+        random.Random(3).shuffle(order)
+        dst.zero_()
+        conn.read_cache(dst, [(keys[i], offs[i]) for i in order], pe)
+        conn.sync()
+        assert torch.equal(src, dst)
+        conn.close()
+    finally:
+        ifs.unregister_server()
