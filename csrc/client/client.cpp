@@ -201,7 +201,7 @@ int ClientConn::shm_wait(uint64_t seq) {
             if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(60)) {
                 ERROR("shm ring response timeout (seq %llu)",
                       static_cast<unsigned long long>(seq));
-                return -1;
+                return kShmErr;
             }
         }
     }
@@ -221,7 +221,7 @@ int ClientConn::shm_request(char op, const uint8_t* body, size_t n, bool want_re
 #if defined(__x86_64__)
         __builtin_ia32_pause();
 #endif
-        if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(60)) return -1;
+        if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(60)) return kShmErr;
         dst = shm_.req->claim(need, &adv);
     }
     shmring::RecHdr h{};
@@ -252,6 +252,7 @@ int ClientConn::wait_local_ticket(uint64_t ticket) {
     std::lock_guard<std::mutex> lk(io_mu_);
     if (!shm_active_) return -1;
     int code = shm_wait(ticket);
+    if (code == kShmErr) return -1;
     if (code != TASK_ACCEPTED && code != FINISH && code != 0) return code < 0 ? code : -code;
     return 0;
 }
@@ -771,6 +772,12 @@ int ClientConn::check_exist(const std::string& key) {
     memcpy(body.data(), &len, 4);
     memcpy(body.data() + 4, key.data(), key.size());
     std::lock_guard<std::mutex> lk(io_mu_);
+    // Query ops ride the shm ring too (the reply status carries the value);
+    // prefix lookups sit on the decode critical path.
+    if (shm_active_) {
+        int code = shm_request(OP_CHECK_EXIST, body.data(), body.size(), true);
+        if (code != kShmNoFit) return code == kShmErr ? -1 : code;
+    }
     if (!send_req(OP_CHECK_EXIST, body.data(), body.size())) return -1;
     int code = -1;
     if (!recv_status(&code)) return -1;
@@ -781,6 +788,10 @@ int ClientConn::delete_keys(const std::vector<std::string>& keys) {
     if (!connected_) return -1;
     auto body = build_match_request(keys);
     std::lock_guard<std::mutex> lk(io_mu_);
+    if (shm_active_) {
+        int code = shm_request(OP_DELETE, body.data(), body.size(), true);
+        if (code != kShmNoFit) return code == kShmErr ? -1 : code;
+    }
     if (!send_req(OP_DELETE, body.data(), body.size())) return -1;
     int n = -1;
     if (!recv_status(&n)) return -1;
@@ -802,6 +813,10 @@ int ClientConn::get_match_last_index(const std::vector<std::string>& keys) {
     if (!connected_) return -1;
     auto body = build_match_request(keys);
     std::lock_guard<std::mutex> lk(io_mu_);
+    if (shm_active_) {
+        int code = shm_request(OP_GET_MATCH_LAST_IDX, body.data(), body.size(), true);
+        if (code != kShmNoFit) return code == kShmErr ? -1 : code;
+    }
     if (!send_req(OP_GET_MATCH_LAST_IDX, body.data(), body.size())) return -1;
     int idx = -1;
     if (!recv_status(&idx)) return -1;

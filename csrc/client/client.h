@@ -128,6 +128,7 @@ class ClientConn {
     // (mapped like the socket path). Returns kShmNoFit when the record does
     // not fit the ring (caller falls back to the socket after a ring sync).
     static constexpr int kShmNoFit = INT32_MIN;
+    static constexpr int kShmErr = INT32_MIN + 1;  // transport failure
     int shm_request(char op, const uint8_t* body, size_t n, bool want_resp,
                     uint64_t* out_ticket = nullptr);
     int shm_wait(uint64_t seq);           // drain responses until seq; io_mu_

@@ -657,11 +657,11 @@ void Server::handle_request(Conn* c, char op, std::vector<uint8_t> body) {
             return op_commit(c, msg);
         }
         case OP_CHECK_EXIST:
-            return op_check_exist(c, body);
+            return op_check_exist(c, body, ReqCtx{});
         case OP_GET_MATCH_LAST_IDX:
-            return op_match_index(c, body);
+            return op_match_index(c, body, ReqCtx{});
         case OP_DELETE:
-            return op_delete(c, body);
+            return op_delete(c, body, ReqCtx{});
         case OP_STATS: {
             std::string js = stats_json();
             return send_status_payload(c, FINISH,
@@ -1306,6 +1306,15 @@ void Server::shm_poll_main(ShmPeer* p) {
                 case OP_SYNC:
                     op_sync(c, ctx);
                     break;
+                case OP_CHECK_EXIST:
+                    op_check_exist(c, std::vector<uint8_t>(body, body + h.body_len), ctx);
+                    break;
+                case OP_GET_MATCH_LAST_IDX:
+                    op_match_index(c, std::vector<uint8_t>(body, body + h.body_len), ctx);
+                    break;
+                case OP_DELETE:
+                    op_delete(c, std::vector<uint8_t>(body, body + h.body_len), ctx);
+                    break;
                 default:
                     reply_local(c, ctx, INVALID_REQ);
             }
@@ -1627,12 +1636,12 @@ void Server::op_tcp_get(Conn* c, const RemoteMetaMsg& msg) {
 }
 
 // ---- queries ---------------------------------------------------------------
-void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body) {
+void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& ctx) {
     // body: u32 len + key bytes (simple framing; no flatbuffer needed)
-    if (body.size() < 4) return send_status(c, INVALID_REQ);
+    if (body.size() < 4) return reply_local(c, ctx, INVALID_REQ);
     uint32_t len;
     memcpy(&len, body.data(), 4);
-    if (body.size() < 4 + len) return send_status(c, INVALID_REQ);
+    if (body.size() < 4 + len) return reply_local(c, ctx, INVALID_REQ);
     std::string key(reinterpret_cast<const char*>(body.data() + 4), len);
     bool exists;
     {
@@ -1642,13 +1651,13 @@ void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body) {
         Ref<BlockEntry>* v = st.map.find_hashed(key, h);
         exists = v && (*v)->committed && !expired(v->get());
     }
-    send_status(c, exists ? 0 : 1);
+    reply_local(c, ctx, exists ? 0 : 1);
 }
 
-void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body) {
+void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& ctx) {
     std::vector<std::string> keys;
     if (!parse_match_request(body.data(), body.size(), &keys) || keys.empty())
-        return send_status(c, -1);
+        return reply_local(c, ctx, -1);
     // Binary search for the last present index, assuming the prefix property
     // (keys[0..i] present iff i <= match). Requires committed entries —
     // divergence from the reference, which counts uncommitted keys as
@@ -1669,13 +1678,13 @@ void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body) {
         else
             right = mid;
     }
-    send_status(c, static_cast<int>(left - 1));
+    reply_local(c, ctx, static_cast<int>(left - 1));
 }
 
-void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
+void Server::op_delete(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& ctx) {
     std::vector<std::string> keys;
     if (!parse_match_request(body.data(), body.size(), &keys))
-        return send_status(c, INVALID_REQ);
+        return reply_local(c, ctx, INVALID_REQ);
     int n = 0;
     {
         std::array<std::vector<uint32_t>, kStripes> by_stripe;
@@ -1697,7 +1706,7 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body) {
             }
         }
     }
-    send_status(c, n);
+    reply_local(c, ctx, n);
 }
 
 // ---------------------------------------------------------------------------

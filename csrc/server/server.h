@@ -244,9 +244,9 @@ class Server {
     void op_tcp_put(Conn* c, std::vector<uint8_t> body);
     void op_tcp_get(Conn* c, const RemoteMetaMsg& msg);
     void op_commit(Conn* c, const RemoteMetaMsg& msg);
-    void op_check_exist(Conn* c, const std::vector<uint8_t>& body);
-    void op_match_index(Conn* c, const std::vector<uint8_t>& body);
-    void op_delete(Conn* c, const std::vector<uint8_t>& body);
+    void op_check_exist(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& ctx);
+    void op_match_index(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& ctx);
+    void op_delete(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& ctx);
 
     Shard* shard_for_device(int device);
     Shard* shard_least_used();
