@@ -476,6 +476,7 @@ int ClientConn::sync_local() {
     if (!local_dirty_ && shm_unacked_ == 0) return 0;
     if (shm_active_) {
         int code = shm_ring_sync();
+        if (code == kShmErr) return -1;
         if (code != 0) return code < 0 ? code : -code;
         local_dirty_ = false;
         if (shm_async_err_) {  // surface an earlier async ring write failure
