@@ -108,3 +108,70 @@ def test_snapshot_restore(cpu_server, tmp_path):
     rn2, _ = ifs.restore_pool(snap)
     assert rn2 == 0
     conn.close()
+
+
+def test_snapshot_http_and_cli_restore(tmp_path):
+    """The full warm-restart story across processes: snapshot over HTTP,
+    then a FRESH server process started with --restore-from serves the
+    same keys."""
+    import subprocess
+    import sys
+    import time
+
+    from conftest import free_port
+
+    repo = __import__("os").path.dirname(__import__("os").path.dirname(
+        __import__("os").path.abspath(__file__)))
+    snap = str(tmp_path / "cli.snap")
+
+    # phase 1: in-process server, write keys, snapshot via the HTTP app
+    port1 = free_port()
+    cfg = ifs.ServerConfig(service_port=port1, manage_port=free_port(),
+                           prealloc_size=1, minimal_allocate_size=16,
+                           cpu_only=True)
+    ifs.register_server(cfg)
+    try:
+        from infinistore_amd import server as srv
+        with TestClient(srv.make_app(cfg)) as tc:
+            conn = make_client(port1)
+            src = torch.arange(2048, dtype=torch.float32)
+            conn.register_mr(src)
+            blocks = conn.allocate_rdma(["cli-a", "cli-b"], 4096)
+            conn.rdma_write_cache(src, [0], 1024, [blocks[0]])
+            conn.rdma_write_cache(src, [1024], 1024, [blocks[1]])
+            conn.sync()
+            conn.close()
+            r = tc.post(f"/snapshot?path={snap}").json()
+            assert r["entries"] == 2
+    finally:
+        ifs.unregister_server()
+
+    # phase 2: fresh server PROCESS restores the snapshot at startup
+    port2 = free_port()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "infinistore_amd.server",
+         "--service-port", str(port2), "--manage-port", str(free_port()),
+         "--prealloc-size", "1", "--minimal-allocate-size", "16",
+         "--cpu-only", "--no-manage", "--restore-from", snap],
+        cwd=repo,
+    )
+    try:
+        t0 = time.time()
+        conn = None
+        while time.time() - t0 < 60:
+            try:
+                conn = make_client(port2)
+                break
+            except Exception:
+                time.sleep(0.3)
+        assert conn is not None
+        dst = torch.zeros(2048, dtype=torch.float32)
+        conn.register_mr(dst)
+        conn.read_cache(dst, [("cli-a", 0), ("cli-b", 1024)], 1024)
+        conn.sync()
+        src = torch.arange(2048, dtype=torch.float32)
+        assert torch.equal(src, dst)
+        conn.close()
+    finally:
+        proc.terminate()
+        proc.wait(timeout=20)
