@@ -128,12 +128,23 @@ def decode_worker(model, tokens, next_token, conn: PagedKVConnector, device):
     cap = len(tokens) + 8
     kv_cache = [torch.zeros(2, cap, model.n_kv, model.hd, device=device, dtype=dt)
                 for _ in range(model.n_layers)]
-    staging = torch.zeros(n_pages * elems, device=device, dtype=dt)
+    stagings = [torch.zeros(n_pages * elems, device=device, dtype=dt)
+                for _ in range(model.n_layers)]
     offsets = [p * elems for p in range(n_pages)]
+    # Prefetch every layer's pages up front (ticketed async reads on the
+    # local path — all gathers are in flight while we unpack layer by
+    # layer), falling back to blocking loads on the fabric path.
+    tickets = [conn.load_layer_async(li, stagings[li], page_keys[:n_pages],
+                                     offsets, elems)
+               for li in range(model.n_layers)]
     for li in range(model.n_layers):
-        ok = conn.load_layer(li, staging, page_keys[:n_pages], offsets, elems)
-        assert ok, f"layer {li}: cached pages missing"
-        paged = staging.view(n_pages, 2, PAGE_TOKENS, model.n_kv, model.hd)
+        if tickets[li] is not None:
+            assert conn.wait_load(tickets[li]), f"layer {li}: async load failed"
+        else:
+            ok = conn.load_layer(li, stagings[li], page_keys[:n_pages], offsets,
+                                 elems)
+            assert ok, f"layer {li}: cached pages missing"
+        paged = stagings[li].view(n_pages, 2, PAGE_TOKENS, model.n_kv, model.hd)
         kv_cache[li][:, : n_pages * PAGE_TOKENS] = (
             paged.transpose(0, 1).reshape(2, -1, model.n_kv, model.hd))
     with torch.no_grad():
