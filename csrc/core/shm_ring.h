@@ -132,7 +132,16 @@ struct Ring {
     // ---- consumer ----
     // Peek the next record; nullptr if empty. *len is the record's total
     // length. Caller must consume(*skip) after copying out.
-    const uint8_t* peek(uint32_t* len, uint64_t* skip) {
+    //
+    // A correct producer publishes whole records with one release store of
+    // head, so once `h - t >= 4` the length field is fully visible and any
+    // invalid value is PROVABLY corrupt (hostile or broken peer), never a
+    // torn write. In particular a record that would straddle the ring's end
+    // (l > until_end) can only come from a peer writing the header by hand —
+    // the producer's claim() always wraps first — and following it would
+    // read past the mapped segment. On corruption we set *corrupt and return
+    // nullptr; the caller must stop consuming this ring for good.
+    const uint8_t* peek(uint32_t* len, uint64_t* skip, bool* corrupt = nullptr) {
         uint64_t t = tail.load(std::memory_order_relaxed);
         uint64_t h = head.load(std::memory_order_acquire);
         if (h == t) return nullptr;
@@ -150,7 +159,10 @@ struct Ring {
             *skip = until_end;
             return data;
         }
-        if (l < sizeof(RecHdr) || l > cap || l > h - t) return nullptr;  // torn/corrupt: wait
+        if (l < sizeof(RecHdr) || l > until_end || l > h - t || (l & 7)) {
+            if (corrupt) *corrupt = true;
+            return nullptr;
+        }
         *len = l;
         *skip = l;
         return data + pos;
@@ -183,6 +195,9 @@ struct Segment {
         if (c->magic != kMagic || c->version != kVersion) return false;
         if (c->req_cap == 0 || c->resp_cap == 0) return false;
         if ((c->req_cap | c->resp_cap) & 7) return false;
+        // Ring::head/tail are std::atomic<uint64_t>: a hostile segment with
+        // misaligned ring offsets would make those accesses UB / torn.
+        if ((c->req_off | c->resp_off) & 7) return false;
         uint64_t req_end = uint64_t(c->req_off) + Ring::footprint(c->req_cap);
         uint64_t resp_end = uint64_t(c->resp_off) + Ring::footprint(c->resp_cap);
         if (c->req_off < sizeof(Ctrl) || req_end > len) return false;

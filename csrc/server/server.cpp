@@ -787,13 +787,14 @@ void* resolve_client_base(Server::Conn* c, const Server::LocalView& msg) {
     if (msg.pid != 0 && msg.pid == static_cast<int32_t>(getpid()) && msg.base_ptr != 0)
         return reinterpret_cast<void*>(msg.base_ptr);
     std::vector<uint8_t> key(msg.ipc, msg.ipc + msg.ipc_len);
+    std::lock_guard<std::mutex> lk(c->ipc_mu);
     auto it = c->ipc_cache.find(key);
     if (it != c->ipc_cache.end()) return it->second.first;
     // Bound the cache: a client that churns tensors would otherwise pin
     // every old allocation via its stale mapping. Only flush when nothing
     // is in flight (a mapping may be read by a queued kernel).
     if (c->ipc_cache.size() >= 64 && c->remain.load() == 0 &&
-        c->fabric_inflight.load() == 0) {
+        c->fabric_inflight.load() == 0 && c->ipc_pin.load() == 1) {
         for (auto& kv2 : c->ipc_cache) gpu::ipc_close(kv2.second.first);
         c->ipc_cache.clear();
     }
@@ -803,6 +804,15 @@ void* resolve_client_base(Server::Conn* c, const Server::LocalView& msg) {
     if (base) c->ipc_cache.emplace(std::move(key), std::make_pair(base, msg.device));
     return base;
 }
+
+// Holds Conn::ipc_pin for a handler's resolve→enqueue window so a concurrent
+// handler on the other transport cannot flush (ipc_close) a base this one
+// resolved but has not yet protected with remain/fabric_inflight.
+struct IpcPinGuard {
+    Server::Conn* c;
+    explicit IpcPinGuard(Server::Conn* conn) : c(conn) { c->ipc_pin.fetch_add(1); }
+    ~IpcPinGuard() { c->ipc_pin.fetch_sub(1); }
+};
 }  // namespace
 
 void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
@@ -811,6 +821,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     if (!gpu::available()) return reply_local(c, ctx, SYSTEM_ERROR);
     if (msg.ipc_len != gpu::kIpcHandleSize || msg.block_size <= 0)
         return reply_local(c, ctx, INVALID_REQ);
+    IpcPinGuard pin(c);
     void* base = resolve_client_base(c, msg);
     if (!base) return reply_local(c, ctx, INTERNAL_ERROR);
     uint8_t* client_ptr = static_cast<uint8_t*>(base) + msg.base_offset;
@@ -1028,6 +1039,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     if (!gpu::available()) return reply_local(c, ctx, SYSTEM_ERROR);
     if (msg.ipc_len != gpu::kIpcHandleSize || msg.block_size <= 0)
         return reply_local(c, ctx, INVALID_REQ);
+    IpcPinGuard pin(c);
     void* base = resolve_client_base(c, msg);
     if (!base) return reply_local(c, ctx, INTERNAL_ERROR);
     uint8_t* client_ptr = static_cast<uint8_t*>(base) + msg.base_offset;
@@ -1252,7 +1264,15 @@ void Server::shm_poll_main(ShmPeer* p) {
     while (!p->stop.load(std::memory_order_acquire)) {
         uint32_t len = 0;
         uint64_t skip = 0;
-        const uint8_t* rec = p->seg.req->peek(&len, &skip);
+        bool corrupt = false;
+        const uint8_t* rec = p->seg.req->peek(&len, &skip, &corrupt);
+        if (corrupt) {
+            // Provably invalid record header (hostile or broken client):
+            // following it could read past the mapped segment. Stop polling
+            // this ring for good; the conn's socket teardown reclaims it.
+            ERROR("shm req ring corrupt (invalid record header); detaching poller");
+            break;
+        }
         if (!rec) {
             // Adaptive idle: spin while recently hot (sub-µs pickup), back
             // off to 50 µs sleeps when the ring has been quiet.

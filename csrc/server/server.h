@@ -350,8 +350,17 @@ struct Server::Conn : RefCounted {
     std::atomic<int> remain{0};
     bool sync_waiting = false;  // loop thread only
 
-    // cached IPC mappings: handle bytes -> base pointer (closed on disconnect)
+    // cached IPC mappings: handle bytes -> base pointer (closed on disconnect).
+    // Guarded by ipc_mu: local ops for one conn can arrive concurrently on the
+    // socket (owner uv-loop thread) and the shm ring (poller thread) — the
+    // stock client serializes them, but server memory safety must not depend
+    // on client behavior.
+    std::mutex ipc_mu;
     std::map<std::vector<uint8_t>, std::pair<void*, int>> ipc_cache;  // base, src_dev
+    // Handlers holding a resolved base that is not yet protected by
+    // remain/fabric_inflight keep this >0; the cache flush in
+    // resolve_client_base only runs when the flusher is the sole pinner.
+    std::atomic<int> ipc_pin{0};
 
     // fabric: blocks allocated for this conn, not yet committed.
     std::unordered_map<uint64_t, Ref<BlockEntry>> pending_rdma;

@@ -30,29 +30,33 @@ WIRE_MAGIC = 0xDEADBEEF
 class RawShmConn:
     """Minimal hand-rolled client: UDS control socket + handcrafted segment."""
 
-    def __init__(self, port):
+    def __init__(self, port, req_cap=REQ_CAP, resp_cap=RESP_CAP, req_off_skew=0,
+                 expect_code=200):
+        self.req_cap = req_cap
+        self.resp_cap = resp_cap
         self.sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
         self.sock.settimeout(30)
         self.sock.connect(f"/tmp/infinistore-amd-{port}.sock")
         name = f"/ifs-fuzz-{os.getpid()}-{uuid.uuid4().hex[:8]}"
         path = "/dev/shm" + name
-        total = CTRL + RING_HDR + REQ_CAP + RING_HDR + RESP_CAP
+        total = CTRL + RING_HDR + req_cap + RING_HDR + resp_cap + req_off_skew
         with open(path, "wb") as f:
             f.write(b"\0" * total)
         self.f = open(path, "r+b")
         self.mm = mmap.mmap(self.f.fileno(), total)
-        req_off = CTRL
-        resp_off = CTRL + RING_HDR + REQ_CAP
-        struct.pack_into("<QIIIIII", self.mm, 0, MAGIC, 1, req_off, REQ_CAP,
-                         resp_off, RESP_CAP, 0)
-        struct.pack_into("<I", self.mm, req_off + 16, REQ_CAP)
-        struct.pack_into("<I", self.mm, resp_off + 16, RESP_CAP)
+        req_off = CTRL + req_off_skew
+        resp_off = req_off + RING_HDR + req_cap
+        struct.pack_into("<QIIIIII", self.mm, 0, MAGIC, 1, req_off, req_cap,
+                         resp_off, resp_cap, 0)
+        struct.pack_into("<I", self.mm, req_off + 16, req_cap)
+        struct.pack_into("<I", self.mm, resp_off + 16, resp_cap)
         self.req_data = req_off + RING_HDR
         self.req_head_off = req_off
+        self.resp_off = resp_off
         # handshake
         self._send(b"h", name.encode())
         code = struct.unpack("<i", self._recv(4))[0]
-        assert code == 200, code  # FINISH
+        assert code == expect_code, code
         os.unlink(path)
 
     def _send(self, op, body):
@@ -69,10 +73,23 @@ class RawShmConn:
     def push_raw(self, raw):
         """Append raw bytes as-is at the ring head and publish."""
         head = struct.unpack_from("<Q", self.mm, self.req_head_off)[0]
-        pos = head % REQ_CAP
-        assert pos + len(raw) <= REQ_CAP  # keep the fuzz simple: no wrap
+        pos = head % self.req_cap
+        assert pos + len(raw) <= self.req_cap  # keep the fuzz simple: no wrap
         self.mm[self.req_data + pos : self.req_data + pos + len(raw)] = raw
         struct.pack_into("<Q", self.mm, self.req_head_off, head + len(raw))
+
+    def set_cursors(self, head, tail):
+        """Hostile: teleport head and tail (tail is normally server-owned)."""
+        struct.pack_into("<Q", self.mm, self.req_head_off, head)
+        struct.pack_into("<Q", self.mm, self.req_head_off + 8, tail)
+
+    def write_at(self, pos, raw):
+        """Write raw bytes at ring offset pos without touching cursors."""
+        self.mm[self.req_data + pos : self.req_data + pos + len(raw)] = raw
+
+    def publish(self, adv):
+        head = struct.unpack_from("<Q", self.mm, self.req_head_off)[0]
+        struct.pack_into("<Q", self.mm, self.req_head_off, head + adv)
 
     def push_record(self, op, body, seq, rec_len=None, body_len=None):
         body_len = len(body) if body_len is None else body_len
@@ -85,7 +102,7 @@ class RawShmConn:
         """Collect (seq, status) pairs until the ring drains."""
         import time
 
-        resp_off = CTRL + RING_HDR + REQ_CAP
+        resp_off = self.resp_off
         out = []
         t0 = time.time()
         while time.time() - t0 < timeout_s:
@@ -96,7 +113,7 @@ class RawShmConn:
                     return out
                 time.sleep(0.01)
                 continue
-            pos = tail % RESP_CAP
+            pos = tail % self.resp_cap
             ln, op = struct.unpack_from("<IB", self.mm, resp_off + RING_HDR + pos)
             seq = struct.unpack_from("<Q", self.mm, resp_off + RING_HDR + pos + 16)[0]
             status = struct.unpack_from("<i", self.mm, resp_off + RING_HDR + pos + 24)[0]
@@ -141,3 +158,58 @@ def test_shm_garbage_records(cpu_server):
     conn = make_client(cpu_server)
     assert conn.check_exist("nope") in (0, False)
     conn.close()
+
+
+def test_shm_straddling_record(cpu_server):
+    """A record whose len runs past the ring's end must NOT be followed.
+
+    The legit producer always wraps before the end, so a straddling header
+    can only be hand-written; following it reads past the ring (with a large
+    ring, past the mapped segment → SIGSEGV pre-fix). The server must detach
+    the poller and keep the socket side of the conn alive.
+    """
+    # 32 MB ring: under the old code len≈cap from a near-end pos reads ~32 MB
+    # past the segment end and crashes the server inside shm_poll_main.
+    big = 32 << 20
+    c = RawShmConn(cpu_server, req_cap=big, resp_cap=RESP_CAP)
+    try:
+        pos = big - 32  # 32 bytes before the end
+        c.set_cursors(pos, pos)  # empty ring, head at the danger zone
+        # header: len = cap-8 (8-aligned, <= cap, <= h-t after publish)
+        ln = big - 8
+        hdr = struct.pack("<IB3sIIQ", ln, ord("w"), b"\0\0\0", ln - REC_HDR, 0, 7)
+        c.write_at(pos, hdr)
+        c.publish(ln)
+        # poller must detach: no response, no crash
+        assert c.pop_responses(2.0) == []
+        # the socket path of this very conn still works
+        c._send(b"C", struct.pack("<I", 3) + b"key")
+        code = struct.unpack("<i", c._recv(4))[0]
+        assert code in (0, 1)
+    finally:
+        c.close()
+    conn = make_client(cpu_server)
+    assert conn.check_exist("nope") in (0, False)
+    conn.close()
+
+
+def test_shm_misaligned_record_len(cpu_server):
+    """A record len that is not 8-byte aligned is provably corrupt: the
+    producer rounds every record to 8 bytes. The poller detaches."""
+    c = RawShmConn(cpu_server)
+    try:
+        hdr = struct.pack("<IB3sIIQ", 44, ord("w"), b"\0\0\0", 20, 0, 3)
+        c.push_raw(hdr + b"\0" * 20)  # publishes 44 bytes... 44 % 8 != 0
+        assert c.pop_responses(2.0) == []
+        c._send(b"C", struct.pack("<I", 3) + b"key")
+        code = struct.unpack("<i", c._recv(4))[0]
+        assert code in (0, 1)
+    finally:
+        c.close()
+
+
+def test_shm_misaligned_ring_offsets_rejected(cpu_server):
+    """req_off/resp_off that are not 8-aligned would put the atomics at
+    misaligned addresses; attach() must reject the segment (handshake 500)."""
+    c = RawShmConn(cpu_server, req_off_skew=4, expect_code=500)
+    c.close()
