@@ -42,6 +42,7 @@ struct VerbsClient::Impl {
     bool msg_ready = false;
     std::unordered_set<uint32_t> done_cookies;
     int imm_count = 0;
+    uint32_t last_imm = 0;  // 0 = success; else server status (e.g. 404)
     uint32_t next_cookie = 1;
     bool error = false;
 
@@ -60,7 +61,6 @@ struct VerbsClient::Impl {
     }
 
     void on_event(vf::Ev ev, uint64_t id, uint32_t imm, uint32_t len) {
-        (void)imm;
         std::lock_guard<std::mutex> lk(mu);
         switch (ev) {
             case vf::Ev::kSendDone:
@@ -74,6 +74,7 @@ struct VerbsClient::Impl {
                 break;
             case vf::Ev::kRecvImm:
                 imm_count++;
+                last_imm = imm;  // nonzero = server-reported read failure
                 break;
             case vf::Ev::kWriteDone: {
                 uint32_t cookie = static_cast<uint32_t>(vf::Endpoint::write_cookie(id));
@@ -313,8 +314,17 @@ int VerbsClient::read_blocks(const std::vector<std::pair<std::string, uint64_t>>
     memcpy(im.ep->send_buf(sb), body.data(), body.size());
     if (!im.ep->post_send_msg(sb, body.size())) return -1;
     if (!im.wait_for([&] { return im.imm_count > imm_before; })) {
-        WARN("verbs read timed out (missing key?)");
+        WARN("verbs read timed out");
         return -1;
+    }
+    {
+        std::lock_guard<std::mutex> lk(im.mu);
+        if (im.last_imm != 0) {
+            int status = static_cast<int>(im.last_imm);
+            WARN("verbs read failed: server status %d", status);
+            im.last_imm = 0;
+            return status == KEY_NOT_FOUND ? -KEY_NOT_FOUND : -1;
+        }
     }
     return 0;
 }

@@ -773,7 +773,15 @@ void Server::maybe_extend(Shard* s) {
     if (!extending_.compare_exchange_strong(expect, 1)) return;
     std::thread([this, s] {
         INFO("extending pool on shard dev=%d", s->device());
-        s->extend();
+        void* arena = nullptr;
+        if (s->extend(&arena) && arena) {
+            // Verbs fabric live: the new arena must carry an MR or verbs
+            // reads from it fail ("pool arena not registered"); the reference
+            // registers an MR per pool at pool creation (mempool.cpp:29-44).
+            std::lock_guard<std::mutex> lk(vdrv_mu_);
+            if (vdrv_ && !vdrv_->reg_region(arena, opt_.extend_bytes, s->on_gpu(), nullptr))
+                ERROR("MR registration of extended arena failed; verbs reads from it will error");
+        }
         extending_.store(0);
     }).detach();
 }
@@ -1375,12 +1383,15 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
     Shard* shard = shard_least_used();
     std::vector<RemoteBlockWire> blocks;
     blocks.reserve(keys.size());
-    std::vector<std::pair<std::string, uint64_t>> created;  // key, hash
+    std::vector<std::tuple<std::string, uint64_t, uint64_t>> created;  // key, hash, ptr
     auto rollback = [&] {
-        for (auto& [k, h] : created) {
-            auto& st = kv_[stripe_of(h)];
-            std::lock_guard<std::shared_mutex> lk(st.mu);
-            st.map.erase(k);
+        for (auto& [k, h, p] : created) {
+            {
+                auto& st = kv_[stripe_of(h)];
+                std::lock_guard<std::shared_mutex> lk(st.mu);
+                st.map.erase(k);
+            }
+            c->pending_rdma.erase(p);  // drop the ref so the block frees now
         }
     };
     for (auto& key : keys) {
@@ -1433,10 +1444,31 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
             blocks.push_back({0, 0, 0});
             continue;  // `ref` releases and frees the block via ~BlockEntry
         }
-        created.push_back({key, h});
+        // rkey: on the verbs fabric the client posts one-sided RDMA_WRITEs
+        // with this key, so it MUST be the registered MR's real rkey for the
+        // arena holding the block (reference returns get_rkey(pool_idx) per
+        // block, infinistore.cpp:382-396 via mempool.h:56-84). On the TCP
+        // fabric the field is unused by the client (addresses key the
+        // pending map); an opaque nonzero marker keeps it distinguishable
+        // from the FAKE dup sentinel {0,0}.
+        uint32_t rkey = static_cast<uint32_t>(shard->device() + 2);
+        if (c->verbs) {
+            vf::MrInfo mr;
+            if (!vdrv_ || !vdrv_->lookup_region(ptr, &mr)) {
+                ERROR("allocate: pool arena %p has no MR on the verbs fabric", ptr);
+                {
+                    std::lock_guard<std::shared_mutex> lk(st.mu);
+                    st.map.erase(key);
+                }
+                rollback();
+                *status = SYSTEM_ERROR;
+                return {};
+            }
+            rkey = mr.rkey;
+        }
+        created.push_back({key, h, reinterpret_cast<uint64_t>(ptr)});
         c->pending_rdma.emplace(reinterpret_cast<uint64_t>(ptr), ref);
-        blocks.push_back({static_cast<uint32_t>(shard->device() + 1), 0,
-                          reinterpret_cast<uint64_t>(ptr)});
+        blocks.push_back({rkey, 0, reinterpret_cast<uint64_t>(ptr)});
     }
     maybe_extend(shard);
     return blocks;

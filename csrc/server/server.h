@@ -269,6 +269,9 @@ class Server {
     ServerOptions opt_;
     std::vector<std::unique_ptr<Shard>> shards_;
     std::unique_ptr<vf::Driver> vdrv_;  // lazy; created at first verbs handshake
+    // Serializes driver init + arena MR registration against pool extension
+    // (the extend thread registers new arenas when the driver is live).
+    std::mutex vdrv_mu_;
 
     IoLoop main_io_;
     std::vector<std::unique_ptr<IoLoop>> workers_;

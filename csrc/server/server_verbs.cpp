@@ -80,10 +80,18 @@ struct Server::VerbsPeer {
             srv->commit_addrs(conn, msg.remote_addrs);
         } else if (msg.op == OP_RDMA_READ) {
             if (msg.keys.size() != msg.remote_addrs.size() || msg.block_size <= 0) return;
+            // The client pre-posted a bare recv for our WRITE_WITH_IMM; on
+            // any failure answer with a zero-WR IMM carrying the status so
+            // it fails fast instead of waiting out its 10 s timeout (the
+            // reference just drops the request, infinistore.cpp:443-446).
+            auto fail = [&](int status) {
+                flow.submit({}, /*with_imm=*/true, static_cast<uint32_t>(status),
+                            /*cookie=*/0);
+            };
             auto held = std::make_shared<std::vector<Ref<BlockEntry>>>();
             if (!srv->collect_read_entries(msg.keys, held.get())) {
-                WARN("verbs read: missing/uncommitted key (client will time out)");
-                return;
+                WARN("verbs read: missing/uncommitted key");
+                return fail(KEY_NOT_FOUND);
             }
             std::vector<WrDesc> wrs;
             wrs.reserve(held->size());
@@ -92,7 +100,7 @@ struct Server::VerbsPeer {
                 vf::MrInfo mr;
                 if (!srv->vdrv_->lookup_region(e->ptr, &mr)) {
                     ERROR("verbs read: pool arena not registered");
-                    return;
+                    return fail(SYSTEM_ERROR);
                 }
                 wrs.push_back({reinterpret_cast<uint64_t>(e->ptr), msg.remote_addrs[i],
                                static_cast<uint32_t>(msg.block_size), mr.lkey, msg.rkey});
@@ -121,6 +129,10 @@ bool Server::verbs_handshake(Conn* c, const std::vector<uint8_t>& body,
             return false;
         }
         // Register every pool arena once (HBM arenas via dmabuf/peer-direct).
+        // Held under vdrv_mu_ so an arena added by a concurrent pool
+        // extension is registered by exactly one side (the extend thread
+        // registers only when it observes vdrv_ set).
+        std::lock_guard<std::mutex> lk(vdrv_mu_);
         bool mr_ok = true;
         for (auto& s : shards_) {
             s->for_each_arena([&](void* base, size_t sz, bool on_gpu) {

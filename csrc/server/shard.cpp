@@ -123,19 +123,21 @@ bool Shard::need_extend() {
     return opt_.auto_extend && mm_.need_extend();
 }
 
-bool Shard::extend() {
+bool Shard::extend(void** arena_out) {
     if (on_gpu()) {
         void* arena = gpu::alloc_device(opt_.device, opt_.extend_bytes);
         if (!arena) return false;
         std::lock_guard<std::mutex> lk(alloc_mu_);
         mm_.add_pool(arena, opt_.extend_bytes, opt_.block_granule,
                      [](void* p, size_t) { gpu::free_device(p); });
+        if (arena_out) *arena_out = arena;
     } else {
         void* arena = nullptr;
         if (posix_memalign(&arena, 4096, opt_.extend_bytes) != 0) return false;
         std::lock_guard<std::mutex> lk(alloc_mu_);
         mm_.add_pool(arena, opt_.extend_bytes, opt_.block_granule,
                      [](void* p, size_t) { free(p); });
+        if (arena_out) *arena_out = arena;
     }
     return true;
 }

@@ -94,7 +94,10 @@ class Shard {
     void submit_fabric(FabricJob&& job);
 
     // Extend pool by one arena (called off the hot path).
-    bool extend();
+    // Adds one arena of extend_bytes; reports the new arena's base so the
+    // caller can MR-register it on the verbs fabric (the reference registers
+    // an MR per pool as pools are created, mempool.cpp:29-44).
+    bool extend(void** arena_out = nullptr);
     bool need_extend();
 
     // Compaction planning: for each (ptr, size), try to find a lower slot in

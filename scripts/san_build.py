@@ -69,11 +69,25 @@ def main():
     if mode.startswith("mockverbs"):
         flags.insert(0, "-Itests/mock_verbs")  # <infiniband/verbs.h> -> mock
         srcs.append("tests/mock_verbs/mock_verbs.cpp")
+    # Rebuild when any header changed too (TUs inline shm_ring/wr_flow/etc.;
+    # mixing object vintages across an inline-function change is ODR UB).
+    hdr_mtime = 0.0
+    for root, _d, files in os.walk(os.path.join(REPO, "csrc")):
+        for f in files:
+            if f.endswith((".h", ".hpp")):
+                hdr_mtime = max(hdr_mtime, os.path.getmtime(os.path.join(root, f)))
+    if mode.startswith("mockverbs"):
+        for root, _d, files in os.walk(os.path.join(REPO, "tests/mock_verbs")):
+            for f in files:
+                if f.endswith(".h"):
+                    hdr_mtime = max(hdr_mtime, os.path.getmtime(os.path.join(root, f)))
+
     objs = []
     for s in srcs:
         o = f"{build}/{s.replace('/', '_')}.o"
         src = os.path.join(REPO, s)
-        if not os.path.exists(o) or os.path.getmtime(o) < os.path.getmtime(src):
+        if (not os.path.exists(o) or os.path.getmtime(o) < os.path.getmtime(src)
+                or os.path.getmtime(o) < hdr_mtime):
             r = subprocess.run(["g++", "-c", src, "-o", o] + flags,
                                capture_output=True, text=True, cwd=REPO)
             if r.returncode:

@@ -203,7 +203,10 @@ struct ibv_recv_wr {
 enum ibv_wc_status {
     IBV_WC_SUCCESS = 0,
     IBV_WC_LOC_LEN_ERR = 1,
+    IBV_WC_LOC_PROT_ERR = 4,
     IBV_WC_WR_FLUSH_ERR = 5,
+    IBV_WC_REM_ACCESS_ERR = 10,
+    IBV_WC_REM_OP_ERR = 11,
 };
 
 enum ibv_wc_opcode {

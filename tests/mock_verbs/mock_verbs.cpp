@@ -84,6 +84,45 @@ std::map<uint32_t, QpImpl*> g_qps;
 std::atomic<uint32_t> g_next_qpn{100};
 std::atomic<uint32_t> g_next_key{1000};
 
+// Registered-MR table: like a real NIC, every RDMA_WRITE's rkey must name an
+// MR that covers [remote_addr, remote_addr+len) with REMOTE_WRITE access, and
+// every SGE's lkey an MR covering the local range. The round-1 mock skipped
+// this, which let a placeholder-rkey bug (allocate responses carrying
+// device+1 instead of the pool MR's rkey) pass the whole loopback suite while
+// guaranteeing remote-access errors on real hardware.
+struct MockMr {
+    uint64_t addr;
+    size_t len;
+    uint32_t lkey;
+    uint32_t rkey;
+    int access;
+};
+std::mutex g_mr_mu;
+std::map<uint32_t, MockMr> g_mrs_by_lkey;  // rkey = lkey + 1 (one entry per MR)
+
+bool check_rkey(uint32_t rkey, uint64_t addr, size_t len) {
+    if (len == 0) return true;  // zero-length writes skip rkey checks (mlx behavior)
+    std::lock_guard<std::mutex> lk(g_mr_mu);
+    for (auto& [k, mr] : g_mrs_by_lkey) {
+        if (mr.rkey == rkey && addr >= mr.addr && addr + len <= mr.addr + mr.len)
+            return (mr.access & IBV_ACCESS_REMOTE_WRITE) != 0;
+    }
+    return false;
+}
+
+bool check_lkeys(const ibv_send_wr* wr) {
+    std::lock_guard<std::mutex> lk(g_mr_mu);
+    for (int i = 0; i < wr->num_sge; i++) {
+        const ibv_sge& s = wr->sg_list[i];
+        if (s.length == 0) continue;
+        auto it = g_mrs_by_lkey.find(s.lkey);
+        if (it == g_mrs_by_lkey.end()) return false;
+        const MockMr& mr = it->second;
+        if (s.addr < mr.addr || s.addr + s.length > mr.addr + mr.len) return false;
+    }
+    return true;
+}
+
 // One fake device.
 ibv_device g_device;
 ibv_device* g_list[2] = {&g_device, nullptr};
@@ -199,16 +238,25 @@ int ibv_modify_qp(ibv_qp* qp, ibv_qp_attr* attr, int mask) {
     return 0;
 }
 
-ibv_mr* ibv_reg_mr(ibv_pd*, void*, size_t, int) {
+ibv_mr* ibv_reg_mr(ibv_pd*, void* addr, size_t len, int access) {
     auto* mr = new ibv_mr();
-    mr->lkey = g_next_key.fetch_add(1);
-    mr->rkey = g_next_key.fetch_add(1);
+    mr->lkey = g_next_key.fetch_add(2);
+    mr->rkey = mr->lkey + 1;
+    {
+        std::lock_guard<std::mutex> lk(g_mr_mu);
+        g_mrs_by_lkey[mr->lkey] =
+            MockMr{reinterpret_cast<uint64_t>(addr), len, mr->lkey, mr->rkey, access};
+    }
     return mr;
 }
 ibv_mr* ibv_reg_dmabuf_mr(ibv_pd*, uint64_t, size_t, uint64_t, int, int) {
     return nullptr;  // force the peer-direct/plain fallback in tests
 }
 int ibv_dereg_mr(ibv_mr* mr) {
+    {
+        std::lock_guard<std::mutex> lk(g_mr_mu);
+        g_mrs_by_lkey.erase(mr->lkey);
+    }
     delete mr;
     return 0;
 }
@@ -240,6 +288,13 @@ int ibv_post_send(ibv_qp* qp, ibv_send_wr* wr, ibv_send_wr** bad) {
         lwc.wr_id = wr->wr_id;
         lwc.status = IBV_WC_SUCCESS;
         lwc.qp_num = q->pub.qp_num;
+        // lkey validation applies to every opcode's local SGEs.
+        if (!check_lkeys(wr)) {
+            lwc.status = IBV_WC_LOC_PROT_ERR;
+            lwc.opcode = wr->opcode == IBV_WR_SEND ? IBV_WC_SEND : IBV_WC_RDMA_WRITE;
+            q->send_cq->push(lwc);  // errored WRs always complete
+            continue;
+        }
         switch (wr->opcode) {
             case IBV_WR_SEND: {
                 RecvSlot slot{};
@@ -251,6 +306,20 @@ int ibv_post_send(ibv_qp* qp, ibv_send_wr* wr, ibv_send_wr** bad) {
                     }
                     slot = peer->recvq.front();
                     peer->recvq.pop_front();
+                }
+                if (sge_len(wr) > slot.length) {
+                    // message longer than the posted recv buffer: receiver
+                    // completes with a local-length error, nothing is copied.
+                    ibv_wc rwc{};
+                    rwc.wr_id = slot.wr_id;
+                    rwc.status = IBV_WC_LOC_LEN_ERR;
+                    rwc.opcode = IBV_WC_RECV;
+                    rwc.qp_num = peer->pub.qp_num;
+                    peer->recv_cq->push(rwc);
+                    lwc.opcode = IBV_WC_SEND;
+                    lwc.status = IBV_WC_REM_OP_ERR;
+                    q->send_cq->push(lwc);
+                    continue;
                 }
                 size_t n =
                     copy_sges(wr, reinterpret_cast<uint8_t*>(slot.addr), slot.length);
@@ -266,6 +335,12 @@ int ibv_post_send(ibv_qp* qp, ibv_send_wr* wr, ibv_send_wr** bad) {
             }
             case IBV_WR_RDMA_WRITE:
             case IBV_WR_RDMA_WRITE_WITH_IMM: {
+                if (!check_rkey(wr->wr.rdma.rkey, wr->wr.rdma.remote_addr, sge_len(wr))) {
+                    lwc.status = IBV_WC_REM_ACCESS_ERR;
+                    lwc.opcode = IBV_WC_RDMA_WRITE;
+                    q->send_cq->push(lwc);
+                    continue;
+                }
                 size_t n = copy_sges(wr, reinterpret_cast<uint8_t*>(wr->wr.rdma.remote_addr),
                                      sge_len(wr));
                 if (wr->opcode == IBV_WR_RDMA_WRITE_WITH_IMM) {

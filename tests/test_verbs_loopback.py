@@ -131,13 +131,99 @@ def test_verbs_async_api(cpu_server):
         conn.close()
 
 
+def test_verbs_bad_rkey_rejected(cpu_server):
+    """The mock provider validates rkeys against registered MRs (like a real
+    NIC): a write posted with a doctored rkey must fail loudly, not silently
+    succeed. Round 1's mock skipped this check, which let the server return
+    placeholder rkeys in allocate responses without any test noticing."""
+    conn = verbs_client(cpu_server)
+    try:
+        src = torch.randn(4096, dtype=torch.float32)
+        conn.register_mr(src)
+        key = f"vk-{uuid.uuid4().hex[:8]}"
+        blocks = conn.allocate_rdma([key], 4096 * 4)
+        assert len(blocks) == 1
+        rkey, addr = tuple(blocks[0])[0], tuple(blocks[0])[1]
+        assert rkey != 0 and addr != 0
+        doctored = [(rkey + 7, addr)]
+        with pytest.raises(Exception):
+            conn.rdma_write_cache(src, [0], 4096, doctored)
+            conn.sync()
+    finally:
+        conn.close()
+
+
+def test_verbs_extended_arena_registered(ports):
+    """Arenas added by pool auto-extension must be MR-registered while the
+    verbs fabric is live: allocate past the first arena's capacity, then
+    write+read a block that can only live in the extended arena. Round 1
+    registered arenas only once at the first verbs handshake, so this failed
+    with 'pool arena not registered'."""
+    import time
+
+    import infinistore_amd as ifs
+
+    service_port, manage_port = ports
+    cfg = ifs.ServerConfig(
+        service_port=service_port,
+        manage_port=manage_port,
+        prealloc_size=1,       # GB
+        extend_size=1,         # GB
+        auto_increase=True,
+        minimal_allocate_size=1024,  # KB -> 1 MB blocks, 1024 per arena
+        cpu_only=True,
+    )
+    ifs.register_server(cfg)
+    try:
+        conn = verbs_client(service_port)
+        try:
+            page_elems = 256 * 1024  # 1 MB of float32
+            run = uuid.uuid4().hex[:8]
+            total = 0
+            target = 1088  # > 1024 => the tail must come from arena 2
+            batch = 64
+            last_keys = None
+            deadline = time.time() + 60
+            while total < target and time.time() < deadline:
+                keys = [f"ve-{run}-{total + i}" for i in range(batch)]
+                try:
+                    blocks = conn.allocate_rdma(keys, page_elems * 4)
+                except Exception:
+                    time.sleep(0.3)  # extension still in flight; retry
+                    continue
+                assert len(blocks) == batch
+                total += batch
+                last_keys = (keys, blocks)
+            assert total >= target, f"only {total} blocks allocated"
+            keys, blocks = last_keys
+            src = torch.randn(page_elems, dtype=torch.float32)
+            dst = torch.zeros_like(src)
+            conn.register_mr(src)
+            conn.register_mr(dst)
+            conn.rdma_write_cache(src, [0], page_elems, [tuple(blocks[0])])
+            conn.sync()
+            conn.read_cache(dst, [(keys[0], 0)], page_elems)
+            conn.sync()
+            assert torch.equal(src, dst)
+        finally:
+            conn.close()
+    finally:
+        ifs.unregister_server()
+
+
 def test_verbs_missing_key(cpu_server):
+    """The server answers a read of a missing key with an error IMM; the
+    client must fail FAST (not wait out its 10 s CQ timeout)."""
+    import time
+
     conn = verbs_client(cpu_server)
     try:
         dst = torch.zeros(1024, dtype=torch.float32)
         conn.register_mr(dst)
+        t0 = time.time()
         with pytest.raises(Exception):
             conn.read_cache(dst, [(f"vm-{uuid.uuid4().hex}", 0)], 1024)
             conn.sync()
+        assert time.time() - t0 < 5.0, "missing-key read should error via IMM, not timeout"
     finally:
         conn.close()
