@@ -487,8 +487,12 @@ int ClientConn::sync_local() {
         return 0;
     }
     if (!send_req(OP_SYNC, nullptr, 0)) return -1;
-    int remain = -1;
-    if (!recv_status(&remain)) return -1;
+    // Reference framing: FINISH + int remain (infinistore.cpp:1070-1075,
+    // libinfinistore.cpp:632-657).
+    int code = -1, remain = -1;
+    if (!recv_status(&code)) return -1;
+    if (code != FINISH) return code > 0 ? -code : code;
+    if (!recv_exact(fd_, &remain, 4)) return -1;
     if (remain == 0) local_dirty_ = false;
     return remain;
 }
@@ -768,21 +772,22 @@ int ClientConn::sync_rdma() {
 // ---------------------------------------------------------------------------
 int ClientConn::check_exist(const std::string& key) {
     if (!connected_) return -1;
-    std::vector<uint8_t> body(4 + key.size());
-    uint32_t len = static_cast<uint32_t>(key.size());
-    memcpy(body.data(), &len, 4);
-    memcpy(body.data() + 4, key.data(), key.size());
+    // Reference framing: body = raw key bytes (body_size IS the key length,
+    // libinfinistore.cpp:659-671); response FINISH + int (0 exists, 1 not).
+    const uint8_t* body = reinterpret_cast<const uint8_t*>(key.data());
     std::lock_guard<std::mutex> lk(io_mu_);
     // Query ops ride the shm ring too (the reply status carries the value);
     // prefix lookups sit on the decode critical path.
     if (shm_active_) {
-        int code = shm_request(OP_CHECK_EXIST, body.data(), body.size(), true);
+        int code = shm_request(OP_CHECK_EXIST, body, key.size(), true);
         if (code != kShmNoFit) return code == kShmErr ? -1 : code;
     }
-    if (!send_req(OP_CHECK_EXIST, body.data(), body.size())) return -1;
-    int code = -1;
+    if (!send_req(OP_CHECK_EXIST, body, key.size())) return -1;
+    int code = -1, exist = -1;
     if (!recv_status(&code)) return -1;
-    return code;  // 0 exists, 1 not
+    if (code != FINISH) return -1;
+    if (!recv_exact(fd_, &exist, 4)) return -1;
+    return exist;  // 0 exists, 1 not
 }
 
 int ClientConn::delete_keys(const std::vector<std::string>& keys) {
@@ -819,8 +824,11 @@ int ClientConn::get_match_last_index(const std::vector<std::string>& keys) {
         if (code != kShmNoFit) return code == kShmErr ? -1 : code;
     }
     if (!send_req(OP_GET_MATCH_LAST_IDX, body.data(), body.size())) return -1;
-    int idx = -1;
-    if (!recv_status(&idx)) return -1;
+    // Reference framing: FINISH + int index (infinistore.cpp:1092-1108).
+    int code = -1, idx = -1;
+    if (!recv_status(&code)) return -1;
+    if (code != FINISH) return -1;
+    if (!recv_exact(fd_, &idx, 4)) return -1;
     return idx;
 }
 

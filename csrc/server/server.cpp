@@ -118,6 +118,17 @@ void send_status_payload(Server::Conn* c, int code, const uint8_t* payload, size
     send_buf(c, std::move(v));
 }
 
+// Reference-framed response: status + raw payload with NO length prefix
+// (the reference's send_resp, infinistore.cpp:1055-1068 — its clients know
+// each op's payload size a priori). Used by the query ops for byte-level
+// wire compatibility.
+void send_status_raw(Server::Conn* c, int code, const void* payload, size_t n) {
+    std::vector<uint8_t> v(4 + n);
+    memcpy(v.data(), &code, 4);
+    if (n) memcpy(v.data() + 4, payload, n);
+    send_buf(c, std::move(v));
+}
+
 void send_raw(Server::Conn* c, std::shared_ptr<uint8_t[]> buf, size_t len) {
     if (c->closed) return;
     auto* wr = new WriteReq();
@@ -1171,7 +1182,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
 void Server::op_sync(Conn* c, const ReqCtx& ctx) {
     std::lock_guard<std::mutex> lk(c->sync_mu);
     if (c->remain.load() == 0) {
-        reply_local(c, ctx, 0);
+        reply_query(c, ctx, 0);  // FINISH + remain (infinistore.cpp:1070-1075)
     } else {
         c->sync_waiting = true;  // answered by finish_task when remain drains
         c->sync_ctx = ctx;
@@ -1193,13 +1204,13 @@ void Server::finish_task(Conn* c, bool on_owner) {
         }
         if (respond) {
             if (ctx.shm || on_owner) {
-                reply_local(c, ctx, 0);
+                reply_query(c, ctx, 0);
             } else {
                 // socket reply from a completion thread: hop to the owner
                 // loop (uv_write is not thread-safe).
                 c->ref();
                 c->owner->post([this, c, ctx] {
-                    reply_local(c, ctx, 0);
+                    reply_query(c, ctx, 0);
                     c->unref();
                 });
             }
@@ -1211,6 +1222,15 @@ void Server::finish_task(Conn* c, bool on_owner) {
 void Server::reply_local(Conn* c, const ReqCtx& ctx, int code) {
     if (ctx.shm && c->shm) return c->shm->push_resp(ctx.seq, code);
     send_status(c, code);
+}
+
+// Query-op responses (check_exist / match / sync): the reference frames
+// these as FINISH + 4-byte value on the socket (libinfinistore.cpp:679-694,
+// infinistore.cpp:1070-1107); the shm ring carries the value in its
+// fixed-size status field instead (ring responses are always 32 bytes).
+void Server::reply_query(Conn* c, const ReqCtx& ctx, int value) {
+    if (ctx.shm && c->shm) return c->shm->push_resp(ctx.seq, value);
+    send_status_raw(c, FINISH, &value, 4);
 }
 
 // ---- shared-memory ring transport -----------------------------------------
@@ -1689,12 +1709,11 @@ void Server::op_tcp_get(Conn* c, const RemoteMetaMsg& msg) {
 
 // ---- queries ---------------------------------------------------------------
 void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& ctx) {
-    // body: u32 len + key bytes (simple framing; no flatbuffer needed)
-    if (body.size() < 4) return reply_local(c, ctx, INVALID_REQ);
-    uint32_t len;
-    memcpy(&len, body.data(), 4);
-    if (body.size() < 4 + len) return reply_local(c, ctx, INVALID_REQ);
-    std::string key(reinterpret_cast<const char*>(body.data() + 4), len);
+    // body: raw key bytes (reference framing — body_size IS the key length,
+    // libinfinistore.cpp:659-671); response FINISH + int, 0 = exists
+    // (infinistore.cpp:1078-1090).
+    if (body.empty()) return reply_local(c, ctx, INVALID_REQ);
+    std::string key(reinterpret_cast<const char*>(body.data()), body.size());
     bool exists;
     {
         uint64_t h = KvMap::hash_of(key);
@@ -1703,13 +1722,13 @@ void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body, const Req
         Ref<BlockEntry>* v = st.map.find_hashed(key, h);
         exists = v && (*v)->committed && !expired(v->get());
     }
-    reply_local(c, ctx, exists ? 0 : 1);
+    reply_query(c, ctx, exists ? 0 : 1);
 }
 
 void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& ctx) {
     std::vector<std::string> keys;
     if (!parse_match_request(body.data(), body.size(), &keys) || keys.empty())
-        return reply_local(c, ctx, -1);
+        return reply_query(c, ctx, -1);
     // Binary search for the last present index, assuming the prefix property
     // (keys[0..i] present iff i <= match). Requires committed entries —
     // divergence from the reference, which counts uncommitted keys as
@@ -1730,7 +1749,7 @@ void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body, const Req
         else
             right = mid;
     }
-    reply_local(c, ctx, static_cast<int>(left - 1));
+    reply_query(c, ctx, static_cast<int>(left - 1));
 }
 
 void Server::op_delete(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& ctx) {

@@ -235,6 +235,7 @@ class Server {
     // Route a status to the request's transport (ring: any thread; socket:
     // owner loop thread only).
     void reply_local(Conn* c, const ReqCtx& ctx, int code);
+    void reply_query(Conn* c, const ReqCtx& ctx, int value);
     void op_shm_setup(Conn* c, const std::vector<uint8_t>& body);
     void shm_teardown(Conn* c);       // owner loop thread; joins the poller
     void shm_poll_main(ShmPeer* p);   // poller thread body

@@ -149,9 +149,9 @@ def test_shm_garbage_records(cpu_server):
                           seq=10 + i)
         assert len(c.pop_responses(20)) >= 1
         # 5. the socket side still works after all of that
-        c._send(b"C", struct.pack("<I", 3) + b"key")
-        code = struct.unpack("<i", c._recv(4))[0]
-        assert code in (0, 1)
+        c._send(b"C", b"key")  # reference framing: raw key bytes
+        code, exist = struct.unpack("<ii", c._recv(8))
+        assert code == 200 and exist in (0, 1)
     finally:
         c.close()
     # 6. and the server accepts fresh clients
@@ -183,9 +183,9 @@ def test_shm_straddling_record(cpu_server):
         # poller must detach: no response, no crash
         assert c.pop_responses(2.0) == []
         # the socket path of this very conn still works
-        c._send(b"C", struct.pack("<I", 3) + b"key")
-        code = struct.unpack("<i", c._recv(4))[0]
-        assert code in (0, 1)
+        c._send(b"C", b"key")  # reference framing: raw key bytes
+        code, exist = struct.unpack("<ii", c._recv(8))
+        assert code == 200 and exist in (0, 1)
     finally:
         c.close()
     conn = make_client(cpu_server)
@@ -201,9 +201,9 @@ def test_shm_misaligned_record_len(cpu_server):
         hdr = struct.pack("<IB3sIIQ", 44, ord("w"), b"\0\0\0", 20, 0, 3)
         c.push_raw(hdr + b"\0" * 20)  # publishes 44 bytes... 44 % 8 != 0
         assert c.pop_responses(2.0) == []
-        c._send(b"C", struct.pack("<I", 3) + b"key")
-        code = struct.unpack("<i", c._recv(4))[0]
-        assert code in (0, 1)
+        c._send(b"C", b"key")  # reference framing: raw key bytes
+        code, exist = struct.unpack("<ii", c._recv(8))
+        assert code == 200 and exist in (0, 1)
     finally:
         c.close()
 
