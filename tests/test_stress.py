@@ -112,11 +112,22 @@ def test_garbage_flatbuffer_body(cpu_server):
     body = b"\xff" * 64
     s.sendall(struct.pack("<IcI", 0xDEADBEEF, b"M", len(body)) + body)
     s.settimeout(5)
-    code = struct.unpack("<i", s.recv(4))[0]
-    assert code < 0 or code == 400  # error, but connection alive
+    # reference query framing: FINISH + i32 value (-1 = no match / bad body)
+    code, val = struct.unpack("<ii", _recv_exact(s, 8))
+    assert code == 200 and val == -1  # error value, but connection alive
     s.sendall(struct.pack("<IcI", 0xDEADBEEF, b"S", 0))
-    assert struct.unpack("<i", s.recv(4))[0] == 0
+    code, remain = struct.unpack("<ii", _recv_exact(s, 8))
+    assert code == 200 and remain == 0
     s.close()
+
+
+def _recv_exact(s, n):
+    out = b""
+    while len(out) < n:
+        chunk = s.recv(n - len(out))
+        assert chunk
+        out += chunk
+    return out
 
 
 def test_kv_index_rehash_churn(ports):
