@@ -407,6 +407,114 @@ void Shard::submit_fabric(FabricJob&& job) {
     fabric_cv_.notify_one();
 }
 
+bool Shard::fabric_stage_init() {
+    if (fab_ready_) return true;
+    fab_stream_ = gpu::stream_create(opt_.device);
+    if (!fab_stream_) return false;
+    for (auto& b : fab_) {
+        b.h_stage = static_cast<uint8_t*>(gpu::alloc_host_pinned(kFabStageBytes));
+        b.d_stage = static_cast<uint8_t*>(gpu::alloc_device(opt_.device, kFabStageBytes));
+        b.h_desc = static_cast<uint64_t*>(
+            gpu::alloc_host_pinned(2 * kFabStageDescs * sizeof(uint64_t)));
+        b.d_desc = static_cast<uint64_t*>(
+            gpu::alloc_device(opt_.device, 2 * kFabStageDescs * sizeof(uint64_t)));
+        b.event = gpu::event_create(opt_.device);
+        if (!b.h_stage || !b.d_stage || !b.h_desc || !b.d_desc || !b.event) {
+            fabric_stage_teardown();
+            return false;
+        }
+    }
+    fab_ready_ = true;
+    return true;
+}
+
+void Shard::fabric_stage_teardown() {
+    for (auto& b : fab_) {
+        if (b.h_stage) gpu::free_host_pinned(b.h_stage);
+        if (b.d_stage) gpu::free_device(b.d_stage);
+        if (b.h_desc) gpu::free_host_pinned(b.h_desc);
+        if (b.d_desc) gpu::free_device(b.d_desc);
+        if (b.event) gpu::event_destroy(b.event);
+        b = FabBuf{};
+    }
+    if (fab_stream_) gpu::stream_destroy(fab_stream_);
+    fab_stream_ = nullptr;
+    fab_ready_ = false;
+}
+
+// Staged GPU fabric transfer. PUT: pageable body -> pinned stage (memcpy),
+// one H2D of the chunk, one scatter-kernel launch stage->pool blocks.
+// GET: one gather-kernel launch pool->stage, one D2H, memcpy out into the
+// response buffer. Chunks alternate between two buffers; the host memcpy
+// of one chunk overlaps the device work of the other, so the sustained
+// rate is bounded by host memcpy bandwidth, not PCIe round trips.
+bool Shard::fabric_run_gpu(FabricJob& job, uint8_t* base) {
+    const size_t bs = job.bytes_per_block;
+    const size_t n = job.block_ptrs.size();
+    const size_t per_chunk = std::min(kFabStageBytes / bs, kFabStageDescs);
+    if (per_chunk == 0) return false;  // block larger than the stage: refuse
+    const bool aligned = bs % 16 == 0;
+    bool ok = true;
+
+    auto flush = [&](FabBuf& b) {  // wait chunk; for GET copy payload out
+        if (!b.in_flight) return;
+        if (ok) ok = gpu::event_sync(b.event);
+        if (ok && !job.is_put) {
+            for (size_t i = 0; i < b.out_n; i++)
+                memcpy(base + job.host_offsets[b.out_first + i], b.h_stage + i * bs, bs);
+        }
+        b.in_flight = false;
+    };
+
+    int cur = 0;
+    for (size_t off = 0; off < n && ok; off += per_chunk) {
+        FabBuf& b = fab_[cur];
+        flush(b);  // buffer must be idle before reuse
+        if (!ok) break;
+        const size_t take = std::min(per_chunk, n - off);
+        if (job.is_put) {
+            for (size_t i = 0; i < take; i++)
+                memcpy(b.h_stage + i * bs, base + job.host_offsets[off + i], bs);
+            for (size_t i = 0; i < take; i++) {
+                b.h_desc[i] = reinterpret_cast<uint64_t>(b.d_stage + i * bs);
+                b.h_desc[kFabStageDescs + i] = job.block_ptrs[off + i];
+            }
+            ok = gpu::memcpy_h2d_async(b.d_stage, b.h_stage, take * bs, fab_stream_) &&
+                 gpu::memcpy_h2d_async(b.d_desc, b.h_desc, take * sizeof(uint64_t),
+                                       fab_stream_) &&
+                 gpu::memcpy_h2d_async(b.d_desc + kFabStageDescs,
+                                       b.h_desc + kFabStageDescs, take * sizeof(uint64_t),
+                                       fab_stream_) &&
+                 gpu::launch_copy_blocks(opt_.device, fab_stream_, b.d_desc,
+                                         b.d_desc + kFabStageDescs, static_cast<int>(take),
+                                         bs, aligned) &&
+                 gpu::event_record(b.event, fab_stream_);
+        } else {
+            for (size_t i = 0; i < take; i++) {
+                b.h_desc[i] = job.block_ptrs[off + i];
+                b.h_desc[kFabStageDescs + i] = reinterpret_cast<uint64_t>(b.d_stage + i * bs);
+            }
+            b.out_first = off;
+            b.out_n = take;
+            ok = gpu::memcpy_h2d_async(b.d_desc, b.h_desc, take * sizeof(uint64_t),
+                                       fab_stream_) &&
+                 gpu::memcpy_h2d_async(b.d_desc + kFabStageDescs,
+                                       b.h_desc + kFabStageDescs, take * sizeof(uint64_t),
+                                       fab_stream_) &&
+                 gpu::launch_copy_blocks(opt_.device, fab_stream_, b.d_desc,
+                                         b.d_desc + kFabStageDescs, static_cast<int>(take),
+                                         bs, aligned) &&
+                 gpu::memcpy_d2h_async(b.h_stage, b.d_stage, take * bs, fab_stream_) &&
+                 gpu::event_record(b.event, fab_stream_);
+        }
+        b.in_flight = ok;
+        cur ^= 1;
+    }
+    flush(fab_[cur]);
+    flush(fab_[cur ^ 1]);
+    return ok;
+}
+
 void Shard::fabric_loop() {
     if (on_gpu()) gpu::set_device(opt_.device);
     for (;;) {
@@ -414,7 +522,7 @@ void Shard::fabric_loop() {
         {
             std::unique_lock<std::mutex> lk(fabric_mu_);
             fabric_cv_.wait(lk, [this] { return stopping_ || !fabric_q_.empty(); });
-            if (stopping_ && fabric_q_.empty()) return;
+            if (stopping_ && fabric_q_.empty()) break;
             if (fabric_q_.empty()) continue;
             job = std::move(fabric_q_.front());
             fabric_q_.pop_front();
@@ -423,27 +531,24 @@ void Shard::fabric_loop() {
                                      : (job.host ? job.host->data() : nullptr);
         bool ok = base != nullptr;
         size_t bs = job.bytes_per_block;
-        if (job.is_put) {
+        if (ok && on_gpu() && bs > 0 && bs <= kFabStageBytes && fabric_stage_init()) {
+            ok = fabric_run_gpu(job, base);
+        } else if (ok) {
+            // CPU shard (or staging alloc failure): plain memcpy per block.
             for (size_t i = 0; i < job.block_ptrs.size() && ok; i++) {
-                const uint8_t* src = base + job.host_offsets[i];
-                void* dst = reinterpret_cast<void*>(job.block_ptrs[i]);
+                uint8_t* hp = base + job.host_offsets[i];
+                void* bp = reinterpret_cast<void*>(job.block_ptrs[i]);
                 if (on_gpu())
-                    ok = gpu::memcpy_h2d(dst, src, bs);
+                    ok = job.is_put ? gpu::memcpy_h2d(bp, hp, bs) : gpu::memcpy_d2h(hp, bp, bs);
+                else if (job.is_put)
+                    memcpy(bp, hp, bs);
                 else
-                    memcpy(dst, src, bs);
-            }
-        } else {
-            for (size_t i = 0; i < job.block_ptrs.size() && ok; i++) {
-                uint8_t* dst = base + job.host_offsets[i];
-                const void* src = reinterpret_cast<const void*>(job.block_ptrs[i]);
-                if (on_gpu())
-                    ok = gpu::memcpy_d2h(dst, src, bs);
-                else
-                    memcpy(dst, src, bs);
+                    memcpy(hp, bp, bs);
             }
         }
         if (job.done) job.done(ok);
     }
+    if (on_gpu()) fabric_stage_teardown();
 }
 
 }  // namespace ifs

@@ -165,6 +165,32 @@ class Shard {
     std::mutex fabric_mu_;
     std::condition_variable fabric_cv_;
     std::thread fabric_thread_;
+
+    // Fabric staging (GPU shards): the TCP data path moves payloads through
+    // a pinned bounce buffer with ONE hipMemcpyAsync + one scatter/gather
+    // kernel launch per ~32 MB chunk, double-buffered so the host-side
+    // memcpy of chunk k+1 overlaps the device work of chunk k. Replaces the
+    // per-block synchronous hipMemcpy loop (the reference's hot-loop shape,
+    // infinistore.cpp:622-625/747-748, which capped the TCP fabric at
+    // ~1.5 GB/s).
+    static constexpr size_t kFabStageBytes = 32ull << 20;
+    static constexpr size_t kFabStageDescs = 8192;
+    struct FabBuf {
+        uint8_t* h_stage = nullptr;  // pinned payload bounce
+        uint8_t* d_stage = nullptr;  // device-side chunk
+        uint64_t* h_desc = nullptr;  // pinned [src[descs] | dst[descs]]
+        uint64_t* d_desc = nullptr;
+        gpu::Event event = nullptr;  // chunk-complete marker
+        bool in_flight = false;
+        // GET bookkeeping for the deferred host copy-out of this chunk
+        size_t out_first = 0, out_n = 0;
+    };
+    FabBuf fab_[2];
+    gpu::Stream fab_stream_ = nullptr;
+    bool fab_ready_ = false;
+    bool fabric_stage_init();          // fabric thread only
+    void fabric_stage_teardown();      // fabric thread only
+    bool fabric_run_gpu(FabricJob& job, uint8_t* base);
 };
 
 }  // namespace ifs
