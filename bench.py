@@ -117,7 +117,20 @@ def main():
     # cross-rank reads (--cross) can still be verified against local src.
     torch.manual_seed(12345)
     src = torch.randn(total_elems, dtype=torch.bfloat16, device=dev)
-    dst = torch.zeros_like(src)
+    # Anti-staleness salting: step s writes srcs[s % 4] whose every block
+    # carries a distinct salt in its first element, and step s reads land in
+    # dsts[s % 2] (adjacent in-flight reads must not share a destination).
+    # Verification then detects a read that returned another step's blocks
+    # for any staleness within 3 steps — deeper than the pipeline ever runs.
+    # (Round 1 wrote identical payloads every step, so returning step s-1's
+    # blocks verified clean.)
+    n_salt = 4
+    srcs = [src.clone() for _ in range(n_salt)]
+    for b in range(n_salt):
+        srcs[b][0::elems_per_block] = float(b + 1)
+    src = srcs[0]
+    dsts = [torch.zeros_like(src) for _ in range(2)]
+    dst = dsts[0]
 
     use_local_path = have_gpu and not external
     ccfg = ifs.ClientConfig(
@@ -129,8 +142,10 @@ def main():
     conn = ifs.InfinityConnection(ccfg)
     conn.connect()
     if not use_local_path:
-        conn.register_mr(src)
-        conn.register_mr(dst)
+        for t in srcs:
+            conn.register_mr(t)
+        for t in dsts:
+            conn.register_mr(t)
 
     # Extra connections for intra-rank parallelism (local path only): the
     # blocking waits release the GIL and the server spreads connections over
@@ -169,21 +184,23 @@ def main():
         owner = rank if owner is None else owner
         return [f"r{owner}-s{step}-{run_id}-{i}" for i in range(args.blocks)]
 
-    def do_put(keys):
+    def do_put(keys, s=0):
+        sb = srcs[s % n_salt]
         if use_local_path:
-            conn.write_pages(src, keys, offsets_np, elems_per_block, sync=True,
+            conn.write_pages(sb, keys, offsets_np, elems_per_block, sync=True,
                              quant=args.quant)
         else:
             blocks = conn.allocate_rdma(keys, block_bytes)
-            conn.rdma_write_cache(src, offsets, elems_per_block, blocks)
+            conn.rdma_write_cache(sb, offsets, elems_per_block, blocks)
             conn.sync()
 
-    def do_get(keys):
+    def do_get(keys, s=0):
+        db = dsts[s % 2]
         if use_local_path:
-            conn.read_pages(dst, keys, offsets_np, elems_per_block)
+            conn.read_pages(db, keys, offsets_np, elems_per_block)
             conn.sync()
         else:
-            conn.read_cache(dst, list(zip(keys, offsets)), elems_per_block)
+            conn.read_cache(db, list(zip(keys, offsets)), elems_per_block)
             conn.sync()
 
     def purge_all():
@@ -225,27 +242,34 @@ def main():
         if use_local_path and not cross:
             wk = [f"warm-{k}" for k in step_keys(w)]
             for c in range(n_conns):  # warm every conn's IPC export + slab
-                wconns[c].write_pages(src, wk[csl[c]], coff[c],
+                wconns[c].write_pages(srcs[w % n_salt], wk[csl[c]], coff[c],
                                       elems_per_block, sync=True,
                                       quant=args.quant)
-                conns[c].read_pages(dst, wk[csl[c]], coff[c], elems_per_block)
+                conns[c].read_pages(dsts[w % 2], wk[csl[c]], coff[c],
+                                    elems_per_block)
                 conns[c].sync()
         else:
-            do_put([f"warm-{k}" for k in step_keys(w)])
+            do_put([f"warm-{k}" for k in step_keys(w)], w)
             if cross:
                 dist.barrier()
-            do_get([f"warm-{k}" for k in step_keys(w, read_rank)])
-    def verify(tag):
+            do_get([f"warm-{k}" for k in step_keys(w, read_rank)], w)
+
+    def verify(tag, s):
+        """dsts[s%2] must hold exactly step s's salted payload; a stale or
+        cross-mixed read from any step within ±3 carries a different salt
+        in the block's first element and fails the comparison."""
+        sb, db = srcs[s % n_salt], dsts[s % 2]
         if args.quant:  # fp8 roundtrip: ~3 mantissa bits, per-page scale
-            ok = torch.allclose(src.float().cpu(), dst.float().cpu(),
-                                atol=float(src.abs().max()) * 0.07)
+            ok = torch.allclose(sb.float().cpu(), db.float().cpu(),
+                                atol=float(sb.abs().max()) * 0.07)
         else:
-            ok = torch.equal(src.cpu(), dst.cpu())
+            ok = torch.equal(sb.cpu(), db.cpu())
         if not ok:
-            print(json.dumps({"error": f"data mismatch in {tag}"}))
+            print(json.dumps({"error": f"data mismatch in {tag} (step {s})"}))
             sys.exit(1)
 
-    verify("warmup")
+    if args.warmup:
+        verify("warmup", args.warmup - 1)
     purge_all()
 
     # ---- timed region ----
@@ -274,73 +298,73 @@ def main():
             # the end of iteration k-1 drains the W(k+1) pushed there, and
             # the prime below covers W(0)/W(1)). While the client waits on
             # R(s)'s ticket, W(s+2) and R(s+1) are already queued server-
-            # side, so the GPU always has copy work in flight.
+            # side, so the GPU always has copy work in flight. W(s) writes
+            # srcs[s%4] (distinct per-step salt) and R(s) lands in dsts[s%2]
+            # so adjacent in-flight reads never share a destination.
             rc_, wc_, o = conns[c], wconns[c], coff[c]
             K = args.steps
-            pt = gt = 0.0
-            t = time.perf_counter()
-            wc_.write_pages(src, put_blobs[0][c], o, elems_per_block, sync=True,
-                            quant=args.quant)
+            wc_.write_pages(srcs[0], put_blobs[0][c], o, elems_per_block,
+                            sync=True, quant=args.quant)
             if K > 1:
-                wc_.write_pages(src, put_blobs[1][c], o, elems_per_block,
-                                sync=False, quant=args.quant)
+                wc_.write_pages(srcs[1 % n_salt], put_blobs[1][c], o,
+                                elems_per_block, sync=False, quant=args.quant)
                 wc_.sync()
-            pt += time.perf_counter() - t
-            tk = rc_.read_pages_async(dst, get_blobs[0][c], o, elems_per_block)
+            tk = rc_.read_pages_async(dsts[0], get_blobs[0][c], o,
+                                      elems_per_block)
             for s in range(K):
-                t = time.perf_counter()
                 if s + 2 < K:
-                    wc_.write_pages(src, put_blobs[s + 2][c], o,
-                                    elems_per_block, sync=False, quant=args.quant)
-                tk_next = (rc_.read_pages_async(dst, get_blobs[s + 1][c], o,
+                    wc_.write_pages(srcs[(s + 2) % n_salt], put_blobs[s + 2][c],
+                                    o, elems_per_block, sync=False,
+                                    quant=args.quant)
+                tk_next = (rc_.read_pages_async(dsts[(s + 1) % 2],
+                                                get_blobs[s + 1][c], o,
                                                 elems_per_block)
                            if s + 1 < K else None)
-                tg = time.perf_counter()
                 rc_.wait_read(tk)
                 wc_.sync()  # commits W(s+2) before R(s+2) is pushed next iter
-                pt += tg - t
-                gt += time.perf_counter() - tg
                 tk = tk_next
-            return pt, gt
 
         if n_conns == 1:
-            put_time, get_time = run_conn(0)
+            run_conn(0)
         else:
             import concurrent.futures as cf
 
             with cf.ThreadPoolExecutor(n_conns) as ex:
-                res = list(ex.map(run_conn, range(n_conns)))
-            put_time = max(r[0] for r in res)
-            get_time = max(r[1] for r in res)
+                list(ex.map(run_conn, range(n_conns)))
     else:
         for s in range(args.steps):
             tp = time.perf_counter()
             if debug and use_local_path:
-                conn.write_pages(src, put_keys[s], offsets_np, elems_per_block, sync=True)
+                conn.write_pages(srcs[s % n_salt], put_keys[s], offsets_np,
+                                 elems_per_block, sync=True)
                 tb = time.perf_counter()
                 conn.sync()
                 tc = time.perf_counter()
                 debug_t["put_req"] += tb - tp
                 debug_t["put_sync"] += tc - tb
             else:
-                do_put(put_keys[s])
+                do_put(put_keys[s], s)
             put_time += time.perf_counter() - tp
             if cross:
                 dist.barrier()  # readers wait for the writer of their keys
             tg = time.perf_counter()
             if debug and use_local_path:
-                conn.read_pages(dst, get_keys[s], offsets_np, elems_per_block)
+                conn.read_pages(dsts[s % 2], get_keys[s], offsets_np,
+                                elems_per_block)
                 tb = time.perf_counter()
                 conn.sync()
                 debug_t["get_req"] += tb - tg
                 debug_t["get_sync"] += time.perf_counter() - tb
             else:
-                do_get(get_keys[s])
+                do_get(get_keys[s], s)
             get_time += time.perf_counter() - tg
     sync_all()
     elapsed = time.perf_counter() - t0
-    if pipeline:
-        verify("pipelined loop")
+    # Verify the last TWO steps (they cover both dst buffers and two salt
+    # values), pipelined or not.
+    verify("timed loop", args.steps - 1)
+    if args.steps > 1:
+        verify("timed loop", args.steps - 2)
     if debug:
         per = {k: round(v / args.steps * 1e6, 1) for k, v in debug_t.items()}
         print(f"rank {rank} per-step us: {per}", file=sys.stderr)
@@ -352,6 +376,30 @@ def main():
         elapsed, put_time, get_time = t.tolist()
 
     purge_all()
+
+    # ---- honest per-direction throughput (sequential, commit-to-commit) ----
+    # The pipelined loop overlaps puts with gets, so splitting its wall time
+    # into put/get components is not meaningful (round 1 reported host-issue
+    # time of async writes as put_GBps — 2-3x the sustainable rate). Measure
+    # put and get separately: each put timed to commit-ACK (sync), each get
+    # to completion.
+    seq_steps = min(args.steps, 4) if pipeline else 0
+    seq_put = seq_get = 0.0
+    for s in range(seq_steps):
+        ks = [f"seq-{k}" for k in put_keys[s]]
+        t1 = time.perf_counter()
+        do_put(ks, s)
+        t2 = time.perf_counter()
+        do_get(ks, s)
+        seq_put += t2 - t1
+        seq_get += time.perf_counter() - t2
+    if seq_steps:
+        verify("sequential phase", seq_steps - 1)
+        purge_all()
+    if dist and seq_steps:
+        t = torch.tensor([seq_put, seq_get], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        seq_put, seq_get = t.tolist()
 
     # ---- latency phase (single-block round trips) ----
     lat_put, lat_get = [], []
@@ -381,8 +429,18 @@ def main():
     bytes_per_step_rank = args.blocks * block_bytes
     total_gb = 2.0 * bytes_per_step_rank * args.steps * world / 1e9  # put+get
     gbps = total_gb / elapsed
-    put_gbps = bytes_per_step_rank * args.steps * world / 1e9 / put_time
-    get_gbps = bytes_per_step_rank * args.steps * world / 1e9 / get_time
+    # Per-direction rates come from sequential commit-to-commit timing only
+    # (the pipelined loop's directions overlap and cannot be split honestly).
+    if pipeline:
+        put_gbps = (bytes_per_step_rank * seq_steps * world / 1e9 / seq_put
+                    if seq_steps and seq_put > 0 else None)
+        get_gbps = (bytes_per_step_rank * seq_steps * world / 1e9 / seq_get
+                    if seq_steps and seq_get > 0 else None)
+    else:
+        put_gbps = (bytes_per_step_rank * args.steps * world / 1e9 / put_time
+                    if put_time > 0 else None)
+        get_gbps = (bytes_per_step_rank * args.steps * world / 1e9 / get_time
+                    if get_time > 0 else None)
 
     if rank == 0:
         result = {
@@ -406,8 +464,9 @@ def main():
                 "seq_len": 0,
                 "parallelism": f"shard{world}",
                 "path": "local_gpu_ipc" if use_local_path else ("fabric_remote" if external else "tcp_fabric_cpu"),
-                "put_GBps": round(put_gbps, 3),
-                "get_GBps": round(get_gbps, 3),
+                # sequential commit-to-commit rates (sustained, unpipelined)
+                "put_GBps": round(put_gbps, 3) if put_gbps else None,
+                "get_GBps": round(get_gbps, 3) if get_gbps else None,
                 "p50_put_us": round(pct(lat_put, 50), 1),
                 "p99_put_us": round(pct(lat_put, 99), 1),
                 "p50_get_us": round(pct(lat_get, 50), 1),
