@@ -189,6 +189,36 @@ def test_gpu_tensor_fabric_roundtrip(gpu_server):
         conn.close()
 
 
+def test_fabric_staged_multichunk_roundtrip(gpu_server):
+    """TCP fabric with a GPU-shard pool across several 32 MB staging chunks:
+    600 x 128 KB = 75 MB exercises the double-buffered pinned-stage pipeline
+    (chunk k+1's host memcpy overlapping chunk k's H2D+kernel) in both
+    directions, including the chunk-boundary descriptor handoff."""
+    conn = make_client(gpu_server)
+    try:
+        n, elems = 600, 32768  # 128 KB blocks of float32
+        src = torch.randn(n * elems)  # CPU tensors: the fabric moves bytes
+        dst = torch.zeros_like(src)
+        conn.register_mr(src)
+        conn.register_mr(dst)
+        pre = uuid.uuid4().hex[:8]
+        keys = [f"fsm-{pre}-{i}" for i in range(n)]
+        offs = [i * elems for i in range(n)]
+        blocks = conn.allocate_rdma(keys, elems * 4)
+        conn.rdma_write_cache(src, offs, elems, blocks)
+        conn.sync()
+        # read back in a shuffled order so stage slots differ from put order
+        import random
+
+        order = list(range(n))
+        random.Random(7).shuffle(order)
+        conn.read_cache(dst, [(keys[i], offs[i]) for i in order], elems)
+        conn.sync()
+        assert torch.equal(src, dst)
+    finally:
+        conn.close()
+
+
 def _mix64(x):
     M = (1 << 64) - 1
     x = (x + 0x9E3779B97F4A7C15) & M
