@@ -167,7 +167,24 @@ int ClientConn::shm_wait(uint64_t seq) {
         return st;
     }
     auto t0 = std::chrono::steady_clock::now();
+    auto finish = [&](int st) {  // fold this wait into the spin-budget EWMA
+        double us = std::chrono::duration<double, std::micro>(
+                        std::chrono::steady_clock::now() - t0)
+                        .count();
+        shm_ewma_us_ = 0.75 * shm_ewma_us_ + 0.25 * std::min(us, 5000.0);
+        return st;
+    };
+    // Adaptive spin: spin through ~2x the typical wait (usleep granularity
+    // is ~50 µs and would dominate the 20-300 µs fast path), then back off
+    // to sleeps so ms-scale waiters (64-client saturation) don't burn a
+    // core each. Clock checked every 256 pauses.
+    // ms-scale typical waits: spinning the first 400 µs is pure waste —
+    // sleep almost immediately (the 50 µs usleep granularity is noise
+    // against the wait itself).
+    const double spin_deadline_us =
+        shm_ewma_us_ > 800.0 ? 60.0 : std::min(400.0, std::max(30.0, shm_ewma_us_ * 2.0));
     int spins = 0;
+    bool spinning = true;
     for (;;) {
         uint32_t len = 0;
         uint64_t skip = 0;
@@ -180,7 +197,7 @@ int ClientConn::shm_wait(uint64_t seq) {
             shmring::RespRec r{};
             memcpy(&r, rec, std::min(sizeof(r), size_t(len)));
             shm_.resp->consume(skip);
-            if (r.h.seq == seq) return r.status;
+            if (r.h.seq == seq) return finish(r.status);
             // Out-of-order response: stash for a later wait (bounded — a
             // runaway map means tickets are being dropped by the caller).
             if (shm_results_.size() < 4096) shm_results_[r.h.seq] = r.status;
@@ -189,15 +206,18 @@ int ClientConn::shm_wait(uint64_t seq) {
                 shm_async_err_ = r.status;
             continue;
         }
-        // Spin through the whole expected wait (a batched op completes in
-        // 100-300 µs; usleep granularity is ~50 µs and would dominate), then
-        // back off for ms-scale waits; give up after the socket path's 60 s.
-        if (++spins < 50000) {
+        if (spinning) {
 #if defined(__x86_64__)
             __builtin_ia32_pause();
 #endif
+            if ((++spins & 255) == 0) {
+                auto us = std::chrono::duration<double, std::micro>(
+                              std::chrono::steady_clock::now() - t0)
+                              .count();
+                if (us > spin_deadline_us) spinning = false;
+            }
         } else {
-            usleep(100);
+            usleep(50);
             if (std::chrono::steady_clock::now() - t0 > std::chrono::seconds(60)) {
                 ERROR("shm ring response timeout (seq %llu)",
                       static_cast<unsigned long long>(seq));

@@ -137,6 +137,12 @@ class ClientConn {
     shmring::Segment shm_;
     bool shm_active_ = false;
     uint64_t shm_seq_ = 0;
+    // EWMA of recent shm response waits (µs): sizes the spin budget so a
+    // latency-bound client spins through its ~20-300 µs waits while a
+    // saturation client stuck behind ms-scale batches backs off to sleeps
+    // instead of burning a core (64 concurrent clients x busy-spin was the
+    // config-5 concern).
+    double shm_ewma_us_ = 50.0;
     uint64_t shm_unacked_ = 0;  // async ring writes since last ring sync
     int shm_async_err_ = 0;     // first error for an unawaited seq (io_mu_)
     std::unordered_map<uint64_t, int> shm_results_;  // ticketed responses (io_mu_)
