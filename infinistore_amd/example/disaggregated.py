@@ -14,12 +14,13 @@ Runs on CPU (TCP fabric path) or GPU (local IPC + HIP gather path):
     python -m infinistore_amd.example.disaggregated [port]
 """
 
+import random
 import sys
 import uuid
 
 import torch
 
-from infinistore_amd.kv_connector import PagedKVConnector, token_page_hashes
+from infinistore_amd.vllm_adapter import InfiniStoreKVAdapter
 
 PAGE_TOKENS = 16
 
@@ -97,60 +98,68 @@ class TinyLlama(torch.nn.Module):
         return blk["wo"](o.transpose(0, 1).reshape(q.shape[0], -1))
 
 
-def prefill_worker(model, tokens, conn: PagedKVConnector, device):
-    """Run the prompt, page the KV, stream it into the store layer by
-    layer (uploads overlap later layers' compute via async writes)."""
+def prefill_worker(model, tokens, adapter: InfiniStoreKVAdapter, device):
+    """Run the prompt, place the paged KV into a vLLM-style block-major
+    physical cache with an arbitrary block table, and stream it into the
+    store through the adapter's per-layer save hook (uploads overlap later
+    layers' compute via async writes)."""
     n_pages = len(tokens) // PAGE_TOKENS
-    page_keys = token_page_hashes(tokens, PAGE_TOKENS, conn.model_tag)
     t = torch.tensor(tokens, device=device)
     with torch.no_grad():
         logits, kvs = model.forward_collect(t)
-    elems = model.kv_elems_per_page()
-    for li, kv in enumerate(kvs):
-        # [2, seq, n_kv, hd] -> page-major [n_pages, 2, T, n_kv, hd]
-        paged = kv[:, : n_pages * PAGE_TOKENS].unflatten(
-            1, (n_pages, PAGE_TOKENS)).transpose(0, 1).contiguous()
-        offsets = [p * elems for p in range(n_pages)]
-        conn.save_layer(li, paged.view(-1), page_keys[:n_pages], offsets, elems)
-    conn.flush()
-    return page_keys
-
-
-def decode_worker(model, tokens, next_token, conn: PagedKVConnector, device):
-    """Reconstruct the KV cache from the store and decode one token.
-    Returns (logits, n_cached_pages)."""
-    page_keys = token_page_hashes(tokens, PAGE_TOKENS, conn.model_tag)
-    hits = conn.cached_pages(page_keys)
-    n_pages = len(tokens) // PAGE_TOKENS
-    assert hits >= n_pages, f"prefix lookup found {hits}/{n_pages} pages"
-    elems = model.kv_elems_per_page()
     dt = next(model.parameters()).dtype
+    # Physical cache: more blocks than pages, pages scattered through it by
+    # a shuffled block table (exactly what a paged-attention allocator does).
+    n_blocks = n_pages + 3
+    block_table = list(range(n_blocks))
+    random.Random(11).shuffle(block_table)
+    block_table = block_table[:n_pages]
+    for li, kv in enumerate(kvs):
+        cache = torch.zeros(n_blocks, 2, PAGE_TOKENS, model.n_kv, model.hd,
+                            device=device, dtype=dt)
+        paged = kv[:, : n_pages * PAGE_TOKENS].unflatten(
+            1, (n_pages, PAGE_TOKENS)).transpose(0, 1)  # [p, 2, T, kv, hd]
+        for p in range(n_pages):
+            cache[block_table[p]] = paged[p]
+        adapter.save_kv_layer(li, cache, tokens, block_table)
+    adapter.wait_for_save()
+
+
+def decode_worker(model, tokens, next_token, adapter: InfiniStoreKVAdapter,
+                  device):
+    """Reconstruct the KV cache from the store through the adapter's load
+    hooks — into this worker's OWN physical cache with its OWN (different)
+    block table — and decode one token. Returns (logits, matched_tokens)."""
+    matched = adapter.get_num_new_matched_tokens(tokens)
+    n_pages = len(tokens) // PAGE_TOKENS
+    assert matched >= n_pages * PAGE_TOKENS, (
+        f"prefix lookup matched {matched} tokens, need {n_pages * PAGE_TOKENS}")
+    dt = next(model.parameters()).dtype
+    # Decode engine's physical placement differs from prefill's on purpose.
+    n_blocks = n_pages + 5
+    block_table = list(range(n_blocks))
+    random.Random(23).shuffle(block_table)
+    block_table = block_table[:n_pages]
+    caches = [torch.zeros(n_blocks, 2, PAGE_TOKENS, model.n_kv, model.hd,
+                          device=device, dtype=dt)
+              for _ in range(model.n_layers)]
+    got = adapter.start_load_kv(caches, tokens, block_table)
+    assert got == n_pages, f"loading {got}/{n_pages} pages"
     cap = len(tokens) + 8
-    kv_cache = [torch.zeros(2, cap, model.n_kv, model.hd, device=device, dtype=dt)
+    kv_cache = [torch.zeros(2, cap, model.n_kv, model.hd, device=device,
+                            dtype=dt)
                 for _ in range(model.n_layers)]
-    stagings = [torch.zeros(n_pages * elems, device=device, dtype=dt)
-                for _ in range(model.n_layers)]
-    offsets = [p * elems for p in range(n_pages)]
-    # Prefetch every layer's pages up front (ticketed async reads on the
-    # local path — all gathers are in flight while we unpack layer by
-    # layer), falling back to blocking loads on the fabric path.
-    tickets = [conn.load_layer_async(li, stagings[li], page_keys[:n_pages],
-                                     offsets, elems)
-               for li in range(model.n_layers)]
     for li in range(model.n_layers):
-        if tickets[li] is not None:
-            assert conn.wait_load(tickets[li]), f"layer {li}: async load failed"
-        else:
-            ok = conn.load_layer(li, stagings[li], page_keys[:n_pages], offsets,
-                                 elems)
-            assert ok, f"layer {li}: cached pages missing"
-        paged = stagings[li].view(n_pages, 2, PAGE_TOKENS, model.n_kv, model.hd)
-        kv_cache[li][:, : n_pages * PAGE_TOKENS] = (
-            paged.transpose(0, 1).reshape(2, -1, model.n_kv, model.hd))
+        assert adapter.wait_for_layer_load(li), f"layer {li}: load failed"
+        # Unpack physical blocks -> linear [2, seq] layout for this toy model
+        # (a real paged-attention kernel would read the blocks directly).
+        for p in range(n_pages):
+            page = caches[li][block_table[p]]  # [2, T, kv, hd]
+            kv_cache[li][:, p * PAGE_TOKENS:(p + 1) * PAGE_TOKENS] = page
     with torch.no_grad():
         logits = model.forward_one(
             torch.tensor(next_token, device=device), len(tokens), kv_cache)
-    return logits, hits
+    return logits, matched
 
 
 def main(port=22345, device=None, seed=7, quant=None):
@@ -165,14 +174,16 @@ def main(port=22345, device=None, seed=7, quant=None):
     tag = f"demo-{uuid.uuid4().hex[:8]}"
 
     local = device.startswith("cuda")
-    pre = PagedKVConnector("127.0.0.1", port, tag, model.n_layers, local=local,
-                           quant=quant)
+    elems = model.kv_elems_per_page()
+    pre = InfiniStoreKVAdapter("127.0.0.1", port, tag, model.n_layers,
+                               PAGE_TOKENS, elems, local=local, quant=quant)
     try:
         prefill_worker(model, tokens, pre, device)
     finally:
         pre.close()
 
-    dec = PagedKVConnector("127.0.0.1", port, tag, model.n_layers, local=local)
+    dec = InfiniStoreKVAdapter("127.0.0.1", port, tag, model.n_layers,
+                               PAGE_TOKENS, elems, local=local)
     try:
         logits, hits = decode_worker(model, tokens, next_token, dec, device)
     finally:
@@ -188,7 +199,8 @@ def main(port=22345, device=None, seed=7, quant=None):
     assert torch.allclose(logits.float(), ref_logits.float(), atol=atol), (
         (logits - ref_logits).abs().max().item())
     print(f"disaggregated decode ok ({'fp8' if quant else 'plain'} pages): "
-          f"{hits} cached pages reused, logits match monolithic forward "
+          f"{hits} cached tokens reused via the vLLM-shaped adapter, logits "
+          f"match monolithic forward "
           f"(max diff {(logits - ref_logits).abs().max().item():.2e})")
 
 
