@@ -1065,8 +1065,32 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     size_t page = static_cast<size_t>(msg.block_size);
 
     auto tr0 = std::chrono::steady_clock::now();
-    // Group blocks by owning shard (keys may live on different GPUs); fp8-
-    // compressed entries go into separate dequantizing jobs.
+    // Group blocks by the shard whose GPU EXECUTES the copy. For same-GPU
+    // hits that is trivially the owner. For cross-shard hits the choice is
+    // deliberate: default OWNER-side (push) — the owner's
+    // kernel reads its local HBM and writes the reader's memory over xGMI.
+    // xGMI writes are posted (fire-and-forget per link), while remote reads
+    // are request/response round trips; and since a multi-shard read groups
+    // per owner, N owners push concurrently over N distinct xGMI links into
+    // the reader (each GPU has 7 links x ~153 GB/s — per-link bound, so the
+    // aggregate scales with the number of pushing GPUs, up to ~1 TB/s into
+    // one reader). Reader-side execution (pull) would serialize the same
+    // traffic through one GPU's remote-read path and its own CUs.
+    // IFS_CROSS_COPY=reader flips to pull for topologies where the owner
+    // GPUs are compute-saturated; it requires a shard on the reader's GPU.
+    // fp8-compressed entries go into separate dequantizing jobs.
+    static const bool cross_pull = [] {
+        const char* v = getenv("IFS_CROSS_COPY");
+        return v && std::string(v) == "reader";
+    }();
+    Shard* reader_shard = nullptr;
+    if (cross_pull) {
+        Shard* s = shard_for_device(msg.device);
+        if (s->device() == msg.device) reader_shard = s;
+    }
+    auto exec_shard = [&](Shard* owner) {
+        return (reader_shard && owner->device() != msg.device) ? reader_shard : owner;
+    };
     std::map<Shard*, Shard::CopyJob> jobs;
     std::map<Shard*, Shard::CopyJob> qjobs;
     auto held = std::make_shared<std::vector<Ref<BlockEntry>>>();
@@ -1098,14 +1122,14 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
                 e->last_access.store(read_tick, std::memory_order_relaxed);
                 if (e->fp8) {
                     if (e->size * 2 != page) return reply_local(c, ctx, INVALID_REQ);
-                    auto& qj = qjobs[e->shard];
+                    auto& qj = qjobs[exec_shard(e->shard)];
                     qj.bytes_per_block = page;
                     qj.xform = Shard::CopyJob::Xform::kDequantFp8Bf16;
                     qj.scales_in.push_back(e->scale);
                     qj.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
                     qj.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.second));
                 } else {
-                    auto& job = jobs[e->shard];
+                    auto& job = jobs[exec_shard(e->shard)];
                     job.bytes_per_block = page;
                     job.src.push_back(reinterpret_cast<uint64_t>(e->ptr));
                     job.dst.push_back(reinterpret_cast<uint64_t>(client_ptr + b.second));

@@ -336,6 +336,47 @@ def test_cross_gpu_write(gpu_server):
         rconn.close()
 
 
+def test_cross_gpu_read_pull_mode(gpu_server):
+    """IFS_CROSS_COPY=reader: the copy kernel runs on the READER's shard,
+    remote-reading the owner's HBM over xGMI (docs/design.md 'Cross-shard
+    copy side'). Own server subprocess because the mode is read once at
+    server start. Skipped on 1-GPU boxes."""
+    if torch.cuda.device_count() < 2:
+        pytest.skip("needs >= 2 GPUs")
+    port = free_port()
+    env = dict(os.environ, IFS_CROSS_COPY="reader")
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "infinistore_amd.server",
+         "--service-port", str(port), "--manage-port", str(free_port()),
+         "--prealloc-size", "2", "--minimal-allocate-size", "64",
+         "--no-manage"],
+        cwd=REPO, env=env,
+    )
+    try:
+        assert _wait_port(port), "pull-mode server did not come up"
+        wconn = local_conn(port)
+        rconn = local_conn(port)
+        try:
+            n = 256 * 1024
+            page = 64 * 1024
+            src = torch.randn(n, device="cuda:0")  # owner shard: GPU 0
+            dst = torch.zeros(n, device="cuda:1")  # reader: GPU 1's shard pulls
+            pre = uuid.uuid4().hex
+            keys = [f"pull-{pre}-{i}" for i in range(n // page)]
+            offs = [i * page for i in range(n // page)]
+            wconn.local_gpu_write_cache(src, list(zip(keys, offs)), page)
+            wconn.sync()
+            rconn.read_cache(dst, list(zip(keys, offs)), page)
+            rconn.sync()
+            assert torch.equal(src.cpu(), dst.cpu())
+        finally:
+            wconn.close()
+            rconn.close()
+    finally:
+        proc.terminate()
+        proc.wait(timeout=20)
+
+
 def test_benchmark_harness_gpu(gpu_server):
     """Reference-style harness end-to-end on the GPU local path."""
     r = subprocess.run(
