@@ -279,7 +279,7 @@ def main():
     # verification.
     o1 = offsets_np[:1]
     for b in range(n_salt):
-        wk1 = [f"warmbuf-{run_id}-{b}"]
+        wk1 = [f"warmbuf-{run_id}-r{rank}-{b}"]
         for c in range(n_conns):
             if use_local_path:
                 wconns[c].write_pages(srcs[b], wk1, o1, elems_per_block,

@@ -71,21 +71,31 @@ def make_conn(args, local):
     return conn
 
 
-def run_once(args, conn, local, device):
-    block_bytes = args.block_size << 10
+def make_buffers(args, conn, local, device):
+    """One src/dst pair per worker, reused across iterations (fresh tensors
+    every iteration would re-export IPC handles and churn the allocator —
+    at 64 clients that dominates the wall)."""
     total_bytes = args.size << 20
-    n_blocks = total_bytes // block_bytes
     dt = torch.bfloat16 if args.quant else torch.float32
     es = 2 if args.quant else 4
     elems = total_bytes // es
-    page_elems = block_bytes // es
-
     src = torch.rand(elems, dtype=dt, device=device)
     dst_dev = f"cuda:{args.dst_gpu}" if local else device
     dst = torch.zeros(elems, dtype=dt, device=dst_dev)
     if not local:
         conn.register_mr(src)
         conn.register_mr(dst)
+    return src, dst
+
+
+def run_once(args, conn, local, device, bufs=None):
+    block_bytes = args.block_size << 10
+    total_bytes = args.size << 20
+    n_blocks = total_bytes // block_bytes
+    es = 2 if args.quant else 4
+    page_elems = block_bytes // es
+
+    src, dst = bufs if bufs is not None else make_buffers(args, conn, local, device)
 
     run = uuid.uuid4().hex
     keys = [f"{run}-{i}" for i in range(n_blocks)]
@@ -171,13 +181,14 @@ def _worker(args_dict, q, barrier, wid):
     conn = None
     try:
         conn = make_conn(ns, local)
-        run_once(ns, conn, local, device)  # warm: IPC opens, allocator, caches
-        barrier.wait(timeout=300)  # start all clients together (steady state)
+        bufs = make_buffers(ns, conn, local, device)
+        run_once(ns, conn, local, device, bufs)  # warm: IPC opens, allocator
+        barrier.wait(timeout=600)  # start all clients together (steady state)
         t0 = time.perf_counter()
         w = r = 0.0
         iters = max(1, ns.iteration)
         for _ in range(iters):
-            w, r = run_once(ns, conn, local, device)
+            w, r = run_once(ns, conn, local, device, bufs)
         wall = time.perf_counter() - t0
         q.put((w, r, wall, iters))
     except Exception as e:  # report instead of leaving the parent hanging
@@ -235,8 +246,13 @@ def main():
         apply_shape(args)
     if args.spawn_server and args.clients > 1:
         # Keys live until each iteration's trailing delete; size the pool
-        # for all clients' live iterations plus slack.
+        # for all clients' live iterations plus slack, capped so the pool +
+        # the clients' own src/dst tensors (2x size each) still fit in HBM.
         need = args.clients * args.size * 2 * 1.3 / 1024 + 1
+        if torch.cuda.is_available() and args.local_gpu:
+            total_gb = torch.cuda.get_device_properties(0).total_memory / (1 << 30)
+            client_gb = args.clients * args.size * 2 / 1024
+            need = min(need, max(4, total_gb * 0.92 - client_gb))
         args.prealloc_size = max(args.prealloc_size, int(need))
     server_proc = _spawn_server(args) if args.spawn_server else None
     local = args.local_gpu and torch.cuda.is_available()
