@@ -120,6 +120,18 @@ void* MemoryPool::allocate(size_t size) {
     if (size == 0 || size > size_) return nullptr;
     size_t nb = (size + block_size_ - 1) / block_size_;
     size_t start;
+    if (nb == 1) {
+        // O(1) churn path: pop cached free indices, skipping stale hints.
+        while (!free_stack_.empty()) {
+            uint32_t idx = free_stack_.back();
+            free_stack_.pop_back();
+            if (!(bits_[idx / 64] & (1ull << (idx % 64)))) {
+                mark(idx, 1, true);
+                used_blocks_ += 1;
+                return static_cast<uint8_t*>(base_) + size_t(idx) * block_size_;
+            }
+        }
+    }
     if (!find_run(nb, &start)) return nullptr;
     mark(start, nb, true);
     used_blocks_ += nb;
@@ -130,9 +142,16 @@ void* MemoryPool::allocate_contiguous(size_t size, size_t n) {
     if (size == 0 || n == 0) return nullptr;
     size_t nb = (size + block_size_ - 1) / block_size_;
     size_t total = nb * n;
-    if (total > n_blocks_) return nullptr;
+    if (total > n_blocks_ || total > n_blocks_ - used_blocks_) return nullptr;
+    if (total > 1 && !contig_ok_) return nullptr;  // fragmented: don't rescan
     size_t start;
-    if (!find_run(total, &start)) return nullptr;
+    if (!find_run(total, &start)) {
+        if (total > 1) {
+            contig_ok_ = false;  // re-armed after enough frees (deallocate)
+            frees_since_contig_fail_ = 0;
+        }
+        return nullptr;
+    }
     mark(start, total, true);
     used_blocks_ += total;
     return static_cast<uint8_t*>(base_) + start * block_size_;
@@ -187,6 +206,10 @@ bool MemoryPool::deallocate(void* ptr, size_t size) {
     }
     mark(start, nb, false);
     used_blocks_ -= nb;
+    if (nb == 1 && free_stack_.size() < n_blocks_) {
+        free_stack_.push_back(static_cast<uint32_t>(start));
+    }
+    if (!contig_ok_ && ++frees_since_contig_fail_ >= 4096) contig_ok_ = true;
     return true;
 }
 

@@ -97,3 +97,35 @@ def test_many_allocs_stress():
     for ptr, size in live.values():
         assert p.deallocate(ptr, size)
     assert p.used_blocks() == 0
+
+
+def test_single_block_churn_fast():
+    """Steady-state KV churn: free/alloc single blocks at high rate on a
+    fragmented pool. The O(1) free-list path must keep this cheap — the
+    round-1 first-fit degraded to O(pool) per request under 64-client
+    churn (25 ms/request measured)."""
+    import time
+
+    p = n._TestPool(64 * MB, 4 * KB)  # 16384 blocks
+    # Fragment: fill completely, then free every other block.
+    ptrs = [p.allocate(4 * KB) for _ in range(16384)]
+    assert all(x != 0 for x in ptrs)
+    for i in range(0, 16384, 2):
+        assert p.deallocate(ptrs[i], 4 * KB)
+    # Churn: 100k single-block alloc+free on the fragmented pool.
+    t0 = time.time()
+    live = []
+    for i in range(100_000):
+        a = p.allocate(4 * KB)
+        assert a != 0
+        live.append(a)
+        if len(live) > 64:
+            assert p.deallocate(live.pop(0), 4 * KB)
+    for a in live:
+        assert p.deallocate(a, 4 * KB)
+    took = time.time() - t0
+    assert took < 5.0, f"churn took {took:.1f}s - free-list path not engaged?"
+    assert p.used_blocks() == 8192  # the odd-index half still allocated
+    for i in range(1, 16384, 2):
+        assert p.deallocate(ptrs[i], 4 * KB)
+    assert p.used_blocks() == 0
