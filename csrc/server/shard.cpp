@@ -186,11 +186,20 @@ size_t Shard::largest_free_run_bytes() {
 }
 
 Shard::Slot* Shard::acquire_slot(StreamCtx& sc) {
+    static const bool dbg = getenv("IFS_SERVER_DEBUG") != nullptr;
+    auto t0 = dbg ? std::chrono::steady_clock::now() : std::chrono::steady_clock::time_point{};
     std::unique_lock<std::mutex> lk(task_mu_);
     for (;;) {
         for (auto& sl : sc.slots) {
             if (!sl.busy) {
                 sl.busy = true;
+                if (dbg) {
+                    auto us = std::chrono::duration<double, std::micro>(
+                                  std::chrono::steady_clock::now() - t0)
+                                  .count();
+                    if (us > 1000)
+                        fprintf(stderr, "[sdbg] slot wait %.0fus (stream backlog)\n", us);
+                }
                 return &sl;
             }
         }

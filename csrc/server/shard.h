@@ -29,8 +29,12 @@ struct ShardOptions {
     size_t pool_bytes = 0;    // initial arena size
     size_t block_granule = 64 << 10;  // bitmap granule (minimal_allocate_size)
     int n_streams = 4;
-    int slots_per_stream = 16;
-    size_t max_descs_per_slot = 65536;  // 64K blocks -> 1 MiB of u64 ptrs/side
+    // Slots bound the number of in-flight copy jobs; a handler out of slots
+    // BLOCKS in acquire_slot, so the count must exceed the worst concurrent
+    // request fan-in (64 saturation clients x ~2 in flight), not match it.
+    // Bigger requests chunk across slots (submit_copy's chunk loop).
+    int slots_per_stream = 48;
+    size_t max_descs_per_slot = 16384;  // 16K blocks/chunk -> 128 KiB ptrs/side
     bool auto_extend = false;
     size_t extend_bytes = 10ull << 30;
 };
