@@ -951,15 +951,43 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     c->ref();
     static const bool sdbg = getenv("IFS_SERVER_DEBUG") != nullptr;
     auto t_start = std::chrono::steady_clock::now();
-    // won[i] is written in phase D after the kernel is submitted. On the
-    // socket path the commit lambda is posted to the owner loop — the same
-    // thread running this handler — so it cannot run before phase D ends.
-    // On the shm path the commit runs on a completion thread; commit_mu
-    // (held here across submit + phase D) provides that ordering instead.
+    // won[i] is written in phase D after the kernel is submitted. The commit
+    // must see phase D's result, so it runs on whichever side finishes LAST:
+    // both the copy-done callback and the end of phase D bump `arrivals`,
+    // and the second arriver performs the commit + reply. Round 1 used a
+    // mutex held across submit + phase D instead, which made completion
+    // threads BLOCK on handlers that were themselves waiting for copy
+    // slots — the only slot-freers waiting on slot-waiters (measured 94 ms
+    // slot waits at 64 saturation clients).
     auto won = std::make_shared<std::vector<uint8_t>>(n_fresh, 0);
-    auto commit_mu = std::make_shared<std::mutex>();
-    job.done = [this, c, entries, won, sync_resp, ctx, commit_mu, scales_out,
-                t_start](bool ok) {
+    auto copy_ok = std::make_shared<std::atomic<bool>>(false);
+    auto arrivals = std::make_shared<std::atomic<int>>(0);
+    auto fin = [this, c, entries, won, sync_resp, ctx, scales_out, copy_ok] {
+        bool ok = copy_ok->load(std::memory_order_acquire);
+        if (ok) {
+            for (size_t i = 0; i < entries->size(); i++)
+                if ((*won)[i]) {
+                    if (scales_out) (*entries)[i]->scale = (*scales_out)[i];
+                    (*entries)[i]->committed = true;
+                }
+        } else {
+            // Copy failed: drop the keys. Always from the owner loop —
+            // erase_entries takes the kv lock exclusively, and compact()
+            // waits on completion-thread futures WHILE holding that
+            // lock, so taking it on a completion thread could deadlock.
+            c->ref();
+            c->owner->post([this, c, entries, won] {
+                std::vector<Ref<BlockEntry>> winners;
+                for (size_t i = 0; i < entries->size(); i++)
+                    if ((*won)[i]) winners.push_back((*entries)[i]);
+                erase_entries(winners);
+                c->unref();
+            });
+        }
+        if (sync_resp) reply_local(c, ctx, ok ? FINISH : INTERNAL_ERROR);
+        finish_task(c, /*on_owner=*/!ctx.shm);
+    };
+    job.done = [this, c, entries, fin, ctx, copy_ok, arrivals, t_start](bool ok) {
         if (sdbg && entries->size() > 64) {
             auto us = std::chrono::duration<double, std::micro>(
                           std::chrono::steady_clock::now() - t_start)
@@ -967,37 +995,16 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
             fprintf(stderr, "[sdbg] write n=%zu submit->complete=%.0fus\n", entries->size(),
                     us);
         }
-        auto fin = [this, c, entries, won, ok, sync_resp, ctx, scales_out] {
-            if (ok) {
-                for (size_t i = 0; i < entries->size(); i++)
-                    if ((*won)[i]) {
-                        if (scales_out) (*entries)[i]->scale = (*scales_out)[i];
-                        (*entries)[i]->committed = true;
-                    }
-            } else {
-                // Copy failed: drop the keys. Always from the owner loop —
-                // erase_entries takes the kv lock exclusively, and compact()
-                // waits on completion-thread futures WHILE holding that
-                // lock, so taking it on a completion thread could deadlock.
-                c->ref();
-                c->owner->post([this, c, entries, won] {
-                    std::vector<Ref<BlockEntry>> winners;
-                    for (size_t i = 0; i < entries->size(); i++)
-                        if ((*won)[i]) winners.push_back((*entries)[i]);
-                    erase_entries(winners);
-                    c->unref();
-                });
-            }
-            if (sync_resp) reply_local(c, ctx, ok ? FINISH : INTERNAL_ERROR);
-            finish_task(c, /*on_owner=*/!ctx.shm);
-        };
+        copy_ok->store(ok, std::memory_order_release);
         if (ctx.shm) {
-            std::lock_guard<std::mutex> lk(*commit_mu);  // wait out phase D
-            fin();
+            // Never block the completion thread: if phase D has not ended,
+            // its own arrival runs fin.
+            if (arrivals->fetch_add(1, std::memory_order_acq_rel) == 1) fin();
         } else {
-            c->owner->post([fin, commit_mu] {
-                std::lock_guard<std::mutex> lk(*commit_mu);
-                fin();
+            // Socket path: both arrivals happen on the owner loop thread
+            // (this post + the handler itself), preserving reply ordering.
+            c->owner->post([fin, arrivals] {
+                if (arrivals->fetch_add(1, std::memory_order_acq_rel) == 1) fin();
             });
         }
     };
@@ -1005,7 +1012,6 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     // overlaps the kernel. Sync-response write (flags&1): one round trip,
     // response sent on completion instead. Ring async writes are unacked.
     if (!sync_resp && !ctx.shm) send_status(c, TASK_ACCEPTED);
-    std::unique_lock<std::mutex> commit_lk(*commit_mu);
     bool submitted = shard->submit_copy(std::move(job));
 
     // Phase D — insert pass, overlapped with the in-flight kernel. Losers
@@ -1037,7 +1043,8 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
             }
         }
     }
-    commit_lk.unlock();  // phase D done: completion may commit
+    // Phase D done: second arriver commits (the copy may already be done).
+    if (submitted && arrivals->fetch_add(1, std::memory_order_acq_rel) == 1) fin();
     auto p3 = std::chrono::steady_clock::now();
     if (pdbg && n_fresh > 64) {
         auto us = [](auto a, auto b) {
