@@ -78,9 +78,12 @@ bool set_device(int dev);
 
 // --- batched block copy ----------------------------------------------------
 // One launch copies n_blocks of `bytes_per_block` from src_ptrs[i] to
-// dst_ptrs[i]. The pointer arrays must be in device-visible memory on the
-// launch device. Chooses the vectorized 16 B/lane kernel when every pointer
-// and the size are 16-byte aligned, else a byte-granular fallback.
+// dst_ptrs[i]. The pointer arrays must be device-VISIBLE: device memory or
+// pinned host (hipHostMalloc — ROCm unified addressing; each workgroup reads
+// its 16 B descriptor once, so keeping descriptors in pinned host memory
+// avoids any per-launch SDMA upload). Chooses the vectorized 16 B/lane
+// kernel when every pointer and the size are 16-byte aligned, else a
+// byte-granular fallback.
 bool launch_copy_blocks(int dev, Stream s, const uint64_t* dev_src_ptrs,
                         const uint64_t* dev_dst_ptrs, int n_blocks, size_t bytes_per_block,
                         bool aligned16);

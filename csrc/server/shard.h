@@ -122,12 +122,12 @@ class Shard {
 
    private:
     struct Slot {
-        uint64_t* h_src = nullptr;  // pinned staging
+        // Pinned host descriptor arrays, read by the kernels directly
+        // (device-visible via ROCm unified addressing) — no per-job SDMA
+        // upload.
+        uint64_t* h_src = nullptr;
         uint64_t* h_dst = nullptr;
-        uint64_t* d_src = nullptr;  // device descriptor buffers
-        uint64_t* d_dst = nullptr;
-        float* h_scale = nullptr;   // fp8 transform scale staging
-        float* d_scale = nullptr;
+        float* h_scale = nullptr;  // fp8 scales (kernel writes/reads in place)
         gpu::Event event = nullptr;
         bool busy = false;
     };
