@@ -11,11 +11,15 @@ namespace ifs {
 Shard::Shard(const ShardOptions& opt) : opt_(opt) {}
 
 Shard::~Shard() {
-    {
-        std::lock_guard<std::mutex> lk(task_mu_);
+    for (auto& sc : streams_) {
+        std::lock_guard<std::mutex> lk(sc.mu);
         stopping_ = true;
     }
-    task_cv_.notify_all();
+    stopping_ = true;  // also when there are no streams (CPU shard)
+    for (auto& sc : streams_) {
+        sc.task_cv.notify_all();
+        sc.slot_cv.notify_all();
+    }
     {
         std::lock_guard<std::mutex> lk(fabric_mu_);
     }
@@ -180,11 +184,14 @@ size_t Shard::largest_free_run_bytes() {
 Shard::Slot* Shard::acquire_slot(StreamCtx& sc) {
     static const bool dbg = getenv("IFS_SERVER_DEBUG") != nullptr;
     auto t0 = dbg ? std::chrono::steady_clock::now() : std::chrono::steady_clock::time_point{};
-    std::unique_lock<std::mutex> lk(task_mu_);
+    std::unique_lock<std::mutex> lk(sc.mu);
     for (;;) {
-        for (auto& sl : sc.slots) {
+        int n = static_cast<int>(sc.slots.size());
+        for (int i = 0; i < n; i++) {
+            Slot& sl = sc.slots[(sc.next_slot + i) % n];
             if (!sl.busy) {
                 sl.busy = true;
+                sc.next_slot = (sc.next_slot + i + 1) % n;
                 if (dbg) {
                     auto us = std::chrono::duration<double, std::micro>(
                                   std::chrono::steady_clock::now() - t0)
@@ -195,7 +202,7 @@ Shard::Slot* Shard::acquire_slot(StreamCtx& sc) {
                 return &sl;
             }
         }
-        slot_cv_.wait(lk);
+        sc.slot_cv.wait(lk);
         if (stopping_) return nullptr;
     }
 }
@@ -264,9 +271,30 @@ bool Shard::submit_copy(CopyJob&& job) {
     // HIP stream enqueues are thread-safe; each job's ops precede its event
     // in stream order regardless of interleaving with other submitters, so
     // no submit-wide lock is needed (a lock here convoyed all clients behind
-    // slot waits).
-    StreamCtx& sc =
-        streams_[next_stream_.fetch_add(1) % static_cast<uint32_t>(streams_.size())];
+    // slot waits). Stream choice: round-robin, but skip ahead to a stream
+    // with a free slot when the nominal one is full (a job's chunks must
+    // stay on ONE stream — completion order per stream is what fires the
+    // done callback after the last chunk).
+    uint32_t rr = next_stream_.fetch_add(1);
+    uint32_t n_streams = static_cast<uint32_t>(streams_.size());
+    uint32_t pick = rr % n_streams;
+    for (uint32_t i = 0; i < n_streams; i++) {
+        StreamCtx& cand = streams_[(rr + i) % n_streams];
+        bool has_free = false;
+        {
+            std::lock_guard<std::mutex> lk(cand.mu);
+            for (auto& sl : cand.slots)
+                if (!sl.busy) {
+                    has_free = true;
+                    break;
+                }
+        }
+        if (has_free) {
+            pick = (rr + i) % n_streams;
+            break;
+        }
+    }
+    StreamCtx& sc = streams_[pick];
 
     // Small aligned batches: descriptors ride in the kernel arguments.
     if (aligned && n <= 16 && job.xform == CopyJob::Xform::kCopy) {
@@ -278,11 +306,11 @@ bool Shard::submit_copy(CopyJob&& job) {
                                                  job.bytes_per_block) &&
                   gpu::event_record(slot->event, sc.stream);
         {
-            std::lock_guard<std::mutex> lk(task_mu_);
+            std::lock_guard<std::mutex> lk(sc.mu);
             if (!ok) slot->busy = false;
             sc.pending.push_back({ok ? slot : nullptr, std::move(job.done), nullptr, 0, 0});
         }
-        task_cv_.notify_all();
+        sc.task_cv.notify_one();
         return true;
     }
 
@@ -328,21 +356,22 @@ bool Shard::submit_copy(CopyJob&& job) {
             // chunks were already queued) and stop submitting.
             ERROR("shard %d submit failed: %s", opt_.device, gpu::last_error());
             {
-                std::lock_guard<std::mutex> lk(task_mu_);
+                std::lock_guard<std::mutex> lk(sc.mu);
                 slot->busy = false;
                 sc.pending.push_back({nullptr, std::move(job.done), nullptr, 0, 0});
             }
-            task_cv_.notify_all();
+            sc.slot_cv.notify_one();
+            sc.task_cv.notify_one();
             return true;
         }
         {
-            std::lock_guard<std::mutex> lk(task_mu_);
+            std::lock_guard<std::mutex> lk(sc.mu);
             bool q = job.xform == CopyJob::Xform::kQuantBf16Fp8;
             sc.pending.push_back({slot,
                                   last ? std::move(job.done) : std::function<void(bool)>(),
                                   q ? job.scales_out : nullptr, job.scales_off + off, take});
         }
-        task_cv_.notify_all();
+        sc.task_cv.notify_one();
         off += take;
     }
     return true;
@@ -358,7 +387,7 @@ void Shard::completion_loop(size_t stream_idx) {
         PendingTask t{};
         bool have = false;
         {
-            std::unique_lock<std::mutex> lk(task_mu_);
+            std::unique_lock<std::mutex> lk(sc.mu);
             while (!have) {
                 if (!sc.pending.empty()) {
                     PendingTask& front = sc.pending.front();
@@ -368,16 +397,16 @@ void Shard::completion_loop(size_t stream_idx) {
                         have = true;
                         break;
                     }
-                    // Event pending but not complete yet: spin politely.
+                    // Event pending but not complete yet: wait off-lock
+                    // (event_sync blocks in the runtime, not on sc.mu, so
+                    // submitters keep flowing).
                     lk.unlock();
-#if defined(__x86_64__)
-                    for (int i = 0; i < 64; i++) __builtin_ia32_pause();
-#endif
+                    gpu::event_sync(front.slot->event);
                     lk.lock();
                     continue;
                 }
                 if (stopping_) return;
-                task_cv_.wait(lk, [&] { return stopping_ || !sc.pending.empty(); });
+                sc.task_cv.wait(lk, [&] { return stopping_ || !sc.pending.empty(); });
                 if (stopping_ && sc.pending.empty()) return;
             }
         }
@@ -390,10 +419,10 @@ void Shard::completion_loop(size_t stream_idx) {
                 memcpy(t.scales_out->data() + t.scales_off, t.slot->h_scale,
                        t.scales_n * sizeof(float));
             }
-            std::lock_guard<std::mutex> lk(task_mu_);
+            std::lock_guard<std::mutex> lk(sc.mu);
             t.slot->busy = false;
         }
-        slot_cv_.notify_all();
+        sc.slot_cv.notify_one();
         if (t.done) t.done(ok);
     }
 }

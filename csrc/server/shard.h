@@ -140,10 +140,19 @@ class Shard {
         size_t scales_off = 0;
         size_t scales_n = 0;
     };
+    // Fully per-stream state: slots, pending FIFO, and their synchronization
+    // live with the stream, so submitters and the completion thread of one
+    // stream never contend with other streams'. (Round 1 used one shard-wide
+    // mutex + notify_all; at 64 saturation clients every completion woke ~60
+    // blocked handlers and the completion threads' event spin-poll hammered
+    // the same lock — task_mu_ was the shard's true bottleneck.)
     struct StreamCtx {
         gpu::Stream stream = nullptr;
+        std::mutex mu;
+        std::condition_variable slot_cv;  // a slot was freed
+        std::condition_variable task_cv;  // a task was enqueued
         std::vector<Slot> slots;
-        std::deque<PendingTask> pending;  // FIFO per stream (task_mu_)
+        std::deque<PendingTask> pending;  // FIFO per stream (mu)
         int next_slot = 0;
     };
 
@@ -155,12 +164,9 @@ class Shard {
     MM mm_;
     std::mutex alloc_mu_;
 
-    std::vector<StreamCtx> streams_;
+    std::deque<StreamCtx> streams_;  // deque: StreamCtx is not movable (mutex)
     std::atomic<uint32_t> next_stream_{0};
 
-    std::mutex task_mu_;
-    std::condition_variable task_cv_;
-    std::condition_variable slot_cv_;
     std::vector<std::thread> completion_threads_;  // one per stream
     std::atomic<bool> stopping_{false};
     bool inited_ = false;
