@@ -397,11 +397,23 @@ void Shard::completion_loop(size_t stream_idx) {
                         have = true;
                         break;
                     }
-                    // Event pending but not complete yet: wait off-lock
-                    // (event_sync blocks in the runtime, not on sc.mu, so
-                    // submitters keep flowing).
+                    // Event pending but not complete yet: spin-poll it
+                    // off-lock. hipEventSynchronize parks the thread in the
+                    // interrupt path (~100 µs wake latency per event — a
+                    // 48-deep slot drain became tens of ms and starved
+                    // submitters); event_query polling completes in ~2 µs.
+                    gpu::Event ev = front.slot->event;
                     lk.unlock();
-                    gpu::event_sync(front.slot->event);
+                    int spins = 0;
+                    while (!gpu::event_query(ev) && !stopping_) {
+                        if (++spins < 2000) {
+#if defined(__x86_64__)
+                            __builtin_ia32_pause();
+#endif
+                        } else {
+                            usleep(20);  // ms-scale kernel: stop burning a core
+                        }
+                    }
                     lk.lock();
                     continue;
                 }
