@@ -1013,6 +1013,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     // response sent on completion instead. Ring async writes are unacked.
     if (!sync_resp && !ctx.shm) send_status(c, TASK_ACCEPTED);
     bool submitted = shard->submit_copy(std::move(job));
+    auto p2b = std::chrono::steady_clock::now();
 
     // Phase D — insert pass, overlapped with the in-flight kernel. Losers
     // (a racing writer inserted the key first) keep their block alive until
@@ -1050,8 +1051,8 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         auto us = [](auto a, auto b) {
             return std::chrono::duration<double, std::micro>(b - a).count();
         };
-        fprintf(stderr, "[pdbg] dedup=%.0f alloc=%.0f build+insert=%.0f\n", us(p0, p1),
-                us(p1, p2), us(p2, p3));
+        fprintf(stderr, "[pdbg] dedup=%.0f alloc=%.0f submit=%.0f insert=%.0f\n", us(p0, p1),
+                us(p1, p2), us(p2, p2b), us(p2b, p3));
     }
     if (!submitted) {
         // The job never launched (e.g. a transform on a CPU shard): the done

@@ -383,59 +383,84 @@ void Shard::completion_loop(size_t stream_idx) {
     // done-callback that blocks (e.g. the shm write commit waiting for the
     // insert pass) stalls only its own stream's FIFO.
     auto& sc = streams_[stream_idx];
+    std::vector<PendingTask> batch;
     for (;;) {
-        PendingTask t{};
-        bool have = false;
+        batch.clear();
         {
             std::unique_lock<std::mutex> lk(sc.mu);
-            while (!have) {
-                if (!sc.pending.empty()) {
-                    PendingTask& front = sc.pending.front();
-                    if (!front.slot || gpu::event_query(front.slot->event)) {
-                        t = std::move(front);
-                        sc.pending.pop_front();
-                        have = true;
-                        break;
-                    }
-                    // Event pending but not complete yet: spin-poll it
-                    // off-lock. hipEventSynchronize parks the thread in the
-                    // interrupt path (~100 µs wake latency per event — a
-                    // 48-deep slot drain became tens of ms and starved
-                    // submitters); event_query polling completes in ~2 µs.
-                    gpu::Event ev = front.slot->event;
-                    lk.unlock();
-                    int spins = 0;
-                    while (!gpu::event_query(ev) && !stopping_) {
-                        if (++spins < 2000) {
-#if defined(__x86_64__)
-                            __builtin_ia32_pause();
-#endif
-                        } else {
-                            usleep(20);  // ms-scale kernel: stop burning a core
-                        }
-                    }
-                    lk.lock();
+            for (;;) {
+                if (stopping_ && sc.pending.empty()) return;
+                if (sc.pending.empty()) {
+                    sc.task_cv.wait(lk, [&] { return stopping_ || !sc.pending.empty(); });
                     continue;
                 }
-                if (stopping_) return;
-                sc.task_cv.wait(lk, [&] { return stopping_ || !sc.pending.empty(); });
-                if (stopping_ && sc.pending.empty()) return;
+                // Events on one stream complete in order, so if pending[k]'s
+                // event fired, 0..k all did: binary-search the completed
+                // prefix and take it in ONE pass. This keeps hipEventQuery
+                // traffic at O(log depth) per sweep instead of hammering the
+                // HIP runtime's locks from a per-task poll (which competed
+                // with the 64 handler threads' kernel launches).
+                size_t n = sc.pending.size();
+                size_t done_n = 0;
+                // Null-slot tasks (submit failures) are trivially complete.
+                while (done_n < n && !sc.pending[done_n].slot) done_n++;
+                if (done_n < n) {
+                    size_t lo = done_n, hi = n;  // invariant: [0,lo) complete
+                    while (lo < hi) {
+                        size_t mid = (lo + hi) / 2;
+                        if (!sc.pending[mid].slot ||
+                            gpu::event_query(sc.pending[mid].slot->event))
+                            lo = mid + 1;
+                        else
+                            hi = mid;
+                    }
+                    done_n = lo;
+                }
+                if (done_n > 0) {
+                    for (size_t i = 0; i < done_n; i++) {
+                        batch.push_back(std::move(sc.pending.front()));
+                        sc.pending.pop_front();
+                    }
+                    break;
+                }
+                // Nothing complete yet: poll the front event off-lock with
+                // a short backoff (hipEventSynchronize's interrupt-path
+                // wake costs ~100 µs per event and serialized slot drains).
+                gpu::Event ev = sc.pending.front().slot->event;
+                lk.unlock();
+                int spins = 0;
+                while (!gpu::event_query(ev) && !stopping_) {
+                    if (++spins < 200) {
+#if defined(__x86_64__)
+                        __builtin_ia32_pause();
+#endif
+                    } else {
+                        usleep(10);
+                    }
+                }
+                lk.lock();
             }
         }
-        bool ok = t.slot != nullptr;
-        if (t.slot) {
-            ok = gpu::event_query(t.slot->event) || gpu::event_sync(t.slot->event);
-            if (ok && t.scales_out && t.scales_n) {
-                // Must precede the slot release: the next submitter reuses
-                // the pinned h_scale buffer.
-                memcpy(t.scales_out->data() + t.scales_off, t.slot->h_scale,
-                       t.scales_n * sizeof(float));
-            }
+        // Copy out fp8 scales and release every slot in one lock hold, then
+        // run the done callbacks in order outside the lock.
+        {
             std::lock_guard<std::mutex> lk(sc.mu);
-            t.slot->busy = false;
+            for (auto& t : batch) {
+                if (t.slot) {
+                    if (t.scales_out && t.scales_n)
+                        memcpy(t.scales_out->data() + t.scales_off, t.slot->h_scale,
+                               t.scales_n * sizeof(float));
+                    t.slot->busy = false;
+                }
+            }
         }
-        sc.slot_cv.notify_one();
-        if (t.done) t.done(ok);
+        if (batch.size() > 1)
+            sc.slot_cv.notify_all();
+        else
+            sc.slot_cv.notify_one();
+        for (auto& t : batch) {
+            if (t.done) t.done(t.slot != nullptr);
+        }
     }
 }
 
