@@ -22,10 +22,6 @@ struct EntrySlab {
     std::vector<std::unique_ptr<char[]>> chunks;
     static constexpr size_t kPerChunk = 4096;
 
-    void* take(size_t n) {
-        std::lock_guard<std::mutex> lk(mu);
-        return take_locked(n);
-    }
     void* take_locked(size_t n) {
         if (free_list.empty()) {
             chunks.emplace_back(new char[n * kPerChunk]);
@@ -37,18 +33,69 @@ struct EntrySlab {
         free_list.pop_back();
         return p;
     }
-    // Pre-size the freelist for a burst (one lock for the whole batch).
-    void prepare(size_t n, size_t count) {
-        std::lock_guard<std::mutex> lk(mu);
-        while (free_list.size() < count) {
-            chunks.emplace_back(new char[n * kPerChunk]);
-            char* base = chunks.back().get();
-            for (size_t i = 0; i < kPerChunk; i++) free_list.push_back(base + i * n);
+
+    // Thread-local magazines: entry alloc/free happens on every handler,
+    // completion and delete path — one global mutex per op (especially
+    // give(), which delete churn hit ~6M times per 64-client run) serialized
+    // the whole server. Each thread caches up to kMagMax entries and
+    // exchanges them with the global list kMagBatch at a time.
+    static constexpr size_t kMagBatch = 256;
+    static constexpr size_t kMagMax = 512;
+    struct Magazine {
+        std::vector<void*> items;
+        ~Magazine();  // flush to the global list (pollers exit per conn)
+    };
+    static Magazine& mag() {
+        thread_local Magazine m;
+        return m;
+    }
+
+    void* take(size_t n) {
+        auto& m = mag();
+        if (m.items.empty()) {
+            std::lock_guard<std::mutex> lk(mu);
+            m.items.reserve(kMagMax);
+            for (size_t i = 0; i < kMagBatch; i++) m.items.push_back(take_locked(n));
+        }
+        void* p = m.items.back();
+        m.items.pop_back();
+        return p;
+    }
+    // Bulk take for a write batch: drain the magazine, top up from the
+    // global list under ONE lock.
+    void take_bulk(size_t n, size_t count, std::vector<void*>* out) {
+        auto& m = mag();
+        while (count && !m.items.empty()) {
+            out->push_back(m.items.back());
+            m.items.pop_back();
+            count--;
+        }
+        if (count) {
+            std::lock_guard<std::mutex> lk(mu);
+            for (size_t i = 0; i < count; i++) out->push_back(take_locked(n));
         }
     }
     void give(void* p) {
-        std::lock_guard<std::mutex> lk(mu);
-        free_list.push_back(p);
+        auto& m = mag();
+        m.items.push_back(p);
+        if (m.items.size() >= kMagMax) {
+            std::lock_guard<std::mutex> lk(mu);
+            for (size_t i = 0; i < kMagBatch; i++) {
+                free_list.push_back(m.items.back());
+                m.items.pop_back();
+            }
+        }
+    }
+    void give_bulk(std::vector<void*>& items, size_t from) {
+        auto& m = mag();
+        for (size_t i = from; i < items.size(); i++) m.items.push_back(items[i]);
+        if (m.items.size() >= kMagMax) {
+            std::lock_guard<std::mutex> lk(mu);
+            while (m.items.size() > kMagBatch) {
+                free_list.push_back(m.items.back());
+                m.items.pop_back();
+            }
+        }
     }
 };
 EntrySlab& entry_slab() {
@@ -56,6 +103,13 @@ EntrySlab& entry_slab() {
     // teardown after static destructors would have run.
     static EntrySlab* s = new EntrySlab();
     return *s;
+}
+
+EntrySlab::Magazine::~Magazine() {
+    if (items.empty()) return;
+    auto& s = entry_slab();  // leaked singleton: safe at thread exit
+    std::lock_guard<std::mutex> lk(s.mu);
+    for (void* p : items) s.free_list.push_back(p);
 }
 }  // namespace
 
@@ -69,15 +123,11 @@ struct SlabBatch {
     size_t next = 0;
     explicit SlabBatch(size_t count) {
         slots.reserve(count);
-        auto& slab = entry_slab();
-        std::lock_guard<std::mutex> lk(slab.mu);
-        for (size_t i = 0; i < count; i++) slots.push_back(slab.take_locked(sizeof(BlockEntry)));
+        entry_slab().take_bulk(sizeof(BlockEntry), count, &slots);
     }
     ~SlabBatch() {
         if (next >= slots.size()) return;
-        auto& slab = entry_slab();
-        std::lock_guard<std::mutex> lk(slab.mu);
-        for (size_t i = next; i < slots.size(); i++) slab.free_list.push_back(slots[i]);
+        entry_slab().give_bulk(slots, next);
     }
     BlockEntry* make() { return new (slots[next++]) BlockEntry(); }
 };
@@ -876,7 +926,11 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     std::array<std::vector<uint32_t>, kStripes> by_stripe;
     for (size_t i = 0; i < nb; i++)
         by_stripe[stripe_of(hashes[i])].push_back(static_cast<uint32_t>(i));
-    for (size_t si = 0; si < kStripes; si++) {
+    // Random start: concurrent requests sweeping stripes in the same
+    // ascending order convoy on every lock in turn.
+    size_t sweep0 = static_cast<size_t>(hashes[0] >> 32) % kStripes;
+    for (size_t sk = 0; sk < kStripes; sk++) {
+        size_t si = (sweep0 + sk) % kStripes;
         auto& list = by_stripe[si];
         if (list.empty()) continue;
         std::shared_lock<std::shared_mutex> lk(kv_[si].mu);
@@ -1029,7 +1083,9 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         std::array<std::vector<uint32_t>, kStripes> ins_by_stripe;  // pos in fresh
         for (uint32_t p = 0; p < n_fresh; p++)
             ins_by_stripe[stripe_of(hashes[fresh[p]])].push_back(p);
-        for (size_t si = 0; si < kStripes; si++) {
+        size_t ins0 = static_cast<size_t>(hashes[fresh[0]] >> 32) % kStripes;
+        for (size_t sk = 0; sk < kStripes; sk++) {
+            size_t si = (ins0 + sk) % kStripes;
             auto& list = ins_by_stripe[si];
             if (list.empty()) continue;
             std::lock_guard<std::shared_mutex> lk(kv_[si].mu);
