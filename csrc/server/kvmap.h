@@ -103,6 +103,31 @@ class KvMap {
         }
     }
 
+    // Erase but hand the value back: callers that hold the stripe lock move
+    // the refs out and release them AFTER unlocking — the entry destructor
+    // frees pool blocks (shard alloc lock) and slab memory, and doing ~160
+    // of those under an exclusive stripe lock blocked every reader/writer
+    // on the stripe for ms at a time (measured: 64-client churn spent
+    // 1.3-2.6 ms per write request waiting for stripe locks, 21 µs using
+    // them).
+    bool extract(std::string_view key, Ref<BlockEntry>* out) {
+        size_t mask = slots_.size() - 1;
+        uint64_t h = hash_key(key);
+        for (size_t i = h & mask;; i = (i + 1) & mask) {
+            Slot& s = slots_[i];
+            if (s.state == kEmpty) return false;
+            if (s.state == kFull && s.hash == h && key_equals(s, key)) {
+                *out = std::move(s.val);
+                s.val.~Ref<BlockEntry>();
+                s.state = kTomb;
+                size_--;
+                tombs_++;
+                arena_waste_ += s.key_len;
+                return true;
+            }
+        }
+    }
+
     template <typename Fn>  // fn(string_view key, Ref<BlockEntry>& val)
     void for_each(Fn&& fn) {
         for (auto& s : slots_) {

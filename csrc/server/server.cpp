@@ -352,14 +352,21 @@ bool Server::start() {
                 lk.unlock();
                 size_t swept = 0;
                 for (auto& st : kv_) {
-                    std::lock_guard<std::shared_mutex> ex(st.mu);
-                    std::vector<std::string> victims;
-                    st.map.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
-                        if (expired(val.get()) && val->ref_count() == 1)
-                            victims.emplace_back(key);
-                    });
-                    for (auto& k : victims) st.map.erase(k);
-                    swept += victims.size();
+                    std::vector<Ref<BlockEntry>> dead;  // dropped after unlock
+                    {
+                        std::lock_guard<std::shared_mutex> ex(st.mu);
+                        std::vector<std::string> victims;
+                        st.map.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
+                            if (expired(val.get()) && val->ref_count() == 1)
+                                victims.emplace_back(key);
+                        });
+                        dead.reserve(victims.size());
+                        for (auto& k : victims) {
+                            Ref<BlockEntry> ref;
+                            if (st.map.extract(k, &ref)) dead.emplace_back(std::move(ref));
+                        }
+                        swept += victims.size();
+                    }
                 }
                 if (swept) DEBUG("ttl sweep: %zu expired entries erased", swept);
                 lk.lock();
@@ -771,13 +778,20 @@ void Server::erase_entries(const std::vector<Ref<BlockEntry>>& entries) {
     // index by identity. O(map) scan, but keys are not kept around on the
     // hot path. One stripe at a time.
     for (auto& st : kv_) {
-        std::lock_guard<std::shared_mutex> lk(st.mu);
-        std::vector<std::string> victims;
-        st.map.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
-            for (auto& e : entries)
-                if (val.get() == e.get()) victims.emplace_back(key);
-        });
-        for (auto& k : victims) st.map.erase(k);
+        std::vector<Ref<BlockEntry>> dead;  // dropped after unlock
+        {
+            std::lock_guard<std::shared_mutex> lk(st.mu);
+            std::vector<std::string> victims;
+            st.map.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
+                for (auto& e : entries)
+                    if (val.get() == e.get()) victims.emplace_back(key);
+            });
+            dead.reserve(victims.size());
+            for (auto& k : victims) {
+                Ref<BlockEntry> ref;
+                if (st.map.extract(k, &ref)) dead.emplace_back(std::move(ref));
+            }
+        }
     }
 }
 
@@ -793,6 +807,7 @@ size_t Server::evict_lru(Shard* shard, size_t bytes) {
     size_t freed = 0;
     for (size_t visited = 0; freed < bytes && visited < kStripes; visited++) {
         auto& st = kv_[evict_stripe_rr_.fetch_add(1) % kStripes];
+        std::vector<Ref<BlockEntry>> dead;  // block frees run after unlock
         std::lock_guard<std::shared_mutex> lk(st.mu);
         size_t scanned = 0;
         const size_t scan_cap = st.map.capacity();  // one revolution max
@@ -822,10 +837,10 @@ size_t Server::evict_lru(Shard* shard, size_t bytes) {
                       [](const auto& a, const auto& b) { return a.first < b.first; });
             size_t take = std::max<size_t>(1, sample.size() / 2);
             for (size_t i = 0; i < take && freed < bytes; i++) {
-                Ref<BlockEntry>* v = st.map.find(sample[i].second);
-                if (!v) continue;
-                freed += (*v)->size;
-                st.map.erase(sample[i].second);
+                Ref<BlockEntry> ref;
+                if (!st.map.extract(sample[i].second, &ref)) continue;
+                freed += ref->size;
+                dead.emplace_back(std::move(ref));
                 n_evicted_.fetch_add(1);
             }
             sample.clear();
@@ -929,11 +944,16 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     // Random start: concurrent requests sweeping stripes in the same
     // ascending order convoy on every lock in turn.
     size_t sweep0 = static_cast<size_t>(hashes[0] >> 32) % kStripes;
+    double lw_us = 0, ul_us = 0;  // lock-wait vs under-lock (pdbg)
     for (size_t sk = 0; sk < kStripes; sk++) {
         size_t si = (sweep0 + sk) % kStripes;
         auto& list = by_stripe[si];
         if (list.empty()) continue;
+        auto tl0 = pdbg ? std::chrono::steady_clock::now()
+                        : std::chrono::steady_clock::time_point{};
         std::shared_lock<std::shared_mutex> lk(kv_[si].mu);
+        auto tl1 = pdbg ? std::chrono::steady_clock::now()
+                        : std::chrono::steady_clock::time_point{};
         auto& m = kv_[si].map;
         size_t ln = list.size();
         for (size_t i = 0; i < std::min(kPf, ln); i++) m.prefetch(hashes[list[i]]);
@@ -943,7 +963,14 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
             Ref<BlockEntry>* v = m.find_hashed(msg.blocks[gi].first, hashes[gi]);
             if (!v || expired(v->get())) fresh.push_back(gi);
         }
+        if (pdbg) {
+            auto tl2 = std::chrono::steady_clock::now();
+            lw_us += std::chrono::duration<double, std::micro>(tl1 - tl0).count();
+            ul_us += std::chrono::duration<double, std::micro>(tl2 - tl1).count();
+        }
     }
+    if (pdbg && nb > 64)
+        fprintf(stderr, "[pdbg2] dedup lockwait=%.0f underlock=%.0f\n", lw_us, ul_us);
     auto p1 = std::chrono::steady_clock::now();
 
     n_writes_.fetch_add(1);
@@ -1084,11 +1111,28 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         for (uint32_t p = 0; p < n_fresh; p++)
             ins_by_stripe[stripe_of(hashes[fresh[p]])].push_back(p);
         size_t ins0 = static_cast<size_t>(hashes[fresh[0]] >> 32) % kStripes;
+        double ilw_us = 0, iul_us = 0;
         for (size_t sk = 0; sk < kStripes; sk++) {
             size_t si = (ins0 + sk) % kStripes;
             auto& list = ins_by_stripe[si];
             if (list.empty()) continue;
+            auto tl0 = pdbg ? std::chrono::steady_clock::now()
+                            : std::chrono::steady_clock::time_point{};
             std::lock_guard<std::shared_mutex> lk(kv_[si].mu);
+            auto tl1 = pdbg ? std::chrono::steady_clock::now()
+                            : std::chrono::steady_clock::time_point{};
+            struct UlAcc {  // accumulate under-lock time at scope exit
+                bool on;
+                std::chrono::steady_clock::time_point t1;
+                double *lw, *ul;
+                std::chrono::steady_clock::time_point t0;
+                ~UlAcc() {
+                    if (!on) return;
+                    auto t2 = std::chrono::steady_clock::now();
+                    *lw += std::chrono::duration<double, std::micro>(t1 - t0).count();
+                    *ul += std::chrono::duration<double, std::micro>(t2 - t1).count();
+                }
+            } acc{pdbg, tl1, &ilw_us, &iul_us, tl0};
             auto& m = kv_[si].map;
             size_t ln = list.size();
             for (size_t i = 0; i < std::min(kPf, ln); i++) m.prefetch(hashes[fresh[list[i]]]);
@@ -1106,6 +1150,8 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
                 (*won)[p] = inserted ? 1 : 0;
             }
         }
+        if (pdbg && n_fresh > 64)
+            fprintf(stderr, "[pdbg2] insert lockwait=%.0f underlock=%.0f\n", ilw_us, iul_us);
     }
     // Phase D done: second arriver commits (the copy may already be done).
     if (submitted && arrivals->fetch_add(1, std::memory_order_acq_rel) == 1) fin();
@@ -1860,16 +1906,30 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& 
             by_stripe[stripe_of(hashes[i])].push_back(static_cast<uint32_t>(i));
         }
         constexpr size_t kPf = 16;
+        // Move the refs out under the lock, destroy them after unlocking:
+        // entry destruction frees pool blocks (shard alloc lock) + slab
+        // memory, which must never run under an exclusive stripe lock (it
+        // blocked every op on the stripe for ms during delete sweeps).
+        std::vector<Ref<BlockEntry>> dead;
+        dead.reserve(keys.size());
         for (size_t si = 0; si < kStripes; si++) {
             auto& list = by_stripe[si];
             if (list.empty()) continue;
-            std::lock_guard<std::shared_mutex> lk(kv_[si].mu);
-            auto& m = kv_[si].map;
-            for (size_t i = 0; i < std::min(kPf, list.size()); i++) m.prefetch(hashes[list[i]]);
-            for (size_t i = 0; i < list.size(); i++) {
-                if (i + kPf < list.size()) m.prefetch(hashes[list[i + kPf]]);
-                n += m.erase(keys[list[i]]) ? 1 : 0;
+            {
+                std::lock_guard<std::shared_mutex> lk(kv_[si].mu);
+                auto& m = kv_[si].map;
+                for (size_t i = 0; i < std::min(kPf, list.size()); i++)
+                    m.prefetch(hashes[list[i]]);
+                for (size_t i = 0; i < list.size(); i++) {
+                    if (i + kPf < list.size()) m.prefetch(hashes[list[i + kPf]]);
+                    Ref<BlockEntry> ref;
+                    if (m.extract(keys[list[i]], &ref)) {
+                        n++;
+                        dead.emplace_back(std::move(ref));
+                    }
+                }
             }
+            dead.clear();  // destructors run lock-free (per-stripe batch)
         }
     }
     reply_local(c, ctx, n);
