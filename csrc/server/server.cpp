@@ -338,6 +338,14 @@ bool Server::start() {
     for (auto& st : kv_) st.map.reserve(kv_initial / kStripes);
     running_.store(true);
     stop_requested_.store(false);
+    // Fast-op workers (ring write/read handlers): enough to keep every HIP
+    // stream fed under a 64-client burst without oversubscribing the box.
+    {
+        unsigned hw = std::thread::hardware_concurrency();
+        unsigned n = std::min(24u, std::max(6u, hw / 8));
+        for (unsigned i = 0; i < n; i++)
+            fast_workers_.emplace_back([this] { fast_worker_main(); });
+    }
     main_io_.start();
     for (auto& w : workers_) w->start();
     if (opt_.ttl_seconds > 0) {
@@ -396,6 +404,12 @@ void Server::stop() {
         if (w->thread.joinable()) w->thread.join();
     if (main_io_.thread.joinable()) main_io_.thread.join();
     workers_.clear();
+    // After the IO loops (and with them every ring poller) are joined no
+    // new fast work can arrive; drain the queue, then join the workers.
+    fast_cv_.notify_all();
+    for (auto& t : fast_workers_)
+        if (t.joinable()) t.join();
+    fast_workers_.clear();
     running_.store(false);
     // Drop all stored blocks.
     purge();
@@ -1427,6 +1441,38 @@ void Server::shm_teardown(Conn* c) {
     // may still write responses into the (now client-less) ring.
 }
 
+void Server::fast_worker_main() {
+    for (;;) {
+        FastWork w;
+        {
+            std::unique_lock<std::mutex> lk(fast_mu_);
+            fast_cv_.wait(lk, [this] { return stop_requested_.load() || !fast_q_.empty(); });
+            if (fast_q_.empty()) {
+                if (stop_requested_.load()) return;
+                continue;
+            }
+            w = std::move(fast_q_.front());
+            fast_q_.pop_front();
+        }
+        run_fast(w);
+    }
+}
+
+void Server::run_fast(FastWork& w) {
+    ReqCtx ctx{w.seq, true};
+    LocalView v;
+    if (!parse_packed_local(w.body.data(), w.body.size(), &v)) {
+        reply_local(w.c, ctx, INVALID_REQ);
+    } else if (w.op == OP_W_FAST) {
+        op_local_write(w.c, v, ctx);
+    } else {
+        op_local_read(w.c, v, ctx);
+    }
+    // Balance the poller's enqueue-time remain bump (+ answer a deferred
+    // sync if this was the last in-flight unit).
+    finish_task(w.c, /*on_owner=*/false);
+}
+
 void Server::shm_poll_main(ShmPeer* p) {
     Conn* c = p->c;
     auto last_work = std::chrono::steady_clock::now();
@@ -1476,18 +1522,38 @@ void Server::shm_poll_main(ShmPeer* p) {
             // The record stays in the ring while the handler runs (views
             // point into it); consume() below frees the space.
             switch (op) {
-                case OP_W_FAST: {
-                    LocalView v;
-                    if (!parse_packed_local(body, h.body_len, &v))
-                        reply_local(c, ctx, INVALID_REQ);
-                    else
-                        op_local_write(c, v, ctx);
-                    break;
-                }
+                case OP_W_FAST:
                 case OP_R_FAST: {
-                    LocalView v;
+                    // Hand off to the fast-op pool so this conn's burst of
+                    // async requests is handled in parallel (inline handling
+                    // serialized each conn behind ~ms handlers). remain is
+                    // bumped HERE so a later OP_SYNC on this ring (FIFO)
+                    // can never miss the request; the worker balances it.
+                    size_t qlen;
+                    {
+                        std::lock_guard<std::mutex> lk(fast_mu_);
+                        qlen = fast_q_.size();
+                    }
+                    if (qlen < 8192 && !fast_workers_.empty()) {
+                        FastWork w;
+                        w.c = c;
+                        w.op = op;
+                        w.seq = h.seq;
+                        w.body.assign(body, body + h.body_len);
+                        c->remain.fetch_add(1);
+                        c->ref();
+                        {
+                            std::lock_guard<std::mutex> lk(fast_mu_);
+                            fast_q_.push_back(std::move(w));
+                        }
+                        fast_cv_.notify_one();
+                        break;
+                    }
+                    LocalView v;  // overload fallback: handle inline
                     if (!parse_packed_local(body, h.body_len, &v))
                         reply_local(c, ctx, INVALID_REQ);
+                    else if (op == OP_W_FAST)
+                        op_local_write(c, v, ctx);
                     else
                         op_local_read(c, v, ctx);
                     break;

@@ -240,6 +240,26 @@ class Server {
     void shm_teardown(Conn* c);       // owner loop thread; joins the poller
     void shm_poll_main(ShmPeer* p);   // poller thread body
     std::atomic<int> shm_peers_{0};
+
+    // Fast-op worker pool: ring pollers hand OP_W_FAST / OP_R_FAST bodies
+    // off so ONE connection's async request burst is handled by many
+    // threads (the poller-inline design serialized each conn's requests
+    // behind ~ms handlers under 64-client saturation). The poller bumps
+    // Conn::remain (and holds a ref) before enqueueing so an OP_SYNC
+    // arriving later on the ring can never observe the writes as absent;
+    // the worker balances it with finish_task after the handler returns.
+    struct FastWork {
+        Conn* c = nullptr;
+        char op = 0;
+        uint64_t seq = 0;
+        std::vector<uint8_t> body;
+    };
+    std::mutex fast_mu_;
+    std::condition_variable fast_cv_;
+    std::deque<FastWork> fast_q_;
+    std::vector<std::thread> fast_workers_;
+    void fast_worker_main();
+    void run_fast(FastWork& w);  // parse + dispatch one fast op
     void op_exchange(Conn* c, const std::vector<uint8_t>& body);
     void op_allocate(Conn* c, const RemoteMetaMsg& msg);
     void op_tcp_put(Conn* c, std::vector<uint8_t> body);
