@@ -92,6 +92,11 @@ bool Shard::allocate(size_t size, size_t n, const AllocationCallback& cb) {
     return mm_.allocate(size, n, cb);
 }
 
+void Shard::deallocate_bulk(const std::vector<BlockFree>& frees) {
+    std::lock_guard<std::mutex> lk(alloc_mu_);
+    for (auto& f : frees) mm_.deallocate(f.ptr, f.size, f.pool_idx);
+}
+
 bool Shard::deallocate(void* ptr, size_t size, int pool_idx) {
     std::lock_guard<std::mutex> lk(alloc_mu_);
     return mm_.deallocate(ptr, size, pool_idx);

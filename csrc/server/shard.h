@@ -74,6 +74,15 @@ class Shard {
     // Allocator (thread-safe).
     bool allocate(size_t size, size_t n, const AllocationCallback& cb);
     bool deallocate(void* ptr, size_t size, int pool_idx);
+    // Bulk free under ONE allocator-lock hold: a 10k-key delete sweep doing
+    // one lock acquisition per block convoyed against concurrent batch
+    // allocations (measured 1.3 s per delete at 64 clients).
+    struct BlockFree {
+        void* ptr;
+        size_t size;
+        int pool_idx;
+    };
+    void deallocate_bulk(const std::vector<BlockFree>& frees);
     size_t used_blocks();
     size_t total_blocks();
     bool contains(const void* p);
