@@ -362,7 +362,7 @@ bool Server::start() {
                 for (auto& st : kv_) {
                     std::vector<Ref<BlockEntry>> dead;  // dropped after unlock
                     {
-                        std::lock_guard<std::mutex> ex(st.mu);
+                        std::lock_guard<std::shared_mutex> ex(st.mu);
                         std::vector<std::string> victims;
                         st.map.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
                             if (expired(val.get()) && val->ref_count() == 1)
@@ -794,7 +794,7 @@ void Server::erase_entries(const std::vector<Ref<BlockEntry>>& entries) {
     for (auto& st : kv_) {
         std::vector<Ref<BlockEntry>> dead;  // dropped after unlock
         {
-            std::lock_guard<std::mutex> lk(st.mu);
+            std::lock_guard<std::shared_mutex> lk(st.mu);
             std::vector<std::string> victims;
             st.map.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
                 for (auto& e : entries)
@@ -822,7 +822,7 @@ size_t Server::evict_lru(Shard* shard, size_t bytes) {
     for (size_t visited = 0; freed < bytes && visited < kStripes; visited++) {
         auto& st = kv_[evict_stripe_rr_.fetch_add(1) % kStripes];
         std::vector<Ref<BlockEntry>> dead;  // block frees run after unlock
-        std::lock_guard<std::mutex> lk(st.mu);
+        std::lock_guard<std::shared_mutex> lk(st.mu);
         size_t scanned = 0;
         const size_t scan_cap = st.map.capacity();  // one revolution max
         std::vector<std::pair<uint64_t, std::string_view>> sample;
@@ -965,7 +965,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         if (list.empty()) continue;
         auto tl0 = pdbg ? std::chrono::steady_clock::now()
                         : std::chrono::steady_clock::time_point{};
-        std::lock_guard<std::mutex> lk(kv_[si].mu);
+        std::shared_lock<std::shared_mutex> lk(kv_[si].mu);
         auto tl1 = pdbg ? std::chrono::steady_clock::now()
                         : std::chrono::steady_clock::time_point{};
         auto& m = kv_[si].map;
@@ -1132,7 +1132,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
             if (list.empty()) continue;
             auto tl0 = pdbg ? std::chrono::steady_clock::now()
                             : std::chrono::steady_clock::time_point{};
-            std::lock_guard<std::mutex> lk(kv_[si].mu);
+            std::lock_guard<std::shared_mutex> lk(kv_[si].mu);
             auto tl1 = pdbg ? std::chrono::steady_clock::now()
                             : std::chrono::steady_clock::time_point{};
             struct UlAcc {  // accumulate under-lock time at scope exit
@@ -1238,7 +1238,7 @@ void Server::op_local_read(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
         for (size_t si = 0; si < kStripes; si++) {
             auto& list = by_stripe[si];
             if (list.empty()) continue;
-            std::lock_guard<std::mutex> lk(kv_[si].mu);
+            std::shared_lock<std::shared_mutex> lk(kv_[si].mu);
             auto& m = kv_[si].map;
             size_t ln = list.size();
             for (size_t i = 0; i < std::min(kPf, ln); i++) m.prefetch(hashes[list[i]]);
@@ -1623,7 +1623,7 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
         for (auto& [k, h, p] : created) {
             {
                 auto& st = kv_[stripe_of(h)];
-                std::lock_guard<std::mutex> lk(st.mu);
+                std::lock_guard<std::shared_mutex> lk(st.mu);
                 st.map.erase(k);
             }
             c->pending_rdma.erase(p);  // drop the ref so the block frees now
@@ -1633,7 +1633,7 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
         uint64_t h = KvMap::hash_of(key);
         auto& st = kv_[stripe_of(h)];
         {
-            std::lock_guard<std::mutex> sl(st.mu);
+            std::shared_lock<std::shared_mutex> sl(st.mu);
             Ref<BlockEntry>* v0 = st.map.find_hashed(key, h);
             if (v0 != nullptr && !expired(v0->get())) {
                 blocks.push_back({0, 0, 0});  // FAKE block: dup key, client skips
@@ -1668,7 +1668,7 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
         Ref<BlockEntry> ref(e);
         bool ins = false;
         {
-            std::lock_guard<std::mutex> lk(st.mu);
+            std::lock_guard<std::shared_mutex> lk(st.mu);
             Ref<BlockEntry>* slot2 = st.map.emplace_hashed(key, h, ref, &ins);
             if (!ins && slot2 && expired(slot2->get())) {
                 *slot2 = ref;  // expired entry: replace
@@ -1692,7 +1692,7 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
             if (!vdrv_ || !vdrv_->lookup_region(ptr, &mr)) {
                 ERROR("allocate: pool arena %p has no MR on the verbs fabric", ptr);
                 {
-                    std::lock_guard<std::mutex> lk(st.mu);
+                    std::lock_guard<std::shared_mutex> lk(st.mu);
                     st.map.erase(key);
                 }
                 rollback();
@@ -1725,7 +1725,7 @@ bool Server::collect_read_entries(const std::vector<std::string>& keys,
     for (auto& key : keys) {
         uint64_t h = KvMap::hash_of(key);
         auto& st = kv_[stripe_of(h)];
-        std::lock_guard<std::mutex> lk(st.mu);
+        std::shared_lock<std::shared_mutex> lk(st.mu);
         Ref<BlockEntry>* v = st.map.find_hashed(key, h);
         if (!v || !(*v)->committed || expired(v->get())) return false;
         // fp8-compressed entries are a local-GPU-path feature: the TCP/verbs
@@ -1933,7 +1933,7 @@ void Server::op_check_exist(Conn* c, const std::vector<uint8_t>& body, const Req
     {
         uint64_t h = KvMap::hash_of(key);
         auto& st = kv_[stripe_of(h)];
-        std::lock_guard<std::mutex> lk(st.mu);
+        std::shared_lock<std::shared_mutex> lk(st.mu);
         Ref<BlockEntry>* v = st.map.find_hashed(key, h);
         exists = v && (*v)->committed && !expired(v->get());
     }
@@ -1952,7 +1952,7 @@ void Server::op_match_index(Conn* c, const std::vector<uint8_t>& body, const Req
     auto present = [&](size_t i) {
         uint64_t h = KvMap::hash_of(keys[i]);
         auto& st = kv_[stripe_of(h)];
-        std::lock_guard<std::mutex> lk(st.mu);
+        std::shared_lock<std::shared_mutex> lk(st.mu);
         Ref<BlockEntry>* v = st.map.find_hashed(keys[i], h);
         return v && (*v)->committed && !expired(v->get());
     };
@@ -1994,7 +1994,7 @@ void Server::op_delete(Conn* c, const std::vector<uint8_t>& body, const ReqCtx& 
             auto& list = by_stripe[si];
             if (list.empty()) continue;
             {
-                std::lock_guard<std::mutex> lk(kv_[si].mu);
+                std::lock_guard<std::shared_mutex> lk(kv_[si].mu);
                 auto& m = kv_[si].map;
                 for (size_t i = 0; i < std::min(kPf, list.size()); i++)
                     m.prefetch(hashes[list[i]]);
@@ -2047,7 +2047,7 @@ std::pair<size_t, size_t> Server::compact() {
         std::vector<Ref<BlockEntry>> held;
         std::vector<std::pair<void*, size_t>> movable;
         {
-            std::vector<std::unique_lock<std::mutex>> locks;
+            std::vector<std::unique_lock<std::shared_mutex>> locks;
             locks.reserve(kStripes);
             for (auto& st : kv_) locks.emplace_back(st.mu);
             for (auto& st : kv_)
@@ -2085,7 +2085,7 @@ std::pair<size_t, size_t> Server::compact() {
         std::map<void*, Shard::Move*> by_old;
         for (auto& m : moves) by_old[m.old_ptr] = &m;
         {
-            std::vector<std::unique_lock<std::mutex>> locks;
+            std::vector<std::unique_lock<std::shared_mutex>> locks;
             locks.reserve(kStripes);
             for (auto& st : kv_) locks.emplace_back(st.mu);
             for (auto& st : kv_)
@@ -2117,7 +2117,7 @@ std::pair<size_t, size_t> Server::compact() {
 size_t Server::kvmap_len() {
     size_t n = 0;
     for (auto& st : kv_) {
-        std::lock_guard<std::mutex> lk(st.mu);
+        std::shared_lock<std::shared_mutex> lk(st.mu);
         n += st.map.size();
     }
     return n;
@@ -2126,7 +2126,7 @@ size_t Server::kvmap_len() {
 size_t Server::purge() {
     size_t n = 0;
     for (auto& st : kv_) {
-        std::lock_guard<std::mutex> lk(st.mu);
+        std::lock_guard<std::shared_mutex> lk(st.mu);
         n += st.map.size();
         st.map.clear();
     }
@@ -2152,7 +2152,7 @@ bool Server::snapshot(const std::string& path, std::pair<size_t, size_t>* out) {
     // them out without holding any lock.
     std::vector<std::pair<std::string, Ref<BlockEntry>>> pinned;
     for (auto& st : kv_) {
-        std::lock_guard<std::mutex> lk(st.mu);
+        std::shared_lock<std::shared_mutex> lk(st.mu);
         st.map.for_each([&](std::string_view key, Ref<BlockEntry>& val) {
             if (val->committed && !expired(val.get()))
                 pinned.emplace_back(std::string(key), val);
@@ -2251,7 +2251,7 @@ bool Server::restore(const std::string& path, std::pair<size_t, size_t>* out) {
         auto& st = kv_[stripe_of(hh)];
         bool ins = false;
         {
-            std::lock_guard<std::mutex> lk(st.mu);
+            std::lock_guard<std::shared_mutex> lk(st.mu);
             st.map.emplace_hashed(key, hh, ref, &ins);
         }
         if (ins) {

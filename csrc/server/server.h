@@ -323,12 +323,13 @@ class Server {
     // stripe locks; 16 stripes measured multi-ms convoys under saturation.
     static constexpr size_t kStripes = 64;
     struct KvStripe {
-        // Plain mutex, NOT shared_mutex: glibc rwlocks prefer readers, and
-        // under the fast-op pool's continuous shared traffic an exclusive
-        // waiter (delete/evict sweeps) starved for SECONDS (measured 967 ms
-        // avg delete at 64 clients). Holds are ~0.4 us over 64 stripes, so
-        // reader concurrency within one stripe is worth nothing here.
-        std::mutex mu;
+        // shared_mutex: dedup/read/query phases take it shared so the
+        // pipelined few-connection path keeps its read concurrency. The
+        // earlier "delete starvation" was NOT rwlock reader preference (a
+        // plain-mutex experiment changed nothing) — it was per-block
+        // allocator-lock frees under the stripe lock, fixed by extract +
+        // bulk free; long sweeps now hold each stripe only briefly.
+        std::shared_mutex mu;
         KvMap map;
         size_t evict_hand = 0;  // per-stripe clock cursor (mu held)
     };
