@@ -283,20 +283,23 @@ bool Shard::submit_copy(CopyJob&& job) {
     uint32_t rr = next_stream_.fetch_add(1);
     uint32_t n_streams = static_cast<uint32_t>(streams_.size());
     uint32_t pick = rr % n_streams;
-    for (uint32_t i = 0; i < n_streams; i++) {
-        StreamCtx& cand = streams_[(rr + i) % n_streams];
-        bool has_free = false;
-        {
-            std::lock_guard<std::mutex> lk(cand.mu);
-            for (auto& sl : cand.slots)
-                if (!sl.busy) {
-                    has_free = true;
-                    break;
-                }
-        }
-        if (has_free) {
-            pick = (rr + i) % n_streams;
-            break;
+    auto has_free_slot = [](StreamCtx& cand) {
+        std::lock_guard<std::mutex> lk(cand.mu);
+        for (auto& sl : cand.slots)
+            if (!sl.busy) return true;
+        return false;
+    };
+    // Round-robin keeps concurrent writers/readers spread across streams
+    // (their kernels overlap); only when the nominal stream is FULL look
+    // for another with room. (An always-scan variant made every submitter
+    // converge on whichever stream had just freed slots, collapsing the
+    // pipelined few-connection bench onto one stream.)
+    if (!has_free_slot(streams_[pick])) {
+        for (uint32_t i = 1; i < n_streams; i++) {
+            if (has_free_slot(streams_[(rr + i) % n_streams])) {
+                pick = (rr + i) % n_streams;
+                break;
+            }
         }
     }
     StreamCtx& sc = streams_[pick];
