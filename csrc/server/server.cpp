@@ -819,6 +819,13 @@ size_t Server::evict_lru(Shard* shard, size_t bytes) {
     // Stripes are visited round-robin (shared cursor) holding one stripe
     // lock at a time.
     size_t freed = 0;
+    // Freshness guard: the most recently accessed slice of the pool is
+    // exempt — at a full pool the LRU would otherwise thrash brand-new
+    // writes (evicting a generation the client is about to read back; seen
+    // in the mixed soak at 98% occupancy). Scaled to occupancy so small
+    // pools can still evict. Computed before any stripe lock is taken
+    // (used_blocks takes the allocator lock).
+    const uint64_t fresh_guard = std::min<uint64_t>(4096, shard->used_blocks() / 8);
     for (size_t visited = 0; freed < bytes && visited < kStripes; visited++) {
         auto& st = kv_[evict_stripe_rr_.fetch_add(1) % kStripes];
         std::vector<Ref<BlockEntry>> dead;  // block frees run after unlock
@@ -829,16 +836,12 @@ size_t Server::evict_lru(Shard* shard, size_t bytes) {
         sample.reserve(128);
         while (freed < bytes && scanned < scan_cap) {
             size_t window = 4096;
-            // Freshness guard: entries accessed within the last ~4096
-            // requests are exempt — at a full pool the LRU would otherwise
-            // thrash brand-new writes (evicting a generation the client is
-            // about to read back; seen in the mixed soak at 98% occupancy).
             const uint64_t now_tick = access_tick_.load(std::memory_order_relaxed);
             st.map.scan_from(&st.evict_hand, window,
                              [&](std::string_view key, Ref<BlockEntry>& val) {
                                  BlockEntry* e = val.get();
                                  uint64_t la = e->last_access.load(std::memory_order_relaxed);
-                                 bool fresh = now_tick - la < 4096;
+                                 bool fresh = now_tick - la < fresh_guard;
                                  if (e->shard == shard && e->committed &&
                                      e->ref_count() == 1 && (!fresh || expired(e)))
                                      sample.push_back(
