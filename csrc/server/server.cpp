@@ -809,7 +809,7 @@ void Server::erase_entries(const std::vector<Ref<BlockEntry>>& entries) {
     }
 }
 
-size_t Server::evict_lru(Shard* shard, size_t bytes) {
+size_t Server::evict_lru(Shard* shard, size_t bytes, bool protect_fresh) {
     // Sampled clock-hand eviction: scan bounded slot windows from a
     // persistent per-stripe cursor, evict the least-recently-accessed half
     // of each sample — O(evicted) amortized instead of a full index scan
@@ -825,7 +825,8 @@ size_t Server::evict_lru(Shard* shard, size_t bytes) {
     // in the mixed soak at 98% occupancy). Scaled to occupancy so small
     // pools can still evict. Computed before any stripe lock is taken
     // (used_blocks takes the allocator lock).
-    const uint64_t fresh_guard = std::min<uint64_t>(4096, shard->used_blocks() / 8);
+    const uint64_t fresh_guard =
+        protect_fresh ? std::min<uint64_t>(4096, shard->used_blocks() / 8) : 0;
     for (size_t visited = 0; freed < bytes && visited < kStripes; visited++) {
         auto& st = kv_[evict_stripe_rr_.fetch_add(1) % kStripes];
         std::vector<Ref<BlockEntry>> dead;  // block frees run after unlock
@@ -1020,7 +1021,11 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     // freed blocks to a racing allocation; keep evicting until the
     // allocation lands or eviction runs dry.
     for (int attempt = 0; !alloc_ok && opt_.auto_evict && attempt < 4; attempt++) {
-        if (evict_lru(shard, page * n_fresh * 2) == 0) break;
+        // attempt 0 protects fresh entries and may find nothing; only an
+        // UNPROTECTED dry run means the pool is truly unreclaimable.
+        if (evict_lru(shard, page * n_fresh * 2, /*protect_fresh=*/attempt == 0) == 0 &&
+            attempt > 0)
+            break;
         alloc_ok = try_alloc();
     }
     if (!alloc_ok) return reply_local(c, ctx, OUT_OF_MEMORY);
@@ -1660,7 +1665,9 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
         };
         bool ok = try_alloc();
         for (int attempt = 0; !ok && opt_.auto_evict && attempt < 4; attempt++) {
-            if (evict_lru(shard, page * 4) == 0) break;
+            if (evict_lru(shard, page * 4, /*protect_fresh=*/attempt == 0) == 0 &&
+                attempt > 0)
+                break;
             ok = try_alloc();
         }
         if (!ok) {
