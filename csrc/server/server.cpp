@@ -809,7 +809,7 @@ void Server::erase_entries(const std::vector<Ref<BlockEntry>>& entries) {
     }
 }
 
-size_t Server::evict_lru(Shard* shard, size_t bytes, bool protect_fresh) {
+size_t Server::evict_lru(Shard* shard, size_t bytes) {
     // Sampled clock-hand eviction: scan bounded slot windows from a
     // persistent per-stripe cursor, evict the least-recently-accessed half
     // of each sample — O(evicted) amortized instead of a full index scan
@@ -819,14 +819,13 @@ size_t Server::evict_lru(Shard* shard, size_t bytes, bool protect_fresh) {
     // Stripes are visited round-robin (shared cursor) holding one stripe
     // lock at a time.
     size_t freed = 0;
-    // Freshness guard: the most recently accessed slice of the pool is
-    // exempt — at a full pool the LRU would otherwise thrash brand-new
-    // writes (evicting a generation the client is about to read back; seen
-    // in the mixed soak at 98% occupancy). Scaled to occupancy so small
-    // pools can still evict. Computed before any stripe lock is taken
-    // (used_blocks takes the allocator lock).
-    const uint64_t fresh_guard =
-        protect_fresh ? std::min<uint64_t>(4096, shard->used_blocks() / 8) : 0;
+    // NOTE on fresh writes: the sample sort below already evicts oldest
+    // first, so recently written keys survive unless a sample is entirely
+    // fresh (possible at ~100% occupancy — the pool is then genuinely too
+    // small for the working set and evicting fresh data is the honest LRU
+    // outcome). An explicit freshness-exemption pass was tried and reverted:
+    // at full occupancy it degenerated to an O(map) rescan per allocation
+    // (13x soak throughput collapse).
     for (size_t visited = 0; freed < bytes && visited < kStripes; visited++) {
         auto& st = kv_[evict_stripe_rr_.fetch_add(1) % kStripes];
         std::vector<Ref<BlockEntry>> dead;  // block frees run after unlock
@@ -837,14 +836,11 @@ size_t Server::evict_lru(Shard* shard, size_t bytes, bool protect_fresh) {
         sample.reserve(128);
         while (freed < bytes && scanned < scan_cap) {
             size_t window = 4096;
-            const uint64_t now_tick = access_tick_.load(std::memory_order_relaxed);
             st.map.scan_from(&st.evict_hand, window,
                              [&](std::string_view key, Ref<BlockEntry>& val) {
                                  BlockEntry* e = val.get();
                                  uint64_t la = e->last_access.load(std::memory_order_relaxed);
-                                 bool fresh = now_tick - la < fresh_guard;
-                                 if (e->shard == shard && e->committed &&
-                                     e->ref_count() == 1 && (!fresh || expired(e)))
+                                 if (e->shard == shard && e->committed && e->ref_count() == 1)
                                      sample.push_back(
                                          {expired(e) ? 0  // expired: evict first
                                                      : la,
@@ -1021,11 +1017,7 @@ void Server::op_local_write(Conn* c, const LocalView& msg, const ReqCtx& ctx) {
     // freed blocks to a racing allocation; keep evicting until the
     // allocation lands or eviction runs dry.
     for (int attempt = 0; !alloc_ok && opt_.auto_evict && attempt < 4; attempt++) {
-        // attempt 0 protects fresh entries and may find nothing; only an
-        // UNPROTECTED dry run means the pool is truly unreclaimable.
-        if (evict_lru(shard, page * n_fresh * 2, /*protect_fresh=*/attempt == 0) == 0 &&
-            attempt > 0)
-            break;
+        if (evict_lru(shard, page * n_fresh * 2) == 0) break;
         alloc_ok = try_alloc();
     }
     if (!alloc_ok) return reply_local(c, ctx, OUT_OF_MEMORY);
@@ -1665,9 +1657,7 @@ std::vector<RemoteBlockWire> Server::allocate_blocks(Conn* c,
         };
         bool ok = try_alloc();
         for (int attempt = 0; !ok && opt_.auto_evict && attempt < 4; attempt++) {
-            if (evict_lru(shard, page * 4, /*protect_fresh=*/attempt == 0) == 0 &&
-                attempt > 0)
-                break;
+            if (evict_lru(shard, page * 4) == 0) break;
             ok = try_alloc();
         }
         if (!ok) {

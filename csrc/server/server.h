@@ -274,10 +274,7 @@ class Server {
     void maybe_extend(Shard* s);
     // Evict >= `bytes` of LRU committed idle entries on `shard`. Takes
     // stripe locks itself (callers must hold none). Returns bytes freed.
-    // protect_fresh: exempt the most recently touched entries (first
-    // attempt); retries pass false so a pool full of fresh data still
-    // makes progress instead of OOMing.
-    size_t evict_lru(Shard* shard, size_t bytes, bool protect_fresh = true);
+    size_t evict_lru(Shard* shard, size_t bytes);
     void erase_entries(const std::vector<Ref<BlockEntry>>& entries);
     uint64_t tick() { return access_tick_.fetch_add(1, std::memory_order_relaxed); }
     static uint32_t now_sec();
