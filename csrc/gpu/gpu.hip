@@ -241,19 +241,31 @@ bool event_query(Event e) {
 // (/root/reference/src/infinistore.cpp:622-625, 747-748) with one launch per
 // request. Lanes walk consecutive uint4 units, so a wave issues fully
 // coalesced 1 KiB transactions; grid-stride covers all blocks.
+// 2D mapping: workgroup (b, chunk) covers block b's units
+// [chunk*T, ...] striding by chunks*T. Each workgroup reads its block's
+// descriptor ONCE (lane 0 -> LDS broadcast), so the descriptor arrays can
+// live in pinned HOST memory with negligible PCIe cost — the earlier
+// per-unit desc[b] indexing either taxed PCIe at ~payload/64 (pinned) or
+// forced per-job uploads that rocclr ran as single-workgroup blits at
+// ~110 µs each (profiles/rocprof_bench_r02.txt).
 __global__ void copy_blocks_vec_kernel(const uint64_t* __restrict__ src_ptrs,
-                                       const uint64_t* __restrict__ dst_ptrs, int n_blocks,
+                                       const uint64_t* __restrict__ dst_ptrs,
+                                       uint32_t chunks_per_block,
                                        uint64_t units_per_block) {
-    uint64_t total = static_cast<uint64_t>(n_blocks) * units_per_block;
-    uint64_t stride = static_cast<uint64_t>(gridDim.x) * blockDim.x;
-    for (uint64_t u = blockIdx.x * static_cast<uint64_t>(blockDim.x) + threadIdx.x; u < total;
-         u += stride) {
-        uint64_t b = u / units_per_block;
-        uint64_t off = u - b * units_per_block;
-        const uint4* s = reinterpret_cast<const uint4*>(src_ptrs[b]) + off;
-        uint4* d = reinterpret_cast<uint4*>(dst_ptrs[b]) + off;
-        *d = *s;
+    uint32_t b = blockIdx.x / chunks_per_block;
+    uint32_t chunk = blockIdx.x % chunks_per_block;
+    __shared__ uint64_t sd[2];
+    if (threadIdx.x == 0) {
+        sd[0] = src_ptrs[b];
+        sd[1] = dst_ptrs[b];
     }
+    __syncthreads();
+    const uint4* s = reinterpret_cast<const uint4*>(sd[0]);
+    uint4* d = reinterpret_cast<uint4*>(sd[1]);
+    uint64_t stride = static_cast<uint64_t>(chunks_per_block) * blockDim.x;
+    for (uint64_t u = static_cast<uint64_t>(chunk) * blockDim.x + threadIdx.x;
+         u < units_per_block; u += stride)
+        d[u] = s[u];
 }
 
 // Small-request variant: descriptors passed by value as kernel arguments —
@@ -297,20 +309,34 @@ bool launch_copy_blocks_inline(int dev, Stream stream, const uint64_t* src_ptrs,
     return true;
 }
 
-// Byte-granular fallback for unaligned pointers / sizes.
+// Byte-granular fallback for unaligned pointers / sizes (same 2D mapping).
 __global__ void copy_blocks_byte_kernel(const uint64_t* __restrict__ src_ptrs,
-                                        const uint64_t* __restrict__ dst_ptrs, int n_blocks,
+                                        const uint64_t* __restrict__ dst_ptrs,
+                                        uint32_t chunks_per_block,
                                         uint64_t bytes_per_block) {
-    uint64_t total = static_cast<uint64_t>(n_blocks) * bytes_per_block;
-    uint64_t stride = static_cast<uint64_t>(gridDim.x) * blockDim.x;
-    for (uint64_t u = blockIdx.x * static_cast<uint64_t>(blockDim.x) + threadIdx.x; u < total;
-         u += stride) {
-        uint64_t b = u / bytes_per_block;
-        uint64_t off = u - b * bytes_per_block;
-        const uint8_t* s = reinterpret_cast<const uint8_t*>(src_ptrs[b]) + off;
-        uint8_t* d = reinterpret_cast<uint8_t*>(dst_ptrs[b]) + off;
-        *d = *s;
+    uint32_t b = blockIdx.x / chunks_per_block;
+    uint32_t chunk = blockIdx.x % chunks_per_block;
+    __shared__ uint64_t sd[2];
+    if (threadIdx.x == 0) {
+        sd[0] = src_ptrs[b];
+        sd[1] = dst_ptrs[b];
     }
+    __syncthreads();
+    const uint8_t* s = reinterpret_cast<const uint8_t*>(sd[0]);
+    uint8_t* d = reinterpret_cast<uint8_t*>(sd[1]);
+    uint64_t stride = static_cast<uint64_t>(chunks_per_block) * blockDim.x;
+    for (uint64_t u = static_cast<uint64_t>(chunk) * blockDim.x + threadIdx.x;
+         u < bytes_per_block; u += stride)
+        d[u] = s[u];
+}
+
+// Enough workgroups to fill 256 CUs across 8 XCDs with headroom, but ONE
+// descriptor read per workgroup.
+static uint32_t copy_chunks_per_block(uint64_t units_per_block, int n_blocks, int threads) {
+    uint64_t max_chunks = (units_per_block + threads - 1) / static_cast<uint64_t>(threads);
+    uint64_t want = (4096 + n_blocks - 1) / static_cast<uint64_t>(n_blocks);
+    uint64_t c = std::min<uint64_t>(std::max<uint64_t>(want, 1), max_chunks);
+    return static_cast<uint32_t>(std::max<uint64_t>(c, 1));
 }
 
 bool launch_copy_blocks(int dev, Stream stream, const uint64_t* dev_src_ptrs,
@@ -325,16 +351,15 @@ bool launch_copy_blocks(int dev, Stream stream, const uint64_t* dev_src_ptrs,
     const int threads = 512;
     if (aligned16 && bytes_per_block % 16 == 0) {
         uint64_t upb = bytes_per_block / 16;
-        uint64_t total = static_cast<uint64_t>(n_blocks) * upb;
-        // Fill 256 CUs (8 XCDs) with headroom; grid-stride the rest.
-        int grid = static_cast<int>(std::min<uint64_t>((total + threads - 1) / threads, 4096));
-        hipLaunchKernelGGL(copy_blocks_vec_kernel, dim3(grid), dim3(threads), 0, s, dev_src_ptrs,
-                           dev_dst_ptrs, n_blocks, upb);
+        uint32_t chunks = copy_chunks_per_block(upb, n_blocks, threads);
+        dim3 grid(static_cast<uint32_t>(n_blocks) * chunks);
+        hipLaunchKernelGGL(copy_blocks_vec_kernel, grid, dim3(threads), 0, s, dev_src_ptrs,
+                           dev_dst_ptrs, chunks, upb);
     } else {
-        uint64_t total = static_cast<uint64_t>(n_blocks) * bytes_per_block;
-        int grid = static_cast<int>(std::min<uint64_t>((total + threads - 1) / threads, 4096));
-        hipLaunchKernelGGL(copy_blocks_byte_kernel, dim3(grid), dim3(threads), 0, s, dev_src_ptrs,
-                           dev_dst_ptrs, n_blocks, bytes_per_block);
+        uint32_t chunks = copy_chunks_per_block(bytes_per_block, n_blocks, threads);
+        dim3 grid(static_cast<uint32_t>(n_blocks) * chunks);
+        hipLaunchKernelGGL(copy_blocks_byte_kernel, grid, dim3(threads), 0, s, dev_src_ptrs,
+                           dev_dst_ptrs, chunks, bytes_per_block);
     }
     HIP_OK(hipGetLastError());
     return true;

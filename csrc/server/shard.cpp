@@ -32,10 +32,7 @@ Shard::~Shard() {
             if (sl.event) gpu::event_destroy(sl.event);
             if (sl.h_src) gpu::free_host_pinned(sl.h_src);
             if (sl.h_dst) gpu::free_host_pinned(sl.h_dst);
-            if (sl.d_src) gpu::free_device(sl.d_src);
-            if (sl.d_dst) gpu::free_device(sl.d_dst);
             if (sl.h_scale) gpu::free_host_pinned(sl.h_scale);
-            if (sl.d_scale) gpu::free_device(sl.d_scale);
         }
         if (sc.stream) gpu::stream_destroy(sc.stream);
     }
@@ -65,14 +62,9 @@ bool Shard::init() {
                 size_t scale_bytes = opt_.max_descs_per_slot * sizeof(float);
                 sl.h_src = static_cast<uint64_t*>(gpu::alloc_host_pinned(desc_bytes));
                 sl.h_dst = static_cast<uint64_t*>(gpu::alloc_host_pinned(desc_bytes));
-                sl.d_src = static_cast<uint64_t*>(gpu::alloc_device(opt_.device, desc_bytes));
-                sl.d_dst = static_cast<uint64_t*>(gpu::alloc_device(opt_.device, desc_bytes));
                 sl.h_scale = static_cast<float*>(gpu::alloc_host_pinned(scale_bytes));
-                sl.d_scale = static_cast<float*>(gpu::alloc_device(opt_.device, scale_bytes));
                 sl.event = gpu::event_create(opt_.device);
-                if (!sl.h_src || !sl.h_dst || !sl.d_src || !sl.d_dst || !sl.h_scale ||
-                    !sl.d_scale || !sl.event)
-                    return false;
+                if (!sl.h_src || !sl.h_dst || !sl.h_scale || !sl.event) return false;
             }
         }
         // One completion thread PER STREAM: done-callbacks can block (the
@@ -342,20 +334,13 @@ bool Shard::submit_copy(CopyJob&& job) {
         memcpy(slot->h_dst, job.dst.data() + off, take * sizeof(uint64_t));
         bool last = off + take >= n;
         bool ok = gpu::set_device(opt_.device);
-        // Descriptor placement (Slot comment): small jobs read pinned host
-        // directly (no SDMA — the per-job upload pair stalled streams at
-        // 64-client saturation); big jobs upload once so the grid-stride
-        // kernel's per-unit desc[b] reads hit HBM, not PCIe.
+        // Descriptors live in pinned host memory for EVERY size: the copy
+        // kernels read each block's descriptor once per workgroup (LDS
+        // broadcast), so there is no per-unit PCIe tax and no per-job
+        // upload (whose rocclr blit cost ~110 µs,
+        // profiles/rocprof_bench_r02.txt).
         const uint64_t* dsrc = slot->h_src;
         const uint64_t* ddst = slot->h_dst;
-        if (take > kPinnedDescMax) {
-            ok = ok && gpu::memcpy_h2d_async(slot->d_src, slot->h_src,
-                                             take * sizeof(uint64_t), sc.stream) &&
-                 gpu::memcpy_h2d_async(slot->d_dst, slot->h_dst, take * sizeof(uint64_t),
-                                       sc.stream);
-            dsrc = slot->d_src;
-            ddst = slot->d_dst;
-        }
         if (job.xform == CopyJob::Xform::kCopy) {
             ok = ok && gpu::launch_copy_blocks(opt_.device, sc.stream, dsrc, ddst,
                                                static_cast<int>(take), job.bytes_per_block,
@@ -366,16 +351,10 @@ bool Shard::submit_copy(CopyJob&& job) {
             ok = ok && gpu::launch_quant_blocks(opt_.device, sc.stream, dsrc, ddst,
                                                 slot->h_scale, static_cast<int>(take),
                                                 job.bytes_per_block / 2);
-        } else {  // kDequantFp8Bf16
+        } else {  // kDequantFp8Bf16 (kernel reads desc + scale once per WG)
             memcpy(slot->h_scale, job.scales_in.data() + off, take * sizeof(float));
-            const float* dscale = slot->h_scale;
-            if (take > kPinnedDescMax) {
-                ok = ok && gpu::memcpy_h2d_async(slot->d_scale, slot->h_scale,
-                                                 take * sizeof(float), sc.stream);
-                dscale = slot->d_scale;
-            }
-            ok = ok && gpu::launch_dequant_blocks(opt_.device, sc.stream, dsrc, ddst, dscale,
-                                                  static_cast<int>(take),
+            ok = ok && gpu::launch_dequant_blocks(opt_.device, sc.stream, dsrc, ddst,
+                                                  slot->h_scale, static_cast<int>(take),
                                                   job.bytes_per_block / 2);
         }
         ok = ok && gpu::event_record(slot->event, sc.stream);

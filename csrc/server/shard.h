@@ -130,19 +130,15 @@ class Shard {
     void for_each_arena(const std::function<void(void*, size_t, bool on_gpu)>& fn);
 
    private:
-    // Small jobs read descriptors straight from pinned host memory (no SDMA
-    // upload — the per-job upload pair backed streams up at 64-client
-    // saturation); big jobs upload once and read from HBM (the grid-stride
-    // kernel re-reads desc[b] per 16 B unit, so pinned-host descs cost
-    // ~payload/64 of PCIe traffic — measurable at TB/s).
-    static constexpr size_t kPinnedDescMax = 512;
+    // Descriptors live in pinned host memory for every job size: the copy
+    // kernels read each block's descriptor once per workgroup (LDS
+    // broadcast), so there is no per-unit PCIe tax and no per-job SDMA/blit
+    // upload (both variants measured and rejected —
+    // profiles/rocprof_bench_r02.txt).
     struct Slot {
         uint64_t* h_src = nullptr;  // pinned
         uint64_t* h_dst = nullptr;
-        uint64_t* d_src = nullptr;  // device (big-job staging)
-        uint64_t* d_dst = nullptr;
         float* h_scale = nullptr;  // fp8 scales (kernel writes/reads in place)
-        float* d_scale = nullptr;  // big-job dequant scale staging
         gpu::Event event = nullptr;
         bool busy = false;
     };
