@@ -58,7 +58,7 @@ def parse_args():
                         "each pair runs its own two-deep pipelined loop over "
                         "1/conns of the blocks in its own thread (3 measured "
                         "best: 1.73-1.80 TB/s vs 1.55 at 2 on one MI355X)")
-    p.add_argument("--procs", type=int, default=3,
+    p.add_argument("--procs", type=int, default=4,
                    help="client worker PROCESSES per rank (local path): each "
                         "runs a pipelined conn pair over 1/procs of the "
                         "blocks. Python threads serialize the per-step "

@@ -1422,7 +1422,10 @@ void Server::op_shm_setup(Conn* c, const std::vector<uint8_t>& body) {
     // set up, name is implausible, or too many pollers are running.
     if (c->shm || body.empty() || body.size() > 100 || body[0] != '/')
         return send_status(c, INVALID_REQ);
-    if (shm_peers_.load() >= 64) return send_status(c, SYSTEM_ERROR);
+    // Cap on ring pollers: sized for 8 ranks x (4 worker pairs + 1 parent)
+    // plus 64 saturation clients with headroom (overflow conns gracefully
+    // stay on the socket).
+    if (shm_peers_.load() >= 160) return send_status(c, SYSTEM_ERROR);
     std::string name(reinterpret_cast<const char*>(body.data()), body.size());
     auto* p = new ShmPeer();
     p->srv = this;
