@@ -187,7 +187,7 @@ def run_procs_mode(args, rank, world, local_rank, dist, port, block_bytes,
         dist.broadcast_object_list(obj, src=0)
         run_id = obj[0]
 
-    P = args.procs
+    P = max(1, min(args.procs, args.blocks))
     base, rem = divmod(args.blocks, P)
     worker_blocks = [base + (1 if w < rem else 0) for w in range(P)]
     ctx = multiprocessing.get_context("spawn")
