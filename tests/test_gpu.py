@@ -621,3 +621,46 @@ print("SNAP OK")
         timeout=180,
     ).returncode
     assert code == 0
+
+
+def test_fast_op_pool_many_conns(gpu_server):
+    """>=8 ring connections engage the server's fast-op worker pool
+    (requests fan across threads instead of per-conn serial handling).
+    Concurrent writers+readers on 10 connections must stay correct."""
+    import threading
+
+    conns = [local_conn(gpu_server) for _ in range(10)]
+    try:
+        import numpy as np
+
+        page_elems = 32768  # 64 KB of bf16
+        nb = 64
+        offs = np.arange(nb, dtype=np.uint64) * page_elems
+        srcs = [torch.randn(nb * page_elems, dtype=torch.bfloat16, device="cuda:0")
+                for _ in range(10)]
+        dsts = [torch.zeros_like(srcs[0]) for _ in range(10)]
+        run = uuid.uuid4().hex[:8]
+        errs = []
+
+        def worker(i):
+            try:
+                c = conns[i]
+                for it in range(6):
+                    keys = [f"pool-{run}-c{i}-it{it}-{j}" for j in range(nb)]
+                    c.write_pages(srcs[i], keys, offs, page_elems, sync=True)
+                    c.read_pages(dsts[i], keys, offs, page_elems)
+                    c.sync()
+                    assert torch.equal(srcs[i], dsts[i]), (i, it)
+                    c.delete_keys(keys)
+            except Exception as e:  # surfaced after join
+                errs.append(f"conn {i}: {e}")
+
+        ts = [threading.Thread(target=worker, args=(i,)) for i in range(10)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(timeout=120)
+        assert not errs, errs
+    finally:
+        for c in conns:
+            c.close()
