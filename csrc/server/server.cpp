@@ -343,6 +343,10 @@ bool Server::start() {
     {
         unsigned hw = std::thread::hardware_concurrency();
         unsigned n = std::min(24u, std::max(6u, hw / 8));
+        if (const char* e = getenv("IFS_FAST_WORKERS")) {
+            unsigned v = static_cast<unsigned>(strtoul(e, nullptr, 10));
+            if (v >= 1 && v <= 128) n = v;
+        }
         for (unsigned i = 0; i < n; i++)
             fast_workers_.emplace_back([this] { fast_worker_main(); });
     }
