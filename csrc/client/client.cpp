@@ -407,20 +407,23 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
     }
     gpu::IpcHandle handle;
     uint64_t base_offset = 0;
+    uint64_t alloc_size = 0;
     {
         std::lock_guard<std::mutex> lk(ipc_mu_);
         auto it = ipc_export_cache_.find(ptr);
         if (it != ipc_export_cache_.end()) {
             memcpy(handle.bytes, it->second.handle, gpu::kIpcHandleSize);
             base_offset = it->second.base_offset;
+            alloc_size = it->second.alloc_size;
         } else {
-            bool have =
-                gpu::ipc_export(reinterpret_cast<void*>(ptr), &handle, &base_offset);
+            bool have = gpu::ipc_export(reinterpret_cast<void*>(ptr), &handle,
+                                        &base_offset, &alloc_size);
             if (!have) memset(handle.bytes, 0, gpu::kIpcHandleSize);
             if (ipc_export_cache_.size() > 4096) ipc_export_cache_.clear();
             IpcExport ent;
             memcpy(ent.handle, handle.bytes, gpu::kIpcHandleSize);
             ent.base_offset = base_offset;
+            ent.alloc_size = alloc_size;
             ent.have_handle = have;
             ipc_export_cache_.emplace(ptr, ent);
         }
@@ -437,7 +440,11 @@ int ClientConn::rw_local_packed(char op, const char* keys_blob, size_t blob_len,
     // writes default to async so uploads overlap compute (prefill pattern),
     // with an opt-in single-round-trip mode (write_pages(sync=True)).
     h.flags = ((op == 'R' || sync_response) ? kLocalFlagSyncResponse : 0) | extra_flags;
-    h.rsvd = 0;
+    // Containing allocation size in MB: the server refuses the IPC open for
+    // allocations >= 2 GiB (hipIpcOpenMemHandle hangs importing them under
+    // dmabuf IPC — scripts/ipc_size_probe.py); the same-process pid fast
+    // path is exempt. 0 = unknown.
+    h.rsvd = static_cast<uint32_t>((alloc_size + (1 << 20) - 1) >> 20);
     memcpy(h.ipc, handle.bytes, gpu::kIpcHandleSize);
 
     std::vector<uint8_t> body(sizeof(h) + n * 8 + blob_len);

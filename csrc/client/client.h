@@ -167,6 +167,7 @@ class ClientConn {
     struct IpcExport {
         uint8_t handle[64];
         uint64_t base_offset;
+        uint64_t alloc_size = 0;
         bool have_handle;
     };
     std::unordered_map<uintptr_t, IpcExport> ipc_export_cache_;

@@ -47,7 +47,12 @@ struct IpcHandle {
 // allocation so importers can reconstruct the exact address (works with the
 // PyTorch caching allocator, unlike the reference which needs
 // PYTORCH_NO_CUDA_MEMORY_CACHING).
-bool ipc_export(const void* ptr, IpcHandle* handle, uint64_t* base_offset);
+// alloc_size (optional): the containing allocation's byte size — callers
+// must refuse the IPC path for allocations >= 2 GiB (hipIpcOpenMemHandle
+// hangs forever importing them under dmabuf IPC on this ROCm/driver stack;
+// scripts/ipc_size_probe.py bisected the threshold).
+bool ipc_export(const void* ptr, IpcHandle* handle, uint64_t* base_offset,
+                uint64_t* alloc_size = nullptr);
 void* ipc_open(const IpcHandle& handle, int src_device);
 bool ipc_close(void* base);
 

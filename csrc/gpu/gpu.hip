@@ -125,7 +125,8 @@ bool memcpy_any_async(void* dst, const void* src, size_t n, Stream s) {
 // --- IPC -------------------------------------------------------------------
 static_assert(sizeof(hipIpcMemHandle_t) == kIpcHandleSize, "hip ipc handle size");
 
-bool ipc_export(const void* ptr, IpcHandle* handle, uint64_t* base_offset) {
+bool ipc_export(const void* ptr, IpcHandle* handle, uint64_t* base_offset,
+                uint64_t* alloc_size) {
     // Find the allocation base so tensors offset inside a caching-allocator
     // segment still round-trip exactly.
     void* base = nullptr;
@@ -135,6 +136,13 @@ bool ipc_export(const void* ptr, IpcHandle* handle, uint64_t* base_offset) {
         base = const_cast<void*>(ptr);  // assume ptr is the base
     }
     *base_offset = reinterpret_cast<uintptr_t>(ptr) - reinterpret_cast<uintptr_t>(base);
+    if (alloc_size) {
+        size_t sz = 0;
+        if (hipPointerGetAttribute(&sz, HIP_POINTER_ATTRIBUTE_RANGE_SIZE,
+                                   reinterpret_cast<hipDeviceptr_t>(base)) != hipSuccess)
+            sz = 0;  // unknown: caller decides
+        *alloc_size = sz;
+    }
     hipIpcMemHandle_t h;
     HIP_OK(hipIpcGetMemHandle(&h, base));
     memcpy(handle->bytes, &h, sizeof(h));

@@ -663,6 +663,7 @@ bool parse_packed_local(const uint8_t* body, size_t body_len, Server::LocalView*
     v->base_offset = h.base_offset;
     v->block_size = h.block_size;
     v->flags = h.flags;
+    v->alloc_mb = h.rsvd;
     v->ipc = body + offsetof(PackedLocalHdr, ipc);
     v->ipc_len = 64;
     const uint64_t* offs = reinterpret_cast<const uint64_t*>(body + sizeof(h));
@@ -902,6 +903,19 @@ namespace {
 void* resolve_client_base(Server::Conn* c, const Server::LocalView& msg) {
     if (msg.pid != 0 && msg.pid == static_cast<int32_t>(getpid()) && msg.base_ptr != 0)
         return reinterpret_cast<void*>(msg.base_ptr);
+    // dmabuf IPC limitation on this ROCm/driver stack: hipIpcOpenMemHandle
+    // HANGS FOREVER importing allocations >= 2 GiB (bisected by
+    // scripts/ipc_size_probe.py: 1.5 GiB fine, 2.0 GiB hangs). Refuse
+    // instead of wedging this conn's poller/worker thread. Clients see a
+    // clear error and can split tensors (< 2 GiB per allocation, e.g.
+    // per-layer KV pools) or use the fabric path for the big tensor.
+    if (msg.alloc_mb >= 2048) {
+        ERROR("refusing IPC import of a %u MB client allocation: "
+              "hipIpcOpenMemHandle hangs for >= 2 GiB allocations under "
+              "dmabuf IPC; split the tensor or use the fabric path",
+              msg.alloc_mb);
+        return nullptr;
+    }
     std::vector<uint8_t> key(msg.ipc, msg.ipc + msg.ipc_len);
     std::lock_guard<std::mutex> lk(c->ipc_mu);
     auto it = c->ipc_cache.find(key);

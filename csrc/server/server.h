@@ -171,6 +171,7 @@ class Server {
         uint64_t base_offset = 0;
         int64_t block_size = 0;
         uint32_t flags = 0;
+        uint32_t alloc_mb = 0;  // client allocation size (MB); 0 = unknown
         const uint8_t* ipc = nullptr;
         size_t ipc_len = 0;
         std::vector<std::pair<std::string_view, uint64_t>> blocks;

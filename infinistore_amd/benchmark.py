@@ -72,20 +72,37 @@ def make_conn(args, local):
 
 
 def make_buffers(args, conn, local, device):
-    """One src/dst pair per worker, reused across iterations (fresh tensors
+    """src/dst buffers per worker, reused across iterations (fresh tensors
     every iteration would re-export IPC handles and churn the allocator —
-    at 64 clients that dominates the wall)."""
+    at 64 clients that dominates the wall). Split into sub-tensors under
+    ~1.25 GiB each: the local path cannot IPC-import allocations >= 2 GiB
+    (hipIpcOpenMemHandle hangs under dmabuf IPC on this driver stack —
+    scripts/ipc_size_probe.py), and engines hold per-layer pools anyway.
+    Returns (srcs, dsts, blocks_per_sub)."""
+    block_bytes = args.block_size << 10
     total_bytes = args.size << 20
+    n_blocks = total_bytes // block_bytes
+    per_step = max(1, n_blocks // args.steps)
     dt = torch.bfloat16 if args.quant else torch.float32
     es = 2 if args.quant else 4
-    elems = total_bytes // es
-    src = torch.rand(elems, dtype=dt, device=device)
+
+    cap = (1 << 30) + (1 << 28)  # 1.25 GiB
+    n_sub = max(1, -(-total_bytes // cap))
+    blocks_per_sub = -(-n_blocks // n_sub)
+    blocks_per_sub = -(-blocks_per_sub // per_step) * per_step  # step-aligned
+    n_sub = -(-n_blocks // blocks_per_sub)
+
     dst_dev = f"cuda:{args.dst_gpu}" if local else device
-    dst = torch.zeros(elems, dtype=dt, device=dst_dev)
-    if not local:
-        conn.register_mr(src)
-        conn.register_mr(dst)
-    return src, dst
+    srcs, dsts = [], []
+    for t in range(n_sub):
+        nb = min(blocks_per_sub, n_blocks - t * blocks_per_sub)
+        elems = nb * block_bytes // es
+        srcs.append(torch.rand(elems, dtype=dt, device=device))
+        dsts.append(torch.zeros(elems, dtype=dt, device=dst_dev))
+        if not local:
+            conn.register_mr(srcs[-1])
+            conn.register_mr(dsts[-1])
+    return srcs, dsts, blocks_per_sub
 
 
 def run_once(args, conn, local, device, bufs=None):
@@ -95,38 +112,51 @@ def run_once(args, conn, local, device, bufs=None):
     es = 2 if args.quant else 4
     page_elems = block_bytes // es
 
-    src, dst = bufs if bufs is not None else make_buffers(args, conn, local, device)
+    srcs, dsts, blocks_per_sub = (bufs if bufs is not None
+                                  else make_buffers(args, conn, local, device))
 
     run = uuid.uuid4().hex
     keys = [f"{run}-{i}" for i in range(n_blocks)]
-    offsets = [i * page_elems for i in range(n_blocks)]
     per_step = max(1, n_blocks // args.steps)
 
-    # ---- write (layer-by-layer batches, like prefill) ----
+    def sub_of(i):  # block index -> (tensor index, element offset within it)
+        return i // blocks_per_sub, (i % blocks_per_sub) * page_elems
+
+    # ---- write (layer-by-layer batches, like prefill; each step's slice
+    # lives in ONE sub-tensor by construction) ----
     t0 = time.perf_counter()
     for s0 in range(0, n_blocks, per_step):
-        sl = slice(s0, min(s0 + per_step, n_blocks))
+        hi = min(s0 + per_step, n_blocks)
+        t_idx, _ = sub_of(s0)
+        pairs = [(keys[i], sub_of(i)[1]) for i in range(s0, hi)]
         if local:
             if args.quant:
-                conn.write_pages(src, keys[sl], offsets[sl], page_elems,
+                conn.write_pages(srcs[t_idx], [p[0] for p in pairs],
+                                 [p[1] for p in pairs], page_elems,
                                  quant=args.quant)
             else:
-                conn.local_gpu_write_cache(src, list(zip(keys[sl], offsets[sl])),
-                                           page_elems)
+                conn.local_gpu_write_cache(srcs[t_idx], pairs, page_elems)
         else:
-            blocks = conn.allocate_rdma(keys[sl], block_bytes)
-            conn.rdma_write_cache(src, offsets[sl], page_elems, blocks)
+            blocks = conn.allocate_rdma([p[0] for p in pairs], block_bytes)
+            conn.rdma_write_cache(srcs[t_idx], [p[1] for p in pairs],
+                                  page_elems, blocks)
     conn.sync()
     w_time = time.perf_counter() - t0
 
-    # ---- read ----
+    # ---- read (one request per sub-tensor) ----
     t0 = time.perf_counter()
-    conn.read_cache(dst, list(zip(keys, offsets)), page_elems)
+    for t_idx in range(len(dsts)):
+        lo = t_idx * blocks_per_sub
+        hi = min(lo + blocks_per_sub, n_blocks)
+        conn.read_cache(dsts[t_idx],
+                        [(keys[i], sub_of(i)[1]) for i in range(lo, hi)],
+                        page_elems)
     conn.sync()
     r_time = time.perf_counter() - t0
 
     if args.verify:
-        assert torch.equal(src.cpu(), dst.cpu()), "verification failed"
+        for a, b in zip(srcs, dsts):
+            assert torch.equal(a.cpu(), b.cpu()), "verification failed"
 
     # Steady-state footprint: drop this iteration's keys (each iteration
     # writes a fresh key set; an engine similarly evicts finished
