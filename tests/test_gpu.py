@@ -664,3 +664,28 @@ def test_fast_op_pool_many_conns(gpu_server):
     finally:
         for c in conns:
             c.close()
+
+
+def test_big_allocation_refused_cleanly(gpu_server):
+    """Allocations >= 2 GiB cannot cross the IPC boundary on this driver
+    stack (hipIpcOpenMemHandle hangs; scripts/ipc_size_probe.py). The
+    server must refuse with an error — NOT hang a poller — and the
+    connection must stay usable."""
+    conn = local_conn(gpu_server)
+    try:
+        big = torch.zeros(1 << 30, dtype=torch.bfloat16, device="cuda:0")  # 2 GiB
+        with pytest.raises(Exception):
+            conn.local_gpu_write_cache(big, [(f"big-{uuid.uuid4()}", 0)], 32768)
+            conn.sync()
+        del big
+        # the conn (and its poller) survived
+        small = torch.randn(32768, dtype=torch.bfloat16, device="cuda:0")
+        out = torch.zeros_like(small)
+        key = f"big-after-{uuid.uuid4()}"
+        conn.local_gpu_write_cache(small, [(key, 0)], 32768)
+        conn.sync()
+        conn.read_cache(out, [(key, 0)], 32768)
+        conn.sync()
+        assert torch.equal(small, out)
+    finally:
+        conn.close()
