@@ -689,3 +689,61 @@ def test_big_allocation_refused_cleanly(gpu_server):
         assert torch.equal(small, out)
     finally:
         conn.close()
+
+
+CLIENT_BURST = r"""
+import sys, time, numpy as np, torch
+sys.path.insert(0, sys.argv[3])
+import infinistore_amd as ifs
+port = int(sys.argv[1]); tag = sys.argv[2]
+conns = []
+for _ in range(9):  # >=8 ring peers engages the fast-op worker pool
+    c = ifs.InfinityConnection(ifs.ClientConfig(
+        host_addr="127.0.0.1", service_port=port,
+        connection_type=ifs.TYPE_LOCAL_GPU))
+    c.connect(); conns.append(c)
+src = torch.randn(256 * 32768, dtype=torch.bfloat16, device="cuda:0")
+offs = np.arange(256, dtype=np.uint64) * 32768
+print("READY", flush=True)
+i = 0
+while True:  # async bursts forever; the parent kill -9s us mid-flight
+    i += 1
+    for j, c in enumerate(conns):
+        c.write_pages(src, [f"{tag}-{j}-{i}-{k}" for k in range(256)], offs,
+                      32768, sync=False)
+"""
+
+
+def test_client_killed_mid_burst(gpu_server):
+    """kill -9 a client while its async write bursts are queued in the
+    fast-op pool: the server must tear the conns down (poller joined, refs
+    drained) and keep serving other clients."""
+    import signal
+
+    tag = uuid.uuid4().hex[:8]
+    proc = subprocess.Popen(
+        [sys.executable, "-c", CLIENT_BURST, str(gpu_server), tag, REPO],
+        cwd=REPO, stdout=subprocess.PIPE, text=True)
+    try:
+        line = proc.stdout.readline()
+        assert "READY" in line, line
+        time.sleep(1.5)  # let bursts queue up
+        proc.send_signal(signal.SIGKILL)
+        proc.wait(timeout=15)
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+    # the server survives and serves a fresh client correctly
+    time.sleep(1.0)
+    conn = local_conn(gpu_server)
+    try:
+        src = torch.randn(32768, dtype=torch.bfloat16, device="cuda:0")
+        dst = torch.zeros_like(src)
+        key = f"after-kill-{uuid.uuid4()}"
+        conn.local_gpu_write_cache(src, [(key, 0)], 32768)
+        conn.sync()
+        conn.read_cache(dst, [(key, 0)], 32768)
+        conn.sync()
+        assert torch.equal(src, dst)
+    finally:
+        conn.close()
