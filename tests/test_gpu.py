@@ -708,8 +708,9 @@ print("READY", flush=True)
 i = 0
 while True:  # async bursts forever; the parent kill -9s us mid-flight
     i += 1
-    for j, c in enumerate(conns):
-        c.write_pages(src, [f"{tag}-{j}-{i}-{k}" for k in range(256)], offs,
+    gen = i % 4  # bounded key space: keys outlive the client (it's a cache),
+    for j, c in enumerate(conns):  # so an unbounded burst would fill the pool
+        c.write_pages(src, [f"{tag}-{j}-{gen}-{k}" for k in range(256)], offs,
                       32768, sync=False)
 """
 
