@@ -414,6 +414,10 @@ void Server::stop() {
     for (auto& t : fast_workers_)
         if (t.joinable()) t.join();
     fast_workers_.clear();
+    {
+        std::lock_guard<std::mutex> jl(extend_mu_);
+        if (extend_thread_.joinable()) extend_thread_.join();
+    }
     running_.store(false);
     // Drop all stored blocks.
     purge();
@@ -880,7 +884,11 @@ void Server::maybe_extend(Shard* s) {
     if (!s->need_extend()) return;
     int expect = 0;
     if (!extending_.compare_exchange_strong(expect, 1)) return;
-    std::thread([this, s] {
+    // Managed (not detached): a detached extender could still be unlocking
+    // vdrv_mu_ while ~Server destroys it (TSAN-caught shutdown race).
+    std::lock_guard<std::mutex> jl(extend_mu_);
+    if (extend_thread_.joinable()) extend_thread_.join();  // prior run done
+    extend_thread_ = std::thread([this, s] {
         INFO("extending pool on shard dev=%d", s->device());
         void* arena = nullptr;
         if (s->extend(&arena) && arena) {
@@ -892,7 +900,7 @@ void Server::maybe_extend(Shard* s) {
                 ERROR("MR registration of extended arena failed; verbs reads from it will error");
         }
         extending_.store(0);
-    }).detach();
+    });
 }
 
 // ---- local (IPC) path -----------------------------------------------------

@@ -342,6 +342,8 @@ class Server {
     std::atomic<uint64_t> n_writes_{0}, n_reads_{0}, n_put_{0}, n_get_{0};
     std::atomic<uint64_t> bytes_in_{0}, bytes_out_{0};
     std::atomic<int> extending_{0};
+    std::thread extend_thread_;  // managed pool extender (extend_mu_)
+    std::mutex extend_mu_;
 
     // per-op handler timing (loop-thread time, µs) — the per-op latency log
     // the reference keeps via INFO prints (infinistore.cpp:1162-1166),
