@@ -54,10 +54,10 @@ def parse_args():
                         "reads dequantize to bf16 — verification switches "
                         "to an fp8-tolerance comparison")
     p.add_argument("--conns", type=int, default=3,
-                   help="write/read connection pairs per rank (local path): "
-                        "each pair runs its own two-deep pipelined loop over "
-                        "1/conns of the blocks in its own thread (3 measured "
-                        "best: 1.73-1.80 TB/s vs 1.55 at 2 on one MI355X)")
+                   help="LEGACY threaded mode only (--procs 0): write/read "
+                        "connection pairs per rank, each a two-deep "
+                        "pipelined loop in its own thread (GIL-bound at "
+                        "~1.4 TB/s; the default process mode replaces it)")
     p.add_argument("--procs", type=int, default=4,
                    help="client worker PROCESSES per rank (local path): each "
                         "runs a pipelined conn pair over 1/procs of the "
