@@ -426,6 +426,18 @@ PYBIND11_MODULE(_native, m) {
         auto v = build_local_meta(msg);
         return py::bytes(reinterpret_cast<const char*>(v.data()), v.size());
     });
+    // Export a device pointer's IPC handle + allocation offset: lets tests
+    // hand-craft reference-framed LocalMetaRequest transcripts (the 'W'/'R'
+    // flatbuffers ops) without going through this client library.
+    m.def("_dbg_ipc_export", [](uintptr_t ptr) {
+        ifs::gpu::IpcHandle h;
+        uint64_t off = 0;
+        if (!ifs::gpu::ipc_export(reinterpret_cast<void*>(ptr), &h, &off))
+            throw std::runtime_error("ipc_export failed");
+        return py::make_tuple(
+            py::bytes(reinterpret_cast<const char*>(h.bytes), sizeof(h.bytes)), off);
+    });
+
     m.def("_dbg_parse_local_meta", [](py::bytes data) {
         std::string s = data;
         LocalMetaMsg msg;

@@ -748,3 +748,49 @@ def test_client_killed_mid_burst(gpu_server):
         assert torch.equal(src, dst)
     finally:
         conn.close()
+
+
+def test_ref_framed_flatbuffers_local_write(gpu_server):
+    """The reference's local-op wire shape: a raw-socket 'W' with a
+    flatbuffers LocalMetaRequest body (cuda IPC handle + [Block{key,
+    byte-offset}]), answered 202, then 'S' sync (FINISH + remain) — and the
+    page read back through the normal client. Covers the flatbuffers local
+    path, which this repo's own client never uses (it speaks the packed
+    extension ops)."""
+    import socket as socklib
+    import struct
+
+    from infinistore_amd import _native as n
+
+    src = torch.randn(32768, dtype=torch.float32, device="cuda:0")
+    handle, base_off = n._dbg_ipc_export(src.data_ptr())
+    key = f"fbw-{uuid.uuid4().hex[:8]}"
+    body = n._dbg_build_local_meta(0, handle, 32768 * 4, [(key, 0)], base_off)
+
+    s = socklib.create_connection(("127.0.0.1", gpu_server), timeout=30)
+    try:
+        s.sendall(struct.pack("<IcI", 0xDEADBEEF, b"W", len(body)) + body)
+        code = struct.unpack("<i", s.recv(4))[0]
+        assert code == 202, code  # TASK_ACCEPTED
+        for _ in range(100):  # reference-framed sync until remain == 0
+            s.sendall(struct.pack("<IcI", 0xDEADBEEF, b"S", 0))
+            buf = b""
+            while len(buf) < 8:
+                buf += s.recv(8 - len(buf))
+            code, remain = struct.unpack("<ii", buf)
+            assert code == 200
+            if remain == 0:
+                break
+            time.sleep(0.01)
+        assert remain == 0
+    finally:
+        s.close()
+
+    conn = local_conn(gpu_server)
+    try:
+        dst = torch.zeros_like(src)
+        conn.read_cache(dst, [(key, 0)], 32768)
+        conn.sync()
+        assert torch.equal(src, dst)
+    finally:
+        conn.close()
