@@ -794,3 +794,28 @@ def test_ref_framed_flatbuffers_local_write(gpu_server):
         assert torch.equal(src, dst)
     finally:
         conn.close()
+
+    # and the reference-framed 'R': async accept + poll-sync, data pushed
+    # into the raw client's tensor through its IPC handle
+    dst2 = torch.zeros_like(src)
+    h2, off2 = n._dbg_ipc_export(dst2.data_ptr())
+    rbody = n._dbg_build_local_meta(0, h2, 32768 * 4, [(key, 0)], off2)
+    s = socklib.create_connection(("127.0.0.1", gpu_server), timeout=30)
+    try:
+        s.sendall(struct.pack("<IcI", 0xDEADBEEF, b"R", len(rbody)) + rbody)
+        code = struct.unpack("<i", s.recv(4))[0]
+        assert code == 202, code
+        for _ in range(100):
+            s.sendall(struct.pack("<IcI", 0xDEADBEEF, b"S", 0))
+            buf = b""
+            while len(buf) < 8:
+                buf += s.recv(8 - len(buf))
+            code, remain = struct.unpack("<ii", buf)
+            assert code == 200
+            if remain == 0:
+                break
+            time.sleep(0.01)
+        assert remain == 0
+        assert torch.equal(src, dst2)
+    finally:
+        s.close()
